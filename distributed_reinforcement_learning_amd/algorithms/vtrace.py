@@ -1,0 +1,121 @@
+"""V-trace importance-weighted returns (Espeholt et al. 2018).
+
+Semantics-parity with reference optimizer/vtrace.py:29-126, redesigned for the
+batched unroll:
+
+* The reference computes values / next_values from *separate* window replicas
+  whose inputs are identical (impala_actor_critic.py:71-114), so
+  ``values[:, 1:]`` == ``next_values[:, :-1]`` exactly; with the batched
+  unroll they are literal views of one tensor and ``values_{t+1}`` is simply
+  ``next_values`` (reference vtrace.py:81-83 concat + bootstrap collapses).
+* Everything is batch-major [B, T]; the reference's transposes to time-major
+  (vtrace.py:53-57) disappear.
+* ``vs`` and ``clipped_rho`` are detached, matching tf.stop_gradient
+  (vtrace.py:103).
+
+The reverse recursion vs_t = delta_t + discount_t * c_t * (vs_{t+1}-V_{t+1})
+runs in a tiny Python loop on CPU (T<=20) and in one fused HIP kernel on
+gfx950 (ops/hip/vtrace.hip) — sequential in T, all B lanes parallel.
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from distributed_reinforcement_learning_amd import ops as _ops
+
+
+def split_data(x: torch.Tensor):
+    """first/middle/last shifted windows over dim 1
+    (reference vtrace.py:3-14)."""
+    return x[:, :-2], x[:, 1:-1], x[:, 2:]
+
+
+def log_probs_from_softmax_and_actions(policy_softmax: torch.Tensor,
+                                       actions: torch.Tensor) -> torch.Tensor:
+    """log pi(a_t | x_t) from a [B,T,A] softmax and [B,T] actions
+    (reference vtrace.py:16-27)."""
+    sel = policy_softmax.gather(2, actions.long().unsqueeze(-1)).squeeze(-1)
+    return torch.log(sel)
+
+
+@torch.no_grad()
+def from_importance_weights(log_rhos: torch.Tensor, discounts: torch.Tensor,
+                            rewards: torch.Tensor, values: torch.Tensor,
+                            bootstrap_value: torch.Tensor,
+                            clip_rho_threshold: float = 1.0,
+                            clip_c_threshold: float = 1.0
+                            ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Batch-major V-trace core (reference vtrace.py:71-103).
+
+    All args [B, T] except bootstrap_value [B]. Returns (vs, clipped_rho),
+    both [B, T] and gradient-free.
+    """
+    rhos = torch.exp(log_rhos)
+    clipped_rhos = torch.clamp(rhos, max=clip_rho_threshold) \
+        if clip_rho_threshold is not None else rhos
+    cs = torch.clamp(rhos, max=clip_c_threshold)
+    values_t_plus_1 = torch.cat(
+        [values[:, 1:], bootstrap_value.unsqueeze(1)], dim=1)
+    deltas = clipped_rhos * (rewards + discounts * values_t_plus_1 - values)
+
+    if values.is_cuda and _ops.available():
+        vs_minus_v = _ops.vtrace_scan(deltas, discounts, cs)
+    else:
+        T = values.shape[1]
+        acc = torch.zeros_like(bootstrap_value)
+        out = []
+        for t in reversed(range(T)):
+            acc = deltas[:, t] + discounts[:, t] * cs[:, t] * acc
+            out.append(acc)
+        vs_minus_v = torch.stack(out[::-1], dim=1)
+    vs = vs_minus_v + values
+    return vs, clipped_rhos
+
+
+@torch.no_grad()
+def from_softmax(behavior_policy_softmax: torch.Tensor,
+                 target_policy_softmax: torch.Tensor,
+                 actions: torch.Tensor, discounts: torch.Tensor,
+                 rewards: torch.Tensor, values: torch.Tensor,
+                 next_values: torch.Tensor,
+                 clip_rho_threshold: float = 1.0,
+                 clip_c_threshold: float = 1.0):
+    """V-trace from policy softmaxes (reference vtrace.py:29-69).
+
+    All [B,T(,A)]; ``next_values[:, t]`` must equal V(x_{t+1}) — with the
+    batched unroll this is the middle-window slice of the same value tensor.
+    """
+    target_lp = log_probs_from_softmax_and_actions(target_policy_softmax,
+                                                   actions)
+    behavior_lp = log_probs_from_softmax_and_actions(behavior_policy_softmax,
+                                                     actions)
+    log_rhos = target_lp - behavior_lp
+    return from_importance_weights(
+        log_rhos=log_rhos, discounts=discounts, rewards=rewards,
+        values=values, bootstrap_value=next_values[:, -1],
+        clip_rho_threshold=clip_rho_threshold,
+        clip_c_threshold=clip_c_threshold)
+
+
+def compute_policy_gradient_loss(softmax: torch.Tensor, actions: torch.Tensor,
+                                 advantages: torch.Tensor) -> torch.Tensor:
+    """-sum log pi(a) * adv (reference vtrace.py:105-112; same 1e-8 guard and
+    sum reduction)."""
+    sel = softmax.gather(2, actions.long().unsqueeze(-1)).squeeze(-1)
+    log_prob = torch.log(sel + 1e-8)
+    return -(log_prob * advantages.detach()).sum()
+
+
+def compute_baseline_loss(vs: torch.Tensor, value: torch.Tensor) -> torch.Tensor:
+    """0.5 * sum (vs - V)^2 (reference vtrace.py:114-118)."""
+    return 0.5 * (vs.detach() - value).pow(2).sum()
+
+
+def compute_entropy_loss(softmax: torch.Tensor) -> torch.Tensor:
+    """-sum_t entropy_t (reference vtrace.py:120-126: the *negative* entropy,
+    added to the total loss with entropy_coef)."""
+    entropy_per_step = (-softmax * torch.log(softmax)).sum(dim=-1)
+    return -entropy_per_step.sum()

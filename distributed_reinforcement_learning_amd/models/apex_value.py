@@ -1,0 +1,62 @@
+"""Ape-X dueling DQN model (reference model/apex_value.py).
+
+Q = value_branch(|A| units) - mean_branch(1 unit), both fed by
+conv+action-embedding features (apex_value.py:22-40; note this is the
+reference's own dueling form — the mean branch is a learned scalar, not the
+advantage-mean subtraction of the dueling paper; kept for parity).
+
+The agent holds two instances (main / target) — the reference's variable
+scopes 'main' / 'target' (apex_value.py:44-63).
+
+``VectorDuelingQ`` mirrors the reference's unused ``simple_network``
+(apex_value.py:67-100) for 1-D observations.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from distributed_reinforcement_learning_amd.models.blocks import (
+    ActionEmbedding, AtariConvStack, MLPHead,
+)
+
+
+class ApexDuelingQ(nn.Module):
+    def __init__(self, input_shape, num_action: int, hidden_list=(256, 256)):
+        super().__init__()
+        h, w, c = input_shape
+        assert (h, w) == (84, 84)
+        self.num_action = num_action
+        self.conv = AtariConvStack(c)
+        self.action_emb = ActionEmbedding(num_action)
+        feat = self.conv.out_features + self.action_emb.out_features
+        self.value_branch = MLPHead(feat, list(hidden_list), num_action, None)
+        self.mean_branch = MLPHead(feat, list(hidden_list), 1, None)
+
+    def forward(self, state: torch.Tensor, prev_action: torch.Tensor):
+        """state [N,84,84,C] normalized NHWC float; returns Q [N,A]."""
+        feat = torch.cat(
+            [self.conv(state), self.action_emb(prev_action)], dim=1)
+        return self.value_branch(feat) - self.mean_branch(feat)
+
+
+class VectorDuelingQ(nn.Module):
+    def __init__(self, input_shape, num_action: int, hidden: int = 256):
+        super().__init__()
+        (obs_dim,) = input_shape
+        self.num_action = num_action
+        self.obs_mlp = nn.Sequential(
+            nn.Linear(obs_dim, hidden), nn.ReLU(),
+            nn.Linear(hidden, hidden), nn.ReLU())
+        self.action_emb = ActionEmbedding(num_action, hidden)
+        self.trunk = nn.Linear(2 * hidden, hidden)
+        self.value_out = nn.Linear(hidden, num_action)
+        self.mean_out = nn.Linear(hidden, 1)
+
+    def forward(self, state: torch.Tensor, prev_action: torch.Tensor):
+        feat = torch.cat(
+            [self.obs_mlp(state), self.action_emb(prev_action)], dim=1)
+        x = F.relu(self.trunk(feat))
+        return self.value_out(x) - self.mean_out(x)

@@ -1,0 +1,121 @@
+"""Shared model building blocks.
+
+The reference copy-pastes these per model file (attention_CNN / action_embedding
+/ fully_connected / lstm in model/actor_critic.py:3-26,
+model/impala_actor_critic.py:5-31, model/apex_value.py:4-20,
+model/r2d2_lstm.py:4-24); here they are defined once.
+
+Layout convention: public model APIs take states as the reference stores them —
+NHWC [N, H, W, C] float (already /255-normalized) — and convert to NCHW
+internally for torch convs. The custom HIP conv path (ops/) consumes NHWC
+uint8 directly and fuses the /255 normalize into the first conv's load stage.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class AtariConvStack(nn.Module):
+    """Conv 8x8/4 -> 32, 4x4/2 -> 64, 3x3/1 -> 64, VALID padding, ReLU,
+    flatten.
+
+    84x84xC input -> 20x20x32 -> 9x9x64 -> 7x7x64 -> 3136 features
+    (reference model/impala_actor_critic.py:5-10).
+    """
+
+    def __init__(self, in_channels: int):
+        super().__init__()
+        self.conv1 = nn.Conv2d(in_channels, 32, kernel_size=8, stride=4)
+        self.conv2 = nn.Conv2d(32, 64, kernel_size=4, stride=2)
+        self.conv3 = nn.Conv2d(64, 64, kernel_size=3, stride=1)
+        self.out_features = 7 * 7 * 64
+
+    def forward(self, x_nhwc: torch.Tensor) -> torch.Tensor:
+        x = x_nhwc.permute(0, 3, 1, 2).contiguous()
+        x = F.relu(self.conv1(x))
+        x = F.relu(self.conv2(x))
+        x = F.relu(self.conv3(x))
+        return x.flatten(1)
+
+
+class ActionEmbedding(nn.Module):
+    """one-hot(prev_action) -> 256 -> 256, ReLU
+    (reference model/impala_actor_critic.py:12-16)."""
+
+    def __init__(self, num_action: int, hidden: int = 256):
+        super().__init__()
+        self.num_action = num_action
+        self.fc1 = nn.Linear(num_action, hidden)
+        self.fc2 = nn.Linear(hidden, hidden)
+        self.out_features = hidden
+
+    def forward(self, prev_action: torch.Tensor) -> torch.Tensor:
+        onehot = F.one_hot(prev_action.long(), self.num_action).to(
+            self.fc1.weight.dtype)
+        x = F.relu(self.fc1(onehot))
+        return F.relu(self.fc2(x))
+
+
+class MLPHead(nn.Module):
+    """Dense stack with ReLU between hidden layers, optional final activation
+    (reference model/impala_actor_critic.py:27-31)."""
+
+    def __init__(self, in_features: int, hidden_list: Sequence[int],
+                 out_features: int,
+                 final_activation: Optional[str] = None):
+        super().__init__()
+        layers: List[nn.Module] = []
+        prev = in_features
+        for h in hidden_list:
+            layers.append(nn.Linear(prev, h))
+            prev = h
+        self.hidden = nn.ModuleList(layers)
+        self.out = nn.Linear(prev, out_features)
+        self.final_activation = final_activation
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        for layer in self.hidden:
+            x = F.relu(layer(x))
+        x = self.out(x)
+        if self.final_activation == "softmax":
+            x = F.softmax(x, dim=-1)
+        return x
+
+
+class LSTMCellTF(nn.Module):
+    """LSTM cell with TF-LSTMCell semantics (forget_bias=1.0, no peepholes) —
+    reference model/impala_actor_critic.py:18-25 runs tf.nn.rnn_cell.LSTMCell
+    over a length-1 sequence, i.e. exactly one cell step.
+
+    Gates are computed as one fused GEMM over concat([x, h]) with gate order
+    [i, g, f, o]; the elementwise tail (sigmoid/tanh/blend) is the piece the
+    HIP fused-gate kernel (ops/hip/lstm_gates.hip) replaces on gfx950.
+    """
+
+    def __init__(self, input_size: int, hidden_size: int,
+                 forget_bias: float = 1.0):
+        super().__init__()
+        self.input_size = input_size
+        self.hidden_size = hidden_size
+        self.forget_bias = forget_bias
+        self.weight = nn.Parameter(
+            torch.empty(input_size + hidden_size, 4 * hidden_size))
+        self.bias = nn.Parameter(torch.zeros(4 * hidden_size))
+        nn.init.xavier_uniform_(self.weight)
+
+    def gates(self, x: torch.Tensor, h: torch.Tensor) -> torch.Tensor:
+        xh = torch.cat([x, h], dim=1)
+        return xh @ self.weight + self.bias
+
+    def forward(self, x: torch.Tensor, h: torch.Tensor, c: torch.Tensor):
+        g = self.gates(x, h)
+        i, j, f, o = g.chunk(4, dim=1)
+        new_c = torch.sigmoid(f + self.forget_bias) * c \
+            + torch.sigmoid(i) * torch.tanh(j)
+        new_h = torch.sigmoid(o) * torch.tanh(new_c)
+        return new_h, new_c

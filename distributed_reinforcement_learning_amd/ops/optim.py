@@ -1,0 +1,157 @@
+"""Fused optimizers with global-norm clipping (K12 in SURVEY.md §2.5).
+
+Replaces the reference's tf.clip_by_global_norm + RMSProp/Adam apply
+(agent/impala.py:96-100, agent/apex.py:73-77) with:
+
+1. one fused squared-norm reduction over ALL grads (flat view), then
+2. one fused update kernel applying clip-scale + RMSProp/Adam in a single
+   HBM pass per parameter bucket.
+
+Parameters and grads live in ONE contiguous flat buffer each (built once at
+optimizer construction; module parameters are views into it), so the whole
+update is two kernel launches — and the DP all-reduce (parallel/dist.py)
+reduces the same flat grad buffer with a single RCCL call.
+
+CPU fallback implements identical math in torch (used by unit tests as the
+golden reference).
+"""
+
+from __future__ import annotations
+
+from typing import Iterable, List, Optional
+
+import torch
+
+from distributed_reinforcement_learning_amd import ops as _ops
+
+
+def flatten_dense_params(params: List[torch.Tensor]) -> torch.Tensor:
+    """Re-home ``params`` as views into one new contiguous flat buffer."""
+    total = sum(p.numel() for p in params)
+    if total == 0:
+        return torch.empty(0)
+    flat = torch.empty(total, dtype=params[0].dtype, device=params[0].device)
+    offset = 0
+    for p in params:
+        n = p.numel()
+        flat[offset:offset + n].copy_(p.detach().reshape(-1))
+        p.data = flat[offset:offset + n].view_as(p)
+        offset += n
+    return flat
+
+
+class _FlatOptimizerBase:
+    """Holds flat param/grad/state buffers; subclasses implement _apply."""
+
+    def __init__(self, params: Iterable[torch.Tensor], lr: float,
+                 clip_norm: Optional[float]):
+        self.params = [p for p in params if p.requires_grad]
+        assert len(self.params) > 0
+        self.flat_params = flatten_dense_params(self.params)
+        self.flat_grads = torch.zeros_like(self.flat_params)
+        # route autograd into the flat grad buffer
+        offset = 0
+        for p in self.params:
+            n = p.numel()
+            p.grad = self.flat_grads[offset:offset + n].view_as(p)
+            offset += n
+        self.lr = lr
+        self.clip_norm = clip_norm
+        self.step_count = 0
+
+    def zero_grad(self) -> None:
+        self.flat_grads.zero_()
+
+    def grad_global_norm(self) -> torch.Tensor:
+        if self.flat_grads.is_cuda and _ops.available():
+            ext = _ops.require_ext()
+            return ext.sq_norm(self.flat_grads).sqrt()
+        return self.flat_grads.norm()
+
+    def _clip_scale(self) -> torch.Tensor:
+        """tf.clip_by_global_norm semantics: scale = clip/max(norm, clip)."""
+        norm = self.grad_global_norm()
+        if self.clip_norm is None:
+            return torch.ones_like(norm)
+        return self.clip_norm / torch.clamp(norm, min=self.clip_norm)
+
+    def state_dict(self) -> dict:
+        return {"step_count": self.step_count,
+                "state": {k: v for k, v in self._state_tensors().items()}}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.step_count = sd["step_count"]
+        for k, v in sd["state"].items():
+            self._state_tensors()[k].copy_(v)
+
+    def _state_tensors(self) -> dict:
+        raise NotImplementedError
+
+    def step(self, lr: Optional[float] = None) -> None:
+        raise NotImplementedError
+
+
+class FusedRMSProp(_FlatOptimizerBase):
+    """TF-RMSProp semantics: ms = rho*ms + (1-rho)*g^2,
+    p -= lr * g / sqrt(ms + eps) — eps inside the sqrt, as TF does.
+    Matches reference agent/impala.py:97 (decay=.99, momentum=0, eps=.1)."""
+
+    def __init__(self, params, lr: float, rho: float = 0.99,
+                 eps: float = 0.1, clip_norm: Optional[float] = None):
+        super().__init__(params, lr, clip_norm)
+        self.rho = rho
+        self.eps = eps
+        self.ms = torch.zeros_like(self.flat_params)
+
+    def _state_tensors(self):
+        return {"ms": self.ms}
+
+    @torch.no_grad()
+    def step(self, lr: Optional[float] = None) -> None:
+        lr = self.lr if lr is None else lr
+        scale = self._clip_scale()
+        if self.flat_params.is_cuda and _ops.available():
+            ext = _ops.require_ext()
+            ext.rmsprop_step(self.flat_params, self.flat_grads, self.ms,
+                             scale, float(lr), self.rho, self.eps)
+        else:
+            g = self.flat_grads * scale
+            self.ms.mul_(self.rho).addcmul_(g, g, value=1 - self.rho)
+            self.flat_params.addcdiv_(g, (self.ms + self.eps).sqrt(),
+                                      value=-lr)
+        self.step_count += 1
+
+
+class FusedAdam(_FlatOptimizerBase):
+    """TF-AdamOptimizer semantics (reference agent/apex.py:73): bias-corrected
+    lr_t = lr * sqrt(1-b2^t)/(1-b1^t); p -= lr_t * m / (sqrt(v) + eps)."""
+
+    def __init__(self, params, lr: float, beta1: float = 0.9,
+                 beta2: float = 0.999, eps: float = 1e-8,
+                 clip_norm: Optional[float] = None):
+        super().__init__(params, lr, clip_norm)
+        self.beta1, self.beta2, self.eps = beta1, beta2, eps
+        self.m = torch.zeros_like(self.flat_params)
+        self.v = torch.zeros_like(self.flat_params)
+
+    def _state_tensors(self):
+        return {"m": self.m, "v": self.v}
+
+    @torch.no_grad()
+    def step(self, lr: Optional[float] = None) -> None:
+        lr = self.lr if lr is None else lr
+        self.step_count += 1
+        t = self.step_count
+        scale = self._clip_scale()
+        lr_t = lr * (1 - self.beta2 ** t) ** 0.5 / (1 - self.beta1 ** t)
+        if self.flat_params.is_cuda and _ops.available():
+            ext = _ops.require_ext()
+            ext.adam_step(self.flat_params, self.flat_grads, self.m, self.v,
+                          scale, float(lr_t), self.beta1, self.beta2,
+                          self.eps)
+        else:
+            g = self.flat_grads * scale
+            self.m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+            self.v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+            self.flat_params.addcdiv_(self.m, self.v.sqrt() + self.eps,
+                                      value=-lr_t)

@@ -1,0 +1,74 @@
+"""Learner data-parallelism over RCCL/xGMI.
+
+One process per GPU; ``torch.distributed`` backend "nccl" (RCCL on ROCm) for
+GPU runs, "gloo" for CPU tests. Gradients live in ONE flat contiguous buffer
+(ops/optim.py), so the per-step all-reduce is a single fused RCCL call — at
+reference model sizes (~16 MB of grads) the all-reduce is latency-bound and
+one bucket beats any bucketing schedule (SURVEY §2.4 xGMI note).
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+
+def is_distributed() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def world_size() -> int:
+    return dist.get_world_size() if is_distributed() else 1
+
+
+def rank() -> int:
+    return dist.get_rank() if is_distributed() else 0
+
+
+def init_distributed(backend: Optional[str] = None,
+                     timeout_s: float = 300.0) -> int:
+    """Initialize from torchrun-style env vars; no-op at WORLD_SIZE<=1.
+    Returns the local rank."""
+    ws = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
+    if ws <= 1:
+        return local_rank
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+    if not dist.is_initialized():
+        dist.init_process_group(
+            backend=backend,
+            timeout=datetime.timedelta(seconds=timeout_s))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+    return local_rank
+
+
+class FlatAllReducer:
+    """Average one flat gradient buffer across ranks with a single
+    all-reduce call."""
+
+    def __init__(self, flat_grads: torch.Tensor):
+        self.flat_grads = flat_grads
+        self._inv_world = 1.0 / world_size()
+
+    def all_reduce(self) -> None:
+        if not is_distributed():
+            return
+        dist.all_reduce(self.flat_grads, op=dist.ReduceOp.SUM)
+        self.flat_grads.mul_(self._inv_world)
+
+
+def broadcast_module(module: torch.nn.Module, src: int = 0) -> None:
+    """Rank-0 init + broadcast (fixes the reference's re-init race, C5 in
+    SURVEY.md §2.4)."""
+    if not is_distributed():
+        return
+    for t in module.state_dict().values():
+        dist.broadcast(t, src=src)
