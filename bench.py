@@ -1,0 +1,150 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: IMPALA deep-conv learner, env-frames/sec (whole job).
+
+BASELINE.json metric: "learner env-frames/sec (whole node), IMPALA deep-conv
+84x84x4 at 1/2/4/8 MI355X". One train step consumes B*T = 32*20 = 640 env
+frames per rank (reference config.json:130,136; global-step accounting at
+agent/impala.py:95-100). Weak scaling: per-GPU batch fixed at 32, global
+batch = 32*N.
+
+Each timed step is the FULL learner update on synthetic uint8 frames with
+random-init weights: H2D upload of the uint8 batch, on-device /255 normalize,
+batched conv+LSTM+heads forward (bf16 autocast), V-trace, backward, flat
+all-reduce (N>1, RCCL), fused global-norm-clip + RMSProp update. Nothing is
+cached or skipped; data is synthetic because this image has no network
+(BASELINE.json mandates synthetic frames).
+
+Launch (driver contract):
+    python bench.py --gpus N --steps K --warmup W
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np
+import torch
+
+
+def main(argv=None) -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=32,
+                   help="per-rank batch (reference config: 32)")
+    p.add_argument("--trajectory", type=int, default=20)
+    p.add_argument("--num-action", type=int, default=18)
+    p.add_argument("--lstm", type=int, default=256)
+    p.add_argument("--pool", type=int, default=8,
+                   help="distinct synthetic batches cycled through")
+    args = p.parse_args(argv)
+
+    from distributed_reinforcement_learning_amd.agents import impala
+    from distributed_reinforcement_learning_amd.parallel import dist as pdist
+
+    local_rank = pdist.init_distributed()
+    world = pdist.world_size()
+    rank = pdist.rank()
+    have_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+
+    B, T, A, H = args.batch, args.trajectory, args.num_action, args.lstm
+    agent = impala.Agent(
+        trajectory=T, input_shape=[84, 84, 4], num_action=A,
+        lstm_hidden_size=H, discount_factor=0.99, start_learning_rate=6e-4,
+        end_learning_rate=0.0, learning_frame=10 ** 9,
+        baseline_loss_coef=1.0, entropy_coef=0.05, gradient_clip_norm=40.0,
+        reward_clipping="abs_one", device=device, seed=1234 + rank)
+    if world > 1:
+        from distributed_reinforcement_learning_amd.parallel.dist import (
+            broadcast_module,
+        )
+        broadcast_module(agent.model)
+        agent.setup_all_reduce()
+
+    # synthetic trajectory batches, shaped exactly like the queue's output
+    rng = np.random.default_rng(99 + rank)
+    pool = []
+    for _ in range(args.pool):
+        pool.append(dict(
+            state=rng.integers(0, 255, (B, T, 84, 84, 4), dtype=np.uint8),
+            reward=rng.normal(size=(B, T)).astype(np.float32),
+            action=rng.integers(0, A, (B, T)).astype(np.int32),
+            done=(rng.random((B, T)) < 0.02),
+            behavior_policy=np.full((B, T, A), 1.0 / A, dtype=np.float32),
+            previous_action=rng.integers(0, A, (B, T)).astype(np.int32),
+            initial_h=(rng.normal(size=(B, T, H)) * 0.1).astype(np.float32),
+            initial_c=(rng.normal(size=(B, T, H)) * 0.1).astype(np.float32),
+        ))
+
+    def step(i: int):
+        b = pool[i % len(pool)]
+        return agent.train(
+            state=b["state"], reward=b["reward"], action=b["action"],
+            done=b["done"], behavior_policy=b["behavior_policy"],
+            previous_action=b["previous_action"], initial_h=b["initial_h"],
+            initial_c=b["initial_c"])
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if have_gpu:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        step(i)
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    barrier_sync()
+    dt = time.perf_counter() - t0
+
+    # MAX over ranks (slowest rank defines job time)
+    if world > 1:
+        dt_t = torch.tensor([dt], dtype=torch.float64,
+                            device=device if have_gpu else "cpu")
+        torch.distributed.all_reduce(dt_t,
+                                     op=torch.distributed.ReduceOp.MAX)
+        dt = float(dt_t.item())
+
+    frames = args.steps * B * T * world
+    value = frames / dt
+    ms_per_step = dt / args.steps * 1e3
+    if rank == 0:
+        print(json.dumps({
+            "metric": "learner env-frames/sec (whole node), IMPALA "
+                      "deep-conv 84x84x4",
+            "value": value,
+            "unit": "frames/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if have_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "impala deep-conv 84x84x4 + LSTM256 V-trace",
+                "global_batch": B * world,
+                "seq_len": T,
+                "parallelism": f"dp{world}",
+                "num_action": A,
+                "frames_per_step_per_rank": B * T,
+            },
+        }), flush=True)
+
+
+if __name__ == "__main__":
+    main()

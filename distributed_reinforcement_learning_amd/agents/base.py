@@ -1,0 +1,153 @@
+"""Shared agent machinery: device/dtype policy, polynomial LR decay,
+reward clipping, checkpointing, weight sync wiring.
+
+The reference's Agent classes (agent/{a3c,impala,apex,r2d2}.py) each rebuild
+this; here it is one base. Checkpoints keep the reference's method names
+(save_weights/load_weights — agent/impala.py:105-109) with a torch.save dict
+{model, optimizer, global_step} as the layout.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import numpy as np
+import torch
+
+from distributed_reinforcement_learning_amd.parallel import dist as _dist_mod
+from distributed_reinforcement_learning_amd.parallel.dist import (
+    FlatAllReducer, is_distributed,
+)
+
+
+def polynomial_decay(start: float, end: float, step: int,
+                     horizon: int) -> float:
+    """tf.train.polynomial_decay(power=1) semantics
+    (reference agent/impala.py:96)."""
+    frac = min(step, horizon) / float(horizon)
+    return (start - end) * (1.0 - frac) + end
+
+
+def clip_rewards(r: torch.Tensor, mode: str) -> torch.Tensor:
+    """abs_one / soft_asymmetric / none (reference agent/impala.py:45-49)."""
+    if mode == "abs_one":
+        return torch.clamp(r, -1.0, 1.0)
+    if mode == "soft_asymmetric":
+        squeezed = torch.tanh(r / 5.0)
+        return torch.where(r < 0, 0.3 * squeezed, squeezed) * 5.0
+    if mode == "none":
+        return r
+    raise ValueError(mode)
+
+
+class AgentBase:
+    """Device/dtype policy + LR schedule + checkpoint + sync plumbing."""
+
+    def __init__(self, device: str = "cpu",
+                 compute_dtype: torch.dtype = torch.bfloat16):
+        self.device = torch.device(device)
+        self.compute_dtype = compute_dtype
+        self.global_step = 0
+        self.num_env_frames = 0
+        # wired by trainers (parallel/weights.py): learner gets a publisher,
+        # actors get a subscriber
+        self.weight_publisher = None
+        self.weight_subscriber = None
+        self._all_reducer: Optional[FlatAllReducer] = None
+
+    # -- autocast ------------------------------------------------------------
+
+    def autocast(self):
+        if self.device.type == "cuda":
+            return torch.autocast(device_type="cuda",
+                                  dtype=self.compute_dtype)
+        import contextlib
+        return contextlib.nullcontext()
+
+    # -- data movement -------------------------------------------------------
+
+    def to_device(self, arr, dtype=None) -> torch.Tensor:
+        t = torch.as_tensor(np.asarray(arr))
+        if dtype is not None:
+            t = t.to(dtype)
+        return t.to(self.device, non_blocking=True)
+
+    def frames_to_device(self, frames) -> torch.Tensor:
+        """uint8 (or float) frames -> normalized float32 on device.
+
+        uint8 stays uint8 across the H2D copy (1 byte/pixel on the bus); the
+        /255 cast runs on-device (ops/preprocess.py, HIP kernel on gfx950).
+        Float inputs (vector obs) pass through unscaled.
+        """
+        t = torch.as_tensor(np.asarray(frames))
+        if t.dtype == torch.uint8:
+            t = t.to(self.device, non_blocking=True)
+            from distributed_reinforcement_learning_amd.ops import normalize_frames
+            return normalize_frames(t)
+        return t.to(self.device, dtype=torch.float32, non_blocking=True)
+
+    # -- model plumbing (subclasses set self.model / self.optimizer) ---------
+
+    model: torch.nn.Module
+    optimizer = None
+
+    def lr_at(self, step: int) -> float:
+        return polynomial_decay(self.start_learning_rate,
+                                self.end_learning_rate, step,
+                                self.learning_frame)
+
+    def setup_all_reduce(self) -> None:
+        if is_distributed():
+            self._all_reducer = FlatAllReducer(self.optimizer.flat_grads)
+
+    def reduce_gradients(self) -> None:
+        if self._all_reducer is not None:
+            self._all_reducer.all_reduce()
+
+    # -- weight sync ---------------------------------------------------------
+
+    def parameter_sync(self) -> Optional[int]:
+        """Actor side: pull the newest published weights (replaces reference
+        copy_src_to_dst assigns, utils.py:6-22). Returns publisher step or
+        None."""
+        if self.weight_subscriber is None:
+            return None
+        sd = self.model.state_dict()
+        step = self.weight_subscriber.pull(sd)
+        if step is not None:
+            self.model.load_state_dict(sd)
+        return step
+
+    def publish_weights(self) -> None:
+        if self.weight_publisher is not None:
+            self.weight_publisher.publish(self.model.state_dict(),
+                                          global_step=self.global_step)
+
+    # -- checkpointing -------------------------------------------------------
+
+    def _checkpoint_extra(self) -> Dict:
+        return {}
+
+    def save_weights(self, path: str) -> None:
+        blob = {
+            "model": self.model.state_dict(),
+            "optimizer": (self.optimizer.state_dict()
+                          if self.optimizer is not None else None),
+            "global_step": self.global_step,
+            "num_env_frames": self.num_env_frames,
+        }
+        blob.update(self._checkpoint_extra())
+        torch.save(blob, path)
+
+    def load_weights(self, path: str) -> None:
+        blob = torch.load(path, map_location=self.device,
+                          weights_only=False)
+        self.model.load_state_dict(blob["model"])
+        if self.optimizer is not None and blob.get("optimizer") is not None:
+            self.optimizer.load_state_dict(blob["optimizer"])
+        self.global_step = blob.get("global_step", 0)
+        self.num_env_frames = blob.get("num_env_frames", 0)
+        self._load_checkpoint_extra(blob)
+
+    def _load_checkpoint_extra(self, blob: Dict) -> None:
+        pass

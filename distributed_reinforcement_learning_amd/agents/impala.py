@@ -1,0 +1,146 @@
+"""IMPALA agent: batched V-trace learner + single-step actor inference.
+
+Capability-parity with reference agent/impala.py (same hyperparameters, same
+loss composition agent/impala.py:63-100, same public API: train /
+get_policy_and_action / parameter_sync / save_weights / load_weights), built
+MI355X-first:
+
+* ONE batched network pass over B*T positions per train step instead of the
+  reference's 3x(T-2) per-window replicas (models/impala_actor_critic.py).
+* frames stay uint8 until the on-device normalize (ops/preprocess.py).
+* RMSProp + global-norm clip run as the fused flat-buffer optimizer
+  (ops/optim.FusedRMSProp), whose flat grad buffer is also the DP all-reduce
+  bucket (parallel/dist.py).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from distributed_reinforcement_learning_amd.agents.base import AgentBase, clip_rewards
+from distributed_reinforcement_learning_amd.algorithms import vtrace
+from distributed_reinforcement_learning_amd.models import ImpalaActorCritic
+from distributed_reinforcement_learning_amd.ops import FusedRMSProp
+
+
+class Agent(AgentBase):
+    def __init__(self, trajectory: int, input_shape, num_action: int,
+                 lstm_hidden_size: int, discount_factor: float,
+                 start_learning_rate: float, end_learning_rate: float,
+                 learning_frame: int, baseline_loss_coef: float,
+                 entropy_coef: float, gradient_clip_norm: float,
+                 reward_clipping: str, device: str = "cpu",
+                 compute_dtype: torch.dtype = torch.bfloat16,
+                 build_optimizer: bool = True, seed: Optional[int] = None):
+        super().__init__(device=device, compute_dtype=compute_dtype)
+        self.trajectory = trajectory
+        self.input_shape = tuple(input_shape)
+        self.num_action = num_action
+        self.lstm_hidden_size = lstm_hidden_size
+        self.discount_factor = discount_factor
+        self.start_learning_rate = start_learning_rate
+        self.end_learning_rate = end_learning_rate
+        self.learning_frame = learning_frame
+        self.baseline_loss_coef = baseline_loss_coef
+        self.entropy_coef = entropy_coef
+        self.gradient_clip_norm = gradient_clip_norm
+        self.reward_clipping = reward_clipping
+        if seed is not None:
+            torch.manual_seed(seed)
+        self.rng = np.random.default_rng(seed)
+
+        self.model = ImpalaActorCritic(self.input_shape, num_action,
+                                       lstm_hidden_size).to(self.device)
+        self.optimizer = None
+        if build_optimizer:
+            # TF RMSProp(decay=.99, momentum=0, eps=.1) + clip_by_global_norm
+            # (reference agent/impala.py:96-99)
+            self.optimizer = FusedRMSProp(
+                self.model.parameters(), lr=start_learning_rate, rho=0.99,
+                eps=0.1, clip_norm=gradient_clip_norm)
+
+    # -- learner -------------------------------------------------------------
+
+    def train(self, state, reward, action, done, behavior_policy,
+              previous_action, initial_h, initial_c):
+        """One V-trace update on a [B, T] batch of unrolls.
+
+        Arrays as stored by the queue: state uint8 [B,T,H,W,C], reward [B,T],
+        action/previous_action [B,T], done [B,T] bool,
+        behavior_policy [B,T,A], initial_h/c [B,T,lstm].
+        Returns (pi_loss, baseline_loss, entropy, learning_rate).
+        """
+        s = self.frames_to_device(state)
+        r = self.to_device(reward, torch.float32)
+        a = self.to_device(action, torch.int64)
+        d = self.to_device(done, torch.bool)
+        mu = self.to_device(behavior_policy, torch.float32)
+        pa = self.to_device(previous_action, torch.int64)
+        h0 = self.to_device(initial_h, torch.float32)
+        c0 = self.to_device(initial_c, torch.float32)
+
+        clipped_r = clip_rewards(r, self.reward_clipping)
+        discounts = (~d).float() * self.discount_factor
+
+        with self.autocast():
+            policy, value = self.model.unroll(s, pa, h0, c0)
+        policy = policy.float()
+        value = value.float()
+
+        # first/middle/last windows are views of the batched unroll
+        p_f, p_m, _ = vtrace.split_data(policy)
+        v_f, v_m, v_l = vtrace.split_data(value)
+        a_f, a_m, _ = vtrace.split_data(a)
+        r_f, r_m, _ = vtrace.split_data(clipped_r)
+        g_f, g_m, _ = vtrace.split_data(discounts)
+        mu_f, mu_m, _ = vtrace.split_data(mu)
+
+        vs, clipped_rho = vtrace.from_softmax(
+            behavior_policy_softmax=mu_f, target_policy_softmax=p_f,
+            actions=a_f, discounts=g_f, rewards=r_f, values=v_f,
+            next_values=v_m)
+        vs_plus_1, _ = vtrace.from_softmax(
+            behavior_policy_softmax=mu_m, target_policy_softmax=p_m,
+            actions=a_m, discounts=g_m, rewards=r_m, values=v_m,
+            next_values=v_l)
+
+        pg_advantage = (clipped_rho
+                        * (r_f + g_f * vs_plus_1 - v_f)).detach()
+
+        pi_loss = vtrace.compute_policy_gradient_loss(p_f, a_f, pg_advantage)
+        baseline_loss = vtrace.compute_baseline_loss(vs, v_f)
+        entropy = vtrace.compute_entropy_loss(p_f)
+        total = (pi_loss + baseline_loss * self.baseline_loss_coef
+                 + entropy * self.entropy_coef)
+
+        self.optimizer.zero_grad()
+        total.backward()
+        self.reduce_gradients()
+        lr = self.lr_at(self.global_step)
+        self.optimizer.step(lr=lr)
+        self.global_step += 1
+        self.num_env_frames += int(np.prod(np.shape(reward)))
+        return (float(pi_loss.detach()), float(baseline_loss.detach()),
+                float(entropy.detach()), lr)
+
+    # -- actor ---------------------------------------------------------------
+
+    @torch.no_grad()
+    def get_policy_and_action(self, state, previous_action, h, c
+                              ) -> Tuple[int, np.ndarray, float,
+                                         np.ndarray, np.ndarray]:
+        """Single-env inference (reference agent/impala.py:118-130).
+        state: uint8 [H,W,C]; returns (action, policy, max_prob, h', c')."""
+        s = self.frames_to_device(np.asarray(state)[None])
+        pa = self.to_device(np.asarray([previous_action]), torch.int64)
+        ht = self.to_device(np.asarray([h]), torch.float32)
+        ct = self.to_device(np.asarray([c]), torch.float32)
+        policy, _, nh, nc = self.model.single_step(s, pa, ht, ct)
+        p = policy[0].float().cpu().numpy().astype(np.float64)
+        p = p / p.sum()
+        action = int(self.rng.choice(self.num_action, p=p))
+        return action, p.astype(np.float32), float(p.max()), \
+            nh[0].float().cpu().numpy(), nc[0].float().cpu().numpy()

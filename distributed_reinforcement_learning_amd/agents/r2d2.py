@@ -1,0 +1,201 @@
+"""R2D2 agent — capability-parity with reference agent/r2d2.py: recurrent
+dueling double-DQN over stored sequences with burn-in, value-function
+rescaling (optimizer/burn_in.py:23-32), per-sequence priorities
+(|mean TD|, agent/r2d2.py:125-126,151-153), IS-weighted loss, and
+main_to_target hard sync every 2500 steps (train_r2d2.py:164-165).
+
+Burn-in redesign (BASELINE "burn_in hidden-state recompute" hot-path item):
+by default the first ``burn_in`` steps run under no_grad purely to recompute
+the hidden state (the R2D2 paper's burn-in), then the trained window unrolls
+with gradient. ``burn_in_gradient=True`` reproduces the reference's exact
+behavior (gradient flows through the burn-in steps; the loss is still sliced
+at burn_in) for parity checks.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from distributed_reinforcement_learning_amd.agents.base import AgentBase, clip_rewards
+from distributed_reinforcement_learning_amd.algorithms import burn_in as rescale
+from distributed_reinforcement_learning_amd.algorithms import dqn
+from distributed_reinforcement_learning_amd.models import R2D2LstmQ
+from distributed_reinforcement_learning_amd.ops import FusedAdam
+
+
+class Agent(AgentBase):
+    def __init__(self, seq_len: int, burn_in: int, input_shape,
+                 num_action: int, lstm_size: int, discount_factor: float,
+                 start_learning_rate: float, end_learning_rate: float,
+                 learning_frame: int, gradient_clip_norm: float,
+                 reward_clipping: str = "abs_one", device: str = "cpu",
+                 compute_dtype: torch.dtype = torch.bfloat16,
+                 build_optimizer: bool = True, seed: Optional[int] = None,
+                 burn_in_gradient: bool = False):
+        super().__init__(device=device, compute_dtype=compute_dtype)
+        self.seq_len = seq_len
+        self.burn_in = burn_in
+        self.input_shape = tuple(input_shape)
+        self.num_action = num_action
+        self.lstm_size = lstm_size
+        self.discount_factor = discount_factor
+        self.start_learning_rate = start_learning_rate
+        self.end_learning_rate = end_learning_rate
+        self.learning_frame = learning_frame
+        self.gradient_clip_norm = gradient_clip_norm
+        self.reward_clipping = reward_clipping
+        self.burn_in_gradient = burn_in_gradient
+        if seed is not None:
+            torch.manual_seed(seed)
+        self.rng = np.random.default_rng(seed)
+
+        self.model = R2D2LstmQ(self.input_shape, num_action,
+                               lstm_size).to(self.device)
+        self.target_model = R2D2LstmQ(self.input_shape, num_action,
+                                      lstm_size).to(self.device)
+        self.target_model.load_state_dict(self.model.state_dict())
+        for p in self.target_model.parameters():
+            p.requires_grad_(False)
+        self.optimizer = None
+        if build_optimizer:
+            # reference agent/r2d2.py:91: Adam(1e-4), no explicit clip — we
+            # keep the constructor's clip_norm wired (pass None to disable)
+            self.optimizer = FusedAdam(self.model.parameters(),
+                                       lr=start_learning_rate,
+                                       clip_norm=gradient_clip_norm)
+
+    def main_to_target(self) -> None:
+        self.target_model.load_state_dict(self.model.state_dict())
+
+    def _checkpoint_extra(self):
+        return {"target_model": self.target_model.state_dict()}
+
+    def _load_checkpoint_extra(self, blob):
+        if "target_model" in blob:
+            self.target_model.load_state_dict(blob["target_model"])
+
+    # -- core sequence evaluation -------------------------------------------
+
+    def _unroll_q(self, model, s, pa, h0, c0, d, with_grad: bool):
+        """Unroll ``model`` over the full sequence; optionally run the first
+        burn_in steps detached (hidden-state recompute only)."""
+        if with_grad and not self.burn_in_gradient and self.burn_in > 0:
+            b = self.burn_in
+            with torch.no_grad():
+                feat = model.features(
+                    s[:, :b].reshape(-1, *s.shape[2:]),
+                    pa[:, :b].reshape(-1)).reshape(s.shape[0], b, -1)
+                h, c = h0, c0
+                q_burn = []
+                for i in range(b):
+                    h, c = model.lstm(feat[:, i], h, c)
+                    q_burn.append(model._head(h))
+                    keep = (~d[:, i]).to(h.dtype).unsqueeze(1)
+                    h, c = h * keep, c * keep
+            q_rest = model.unroll_sequence(s[:, b:], pa[:, b:], h.detach(),
+                                           c.detach(), d[:, b:])
+            return torch.cat([torch.stack(q_burn, 1), q_rest], dim=1)
+        ctx = torch.enable_grad() if with_grad else torch.no_grad()
+        with ctx:
+            return model.unroll_sequence(s, pa, h0, c0, d)
+
+    def _sequence_losses(self, state, previous_action, action, h0, c0,
+                         reward, done, with_grad: bool):
+        """Returns (per-sequence unweighted loss [B], target_value,
+        state_action_value) over the post-burn-in window — the math of
+        reference agent/r2d2.py:62-93."""
+        s = self.frames_to_device(state)
+        pa = self.to_device(previous_action, torch.int64)
+        a = self.to_device(action, torch.int64)
+        r = self.to_device(reward, torch.float32)
+        d = self.to_device(done, torch.bool)
+        h = self.to_device(h0, torch.float32)
+        c = self.to_device(c0, torch.float32)
+
+        with self.autocast():
+            main_q = self._unroll_q(self.model, s, pa, h, c, d,
+                                    with_grad).float()
+            target_q = self._unroll_q(self.target_model, s, pa, h, c, d,
+                                      False).float()
+
+        clipped_r = clip_rewards(r, self.reward_clipping)
+        discounts = (~d).float() * self.discount_factor
+
+        b = self.burn_in
+        bm, bt = main_q[:, b:], target_q[:, b:]
+        br, bg, ba = clipped_r[:, b:], discounts[:, b:], a[:, b:]
+
+        state_main_q = bm[:, :-1]
+        next_main_q = bm[:, 1:]
+        next_target_q = bt[:, 1:]
+        act = ba[:, :-1]
+        rew, dis = br[:, :-1], bg[:, :-1]
+
+        sav = dqn.take_state_action_value(state_main_q, act)
+        next_action = next_main_q.argmax(dim=2)
+        nsav = dqn.take_state_action_value(next_target_q, next_action)
+        rescaled_next = rescale.inverse_value_function_rescaling(nsav)
+        rescaled_target = (rescaled_next * dis + rew).detach()
+        target_value = rescale.value_function_rescaling(rescaled_target)
+        unweighted = ((target_value - sav) ** 2).mean(dim=1)
+        return unweighted, target_value, sav
+
+    # -- public API ----------------------------------------------------------
+
+    @torch.no_grad()
+    def get_td_error(self, state, previous_action, action, h, c, reward,
+                     done) -> float:
+        """Priority for ONE freshly-arrived sequence: |mean TD| (reference
+        agent/r2d2.py:97-127). h/c are the per-step stored states; step 0's
+        is the sequence's initial state."""
+        h0 = np.asarray(h)[0]
+        c0 = np.asarray(c)[0]
+        _, target_value, sav = self._sequence_losses(
+            np.asarray(state)[None], np.asarray(previous_action)[None],
+            np.asarray(action)[None], h0[None], c0[None],
+            np.asarray(reward)[None], np.asarray(done)[None],
+            with_grad=False)
+        return float((target_value - sav).mean().abs())
+
+    def train(self, state, previous_action, action, h, c, reward, done,
+              weight) -> Tuple[float, np.ndarray]:
+        """IS-weighted batch update; h/c [B, L, H] stored per step — the
+        sequence-start state is index 0 (reference train_r2d2.py:135-136).
+        Returns (loss, per-sequence |mean TD|)."""
+        h0 = np.asarray(h)[:, 0]
+        c0 = np.asarray(c)[:, 0]
+        unweighted, target_value, sav = self._sequence_losses(
+            state, previous_action, action, h0, c0, reward, done,
+            with_grad=True)
+        w = self.to_device(weight, torch.float32)
+        loss = (unweighted * w).mean()
+
+        self.optimizer.zero_grad()
+        loss.backward()
+        self.reduce_gradients()
+        lr = self.lr_at(self.global_step)
+        self.optimizer.step(lr=lr)
+        self.global_step += 1
+        self.num_env_frames += int(np.prod(np.shape(reward)))
+        td = (target_value - sav).mean(dim=1).abs().detach().cpu().numpy()
+        return float(loss.detach()), td
+
+    @torch.no_grad()
+    def get_action(self, state, h, c, previous_action, epsilon: float):
+        """Epsilon-greedy single-step acting (reference agent/r2d2.py:166-186).
+        Returns (action, q[action], h', c')."""
+        s = self.frames_to_device(np.asarray(state)[None])
+        pa = self.to_device(np.asarray([previous_action]), torch.int64)
+        ht = self.to_device(np.asarray([h]), torch.float32)
+        ct = self.to_device(np.asarray([c]), torch.float32)
+        q, nh, nc = self.model.single_step(s, pa, ht, ct)
+        qv = q[0].float().cpu().numpy()
+        if self.rng.random() > epsilon:
+            action = int(qv.argmax())
+        else:
+            action = int(self.rng.integers(self.num_action))
+        return action, float(qv[action]), nh[0].float().cpu().numpy(), \
+            nc[0].float().cpu().numpy()

@@ -109,13 +109,12 @@ class LSTMCellTF(nn.Module):
         nn.init.xavier_uniform_(self.weight)
 
     def gates(self, x: torch.Tensor, h: torch.Tensor) -> torch.Tensor:
-        xh = torch.cat([x, h], dim=1)
+        xh = torch.cat([x, h.to(x.dtype)], dim=1)
         return xh @ self.weight + self.bias
 
     def forward(self, x: torch.Tensor, h: torch.Tensor, c: torch.Tensor):
         g = self.gates(x, h)
-        i, j, f, o = g.chunk(4, dim=1)
-        new_c = torch.sigmoid(f + self.forget_bias) * c \
-            + torch.sigmoid(i) * torch.tanh(j)
-        new_h = torch.sigmoid(o) * torch.tanh(new_c)
-        return new_h, new_c
+        # elementwise tail: fused HIP kernel on gfx950 (fp32 states), eager
+        # math on CPU — ops/lstm_op.py
+        from distributed_reinforcement_learning_amd.ops import lstm_fused_step
+        return lstm_fused_step(g.float(), c.float(), self.forget_bias)

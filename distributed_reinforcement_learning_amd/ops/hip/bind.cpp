@@ -1,0 +1,215 @@
+// PyTorch bindings for the drla gfx950 kernels (native HIP — no CUDA
+// masquerade: ATen/hip API, hipLaunchKernelGGL on the current HIP stream).
+// Compiled by hipcc together with the .hip translation units (see
+// build_ext.py); loaded as distributed_reinforcement_learning_amd.ops._drla_hip.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include <tuple>
+
+#include "drla_common.h"
+
+// kernel decls (defined in the .hip TUs)
+extern "C" __global__ void drla_u8_normalize_f32(const uchar4*, float4*,
+                                                 long long);
+typedef __attribute__((ext_vector_type(4))) unsigned short ushort4v;
+extern "C" __global__ void drla_u8_normalize_bf16(const uchar4*, ushort4v*,
+                                                  long long);
+extern "C" __global__ void drla_u8_normalize_f32_tail(const unsigned char*,
+                                                      float*, long long,
+                                                      long long);
+extern "C" __global__ void drla_u8_normalize_bf16_tail(const unsigned char*,
+                                                       unsigned short*,
+                                                       long long, long long);
+extern "C" __global__ void drla_vtrace_scan(const float*, const float*,
+                                            const float*, float*, int, int);
+extern "C" __global__ void drla_lstm_tail_fwd(const float*, const float*,
+                                              float*, float*, float*, float,
+                                              long long, int);
+extern "C" __global__ void drla_lstm_tail_bwd(const float*, const float*,
+                                              const float*, const float*,
+                                              const float*, float*, float*,
+                                              long long, int);
+extern "C" __global__ void drla_sq_norm(const float*, float*, long long);
+extern "C" __global__ void drla_rmsprop_step(float*, const float*, float*,
+                                             const float*, float, float,
+                                             float, float, long long);
+extern "C" __global__ void drla_adam_step(float*, const float*, float*,
+                                          float*, const float*, float, float,
+                                          float, float, float, long long);
+
+namespace {
+
+hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void check_gpu_contig(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on the GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+torch::Tensor normalize_frames_f32(torch::Tensor frames) {
+  check_gpu_contig(frames, "frames");
+  TORCH_CHECK(frames.scalar_type() == torch::kUInt8, "frames must be uint8");
+  auto out = torch::empty_like(frames, frames.options().dtype(torch::kFloat));
+  const long long n = frames.numel();
+  const long long n4 = n / 4;
+  if (n4 > 0) {
+    hipLaunchKernelGGL(drla_u8_normalize_f32, dim3(drla_grid(n4)),
+                       dim3(DRLA_BLOCK), 0, cur_stream(),
+                       reinterpret_cast<const uchar4*>(frames.data_ptr<uint8_t>()),
+                       reinterpret_cast<float4*>(out.data_ptr<float>()), n4);
+  }
+  if (n % 4) {
+    hipLaunchKernelGGL(drla_u8_normalize_f32_tail, dim3(1), dim3(DRLA_BLOCK),
+                       0, cur_stream(), frames.data_ptr<uint8_t>(),
+                       out.data_ptr<float>(), n4 * 4, n);
+  }
+  return out;
+}
+
+torch::Tensor normalize_frames_bf16(torch::Tensor frames) {
+  check_gpu_contig(frames, "frames");
+  TORCH_CHECK(frames.scalar_type() == torch::kUInt8, "frames must be uint8");
+  auto out = torch::empty_like(frames,
+                               frames.options().dtype(torch::kBFloat16));
+  const long long n = frames.numel();
+  const long long n4 = n / 4;
+  if (n4 > 0) {
+    hipLaunchKernelGGL(drla_u8_normalize_bf16, dim3(drla_grid(n4)),
+                       dim3(DRLA_BLOCK), 0, cur_stream(),
+                       reinterpret_cast<const uchar4*>(frames.data_ptr<uint8_t>()),
+                       reinterpret_cast<ushort4v*>(out.data_ptr()), n4);
+  }
+  if (n % 4) {
+    hipLaunchKernelGGL(drla_u8_normalize_bf16_tail, dim3(1), dim3(DRLA_BLOCK),
+                       0, cur_stream(), frames.data_ptr<uint8_t>(),
+                       reinterpret_cast<unsigned short*>(out.data_ptr()),
+                       n4 * 4, n);
+  }
+  return out;
+}
+
+torch::Tensor vtrace_scan(torch::Tensor deltas, torch::Tensor discounts,
+                          torch::Tensor cs) {
+  for (auto* t : {&deltas, &discounts, &cs}) {
+    check_gpu_contig(*t, "vtrace input");
+    TORCH_CHECK(t->scalar_type() == torch::kFloat, "vtrace wants float32");
+  }
+  const int B = deltas.size(0);
+  const int T = deltas.size(1);
+  auto out = torch::empty_like(deltas);
+  hipLaunchKernelGGL(drla_vtrace_scan, dim3(drla_grid(B)), dim3(DRLA_BLOCK),
+                     0, cur_stream(), deltas.data_ptr<float>(),
+                     discounts.data_ptr<float>(), cs.data_ptr<float>(),
+                     out.data_ptr<float>(), B, T);
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_tail_fwd(
+    torch::Tensor gates, torch::Tensor c_prev, double forget_bias) {
+  check_gpu_contig(gates, "gates");
+  check_gpu_contig(c_prev, "c_prev");
+  TORCH_CHECK(gates.scalar_type() == torch::kFloat, "gates must be float32");
+  TORCH_CHECK(c_prev.scalar_type() == torch::kFloat, "c_prev must be float32");
+  const long long N = gates.size(0);
+  const int H4 = gates.size(1);
+  TORCH_CHECK(H4 % 4 == 0, "gates dim1 must be 4*H");
+  const int H = H4 / 4;
+  TORCH_CHECK(c_prev.size(1) == H, "c_prev width mismatch");
+  auto new_h = torch::empty_like(c_prev);
+  auto new_c = torch::empty_like(c_prev);
+  auto stash = torch::empty_like(gates);
+  hipLaunchKernelGGL(drla_lstm_tail_fwd, dim3(drla_grid(N * H)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     gates.data_ptr<float>(), c_prev.data_ptr<float>(),
+                     new_h.data_ptr<float>(), new_c.data_ptr<float>(),
+                     stash.data_ptr<float>(),
+                     static_cast<float>(forget_bias), N, H);
+  return {new_h, new_c, stash};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
+    torch::Tensor grad_h, torch::Tensor grad_c, torch::Tensor stash,
+    torch::Tensor c_prev, torch::Tensor new_c) {
+  for (auto* t : {&grad_h, &grad_c, &stash, &c_prev, &new_c})
+    check_gpu_contig(*t, "lstm bwd input");
+  const long long N = stash.size(0);
+  const int H = stash.size(1) / 4;
+  auto grad_gates = torch::empty_like(stash);
+  auto grad_c_prev = torch::empty_like(c_prev);
+  hipLaunchKernelGGL(drla_lstm_tail_bwd, dim3(drla_grid(N * H)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     grad_h.data_ptr<float>(), grad_c.data_ptr<float>(),
+                     stash.data_ptr<float>(), c_prev.data_ptr<float>(),
+                     new_c.data_ptr<float>(), grad_gates.data_ptr<float>(),
+                     grad_c_prev.data_ptr<float>(), N, H);
+  return {grad_gates, grad_c_prev};
+}
+
+torch::Tensor sq_norm(torch::Tensor x) {
+  check_gpu_contig(x, "x");
+  TORCH_CHECK(x.scalar_type() == torch::kFloat, "sq_norm wants float32");
+  auto out = torch::zeros({1}, x.options());
+  hipLaunchKernelGGL(drla_sq_norm, dim3(drla_grid(x.numel() / 4 + 1)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(), x.data_ptr<float>(),
+                     out.data_ptr<float>(), (long long)x.numel());
+  return out;
+}
+
+void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
+                  double clip, double lr, double rho, double eps) {
+  for (auto* t : {&p, &g, &ms}) check_gpu_contig(*t, "rmsprop tensor");
+  const long long n = p.numel();
+  torch::Tensor norm_buf;
+  if (clip > 0) {
+    norm_buf = sq_norm(g);
+  } else {
+    norm_buf = torch::zeros({1}, p.options());
+  }
+  hipLaunchKernelGGL(drla_rmsprop_step, dim3(drla_grid(n)), dim3(DRLA_BLOCK),
+                     0, cur_stream(), p.data_ptr<float>(),
+                     g.data_ptr<float>(), ms.data_ptr<float>(),
+                     norm_buf.data_ptr<float>(), static_cast<float>(clip),
+                     static_cast<float>(lr), static_cast<float>(rho),
+                     static_cast<float>(eps), n);
+}
+
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, double clip, double lr_t, double beta1,
+               double beta2, double eps) {
+  for (auto* t : {&p, &g, &m, &v}) check_gpu_contig(*t, "adam tensor");
+  const long long n = p.numel();
+  torch::Tensor norm_buf;
+  if (clip > 0) {
+    norm_buf = sq_norm(g);
+  } else {
+    norm_buf = torch::zeros({1}, p.options());
+  }
+  hipLaunchKernelGGL(drla_adam_step, dim3(drla_grid(n)), dim3(DRLA_BLOCK), 0,
+                     cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     norm_buf.data_ptr<float>(), static_cast<float>(clip),
+                     static_cast<float>(lr_t), static_cast<float>(beta1),
+                     static_cast<float>(beta2), static_cast<float>(eps), n);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("normalize_frames_f32", &normalize_frames_f32,
+        "uint8 frames -> float32/255 (K11)");
+  m.def("normalize_frames_bf16", &normalize_frames_bf16,
+        "uint8 frames -> bf16/255 (K11)");
+  m.def("vtrace_scan", &vtrace_scan, "fused V-trace reverse scan (K5)");
+  m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");
+  m.def("lstm_tail_bwd", &lstm_tail_bwd, "fused LSTM gate tail bwd (K3)");
+  m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");
+  m.def("rmsprop_step", &rmsprop_step,
+        "fused global-norm-clip + TF-RMSProp update (K12)");
+  m.def("adam_step", &adam_step,
+        "fused global-norm-clip + TF-Adam update (K12)");
+}

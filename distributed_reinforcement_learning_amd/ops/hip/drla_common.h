@@ -1,0 +1,30 @@
+// Common helpers for the drla gfx950 kernels.
+// CDNA4: wave64, 256 CUs / 8 XCDs; memory-bound kernels follow the
+// grid-stride + vectorized-access pattern (guide §6 G11/G13).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define DRLA_WAVE 64
+#define DRLA_BLOCK 256
+// cap grids at ~8 blocks/CU x 256 CUs and grid-stride the rest
+#define DRLA_MAX_BLOCKS 2048
+
+static inline int drla_grid(long long work_items, int block = DRLA_BLOCK) {
+  long long blocks = (work_items + block - 1) / block;
+  if (blocks > DRLA_MAX_BLOCKS) blocks = DRLA_MAX_BLOCKS;
+  if (blocks < 1) blocks = 1;
+  return static_cast<int>(blocks);
+}
+
+__device__ __forceinline__ float drla_sigmoid(float x) {
+  return 1.0f / (1.0f + __expf(-x));
+}
+
+// round-to-nearest-even f32 -> bf16 raw bits (inputs here are finite)
+__device__ __forceinline__ unsigned short drla_f32_to_bf16(float f) {
+  unsigned int u = __float_as_uint(f);
+  unsigned int rounding = 0x7FFFu + ((u >> 16) & 1u);
+  return static_cast<unsigned short>((u + rounding) >> 16);
+}

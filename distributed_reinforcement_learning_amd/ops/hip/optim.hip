@@ -1,0 +1,82 @@
+// K12: fused optimizer over ONE flat parameter buffer.
+//
+// Replaces tf.clip_by_global_norm + RMSProp/Adam apply (reference
+// agent/impala.py:96-100) with two launches per step:
+//   1. drla_sq_norm        — grid-stride float4 squared-norm reduction
+//                            (wave shuffle -> LDS -> one atomic per block)
+//   2. drla_{rmsprop,adam} — clip-scale folded into the update, one fused
+//                            HBM pass over params+grads+state.
+// The clip scale is computed on-device from the norm result (no host sync):
+// scale = clip / max(norm, clip), passed as a 1-element tensor.
+
+#include "drla_common.h"
+
+extern "C" __global__ void drla_sq_norm(
+    const float* __restrict__ x, float* __restrict__ out, long long n) {
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  float acc = 0.0f;
+  const long long n4 = n / 4;
+  const float4* x4 = reinterpret_cast<const float4*>(x);
+  for (long long k = i; k < n4; k += stride) {
+    float4 v = x4[k];
+    acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+  }
+  // scalar tail handled by the first lanes
+  for (long long k = n4 * 4 + i; k < n; k += stride) acc += x[k] * x[k];
+  // wave64 reduce
+  for (int off = DRLA_WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, DRLA_WAVE);
+  __shared__ float lds[DRLA_BLOCK / DRLA_WAVE];
+  const int wave = threadIdx.x / DRLA_WAVE;
+  const int lane = threadIdx.x % DRLA_WAVE;
+  if (lane == 0) lds[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.0f;
+    for (int w = 0; w < blockDim.x / DRLA_WAVE; ++w) s += lds[w];
+    atomicAdd(out, s);
+  }
+}
+
+// scale_buf: 1-element squared-norm; computes the tf.clip_by_global_norm
+// factor in-kernel so no host round trip happens.
+__device__ __forceinline__ float drla_clip_scale(const float* sq_norm_buf,
+                                                 float clip) {
+  if (clip <= 0.0f) return 1.0f;
+  const float norm = sqrtf(*sq_norm_buf);
+  return norm > clip ? clip / norm : 1.0f;
+}
+
+extern "C" __global__ void drla_rmsprop_step(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ ms, const float* __restrict__ sq_norm_buf,
+    float clip, float lr, float rho, float eps, long long n) {
+  const float scale = drla_clip_scale(sq_norm_buf, clip);
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < n; i += stride) {
+    const float gc = g[i] * scale;
+    const float m = rho * ms[i] + (1.0f - rho) * gc * gc;
+    ms[i] = m;
+    p[i] -= lr * gc * rsqrtf(m + eps);
+  }
+}
+
+extern "C" __global__ void drla_adam_step(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ m, float* __restrict__ v,
+    const float* __restrict__ sq_norm_buf, float clip, float lr_t,
+    float beta1, float beta2, float eps, long long n) {
+  const float scale = drla_clip_scale(sq_norm_buf, clip);
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < n; i += stride) {
+    const float gc = g[i] * scale;
+    const float mi = beta1 * m[i] + (1.0f - beta1) * gc;
+    const float vi = beta2 * v[i] + (1.0f - beta2) * gc * gc;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] -= lr_t * mi / (sqrtf(vi) + eps);
+  }
+}

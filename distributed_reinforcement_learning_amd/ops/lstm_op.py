@@ -22,7 +22,9 @@ class _FusedLstmTail(torch.autograd.Function):
     def forward(ctx, gates: torch.Tensor, c_prev: torch.Tensor,
                 forget_bias: float):
         ext = _ops.require_ext()
-        new_h, new_c, stash = ext.lstm_tail_fwd(gates, c_prev, forget_bias)
+        new_h, new_c, stash = ext.lstm_tail_fwd(gates.contiguous(),
+                                                c_prev.contiguous(),
+                                                forget_bias)
         ctx.save_for_backward(stash, c_prev, new_c)
         return new_h, new_c
 
@@ -31,7 +33,8 @@ class _FusedLstmTail(torch.autograd.Function):
         stash, c_prev, new_c = ctx.saved_tensors
         ext = _ops.require_ext()
         grad_gates, grad_c_prev = ext.lstm_tail_bwd(
-            grad_h.contiguous(), grad_c.contiguous(), stash, c_prev, new_c)
+            grad_h.contiguous(), grad_c.contiguous(), stash,
+            c_prev.contiguous(), new_c)
         return grad_gates, grad_c_prev, None
 
 

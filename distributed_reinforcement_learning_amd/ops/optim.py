@@ -63,9 +63,9 @@ class _FlatOptimizerBase:
         self.flat_grads.zero_()
 
     def grad_global_norm(self) -> torch.Tensor:
-        if self.flat_grads.is_cuda and _ops.available():
+        if self.flat_grads.is_cuda:
             ext = _ops.require_ext()
-            return ext.sq_norm(self.flat_grads).sqrt()
+            return ext.sq_norm(self.flat_grads).sqrt()[0]
         return self.flat_grads.norm()
 
     def _clip_scale(self) -> torch.Tensor:
@@ -109,13 +109,13 @@ class FusedRMSProp(_FlatOptimizerBase):
     @torch.no_grad()
     def step(self, lr: Optional[float] = None) -> None:
         lr = self.lr if lr is None else lr
-        scale = self._clip_scale()
-        if self.flat_params.is_cuda and _ops.available():
+        if self.flat_params.is_cuda:
             ext = _ops.require_ext()
             ext.rmsprop_step(self.flat_params, self.flat_grads, self.ms,
-                             scale, float(lr), self.rho, self.eps)
+                             float(self.clip_norm or -1.0), float(lr),
+                             self.rho, self.eps)
         else:
-            g = self.flat_grads * scale
+            g = self.flat_grads * self._clip_scale()
             self.ms.mul_(self.rho).addcmul_(g, g, value=1 - self.rho)
             self.flat_params.addcdiv_(g, (self.ms + self.eps).sqrt(),
                                       value=-lr)
@@ -142,15 +142,14 @@ class FusedAdam(_FlatOptimizerBase):
         lr = self.lr if lr is None else lr
         self.step_count += 1
         t = self.step_count
-        scale = self._clip_scale()
         lr_t = lr * (1 - self.beta2 ** t) ** 0.5 / (1 - self.beta1 ** t)
-        if self.flat_params.is_cuda and _ops.available():
+        if self.flat_params.is_cuda:
             ext = _ops.require_ext()
             ext.adam_step(self.flat_params, self.flat_grads, self.m, self.v,
-                          scale, float(lr_t), self.beta1, self.beta2,
-                          self.eps)
+                          float(self.clip_norm or -1.0), float(lr_t),
+                          self.beta1, self.beta2, self.eps)
         else:
-            g = self.flat_grads * scale
+            g = self.flat_grads * self._clip_scale()
             self.m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
             self.v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
             self.flat_params.addcdiv_(self.m, self.v.sqrt() + self.eps,

@@ -1,0 +1,162 @@
+"""A3C trainer loops (reference train_a3c.py).
+
+Learner (reference :83-110): pull one unrolled trajectory at a time and train
+on its T transitions (the reference's capacity-1 queue is a rendezvous; here
+the ring decouples actors from the learner but the per-trajectory update is
+kept). Actor (reference :112-183): T-step unroll, weight pull per unroll.
+
+Works for both the Atari conv config ("a3c") and the CartPole plumbing config
+("a3c_cartpole", vector obs — BASELINE config #1).
+"""
+
+from __future__ import annotations
+
+import time
+
+import numpy as np
+
+from distributed_reinforcement_learning_amd.agents import a3c as a3c_agent
+from distributed_reinforcement_learning_amd.envs import make_env
+from distributed_reinforcement_learning_amd.parallel.queue import (
+    TrajectoryQueue, queue_schema_for,
+)
+from distributed_reinforcement_learning_amd.parallel.weights import (
+    WeightPublisher, WeightSubscriber,
+)
+from distributed_reinforcement_learning_amd.trainers import common
+from distributed_reinforcement_learning_amd.utils import (
+    StageTimer, SummaryWriter, UnrolledA3CTrajectory,
+)
+
+
+def build_agent(ctx, device: str, build_optimizer: bool, seed=None):
+    cfg = ctx.cfg
+    return a3c_agent.Agent(
+        input_shape=cfg.model_input, num_action=cfg.model_output,
+        discount_factor=cfg.discount_factor,
+        start_learning_rate=cfg.start_learning_rate,
+        end_learning_rate=cfg.end_learning_rate,
+        learning_frame=cfg.learning_frame,
+        baseline_loss_coef=cfg.baseline_loss_coef,
+        entropy_coef=cfg.entropy_coef,
+        gradient_clip_norm=cfg.gradient_clip_norm,
+        reward_clipping=cfg.reward_clipping, device=device,
+        build_optimizer=build_optimizer, seed=seed)
+
+
+def learner(ctx: common.TrainerContext, supervisor=None) -> None:
+    cfg, args = ctx.cfg, ctx.args
+    queue = TrajectoryQueue(
+        queue_schema_for("a3c", cfg), cfg.num_actors, cfg.queue_size,
+        role="learner", namespace=ctx.namespace, rank=ctx.rank,
+        world_size=ctx.world_size)
+    agent = build_agent(ctx, ctx.device, True, args.seed)
+    if args.restore:
+        agent.load_weights(args.restore)
+    from distributed_reinforcement_learning_amd.parallel.dist import broadcast_module
+    broadcast_module(agent.model)
+    agent.setup_all_reduce()
+    if ctx.rank == 0:
+        agent.weight_publisher = WeightPublisher(ctx.weights_name,
+                                                 agent.model.state_dict())
+        agent.publish_weights()
+    monitor = common.learner_supervision(ctx, queue, supervisor)
+    writer = SummaryWriter(ctx.learner_logdir())
+    timer = StageTimer()
+    train_step = 0
+    try:
+        while args.max_steps <= 0 or train_step < args.max_steps:
+            with timer.track("ingest"):
+                batch = queue.sample_batch(1)
+            t0 = time.time()
+            # one trajectory = a batch of T transitions (reference
+            # train_a3c.py:92-98)
+            with timer.track("train"):
+                pi_loss, v_loss, entropy, lr = agent.train(
+                    state=batch["state"][0],
+                    next_state=batch["next_state"][0],
+                    previous_action=batch["previous_action"][0],
+                    action=batch["action"][0], reward=batch["reward"][0],
+                    done=batch["done"][0])
+            train_step += 1
+            if ctx.rank == 0 and train_step % args.publish_every == 0:
+                agent.publish_weights()
+            ctx.maybe_checkpoint(agent)
+            if monitor is not None and train_step % 50 == 0:
+                supervisor.check()
+            if ctx.rank == 0:
+                step = agent.global_step
+                writer.add_scalar("data/pi_loss", pi_loss, step)
+                writer.add_scalar("data/value_loss", v_loss, step)
+                writer.add_scalar("data/entropy", entropy, step)
+                writer.add_scalar("data/lr", lr, step)
+                writer.add_scalar("data/time", time.time() - t0, step)
+    finally:
+        writer.close()
+        queue.close()
+        if agent.weight_publisher:
+            agent.weight_publisher.close()
+
+
+def actor(ctx: common.TrainerContext, task: int) -> None:
+    import torch
+    torch.set_num_threads(1)  # batch-1 CPU inference; also avoids
+    # the forked-child OpenMP deadlock (see trainers/common.py)
+    cfg, args = ctx.cfg, ctx.args
+    env = make_env(cfg.env[task], kind="uint8",
+                   num_actions=cfg.model_output, seed=(args.seed or 0) + task)
+    available_action = cfg.available_action[task]
+    queue = TrajectoryQueue(
+        queue_schema_for("a3c", cfg), cfg.num_actors, cfg.queue_size,
+        role="actor", namespace=ctx.namespace, actor_task=task,
+        world_size=ctx.world_size)
+    agent = build_agent(ctx, "cpu", False, (args.seed or 0) + 1000 + task)
+    agent.weight_subscriber = WeightSubscriber(ctx.weights_name,
+                                               agent.model.state_dict())
+    agent.weight_subscriber.wait_for_first()
+    writer = SummaryWriter(ctx.actor_logdir(task))
+    traj = UnrolledA3CTrajectory()
+
+    state = env.reset()
+    previous_action = 0
+    episode, score, episode_step = 0, 0.0, 0
+    prob_sum, prob_n = 0.0, 0
+    unrolls = 0
+    try:
+        while args.max_unrolls <= 0 or unrolls < args.max_unrolls:
+            traj.initialize()
+            agent.parameter_sync()
+            for _ in range(cfg.trajectory):
+                action, policy, max_prob = agent.get_policy_and_action(
+                    state, previous_action)
+                env_action = action % available_action
+                next_state, reward, done, info = env.step(env_action)
+                if info.get("life_lost"):
+                    reward, done = -1.0, True
+                score += reward
+                episode_step += 1
+                prob_sum += max_prob
+                prob_n += 1
+                traj.append(state=state, next_state=next_state,
+                            previous_action=previous_action, action=action,
+                            reward=reward, done=done)
+                state, previous_action = next_state, action
+                if done:
+                    writer.add_scalar("data/score", score, episode)
+                    writer.add_scalar("data/episode_step", episode_step,
+                                      episode)
+                    writer.add_scalar("data/max_prob",
+                                      prob_sum / max(prob_n, 1), episode)
+                    episode += 1
+                    score, episode_step, prob_sum, prob_n = 0.0, 0, 0.0, 0
+                    state = env.reset()
+                    previous_action = 0
+            queue.append_to_queue(task, **traj.stacked())
+            unrolls += 1
+    finally:
+        writer.close()
+        queue.close()
+
+
+def main(argv=None) -> None:
+    common.run("a3c", learner, actor, argv)

@@ -1,0 +1,190 @@
+"""R2D2 trainer loops (reference train_r2d2.py).
+
+Learner (reference :87-165): ingest arrived sequences, score per-sequence
+priority |mean TD| in one forward, PER sample of batch_size sequences,
+IS-weighted recurrent update with burn-in, priority refresh, main->target
+sync every 2500 steps (:164-165).
+
+Actor (reference :167-251): POMDP env (20% blanked single frames), carries
+h/c across steps, stores per-step state for the sequence, epsilon =
+1/(0.01*episode+1) (:233); no action modulo (reference quirk, :200 — R2D2
+acts directly in model action space).
+"""
+
+from __future__ import annotations
+
+import time
+
+import numpy as np
+
+from distributed_reinforcement_learning_amd.agents import r2d2 as r2d2_agent
+from distributed_reinforcement_learning_amd.envs import pomdp_uint8_env
+from distributed_reinforcement_learning_amd.parallel.queue import (
+    TrajectoryQueue, queue_schema_for,
+)
+from distributed_reinforcement_learning_amd.parallel.weights import (
+    WeightPublisher, WeightSubscriber,
+)
+from distributed_reinforcement_learning_amd.replay import Memory
+from distributed_reinforcement_learning_amd.trainers import common
+from distributed_reinforcement_learning_amd.utils import StageTimer, SummaryWriter
+from distributed_reinforcement_learning_amd.utils.trajectory import FieldTrajectory
+
+TARGET_SYNC_EVERY = 2500     # reference train_r2d2.py:164
+MEMORY_CAPACITY = 100_000    # reference train_r2d2.py:91-92
+
+
+def build_agent(ctx, device: str, build_optimizer: bool, seed=None):
+    cfg = ctx.cfg
+    return r2d2_agent.Agent(
+        seq_len=cfg.seq_len, burn_in=cfg.burn_in,
+        input_shape=cfg.model_input, num_action=cfg.model_output,
+        lstm_size=cfg.lstm_size, discount_factor=cfg.discount_factor,
+        start_learning_rate=cfg.start_learning_rate,
+        end_learning_rate=cfg.end_learning_rate,
+        learning_frame=cfg.learning_frame,
+        gradient_clip_norm=cfg.gradient_clip_norm,
+        reward_clipping=cfg.reward_clipping, device=device,
+        build_optimizer=build_optimizer, seed=seed)
+
+
+def learner(ctx: common.TrainerContext, supervisor=None) -> None:
+    cfg, args = ctx.cfg, ctx.args
+    queue = TrajectoryQueue(
+        queue_schema_for("r2d2", cfg), cfg.num_actors, cfg.queue_size,
+        role="learner", namespace=ctx.namespace, rank=ctx.rank,
+        world_size=ctx.world_size)
+    agent = build_agent(ctx, ctx.device, True, args.seed)
+    if args.restore:
+        agent.load_weights(args.restore)
+    from distributed_reinforcement_learning_amd.parallel.dist import broadcast_module
+    broadcast_module(agent.model)
+    broadcast_module(agent.target_model)
+    agent.setup_all_reduce()
+    agent.main_to_target()
+    if ctx.rank == 0:
+        agent.weight_publisher = WeightPublisher(ctx.weights_name,
+                                                 agent.model.state_dict())
+        agent.publish_weights()
+    monitor = common.learner_supervision(ctx, queue, supervisor)
+    writer = SummaryWriter(ctx.learner_logdir())
+    timer = StageTimer()
+    memory = Memory(MEMORY_CAPACITY, seed=args.seed)
+    train_step, buffer_steps = 0, 0
+    min_warm = 2 * cfg.batch_size  # reference :122
+    try:
+        while args.max_steps <= 0 or train_step < args.max_steps:
+            # one ingest per iteration (see trainers/apex.py note)
+            need_data = buffer_steps < min_warm
+            if queue.get_size() > 0 or need_data:
+                with timer.track("ingest"):
+                    u = queue.sample_batch(1)
+                    td = agent.get_td_error(
+                        u["state"][0], u["previous_action"][0],
+                        u["action"][0], u["initial_h"][0],
+                        u["initial_c"][0], u["reward"][0], u["done"][0])
+                    memory.add(td, {k: v[0] for k, v in u.items()})
+                    buffer_steps += 1
+            if buffer_steps < min_warm:
+                continue
+            t0 = time.time()
+            with timer.track("sample"):
+                batch, idxs, is_weight = memory.sample(cfg.batch_size)
+                stacked = {k: np.stack([b[k] for b in batch])
+                           for k in batch[0]}
+            with timer.track("train"):
+                loss, td_error = agent.train(
+                    state=stacked["state"],
+                    previous_action=stacked["previous_action"],
+                    action=stacked["action"], h=stacked["initial_h"],
+                    c=stacked["initial_c"], reward=stacked["reward"],
+                    done=stacked["done"], weight=is_weight)
+            with timer.track("per_update"):
+                memory.update_batch(idxs, td_error)
+            train_step += 1
+            if train_step % TARGET_SYNC_EVERY == 0:
+                agent.main_to_target()
+            if ctx.rank == 0 and train_step % args.publish_every == 0:
+                agent.publish_weights()
+            ctx.maybe_checkpoint(agent)
+            if monitor is not None and train_step % 50 == 0:
+                supervisor.check()
+            if ctx.rank == 0:
+                step = agent.global_step
+                writer.add_scalar("data/loss", loss, step)
+                writer.add_scalar("data/time", time.time() - t0, step)
+                if train_step % 50 == 0:
+                    print(f"[r2d2 learner] step={step} loss={loss:.5f} "
+                          f"{timer.report()}", flush=True)
+    finally:
+        writer.close()
+        queue.close()
+        if agent.weight_publisher:
+            agent.weight_publisher.close()
+
+
+def actor(ctx: common.TrainerContext, task: int) -> None:
+    import torch
+    torch.set_num_threads(1)  # batch-1 CPU inference; also avoids
+    # the forked-child OpenMP deadlock (see trainers/common.py)
+    cfg, args = ctx.cfg, ctx.args
+    env = pomdp_uint8_env(cfg.env[task], num_actions=cfg.model_output,
+                          seed=(args.seed or 0) + task)
+    queue = TrajectoryQueue(
+        queue_schema_for("r2d2", cfg), cfg.num_actors, cfg.queue_size,
+        role="actor", namespace=ctx.namespace, actor_task=task,
+        world_size=ctx.world_size)
+    agent = build_agent(ctx, "cpu", False, (args.seed or 0) + 1000 + task)
+    agent.weight_subscriber = WeightSubscriber(ctx.weights_name,
+                                               agent.model.state_dict())
+    agent.weight_subscriber.wait_for_first()
+    writer = SummaryWriter(ctx.actor_logdir(task))
+    seq = FieldTrajectory(["state", "previous_action", "action", "reward",
+                           "done", "initial_h", "initial_c"])
+
+    state = env.reset()
+    previous_action = 0
+    h = np.zeros(cfg.lstm_size, dtype=np.float32)
+    c = np.zeros(cfg.lstm_size, dtype=np.float32)
+    episode, score, episode_step = 0, 0.0, 0
+    q_sum = 0.0
+    enqueued = 0
+    try:
+        while args.max_unrolls <= 0 or enqueued < args.max_unrolls:
+            epsilon = 1.0 / (0.01 * episode + 1)  # reference :233
+            action, q_a, nh, nc = agent.get_action(state, h, c,
+                                                   previous_action, epsilon)
+            next_state, reward, done, info = env.step(action)
+            if info.get("life_lost"):
+                reward, done = -1.0, True
+            score += reward
+            episode_step += 1
+            q_sum += q_a
+            seq.append(state=state, previous_action=previous_action,
+                       action=action, reward=reward, done=done,
+                       initial_h=h, initial_c=c)
+            state, previous_action, h, c = next_state, action, nh, nc
+            if len(seq) == cfg.seq_len:
+                queue.append_to_queue(task, **seq.stacked())
+                seq.initialize()
+                enqueued += 1
+                agent.parameter_sync()
+            if done:
+                writer.add_scalar("data/score", score, episode)
+                writer.add_scalar("data/episode_step", episode_step, episode)
+                writer.add_scalar("data/epsilon", epsilon, episode)
+                writer.add_scalar("data/avg_q",
+                                  q_sum / max(episode_step, 1), episode)
+                episode += 1
+                score, episode_step, q_sum = 0.0, 0, 0.0
+                state = env.reset()
+                previous_action = 0
+                h = np.zeros(cfg.lstm_size, dtype=np.float32)
+                c = np.zeros(cfg.lstm_size, dtype=np.float32)
+    finally:
+        writer.close()
+        queue.close()
+
+
+def main(argv=None) -> None:
+    common.run("r2d2", learner, actor, argv)

@@ -1,0 +1,193 @@
+import numpy as np
+import pytest
+import torch
+
+from distributed_reinforcement_learning_amd.agents import (
+    a3c as a3c_agent, apex as apex_agent, impala as impala_agent,
+    r2d2 as r2d2_agent,
+)
+from distributed_reinforcement_learning_amd.agents.base import (
+    clip_rewards, polynomial_decay,
+)
+
+
+def _impala(T=6, A=5, lstm=16):
+    return impala_agent.Agent(
+        trajectory=T, input_shape=[84, 84, 4], num_action=A,
+        lstm_hidden_size=lstm, discount_factor=0.99,
+        start_learning_rate=1e-3, end_learning_rate=0.0,
+        learning_frame=10 ** 9, baseline_loss_coef=1.0, entropy_coef=0.05,
+        gradient_clip_norm=40.0, reward_clipping="abs_one", seed=0)
+
+
+def _impala_batch(B=2, T=6, A=5, lstm=16, seed=0):
+    rng = np.random.default_rng(seed)
+    return dict(
+        state=rng.integers(0, 255, (B, T, 84, 84, 4), dtype=np.uint8),
+        reward=rng.normal(size=(B, T)).astype(np.float32),
+        action=rng.integers(0, A, (B, T)).astype(np.int32),
+        done=np.zeros((B, T), dtype=bool),
+        behavior_policy=np.full((B, T, A), 1 / A, dtype=np.float32),
+        previous_action=rng.integers(0, A, (B, T)).astype(np.int32),
+        initial_h=rng.normal(size=(B, T, lstm)).astype(np.float32) * 0.1,
+        initial_c=rng.normal(size=(B, T, lstm)).astype(np.float32) * 0.1,
+    )
+
+
+def test_polynomial_decay():
+    assert polynomial_decay(1.0, 0.0, 0, 100) == 1.0
+    assert polynomial_decay(1.0, 0.0, 50, 100) == pytest.approx(0.5)
+    assert polynomial_decay(1.0, 0.0, 200, 100) == 0.0
+    assert polynomial_decay(6e-4, 0.0, 10 ** 8, 10 ** 9) == pytest.approx(
+        6e-4 * 0.9)
+
+
+def test_clip_rewards_modes():
+    r = torch.tensor([-10.0, -0.5, 0.5, 10.0])
+    assert clip_rewards(r, "abs_one").tolist() == [-1.0, -0.5, 0.5, 1.0]
+    soft = clip_rewards(r, "soft_asymmetric")
+    assert soft[0] == pytest.approx(0.3 * np.tanh(-2.0) * 5, abs=1e-5)
+    assert soft[3] == pytest.approx(np.tanh(2.0) * 5, abs=1e-5)
+    assert (clip_rewards(r, "none") == r).all()
+
+
+def test_impala_train_step_updates_weights():
+    agent = _impala()
+    before = agent.optimizer.flat_params.clone()
+    pi, bl, ent, lr = agent.train(**_impala_batch())
+    assert np.isfinite([pi, bl, ent, lr]).all()
+    assert agent.global_step == 1
+    assert agent.num_env_frames == 12
+    assert not torch.equal(before, agent.optimizer.flat_params)
+    assert lr == pytest.approx(1e-3, rel=1e-5)
+
+
+def test_impala_actor_inference():
+    agent = _impala()
+    state = np.random.default_rng(0).integers(
+        0, 255, (84, 84, 4), dtype=np.uint8)
+    h = np.zeros(16, np.float32)
+    c = np.zeros(16, np.float32)
+    action, policy, max_prob, nh, nc = agent.get_policy_and_action(
+        state, 0, h, c)
+    assert 0 <= action < 5
+    assert policy.shape == (5,)
+    assert np.isclose(policy.sum(), 1.0, atol=1e-5)
+    assert nh.shape == (16,)
+
+
+def test_impala_checkpoint_roundtrip(tmp_path):
+    agent = _impala()
+    agent.train(**_impala_batch())
+    path = str(tmp_path / "ck.pt")
+    agent.save_weights(path)
+    agent2 = _impala()
+    agent2.load_weights(path)
+    assert agent2.global_step == 1
+    for a, b in zip(agent.model.parameters(), agent2.model.parameters()):
+        assert torch.equal(a, b)
+    assert torch.equal(agent.optimizer.ms, agent2.optimizer.ms)
+
+
+def test_a3c_train_and_act():
+    agent = a3c_agent.Agent(
+        input_shape=[4], num_action=2, discount_factor=0.99,
+        start_learning_rate=1e-3, end_learning_rate=0.0,
+        learning_frame=10 ** 9, baseline_loss_coef=1.0, entropy_coef=0.01,
+        gradient_clip_norm=40.0, reward_clipping="none", seed=0)
+    rng = np.random.default_rng(0)
+    N = 8
+    out = agent.train(
+        state=rng.normal(size=(N, 4)).astype(np.float32),
+        next_state=rng.normal(size=(N, 4)).astype(np.float32),
+        previous_action=rng.integers(0, 2, N), action=rng.integers(0, 2, N),
+        reward=rng.normal(size=N).astype(np.float32),
+        done=np.zeros(N, dtype=bool))
+    assert np.isfinite(out).all()
+    action, policy, mp = agent.get_policy_and_action(
+        np.zeros(4, np.float32), 0)
+    assert 0 <= action < 2 and policy.shape == (2,)
+
+
+def test_apex_td_error_and_train():
+    agent = apex_agent.Agent(
+        input_shape=[84, 84, 4], num_action=4, discount_factor=0.99,
+        gradient_clip_norm=40.0, reward_clipping="abs_one",
+        start_learning_rate=1e-4, end_learning_rate=0.0,
+        learning_frame=10 ** 9, seed=0)
+    rng = np.random.default_rng(1)
+    N = 4
+    args = (rng.integers(0, 255, (N, 84, 84, 4), dtype=np.uint8),
+            rng.integers(0, 255, (N, 84, 84, 4), dtype=np.uint8),
+            rng.integers(0, 4, N), rng.integers(0, 4, N),
+            rng.normal(size=N).astype(np.float32),
+            np.zeros(N, dtype=bool))
+    td = agent.get_td_error(*args)
+    assert td.shape == (N,) and (td >= 0).all()
+    loss, td2 = agent.distributed_train(*args, np.ones(N, np.float32))
+    assert np.isfinite(loss) and td2.shape == (N,)
+    # target sync makes target == main
+    agent.target_to_main()
+    for a, b in zip(agent.model.parameters(),
+                    agent.target_model.parameters()):
+        assert torch.equal(a, b)
+
+
+def test_apex_epsilon_greedy_extremes():
+    agent = apex_agent.Agent(
+        input_shape=[4], num_action=3, discount_factor=0.99,
+        gradient_clip_norm=40.0, reward_clipping="abs_one",
+        start_learning_rate=1e-4, end_learning_rate=0.0,
+        learning_frame=10 ** 9, seed=0)
+    s = np.zeros(4, np.float32)
+    # epsilon=0: always argmax -> deterministic across calls
+    acts = {agent.get_policy_and_action(s, 0, 0.0)[0] for _ in range(5)}
+    assert len(acts) == 1
+
+
+def _r2d2(burn_in_gradient=False):
+    return r2d2_agent.Agent(
+        seq_len=6, burn_in=2, input_shape=[84, 84, 1], num_action=4,
+        lstm_size=8, discount_factor=0.997, start_learning_rate=1e-4,
+        end_learning_rate=0.0, learning_frame=10 ** 9,
+        gradient_clip_norm=40.0, seed=0,
+        burn_in_gradient=burn_in_gradient)
+
+
+def _r2d2_batch(B=2, L=6, lstm=8, seed=0):
+    rng = np.random.default_rng(seed)
+    return dict(
+        state=rng.integers(0, 255, (B, L, 84, 84, 1), dtype=np.uint8),
+        previous_action=rng.integers(0, 4, (B, L)).astype(np.int32),
+        action=rng.integers(0, 4, (B, L)).astype(np.int32),
+        h=rng.normal(size=(B, L, lstm)).astype(np.float32) * 0.1,
+        c=rng.normal(size=(B, L, lstm)).astype(np.float32) * 0.1,
+        reward=rng.normal(size=(B, L)).astype(np.float32),
+        done=np.zeros((B, L), dtype=bool),
+    )
+
+
+@pytest.mark.parametrize("burn_grad", [False, True])
+def test_r2d2_train(burn_grad):
+    agent = _r2d2(burn_in_gradient=burn_grad)
+    batch = _r2d2_batch()
+    loss, td = agent.train(**batch, weight=np.ones(2, np.float32))
+    assert np.isfinite(loss)
+    assert td.shape == (2,) and (td >= 0).all()
+
+
+def test_r2d2_td_error_single_sequence():
+    agent = _r2d2()
+    b = _r2d2_batch(B=1)
+    td = agent.get_td_error(
+        b["state"][0], b["previous_action"][0], b["action"][0],
+        b["h"][0], b["c"][0], b["reward"][0], b["done"][0])
+    assert np.isfinite(td) and td >= 0
+
+
+def test_r2d2_act():
+    agent = _r2d2()
+    s = np.zeros((84, 84, 1), np.uint8)
+    a, q, h, c = agent.get_action(s, np.zeros(8, np.float32),
+                                  np.zeros(8, np.float32), 0, epsilon=0.0)
+    assert 0 <= a < 4 and h.shape == (8,)

@@ -1,0 +1,100 @@
+"""End-to-end spawn-mode runs of all four trainers on tiny configs:
+learner + real actor processes + shm transport + weight sync + TB logging +
+checkpointing, on CPU."""
+
+import json
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pytest
+
+
+def _tiny_config(tmp_path):
+    def block(env, n_actors, avail, **kw):
+        d = {"server_ip": "127.0.0.1", "server_port": 8123,
+             "num_actors": n_actors, "env": [env] * n_actors,
+             "available_action": [avail] * n_actors,
+             "start_learning_rate": 1e-3, "end_learning_rate": 0.0,
+             "learning_frame": 10 ** 9, "gradient_clip_norm": 40.0,
+             "baseline_loss_coef": 1.0, "entropy_coef": 0.01,
+             "discount_factor": 0.99, "reward_clipping": "abs_one"}
+        d.update(kw)
+        return d
+
+    cfg = {
+        "a3c_cartpole": block("CartPole-v0", 2, 2, model_input=[4],
+                              model_output=2, queue_size=16, batch_size=4,
+                              trajectory=8, reward_clipping="none"),
+        "impala": block("SyntheticAtari", 2, 4, model_input=[84, 84, 4],
+                        model_output=4, queue_size=16, batch_size=2,
+                        trajectory=6, lstm_size=8),
+        "apex": block("SyntheticAtari", 1, 4, model_input=[84, 84, 4],
+                      model_output=4, queue_size=8, batch_size=4,
+                      trajectory=4),
+        "r2d2": block("SyntheticAtari", 1, 4, model_input=[84, 84, 1],
+                      model_output=4, queue_size=8, batch_size=2,
+                      seq_len=5, burn_in=2, lstm_size=8),
+    }
+    p = tmp_path / "config.json"
+    p.write_text(json.dumps(cfg))
+    return str(p)
+
+
+def _run_main(algo, block, cfg_path, max_steps, q):
+    try:
+        os.environ.pop("WORLD_SIZE", None)
+        import importlib
+        mod = importlib.import_module(
+            f"distributed_reinforcement_learning_amd.trainers.{algo}")
+        mod.main(["--spawn", "--config", cfg_path,
+                  "--algorithm_block", block,
+                  "--max_steps", str(max_steps), "--seed", "0",
+                  "--checkpoint_every", str(max_steps),
+                  "--publish_every", "1"])
+        q.put("ok")
+    except Exception as e:  # pragma: no cover
+        import traceback
+        q.put("fail: " + traceback.format_exc())
+
+
+@pytest.mark.parametrize("algo,block,max_steps,timeout", [
+    ("a3c", "a3c_cartpole", 4, 120),
+    ("impala", "impala", 3, 240),
+    ("apex", "apex", 3, 300),
+    ("r2d2", "r2d2", 3, 300),
+])
+def test_trainer_end_to_end(tmp_path, algo, block, max_steps, timeout):
+    cfg_path = _tiny_config(tmp_path)
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        q = mp.Queue()
+        p = mp.Process(target=_run_main,
+                       args=(algo, block, cfg_path, max_steps, q))
+        p.start()
+        try:
+            result = q.get(timeout=timeout)
+        finally:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+                p.join()
+        assert result == "ok", result
+        # learner TB run dir + scalars exist (reference layout: runs/learner)
+        scalars = tmp_path / "runs" / "learner" / "scalars.jsonl"
+        assert scalars.exists()
+        tags = {json.loads(l)["tag"] for l in scalars.read_text().splitlines()}
+        expected = {"data/time"} | (
+            {"data/pi_loss", "data/value_loss", "data/entropy", "data/lr"}
+            if algo in ("a3c", "impala") else {"data/loss"})
+        assert expected <= tags, f"missing scalars: {expected - tags}"
+        # checkpoint written with the standard layout
+        ck = tmp_path / "checkpoints" / f"{block}.pt"
+        assert ck.exists()
+        import torch
+        blob = torch.load(str(ck), map_location="cpu", weights_only=False)
+        assert {"model", "optimizer", "global_step"} <= set(blob)
+        assert blob["global_step"] == max_steps
+    finally:
+        os.chdir(cwd)
