@@ -10,8 +10,8 @@ respawn is cheap and sound).
 
 from __future__ import annotations
 
+import multiprocessing as mp
 import time
-from multiprocessing import Process
 from typing import Callable, Dict, List, Optional
 
 
@@ -30,11 +30,15 @@ class ActorSupervisor:
     """Spawn + watch actor processes; restart any that die or go stale."""
 
     def __init__(self, actor_fn: Callable[[int], None], actor_ids: List[int],
-                 monitor: Optional[HeartbeatMonitor] = None):
+                 monitor: Optional[HeartbeatMonitor] = None,
+                 start_method: str = "spawn"):
+        # spawn by default: a forked child of a learner that already ran
+        # parallel torch ops deadlocks in the inherited OpenMP pool
         self.actor_fn = actor_fn
         self.actor_ids = list(actor_ids)
         self.monitor = monitor
-        self.procs: Dict[int, Process] = {}
+        self._ctx = mp.get_context(start_method)
+        self.procs: Dict[int, mp.process.BaseProcess] = {}
         self.restarts: Dict[int, int] = {i: 0 for i in self.actor_ids}
 
     def start(self) -> None:
@@ -42,7 +46,8 @@ class ActorSupervisor:
             self._spawn(i)
 
     def _spawn(self, actor_id: int) -> None:
-        p = Process(target=self.actor_fn, args=(actor_id,), daemon=True)
+        p = self._ctx.Process(target=self.actor_fn, args=(actor_id,),
+                              daemon=True)
         p.start()
         self.procs[actor_id] = p
 

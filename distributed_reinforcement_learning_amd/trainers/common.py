@@ -97,30 +97,34 @@ class TrainerContext:
             agent.save_weights(self.checkpoint_path())
 
 
+def _actor_process_entry(algorithm: str, actor_fn: Callable, args,
+                         task: int) -> None:
+    """Module-level actor entry (picklable for the spawn start method).
+
+    Runs outside the RCCL process group (scrub torchrun env) and
+    single-threaded: batch-1 CPU inference wants one thread, and 20+ actors
+    would oversubscribe the host otherwise.
+    """
+    for var in ("WORLD_SIZE", "RANK", "LOCAL_RANK"):
+        os.environ.pop(var, None)
+    torch.set_num_threads(1)
+    actor_args = argparse.Namespace(**vars(args))
+    actor_args.spawn = False
+    actor_args.job_name = "actor"
+    actor_args.task = task
+    actor_ctx = TrainerContext(algorithm, actor_args)
+    actor_fn(actor_ctx, task)
+
+
 def run(algorithm: str, learner_fn: Callable, actor_fn: Callable,
         argv=None) -> None:
     args = build_parser(algorithm).parse_args(argv)
     ctx = TrainerContext(algorithm, args)
     if args.spawn and ctx.rank == 0:
-        def _actor_entry(task: int) -> None:
-            # each actor runs in its own process, outside the RCCL process
-            # group: scrub torchrun env the child inherited
-            for var in ("WORLD_SIZE", "RANK", "LOCAL_RANK"):
-                os.environ.pop(var, None)
-            # forked children must not re-enter the parent's OpenMP pool
-            # (deadlock on the first at::parallel_for); batch-1 actor
-            # inference wants one thread anyway — 20+ actors would
-            # oversubscribe the host otherwise.
-            torch.set_num_threads(1)
-            actor_args = argparse.Namespace(**vars(args))
-            actor_args.spawn = False
-            actor_args.job_name = "actor"
-            actor_args.task = task
-            actor_ctx = TrainerContext(algorithm, actor_args)
-            actor_fn(actor_ctx, task)
-
-        sup = ActorSupervisor(_actor_entry,
-                              list(range(ctx.cfg.num_actors)))
+        import functools
+        entry = functools.partial(_actor_process_entry, algorithm, actor_fn,
+                                  args)
+        sup = ActorSupervisor(entry, list(range(ctx.cfg.num_actors)))
         try:
             learner_fn(ctx, supervisor=sup)
         finally:

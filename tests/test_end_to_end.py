@@ -69,9 +69,13 @@ def test_trainer_end_to_end(tmp_path, algo, block, max_steps, timeout):
     cwd = os.getcwd()
     os.chdir(tmp_path)
     try:
-        q = mp.Queue()
-        p = mp.Process(target=_run_main,
-                       args=(algo, block, cfg_path, max_steps, q))
+        # spawn (not fork): the pytest process has already run parallel torch
+        # ops, and a forked learner would deadlock in the inherited OpenMP
+        # pool on its first at::parallel_for.
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        p = ctx.Process(target=_run_main,
+                        args=(algo, block, cfg_path, max_steps, q))
         p.start()
         try:
             result = q.get(timeout=timeout)
