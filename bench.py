@@ -46,6 +46,8 @@ def main(argv=None) -> None:
     p.add_argument("--lstm", type=int, default=256)
     p.add_argument("--pool", type=int, default=8,
                    help="distinct synthetic batches cycled through")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph capture (eager launches)")
     args = p.parse_args(argv)
 
     from distributed_reinforcement_learning_amd.agents import impala
@@ -86,8 +88,17 @@ def main(argv=None) -> None:
             initial_c=(rng.normal(size=(B, T, H)) * 0.1).astype(np.float32),
         ))
 
+    graphed = None
+    if have_gpu and not args.no_graph:
+        from distributed_reinforcement_learning_amd.runtime import (
+            GraphedImpalaStep,
+        )
+        graphed = GraphedImpalaStep(agent, B)
+
     def step(i: int):
         b = pool[i % len(pool)]
+        if graphed is not None:
+            return graphed.step(b)
         return agent.train(
             state=b["state"], reward=b["reward"], action=b["action"],
             done=b["done"], behavior_policy=b["behavior_policy"],

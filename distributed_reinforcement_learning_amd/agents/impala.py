@@ -64,24 +64,13 @@ class Agent(AgentBase):
 
     # -- learner -------------------------------------------------------------
 
-    def train(self, state, reward, action, done, behavior_policy,
-              previous_action, initial_h, initial_c):
-        """One V-trace update on a [B, T] batch of unrolls.
+    def compute_losses(self, s, r, a, d, mu, pa, h0, c0):
+        """Pure loss computation over device tensors (shared by the eager
+        train() path and the hipGraph-captured step — runtime/graphed.py).
 
-        Arrays as stored by the queue: state uint8 [B,T,H,W,C], reward [B,T],
-        action/previous_action [B,T], done [B,T] bool,
-        behavior_policy [B,T,A], initial_h/c [B,T,lstm].
-        Returns (pi_loss, baseline_loss, entropy, learning_rate).
+        s: normalized float frames [B,T,H,W,C]; everything else as in
+        train(). Returns (pi_loss, baseline_loss, entropy, total_loss).
         """
-        s = self.frames_to_device(state)
-        r = self.to_device(reward, torch.float32)
-        a = self.to_device(action, torch.int64)
-        d = self.to_device(done, torch.bool)
-        mu = self.to_device(behavior_policy, torch.float32)
-        pa = self.to_device(previous_action, torch.int64)
-        h0 = self.to_device(initial_h, torch.float32)
-        c0 = self.to_device(initial_c, torch.float32)
-
         clipped_r = clip_rewards(r, self.reward_clipping)
         discounts = (~d).float() * self.discount_factor
 
@@ -115,6 +104,28 @@ class Agent(AgentBase):
         entropy = vtrace.compute_entropy_loss(p_f)
         total = (pi_loss + baseline_loss * self.baseline_loss_coef
                  + entropy * self.entropy_coef)
+        return pi_loss, baseline_loss, entropy, total
+
+    def train(self, state, reward, action, done, behavior_policy,
+              previous_action, initial_h, initial_c):
+        """One V-trace update on a [B, T] batch of unrolls (eager path).
+
+        Arrays as stored by the queue: state uint8 [B,T,H,W,C], reward [B,T],
+        action/previous_action [B,T], done [B,T] bool,
+        behavior_policy [B,T,A], initial_h/c [B,T,lstm].
+        Returns (pi_loss, baseline_loss, entropy, learning_rate).
+        """
+        s = self.frames_to_device(state)
+        r = self.to_device(reward, torch.float32)
+        a = self.to_device(action, torch.int64)
+        d = self.to_device(done, torch.bool)
+        mu = self.to_device(behavior_policy, torch.float32)
+        pa = self.to_device(previous_action, torch.int64)
+        h0 = self.to_device(initial_h, torch.float32)
+        c0 = self.to_device(initial_c, torch.float32)
+
+        pi_loss, baseline_loss, entropy, total = self.compute_losses(
+            s, r, a, d, mu, pa, h0, c0)
 
         self.optimizer.zero_grad()
         total.backward()

@@ -34,13 +34,30 @@ class AtariConvStack(nn.Module):
         self.conv2 = nn.Conv2d(32, 64, kernel_size=4, stride=2)
         self.conv3 = nn.Conv2d(64, 64, kernel_size=3, stride=1)
         self.out_features = 7 * 7 * 64
+        # frames arrive NHWC (that is how the queue stores them); running the
+        # whole stack channels_last keeps MIOpen on its native NHWC kernels
+        # and removes the batched_transpose launches around every conv
+        # (measured: they were ~8% of step GPU time in eager NCHW).
+        self.conv1.to(memory_format=torch.channels_last)
+        self.conv2.to(memory_format=torch.channels_last)
+        self.conv3.to(memory_format=torch.channels_last)
 
     def forward(self, x_nhwc: torch.Tensor) -> torch.Tensor:
-        x = x_nhwc.permute(0, 3, 1, 2).contiguous()
+        # NHWC [N,H,W,C] -> logical NCHW with channels_last layout: a view,
+        # no copy, no transpose kernel.
+        x = x_nhwc.permute(0, 3, 1, 2)
+        if x.device.type == "cuda":
+            x = x.contiguous(memory_format=torch.channels_last)
+        else:
+            x = x.contiguous()
         x = F.relu(self.conv1(x))
         x = F.relu(self.conv2(x))
         x = F.relu(self.conv3(x))
-        return x.flatten(1)
+        # flatten in NHWC order (TF semantics, reference
+        # impala_actor_critic.py:8-9) — identical math on CPU and GPU so
+        # weights transfer between learner and actors; on channels_last this
+        # permute+flatten is a free view.
+        return x.permute(0, 2, 3, 1).flatten(1)
 
 
 class ActionEmbedding(nn.Module):

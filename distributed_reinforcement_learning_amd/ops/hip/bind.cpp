@@ -34,11 +34,13 @@ extern "C" __global__ void drla_lstm_tail_bwd(const float*, const float*,
                                               long long, int);
 extern "C" __global__ void drla_sq_norm(const float*, float*, long long);
 extern "C" __global__ void drla_rmsprop_step(float*, const float*, float*,
+                                             const float*, float,
                                              const float*, float, float,
-                                             float, float, long long);
+                                             long long);
 extern "C" __global__ void drla_adam_step(float*, const float*, float*,
-                                          float*, const float*, float, float,
-                                          float, float, float, long long);
+                                          float*, const float*, float,
+                                          const float*, float, float, float,
+                                          long long);
 
 namespace {
 
@@ -160,9 +162,11 @@ torch::Tensor sq_norm(torch::Tensor x) {
   return out;
 }
 
-void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
-                  double clip, double lr, double rho, double eps) {
-  for (auto* t : {&p, &g, &ms}) check_gpu_contig(*t, "rmsprop tensor");
+void rmsprop_step_t(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
+                    double clip, torch::Tensor lr_buf, double rho,
+                    double eps) {
+  for (auto* t : {&p, &g, &ms, &lr_buf})
+    check_gpu_contig(*t, "rmsprop tensor");
   const long long n = p.numel();
   torch::Tensor norm_buf;
   if (clip > 0) {
@@ -174,14 +178,21 @@ void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
                      0, cur_stream(), p.data_ptr<float>(),
                      g.data_ptr<float>(), ms.data_ptr<float>(),
                      norm_buf.data_ptr<float>(), static_cast<float>(clip),
-                     static_cast<float>(lr), static_cast<float>(rho),
+                     lr_buf.data_ptr<float>(), static_cast<float>(rho),
                      static_cast<float>(eps), n);
 }
 
-void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
-               torch::Tensor v, double clip, double lr_t, double beta1,
-               double beta2, double eps) {
-  for (auto* t : {&p, &g, &m, &v}) check_gpu_contig(*t, "adam tensor");
+void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
+                  double clip, double lr, double rho, double eps) {
+  auto lr_buf = torch::full({1}, lr, p.options());
+  rmsprop_step_t(p, g, ms, clip, lr_buf, rho, eps);
+}
+
+void adam_step_t(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                 torch::Tensor v, double clip, torch::Tensor lr_buf,
+                 double beta1, double beta2, double eps) {
+  for (auto* t : {&p, &g, &m, &v, &lr_buf})
+    check_gpu_contig(*t, "adam tensor");
   const long long n = p.numel();
   torch::Tensor norm_buf;
   if (clip > 0) {
@@ -193,8 +204,15 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      norm_buf.data_ptr<float>(), static_cast<float>(clip),
-                     static_cast<float>(lr_t), static_cast<float>(beta1),
+                     lr_buf.data_ptr<float>(), static_cast<float>(beta1),
                      static_cast<float>(beta2), static_cast<float>(eps), n);
+}
+
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, double clip, double lr_t, double beta1,
+               double beta2, double eps) {
+  auto lr_buf = torch::full({1}, lr_t, p.options());
+  adam_step_t(p, g, m, v, clip, lr_buf, beta1, beta2, eps);
 }
 
 }  // namespace
@@ -210,6 +228,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");
   m.def("rmsprop_step", &rmsprop_step,
         "fused global-norm-clip + TF-RMSProp update (K12)");
+  m.def("rmsprop_step_t", &rmsprop_step_t,
+        "RMSProp update with device-tensor lr (hipGraph-safe)");
   m.def("adam_step", &adam_step,
         "fused global-norm-clip + TF-Adam update (K12)");
+  m.def("adam_step_t", &adam_step_t,
+        "Adam update with device-tensor lr_t (hipGraph-safe)");
 }

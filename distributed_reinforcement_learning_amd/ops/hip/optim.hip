@@ -48,11 +48,16 @@ __device__ __forceinline__ float drla_clip_scale(const float* sq_norm_buf,
   return norm > clip ? clip / norm : 1.0f;
 }
 
+// lr comes through a 1-element device buffer so the kernel is
+// hipGraph-replay safe with a per-step decayed LR (the host rewrites the
+// buffer before each replay; a by-value lr would be baked into the graph).
 extern "C" __global__ void drla_rmsprop_step(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ ms, const float* __restrict__ sq_norm_buf,
-    float clip, float lr, float rho, float eps, long long n) {
+    float clip, const float* __restrict__ lr_buf, float rho, float eps,
+    long long n) {
   const float scale = drla_clip_scale(sq_norm_buf, clip);
+  const float lr = *lr_buf;
   long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
   const long long stride = gridDim.x * (long long)blockDim.x;
   for (; i < n; i += stride) {
@@ -66,9 +71,11 @@ extern "C" __global__ void drla_rmsprop_step(
 extern "C" __global__ void drla_adam_step(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ m, float* __restrict__ v,
-    const float* __restrict__ sq_norm_buf, float clip, float lr_t,
-    float beta1, float beta2, float eps, long long n) {
+    const float* __restrict__ sq_norm_buf, float clip,
+    const float* __restrict__ lr_buf, float beta1, float beta2, float eps,
+    long long n) {
   const float scale = drla_clip_scale(sq_norm_buf, clip);
+  const float lr_t = *lr_buf;
   long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
   const long long stride = gridDim.x * (long long)blockDim.x;
   for (; i < n; i += stride) {

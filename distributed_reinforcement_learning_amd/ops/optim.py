@@ -90,6 +90,11 @@ class _FlatOptimizerBase:
     def step(self, lr: Optional[float] = None) -> None:
         raise NotImplementedError
 
+    def step_tensor_lr(self, lr_buf: torch.Tensor) -> None:
+        """GPU-only, hipGraph-capturable update: lr is read from a 1-element
+        device buffer the host rewrites before each replay."""
+        raise NotImplementedError
+
 
 class FusedRMSProp(_FlatOptimizerBase):
     """TF-RMSProp semantics: ms = rho*ms + (1-rho)*g^2,
@@ -120,6 +125,12 @@ class FusedRMSProp(_FlatOptimizerBase):
             self.flat_params.addcdiv_(g, (self.ms + self.eps).sqrt(),
                                       value=-lr)
         self.step_count += 1
+
+    def step_tensor_lr(self, lr_buf: torch.Tensor) -> None:
+        ext = _ops.require_ext()
+        ext.rmsprop_step_t(self.flat_params, self.flat_grads, self.ms,
+                           float(self.clip_norm or -1.0), lr_buf, self.rho,
+                           self.eps)
 
 
 class FusedAdam(_FlatOptimizerBase):
@@ -154,3 +165,13 @@ class FusedAdam(_FlatOptimizerBase):
             self.v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
             self.flat_params.addcdiv_(self.m, self.v.sqrt() + self.eps,
                                       value=-lr_t)
+
+    def lr_t_for(self, lr: float, t: int) -> float:
+        return lr * (1 - self.beta2 ** t) ** 0.5 / (1 - self.beta1 ** t)
+
+    def step_tensor_lr(self, lr_buf: torch.Tensor) -> None:
+        """lr_buf must already hold the bias-corrected lr_t (lr_t_for)."""
+        ext = _ops.require_ext()
+        ext.adam_step_t(self.flat_params, self.flat_grads, self.m, self.v,
+                        float(self.clip_norm or -1.0), lr_buf, self.beta1,
+                        self.beta2, self.eps)

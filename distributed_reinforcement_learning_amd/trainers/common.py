@@ -54,6 +54,8 @@ def build_parser(algorithm: str) -> argparse.ArgumentParser:
                    help="publish weights every N train steps")
     p.add_argument("--restore", default=None)
     p.add_argument("--seed", type=int, default=None)
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph capture of the learner step")
     return p
 
 

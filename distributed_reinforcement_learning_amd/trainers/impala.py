@@ -67,6 +67,12 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
     monitor = common.learner_supervision(ctx, queue, supervisor)
     writer = SummaryWriter(ctx.learner_logdir())
     timer = StageTimer()
+    graphed = None
+    if torch.cuda.is_available() and not getattr(args, "no_graph", False):
+        from distributed_reinforcement_learning_amd.runtime import (
+            GraphedImpalaStep,
+        )
+        graphed = GraphedImpalaStep(agent, cfg.batch_size)
     train_step = 0
     try:
         while args.max_steps <= 0 or train_step < args.max_steps:
@@ -74,13 +80,16 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
                 batch = queue.sample_batch(cfg.batch_size)
             t0 = time.time()
             with timer.track("train"):
-                pi_loss, v_loss, entropy, lr = agent.train(
-                    state=batch["state"], reward=batch["reward"],
-                    action=batch["action"], done=batch["done"],
-                    behavior_policy=batch["behavior_policy"],
-                    previous_action=batch["previous_action"],
-                    initial_h=batch["initial_h"],
-                    initial_c=batch["initial_c"])
+                if graphed is not None:
+                    pi_loss, v_loss, entropy, lr = graphed.step(batch)
+                else:
+                    pi_loss, v_loss, entropy, lr = agent.train(
+                        state=batch["state"], reward=batch["reward"],
+                        action=batch["action"], done=batch["done"],
+                        behavior_policy=batch["behavior_policy"],
+                        previous_action=batch["previous_action"],
+                        initial_h=batch["initial_h"],
+                        initial_c=batch["initial_c"])
             train_step += 1
             if ctx.rank == 0 and train_step % args.publish_every == 0:
                 with timer.track("publish"):
