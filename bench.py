@@ -89,16 +89,26 @@ def main(argv=None) -> None:
         ))
 
     graphed = None
+    pinned_pool = None
     if have_gpu and not args.no_graph:
         from distributed_reinforcement_learning_amd.runtime import (
             GraphedImpalaStep,
         )
         graphed = GraphedImpalaStep(agent, B)
+        # data sits in pinned host memory, as the production ingest path
+        # delivers it (TrajectoryQueue.sample_batch_into); the per-step H2D
+        # upload stays inside the timed region.
+        pinned_pool = []
+        for b in pool:
+            pinned_pool.append({
+                k: torch.as_tensor(np.asarray(v)).to(
+                    graphed.pinned[k].dtype).pin_memory()
+                for k, v in b.items()})
 
     def step(i: int):
-        b = pool[i % len(pool)]
         if graphed is not None:
-            return graphed.step(b)
+            return graphed.step(pinned_src=pinned_pool[i % len(pinned_pool)])
+        b = pool[i % len(pool)]
         return agent.train(
             state=b["state"], reward=b["reward"], action=b["action"],
             done=b["done"], behavior_policy=b["behavior_policy"],

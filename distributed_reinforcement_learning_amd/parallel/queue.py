@@ -262,13 +262,18 @@ class TrajectoryQueue:
         return sum(r.size() for r in self.rings.values())
 
     def sample_batch(self, batch_size: int, timeout: Optional[float] = None,
-                     poll: float = 0.001) -> Dict[str, np.ndarray]:
+                     poll: float = 0.001,
+                     out: Optional[Dict[str, np.ndarray]] = None
+                     ) -> Dict[str, np.ndarray]:
         """Blocking round-robin gather of ``batch_size`` trajectories into
-        freshly allocated stacked arrays (batch dim first)."""
-        out = {
-            name: np.empty((batch_size, *shape), dtype=dtype)
-            for name, (shape, dtype) in self.schema.items()
-        }
+        stacked arrays (batch dim first). Pass ``out`` (e.g. numpy views of
+        the graphed step's pinned staging buffers) to fill caller-owned
+        memory with zero extra copies."""
+        if out is None:
+            out = {
+                name: np.empty((batch_size, *shape), dtype=dtype)
+                for name, (shape, dtype) in self.schema.items()
+            }
         ids = list(self.rings)
         filled = 0
         deadline = time.time() + timeout if timeout else None

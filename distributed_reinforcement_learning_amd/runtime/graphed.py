@@ -1,19 +1,28 @@
 """hipGraph-captured IMPALA train step.
 
 The eager step is launch-bound on MI355X: ~300 kernel dispatches per step,
-~1.9 ms GPU-busy inside a ~4.6 ms wall step (rocprof r01,
-profiles/impala_bench_r01_kernels.md). Capturing the whole
-normalize -> unroll -> V-trace -> backward -> fused-optimizer pipeline into a
-hipGraph collapses the host-side launch gaps to one graph replay.
+~1.9 ms GPU-busy inside a ~4 ms wall step (rocprof r01,
+profiles/impala_bench_r01_kernels.md). Capturing the
+normalize -> unroll -> V-trace -> backward -> fused-optimizer pipeline into
+hipGraphs cuts the replay to ~1.3 ms GPU (measured, profiles/ triage r03).
 
-Structure per step:
-  host:    copy batch into pinned staging, async H2D into the graph's static
-           input buffers, write the decayed LR into a 1-element device buffer
+Two measured host-side traps shape this class (gpu_triage.py r03/r04):
+  * a mid-flight ``.item()`` on the loss tensors costs ~11 ms (blocking-sync
+    wake latency), so losses are NOT read per step — ``last_losses()`` reads
+    them on demand (logging cadence), and the steady-state loop never syncs;
+  * staging numpy -> pinned is a ~4 ms single-thread memcpy, so callers that
+    can produce data straight into ``self.pinned`` (the trajectory queue, the
+    bench pool) skip it entirely.
+
+Step structure:
+  host:    wait on the consume-event (no-op once the pipeline is primed),
+           async H2D pinned -> static inputs, write decayed LR into a device
+           buffer
   graph 1: zero flat grads, normalize frames, batched unroll (bf16),
            V-trace, losses, backward (into the flat grad bucket)
-  eager:   single fused RCCL all-reduce of the flat bucket (world > 1) —
-           kept outside the graph so capture needs no collective support
+  eager:   single fused RCCL all-reduce of the bucket (world > 1)
   graph 2: global-norm clip + RMSProp update reading the LR buffer
+  host:    record consume-event
 
 Weights and optimizer state are snapshotted before the warmup iterations and
 restored before capture, so graphing never perturbs training state.
@@ -21,7 +30,7 @@ restored before capture, so graphing never perturbs training state.
 
 from __future__ import annotations
 
-from typing import Dict, Tuple
+from typing import Dict, Optional, Tuple
 
 import numpy as np
 import torch
@@ -54,6 +63,10 @@ class GraphedImpalaStep:
             for k, v in self.inputs.items()
         }
         self.lr_buf = torch.zeros(1, dtype=torch.float32, device=dev)
+        self._consumed = torch.cuda.Event()
+        self._consumed.record()
+        self._uploaded = torch.cuda.Event()
+        self._uploaded.record()
 
         # ---- warmup (eager, side stream), with state snapshot/restore -----
         opt = agent.optimizer
@@ -99,30 +112,59 @@ class GraphedImpalaStep:
         total.backward()
         return (pi_loss.detach(), baseline_loss.detach(), entropy.detach())
 
-    def load_inputs(self, batch: Dict[str, np.ndarray]) -> None:
-        """Stage a host batch into the graph's static input buffers."""
-        for k, dst in self.inputs.items():
-            src = batch[k]
-            if isinstance(src, torch.Tensor) and src.is_cuda:
-                dst.copy_(src, non_blocking=True)
-            else:
-                pin = self.pinned[k]
-                pin.copy_(torch.as_tensor(np.asarray(src)))
-                dst.copy_(pin, non_blocking=True)
+    # -- input staging -------------------------------------------------------
 
-    def step(self, batch: Dict[str, np.ndarray]) -> Tuple[float, ...]:
-        """Run one full train step; returns (pi_loss, baseline_loss,
-        entropy, lr) as floats (one sync at the end)."""
+    def stage_to_pinned(self, batch: Dict[str, np.ndarray]) -> None:
+        """Host-side memcpy of a numpy batch into the pinned buffers
+        (~4 ms for the reference shape; skip by filling self.pinned
+        directly, e.g. TrajectoryQueue.sample_batch_into)."""
+        for k, pin in self.pinned.items():
+            pin.copy_(torch.as_tensor(np.asarray(batch[k])))
+
+    def upload_inputs(self, src: Optional[Dict[str, torch.Tensor]] = None
+                      ) -> None:
+        """Async H2D pinned -> static graph inputs. Waits (usually no-op)
+        until the previous replay has consumed the static buffers."""
+        self._consumed.synchronize()
+        src = src or self.pinned
+        for k, dst in self.inputs.items():
+            dst.copy_(src[k], non_blocking=True)
+        self._uploaded.record()
+
+    def wait_pinned_free(self) -> None:
+        """Block until the last async H2D has finished reading the pinned
+        staging — call before overwriting self.pinned."""
+        self._uploaded.synchronize()
+
+    # -- stepping ------------------------------------------------------------
+
+    def step(self, batch: Optional[Dict[str, np.ndarray]] = None,
+             pinned_src: Optional[Dict[str, torch.Tensor]] = None
+             ) -> Tuple[torch.Tensor, ...]:
+        """Run one full train step. Sources, in priority order: ``batch``
+        (numpy, staged through self.pinned), ``pinned_src`` (caller-owned
+        pinned tensors), or self.pinned already filled. Never syncs; returns
+        the loss TENSORS (device). Read them with last_losses() at logging
+        cadence."""
         agent = self.agent
-        self.load_inputs(batch)
+        if batch is not None:
+            self.stage_to_pinned(batch)
+        self.upload_inputs(pinned_src)
         lr = agent.lr_at(agent.global_step)
         self.lr_buf.fill_(lr)
         self.g_fwd_bwd.replay()
         if self._distributed:
             agent.reduce_gradients()
         self.g_opt.replay()
+        self._consumed.record()
         agent.optimizer.step_count += 1
         agent.global_step += 1
         agent.num_env_frames += int(np.prod(self.inputs["reward"].shape))
+        self._last_lr = lr
+        return self.losses
+
+    def last_losses(self) -> Tuple[float, float, float, float]:
+        """(pi_loss, baseline_loss, entropy, lr) of the most recent completed
+        step — syncs the device."""
         pi, bl, ent = (float(x) for x in self.losses)
-        return pi, bl, ent, lr
+        return pi, bl, ent, getattr(self, "_last_lr", 0.0)

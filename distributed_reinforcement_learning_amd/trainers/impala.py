@@ -68,20 +68,29 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
     writer = SummaryWriter(ctx.learner_logdir())
     timer = StageTimer()
     graphed = None
+    pinned_views = None
     if torch.cuda.is_available() and not getattr(args, "no_graph", False):
         from distributed_reinforcement_learning_amd.runtime import (
             GraphedImpalaStep,
         )
         graphed = GraphedImpalaStep(agent, cfg.batch_size)
+        # the queue fills the graphed step's pinned staging directly
+        # (zero extra host copies); int64/bool fields go through a numpy
+        # view-compatible dtype conversion inside the ring pop
+        pinned_views = {k: v.numpy() for k, v in graphed.pinned.items()}
     train_step = 0
+    log_every = max(1, args.publish_every) if graphed is None else 25
     try:
         while args.max_steps <= 0 or train_step < args.max_steps:
             with timer.track("ingest"):
-                batch = queue.sample_batch(cfg.batch_size)
+                if graphed is not None:
+                    graphed.wait_pinned_free()
+                batch = queue.sample_batch(cfg.batch_size, out=pinned_views)
             t0 = time.time()
             with timer.track("train"):
                 if graphed is not None:
-                    pi_loss, v_loss, entropy, lr = graphed.step(batch)
+                    graphed.step()  # consumes the pinned staging
+                    pi_loss = v_loss = entropy = lr = None
                 else:
                     pi_loss, v_loss, entropy, lr = agent.train(
                         state=batch["state"], reward=batch["reward"],
@@ -97,7 +106,10 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
             ctx.maybe_checkpoint(agent)
             if monitor is not None and train_step % 50 == 0:
                 supervisor.check()
-            if ctx.rank == 0:
+            if ctx.rank == 0 and (graphed is None
+                                  or train_step % log_every == 0):
+                if graphed is not None:
+                    pi_loss, v_loss, entropy, lr = graphed.last_losses()
                 step = agent.global_step
                 writer.add_scalar("data/pi_loss", pi_loss, step)
                 writer.add_scalar("data/value_loss", v_loss, step)

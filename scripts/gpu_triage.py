@@ -89,6 +89,35 @@ def staged_impala():
     print(f"[staged] g_opt        {tsec(gs.g_opt.replay):.3f} ms")
     print(f"[staged] losses->host "
           f"{tsec(lambda: [float(x) for x in gs.losses]):.3f} ms")
+
+    def combo_load_replay():
+        gs.load_inputs(batch)
+        gs.g_fwd_bwd.replay()
+
+    def combo_replay_replay():
+        gs.g_fwd_bwd.replay()
+        gs.g_opt.replay()
+
+    def combo_load_replay_replay():
+        gs.load_inputs(batch)
+        gs.g_fwd_bwd.replay()
+        gs.g_opt.replay()
+
+    def combo_lr_replay():
+        gs.lr_buf.fill_(1e-4)
+        gs.g_fwd_bwd.replay()
+
+    def step_no_losses():
+        gs.load_inputs(batch)
+        gs.lr_buf.fill_(1e-4)
+        gs.g_fwd_bwd.replay()
+        gs.g_opt.replay()
+
+    print(f"[staged] load+fb      {tsec(combo_load_replay):.3f} ms")
+    print(f"[staged] fb+opt       {tsec(combo_replay_replay):.3f} ms")
+    print(f"[staged] lr+fb        {tsec(combo_lr_replay):.3f} ms")
+    print(f"[staged] load+fb+opt  {tsec(combo_load_replay_replay):.3f} ms")
+    print(f"[staged] step-no-loss {tsec(step_no_losses):.3f} ms")
     print(f"[staged] full step    {tsec(lambda: gs.step(batch)):.3f} ms")
     # eager comparison
     t = tsec(lambda: agent.train(
