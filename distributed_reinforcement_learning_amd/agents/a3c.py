@@ -44,7 +44,7 @@ class Agent(AgentBase):
         self.rng = np.random.default_rng(seed)
 
         cls = ActorCritic if len(self.input_shape) == 3 else VectorActorCritic
-        self.model = cls(self.input_shape, num_action).to(self.device)
+        self.model = self.finalize_model(cls(self.input_shape, num_action))
         self.optimizer = None
         if build_optimizer:
             self.optimizer = FusedAdam(self.model.parameters(),

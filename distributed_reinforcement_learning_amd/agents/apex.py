@@ -43,8 +43,9 @@ class Agent(AgentBase):
         self.rng = np.random.default_rng(seed)
 
         cls = ApexDuelingQ if len(self.input_shape) == 3 else VectorDuelingQ
-        self.model = cls(self.input_shape, num_action).to(self.device)
-        self.target_model = cls(self.input_shape, num_action).to(self.device)
+        self.model = self.finalize_model(cls(self.input_shape, num_action))
+        self.target_model = self.finalize_model(
+            cls(self.input_shape, num_action))
         self.target_model.load_state_dict(self.model.state_dict())
         for p in self.target_model.parameters():
             p.requires_grad_(False)

@@ -55,12 +55,25 @@ class AgentBase:
         self.weight_subscriber = None
         self._all_reducer: Optional[FlatAllReducer] = None
 
-    # -- autocast ------------------------------------------------------------
+    # -- dtype policy ----------------------------------------------------------
+    # GPU learners run bf16-NATIVE models (weights stored bf16, fp32 master in
+    # the fused optimizer) — no autocast, no per-layer weight casts. CPU
+    # actors/learners stay fp32.
+
+    @property
+    def model_dtype(self) -> torch.dtype:
+        if self.device.type == "cuda" and self.compute_dtype == torch.bfloat16:
+            return torch.bfloat16
+        return torch.float32
+
+    def finalize_model(self, model: torch.nn.Module) -> torch.nn.Module:
+        model = model.to(self.device)
+        if self.model_dtype != torch.float32:
+            model = model.to(self.model_dtype)
+        return model
 
     def autocast(self):
-        if self.device.type == "cuda":
-            return torch.autocast(device_type="cuda",
-                                  dtype=self.compute_dtype)
+        # kept for API compatibility: bf16 is native now, nothing to autocast
         import contextlib
         return contextlib.nullcontext()
 
@@ -83,7 +96,7 @@ class AgentBase:
         if t.dtype == torch.uint8:
             t = t.to(self.device, non_blocking=True)
             from distributed_reinforcement_learning_amd.ops import normalize_frames
-            return normalize_frames(t)
+            return normalize_frames(t, out_dtype=self.model_dtype)
         return t.to(self.device, dtype=torch.float32, non_blocking=True)
 
     # -- model plumbing (subclasses set self.model / self.optimizer) ---------

@@ -52,8 +52,9 @@ class Agent(AgentBase):
             torch.manual_seed(seed)
         self.rng = np.random.default_rng(seed)
 
-        self.model = ImpalaActorCritic(self.input_shape, num_action,
-                                       lstm_hidden_size).to(self.device)
+        self.model = self.finalize_model(
+            ImpalaActorCritic(self.input_shape, num_action,
+                              lstm_hidden_size))
         self.optimizer = None
         if build_optimizer:
             # TF RMSProp(decay=.99, momentum=0, eps=.1) + clip_by_global_norm

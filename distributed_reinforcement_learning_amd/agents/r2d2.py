@@ -52,10 +52,10 @@ class Agent(AgentBase):
             torch.manual_seed(seed)
         self.rng = np.random.default_rng(seed)
 
-        self.model = R2D2LstmQ(self.input_shape, num_action,
-                               lstm_size).to(self.device)
-        self.target_model = R2D2LstmQ(self.input_shape, num_action,
-                                      lstm_size).to(self.device)
+        self.model = self.finalize_model(
+            R2D2LstmQ(self.input_shape, num_action, lstm_size))
+        self.target_model = self.finalize_model(
+            R2D2LstmQ(self.input_shape, num_action, lstm_size))
         self.target_model.load_state_dict(self.model.state_dict())
         for p in self.target_model.parameters():
             p.requires_grad_(False)

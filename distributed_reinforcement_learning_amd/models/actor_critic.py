@@ -57,6 +57,7 @@ class VectorActorCritic(nn.Module):
         self.value_head = MLPHead(2 * hidden, [256, 256], 1, None)
 
     def forward(self, state: torch.Tensor, prev_action: torch.Tensor):
+        state = state.to(self.obs_mlp[0].weight.dtype)
         feat = torch.cat(
             [self.obs_mlp(state), self.action_emb(prev_action)], dim=1)
         policy = self.policy_head(feat)

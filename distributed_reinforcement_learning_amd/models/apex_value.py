@@ -56,6 +56,7 @@ class VectorDuelingQ(nn.Module):
         self.mean_out = nn.Linear(hidden, 1)
 
     def forward(self, state: torch.Tensor, prev_action: torch.Tensor):
+        state = state.to(self.obs_mlp[0].weight.dtype)
         feat = torch.cat(
             [self.obs_mlp(state), self.action_emb(prev_action)], dim=1)
         x = F.relu(self.trunk(feat))

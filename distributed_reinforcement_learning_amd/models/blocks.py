@@ -45,7 +45,7 @@ class AtariConvStack(nn.Module):
     def forward(self, x_nhwc: torch.Tensor) -> torch.Tensor:
         # NHWC [N,H,W,C] -> logical NCHW with channels_last layout: a view,
         # no copy, no transpose kernel.
-        x = x_nhwc.permute(0, 3, 1, 2)
+        x = x_nhwc.to(self.conv1.weight.dtype).permute(0, 3, 1, 2)
         if x.device.type == "cuda":
             x = x.contiguous(memory_format=torch.channels_last)
         else:
@@ -62,19 +62,25 @@ class AtariConvStack(nn.Module):
 
 class ActionEmbedding(nn.Module):
     """one-hot(prev_action) -> 256 -> 256, ReLU
-    (reference model/impala_actor_critic.py:12-16)."""
+    (reference model/impala_actor_critic.py:12-16).
+
+    one-hot @ W + b is a row gather of W — computed as an embedding lookup
+    (SURVEY K2): one index_select instead of a one-hot scatter + skinny GEMM.
+    ``table`` is stored [num_action, hidden], i.e. the transpose of the
+    nn.Linear weight it replaces; identical math.
+    """
 
     def __init__(self, num_action: int, hidden: int = 256):
         super().__init__()
         self.num_action = num_action
-        self.fc1 = nn.Linear(num_action, hidden)
+        self.table = nn.Parameter(torch.empty(num_action, hidden))
+        self.bias1 = nn.Parameter(torch.zeros(hidden))
+        nn.init.kaiming_uniform_(self.table, a=5 ** 0.5)
         self.fc2 = nn.Linear(hidden, hidden)
         self.out_features = hidden
 
     def forward(self, prev_action: torch.Tensor) -> torch.Tensor:
-        onehot = F.one_hot(prev_action.long(), self.num_action).to(
-            self.fc1.weight.dtype)
-        x = F.relu(self.fc1(onehot))
+        x = F.relu(F.embedding(prev_action.long(), self.table) + self.bias1)
         return F.relu(self.fc2(x))
 
 
@@ -96,11 +102,13 @@ class MLPHead(nn.Module):
         self.final_activation = final_activation
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = x.to(self.out.weight.dtype)
         for layer in self.hidden:
             x = F.relu(layer(x))
         x = self.out(x)
         if self.final_activation == "softmax":
-            x = F.softmax(x, dim=-1)
+            # softmax/log run in fp32 regardless of the compute dtype
+            x = F.softmax(x.float(), dim=-1)
         return x
 
 

@@ -36,7 +36,7 @@ class R2D2LstmQ(nn.Module):
         self.mean_out = nn.Linear(128, 1)
 
     def _head(self, h: torch.Tensor) -> torch.Tensor:
-        x = F.relu(self.trunk(h))
+        x = F.relu(self.trunk(h.to(self.trunk.weight.dtype)))
         return self.value_out(x) - self.mean_out(x)
 
     def features(self, state: torch.Tensor,

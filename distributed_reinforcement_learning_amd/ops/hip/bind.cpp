@@ -33,6 +33,14 @@ extern "C" __global__ void drla_lstm_tail_bwd(const float*, const float*,
                                               const float*, float*, float*,
                                               long long, int);
 extern "C" __global__ void drla_sq_norm(const float*, float*, long long);
+extern "C" __global__ void drla_sq_norm_bf16(const unsigned short*, float*,
+                                             long long);
+extern "C" __global__ void drla_rmsprop_step_bf16(
+    unsigned short*, const unsigned short*, float*, float*, const float*,
+    float, const float*, float, float, long long);
+extern "C" __global__ void drla_adam_step_bf16(
+    unsigned short*, const unsigned short*, float*, float*, float*,
+    const float*, float, const float*, float, float, float, long long);
 extern "C" __global__ void drla_rmsprop_step(float*, const float*, float*,
                                              const float*, float,
                                              const float*, float, float,
@@ -162,6 +170,66 @@ torch::Tensor sq_norm(torch::Tensor x) {
   return out;
 }
 
+torch::Tensor sq_norm_bf16(torch::Tensor x) {
+  check_gpu_contig(x, "x");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "wants bf16");
+  auto out = torch::zeros({1}, x.options().dtype(torch::kFloat));
+  hipLaunchKernelGGL(drla_sq_norm_bf16, dim3(drla_grid(x.numel() / 4 + 1)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     reinterpret_cast<const unsigned short*>(x.data_ptr()),
+                     out.data_ptr<float>(), (long long)x.numel());
+  return out;
+}
+
+void rmsprop_step_bf16_t(torch::Tensor p, torch::Tensor g,
+                         torch::Tensor master, torch::Tensor ms, double clip,
+                         torch::Tensor lr_buf, double rho, double eps) {
+  for (auto* t : {&p, &g, &master, &ms, &lr_buf})
+    check_gpu_contig(*t, "rmsprop bf16 tensor");
+  TORCH_CHECK(p.scalar_type() == torch::kBFloat16 &&
+                  g.scalar_type() == torch::kBFloat16,
+              "params/grads must be bf16");
+  const long long n = p.numel();
+  torch::Tensor norm_buf;
+  if (clip > 0) {
+    norm_buf = sq_norm_bf16(g);
+  } else {
+    norm_buf = torch::zeros({1}, master.options());
+  }
+  hipLaunchKernelGGL(drla_rmsprop_step_bf16, dim3(drla_grid(n)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     reinterpret_cast<unsigned short*>(p.data_ptr()),
+                     reinterpret_cast<const unsigned short*>(g.data_ptr()),
+                     master.data_ptr<float>(), ms.data_ptr<float>(),
+                     norm_buf.data_ptr<float>(), static_cast<float>(clip),
+                     lr_buf.data_ptr<float>(), static_cast<float>(rho),
+                     static_cast<float>(eps), n);
+}
+
+void adam_step_bf16_t(torch::Tensor p, torch::Tensor g, torch::Tensor master,
+                      torch::Tensor m, torch::Tensor v, double clip,
+                      torch::Tensor lr_buf, double beta1, double beta2,
+                      double eps) {
+  for (auto* t : {&p, &g, &master, &m, &v, &lr_buf})
+    check_gpu_contig(*t, "adam bf16 tensor");
+  const long long n = p.numel();
+  torch::Tensor norm_buf;
+  if (clip > 0) {
+    norm_buf = sq_norm_bf16(g);
+  } else {
+    norm_buf = torch::zeros({1}, master.options());
+  }
+  hipLaunchKernelGGL(drla_adam_step_bf16, dim3(drla_grid(n)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     reinterpret_cast<unsigned short*>(p.data_ptr()),
+                     reinterpret_cast<const unsigned short*>(g.data_ptr()),
+                     master.data_ptr<float>(), m.data_ptr<float>(),
+                     v.data_ptr<float>(), norm_buf.data_ptr<float>(),
+                     static_cast<float>(clip), lr_buf.data_ptr<float>(),
+                     static_cast<float>(beta1), static_cast<float>(beta2),
+                     static_cast<float>(eps), n);
+}
+
 void rmsprop_step_t(torch::Tensor p, torch::Tensor g, torch::Tensor ms,
                     double clip, torch::Tensor lr_buf, double rho,
                     double eps) {
@@ -234,4 +302,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused global-norm-clip + TF-Adam update (K12)");
   m.def("adam_step_t", &adam_step_t,
         "Adam update with device-tensor lr_t (hipGraph-safe)");
+  m.def("sq_norm_bf16", &sq_norm_bf16, "squared L2 norm of a bf16 tensor");
+  m.def("rmsprop_step_bf16_t", &rmsprop_step_bf16_t,
+        "bf16-model/fp32-master RMSProp (K12, mixed precision)");
+  m.def("adam_step_bf16_t", &adam_step_bf16_t,
+        "bf16-model/fp32-master Adam (K12, mixed precision)");
 }

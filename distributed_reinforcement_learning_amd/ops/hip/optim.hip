@@ -68,6 +68,90 @@ extern "C" __global__ void drla_rmsprop_step(
   }
 }
 
+// ---- bf16-model / fp32-master variants (mixed precision) ----
+// model params + grads are bf16 (raw ushort), master + state fp32; the
+// update runs on the master and writes the rounded bf16 copy the model
+// reads — one fused HBM pass, no per-layer weight casts anywhere else.
+
+__device__ __forceinline__ float drla_bf16_to_f32(unsigned short u) {
+  unsigned int x = ((unsigned int)u) << 16;
+  return __uint_as_float(x);
+}
+
+extern "C" __global__ void drla_sq_norm_bf16(
+    const unsigned short* __restrict__ x, float* __restrict__ out,
+    long long n) {
+  typedef __attribute__((ext_vector_type(4))) unsigned short ushort4v;
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  float acc = 0.0f;
+  const long long n4 = n / 4;
+  const ushort4v* x4 = reinterpret_cast<const ushort4v*>(x);
+  for (long long k = i; k < n4; k += stride) {
+    ushort4v v = x4[k];
+    float a = drla_bf16_to_f32(v.x), b = drla_bf16_to_f32(v.y);
+    float c = drla_bf16_to_f32(v.z), d = drla_bf16_to_f32(v.w);
+    acc += a * a + b * b + c * c + d * d;
+  }
+  for (long long k = n4 * 4 + i; k < n; k += stride) {
+    float a = drla_bf16_to_f32(x[k]);
+    acc += a * a;
+  }
+  for (int off = DRLA_WAVE / 2; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off, DRLA_WAVE);
+  __shared__ float lds[DRLA_BLOCK / DRLA_WAVE];
+  const int wave = threadIdx.x / DRLA_WAVE;
+  const int lane = threadIdx.x % DRLA_WAVE;
+  if (lane == 0) lds[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.0f;
+    for (int w = 0; w < blockDim.x / DRLA_WAVE; ++w) s += lds[w];
+    atomicAdd(out, s);
+  }
+}
+
+extern "C" __global__ void drla_rmsprop_step_bf16(
+    unsigned short* __restrict__ p_bf16, const unsigned short* __restrict__ g,
+    float* __restrict__ master, float* __restrict__ ms,
+    const float* __restrict__ sq_norm_buf, float clip,
+    const float* __restrict__ lr_buf, float rho, float eps, long long n) {
+  const float scale = drla_clip_scale(sq_norm_buf, clip);
+  const float lr = *lr_buf;
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < n; i += stride) {
+    const float gc = drla_bf16_to_f32(g[i]) * scale;
+    const float m = rho * ms[i] + (1.0f - rho) * gc * gc;
+    ms[i] = m;
+    const float w = master[i] - lr * gc * rsqrtf(m + eps);
+    master[i] = w;
+    p_bf16[i] = drla_f32_to_bf16(w);
+  }
+}
+
+extern "C" __global__ void drla_adam_step_bf16(
+    unsigned short* __restrict__ p_bf16, const unsigned short* __restrict__ g,
+    float* __restrict__ master, float* __restrict__ m, float* __restrict__ v,
+    const float* __restrict__ sq_norm_buf, float clip,
+    const float* __restrict__ lr_buf, float beta1, float beta2, float eps,
+    long long n) {
+  const float scale = drla_clip_scale(sq_norm_buf, clip);
+  const float lr_t = *lr_buf;
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < n; i += stride) {
+    const float gc = drla_bf16_to_f32(g[i]) * scale;
+    const float mi = beta1 * m[i] + (1.0f - beta1) * gc;
+    const float vi = beta2 * v[i] + (1.0f - beta2) * gc * gc;
+    m[i] = mi;
+    v[i] = vi;
+    const float w = master[i] - lr_t * mi / (sqrtf(vi) + eps);
+    master[i] = w;
+    p_bf16[i] = drla_f32_to_bf16(w);
+  }
+}
+
 extern "C" __global__ void drla_adam_step(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ m, float* __restrict__ v,

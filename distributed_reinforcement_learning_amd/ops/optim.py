@@ -12,8 +12,14 @@ optimizer construction; module parameters are views into it), so the whole
 update is two kernel launches — and the DP all-reduce (parallel/dist.py)
 reduces the same flat grad buffer with a single RCCL call.
 
-CPU fallback implements identical math in torch (used by unit tests as the
-golden reference).
+Mixed precision: when the model's parameters are bf16 (the learner's GPU
+models), the optimizer keeps an fp32 MASTER copy; the fused kernel updates
+the master and writes the rounded bf16 copy the model computes with — no
+per-layer weight casts anywhere in the step, and the all-reduce moves bf16
+(half the xGMI bytes).
+
+CPU fallback implements identical math in torch (used by the unit tests as
+the golden reference).
 """
 
 from __future__ import annotations
@@ -41,7 +47,11 @@ def flatten_dense_params(params: List[torch.Tensor]) -> torch.Tensor:
 
 
 class _FlatOptimizerBase:
-    """Holds flat param/grad/state buffers; subclasses implement _apply."""
+    """Holds flat param/grad/state buffers; subclasses implement the update.
+
+    ``flat_params`` is the model's compute copy (fp32 or bf16);
+    ``master`` is the fp32 copy the update math runs on (aliases
+    flat_params when params are already fp32)."""
 
     def __init__(self, params: Iterable[torch.Tensor], lr: float,
                  clip_norm: Optional[float]):
@@ -49,6 +59,9 @@ class _FlatOptimizerBase:
         assert len(self.params) > 0
         self.flat_params = flatten_dense_params(self.params)
         self.flat_grads = torch.zeros_like(self.flat_params)
+        self.mixed = self.flat_params.dtype != torch.float32
+        self.master = (self.flat_params.detach().float()
+                       if self.mixed else self.flat_params)
         # route autograd into the flat grad buffer
         offset = 0
         for p in self.params:
@@ -65,8 +78,10 @@ class _FlatOptimizerBase:
     def grad_global_norm(self) -> torch.Tensor:
         if self.flat_grads.is_cuda:
             ext = _ops.require_ext()
+            if self.flat_grads.dtype == torch.bfloat16:
+                return ext.sq_norm_bf16(self.flat_grads).sqrt()[0]
             return ext.sq_norm(self.flat_grads).sqrt()[0]
-        return self.flat_grads.norm()
+        return self.flat_grads.float().norm()
 
     def _clip_scale(self) -> torch.Tensor:
         """tf.clip_by_global_norm semantics: scale = clip/max(norm, clip)."""
@@ -75,14 +90,25 @@ class _FlatOptimizerBase:
             return torch.ones_like(norm)
         return self.clip_norm / torch.clamp(norm, min=self.clip_norm)
 
+    def _sync_model_from_master(self) -> None:
+        if self.mixed:
+            with torch.no_grad():
+                self.flat_params.copy_(self.master)
+
     def state_dict(self) -> dict:
-        return {"step_count": self.step_count,
-                "state": {k: v for k, v in self._state_tensors().items()}}
+        sd = {"step_count": self.step_count,
+              "state": {k: v for k, v in self._state_tensors().items()}}
+        if self.mixed:
+            sd["master"] = self.master
+        return sd
 
     def load_state_dict(self, sd: dict) -> None:
         self.step_count = sd["step_count"]
         for k, v in sd["state"].items():
             self._state_tensors()[k].copy_(v)
+        if self.mixed and "master" in sd:
+            self.master.copy_(sd["master"])
+            self._sync_model_from_master()
 
     def _state_tensors(self) -> dict:
         raise NotImplementedError
@@ -106,7 +132,7 @@ class FusedRMSProp(_FlatOptimizerBase):
         super().__init__(params, lr, clip_norm)
         self.rho = rho
         self.eps = eps
-        self.ms = torch.zeros_like(self.flat_params)
+        self.ms = torch.zeros_like(self.master)
 
     def _state_tensors(self):
         return {"ms": self.ms}
@@ -115,22 +141,26 @@ class FusedRMSProp(_FlatOptimizerBase):
     def step(self, lr: Optional[float] = None) -> None:
         lr = self.lr if lr is None else lr
         if self.flat_params.is_cuda:
-            ext = _ops.require_ext()
-            ext.rmsprop_step(self.flat_params, self.flat_grads, self.ms,
-                             float(self.clip_norm or -1.0), float(lr),
-                             self.rho, self.eps)
+            lr_buf = torch.full((1,), lr, dtype=torch.float32,
+                                device=self.flat_params.device)
+            self.step_tensor_lr(lr_buf)
         else:
-            g = self.flat_grads * self._clip_scale()
+            g = self.flat_grads.float() * self._clip_scale()
             self.ms.mul_(self.rho).addcmul_(g, g, value=1 - self.rho)
-            self.flat_params.addcdiv_(g, (self.ms + self.eps).sqrt(),
-                                      value=-lr)
+            self.master.addcdiv_(g, (self.ms + self.eps).sqrt(), value=-lr)
+            self._sync_model_from_master()
         self.step_count += 1
 
     def step_tensor_lr(self, lr_buf: torch.Tensor) -> None:
         ext = _ops.require_ext()
-        ext.rmsprop_step_t(self.flat_params, self.flat_grads, self.ms,
-                           float(self.clip_norm or -1.0), lr_buf, self.rho,
-                           self.eps)
+        clip = float(self.clip_norm or -1.0)
+        if self.mixed:
+            ext.rmsprop_step_bf16_t(self.flat_params, self.flat_grads,
+                                    self.master, self.ms, clip, lr_buf,
+                                    self.rho, self.eps)
+        else:
+            ext.rmsprop_step_t(self.flat_params, self.flat_grads, self.ms,
+                               clip, lr_buf, self.rho, self.eps)
 
 
 class FusedAdam(_FlatOptimizerBase):
@@ -142,36 +172,41 @@ class FusedAdam(_FlatOptimizerBase):
                  clip_norm: Optional[float] = None):
         super().__init__(params, lr, clip_norm)
         self.beta1, self.beta2, self.eps = beta1, beta2, eps
-        self.m = torch.zeros_like(self.flat_params)
-        self.v = torch.zeros_like(self.flat_params)
+        self.m = torch.zeros_like(self.master)
+        self.v = torch.zeros_like(self.master)
 
     def _state_tensors(self):
         return {"m": self.m, "v": self.v}
+
+    def lr_t_for(self, lr: float, t: int) -> float:
+        return lr * (1 - self.beta2 ** t) ** 0.5 / (1 - self.beta1 ** t)
 
     @torch.no_grad()
     def step(self, lr: Optional[float] = None) -> None:
         lr = self.lr if lr is None else lr
         self.step_count += 1
-        t = self.step_count
-        lr_t = lr * (1 - self.beta2 ** t) ** 0.5 / (1 - self.beta1 ** t)
+        lr_t = self.lr_t_for(lr, self.step_count)
         if self.flat_params.is_cuda:
-            ext = _ops.require_ext()
-            ext.adam_step(self.flat_params, self.flat_grads, self.m, self.v,
-                          float(self.clip_norm or -1.0), float(lr_t),
-                          self.beta1, self.beta2, self.eps)
+            lr_buf = torch.full((1,), lr_t, dtype=torch.float32,
+                                device=self.flat_params.device)
+            self.step_tensor_lr(lr_buf)
         else:
-            g = self.flat_grads * self._clip_scale()
+            g = self.flat_grads.float() * self._clip_scale()
             self.m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
             self.v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
-            self.flat_params.addcdiv_(self.m, self.v.sqrt() + self.eps,
-                                      value=-lr_t)
-
-    def lr_t_for(self, lr: float, t: int) -> float:
-        return lr * (1 - self.beta2 ** t) ** 0.5 / (1 - self.beta1 ** t)
+            self.master.addcdiv_(self.m, self.v.sqrt() + self.eps,
+                                 value=-lr_t)
+            self._sync_model_from_master()
 
     def step_tensor_lr(self, lr_buf: torch.Tensor) -> None:
         """lr_buf must already hold the bias-corrected lr_t (lr_t_for)."""
         ext = _ops.require_ext()
-        ext.adam_step_t(self.flat_params, self.flat_grads, self.m, self.v,
-                        float(self.clip_norm or -1.0), lr_buf, self.beta1,
-                        self.beta2, self.eps)
+        clip = float(self.clip_norm or -1.0)
+        if self.mixed:
+            ext.adam_step_bf16_t(self.flat_params, self.flat_grads,
+                                 self.master, self.m, self.v, clip, lr_buf,
+                                 self.beta1, self.beta2, self.eps)
+        else:
+            ext.adam_step_t(self.flat_params, self.flat_grads, self.m,
+                            self.v, clip, lr_buf, self.beta1, self.beta2,
+                            self.eps)

@@ -105,7 +105,7 @@ class GraphedImpalaStep:
         agent = self.agent
         agent.optimizer.flat_grads.zero_()
         i = self.inputs
-        s = normalize_frames(i["state"])
+        s = normalize_frames(i["state"], out_dtype=agent.model_dtype)
         pi_loss, baseline_loss, entropy, total = agent.compute_losses(
             s, i["reward"], i["action"], i["done"], i["behavior_policy"],
             i["previous_action"], i["initial_h"], i["initial_c"])

@@ -189,3 +189,61 @@ def test_impala_gpu_matches_cpu_reference_forward():
                                         c.cuda())
     assert torch.allclose(P_cpu, P_gpu.float().cpu(), atol=0.03)
     assert torch.allclose(V_cpu, V_gpu.float().cpu(), atol=0.15, rtol=0.1)
+
+
+def test_mixed_precision_rmsprop_master(ext):
+    """bf16-model / fp32-master fused RMSProp vs manual fp32 math."""
+    torch.manual_seed(7)
+    n = 10_000
+    master_ref = torch.randn(n)
+    p = master_ref.bfloat16().cuda()
+    master = master_ref.clone().cuda()
+    g = (torch.randn(n) * 10).bfloat16().cuda()
+    ms = torch.rand(n).cuda()
+    lr_buf = torch.full((1,), 1e-3, device="cuda")
+    ms_ref = ms.cpu().clone()
+    ext.rmsprop_step_bf16_t(p, g, master, ms, 40.0, lr_buf, 0.99, 0.1)
+    # golden: clip on bf16 grads, update fp32 master, round to bf16
+    gf = g.float().cpu()
+    norm = gf.norm()
+    scale = float(40.0 / torch.clamp(norm, min=40.0))
+    gc = gf * scale
+    ms2 = 0.99 * ms_ref + 0.01 * gc * gc
+    master2 = master_ref - 1e-3 * gc / (ms2 + 0.1).sqrt()
+    assert torch.allclose(master.cpu(), master2, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(ms.cpu(), ms2, atol=1e-5, rtol=1e-5)
+    assert torch.equal(p.cpu(), master2.bfloat16())
+
+
+def test_bf16_model_agents_train_on_gpu():
+    """bf16-native agents: one train step each for apex and r2d2 on GPU."""
+    from distributed_reinforcement_learning_amd.agents import apex, r2d2
+    rng = np.random.default_rng(3)
+    ag = apex.Agent(input_shape=[84, 84, 4], num_action=4,
+                    discount_factor=0.99, gradient_clip_norm=40.0,
+                    reward_clipping="abs_one", start_learning_rate=1e-4,
+                    end_learning_rate=0.0, learning_frame=10 ** 9,
+                    device="cuda:0", seed=0)
+    assert next(ag.model.parameters()).dtype == torch.bfloat16
+    N = 8
+    loss, td = ag.train(
+        rng.integers(0, 255, (N, 84, 84, 4), dtype=np.uint8),
+        rng.integers(0, 255, (N, 84, 84, 4), dtype=np.uint8),
+        rng.integers(0, 4, N), rng.integers(0, 4, N),
+        rng.normal(size=N).astype(np.float32), np.zeros(N, bool))
+    assert np.isfinite(loss) and np.isfinite(td).all()
+
+    r = r2d2.Agent(seq_len=6, burn_in=2, input_shape=[84, 84, 1],
+                   num_action=4, lstm_size=8, discount_factor=0.997,
+                   start_learning_rate=1e-4, end_learning_rate=0.0,
+                   learning_frame=10 ** 9, gradient_clip_norm=40.0,
+                   device="cuda:0", seed=0)
+    B, L = 2, 6
+    loss, td = r.train(
+        state=rng.integers(0, 255, (B, L, 84, 84, 1), dtype=np.uint8),
+        previous_action=rng.integers(0, 4, (B, L)).astype(np.int32),
+        action=rng.integers(0, 4, (B, L)).astype(np.int32),
+        h=np.zeros((B, L, 8), np.float32), c=np.zeros((B, L, 8), np.float32),
+        reward=rng.normal(size=(B, L)).astype(np.float32),
+        done=np.zeros((B, L), bool), weight=np.ones(B, np.float32))
+    assert np.isfinite(loss) and np.isfinite(td).all()
