@@ -71,12 +71,28 @@ class Agent(AgentBase):
 
         s: normalized float frames [B,T,H,W,C]; everything else as in
         train(). Returns (pi_loss, baseline_loss, entropy, total_loss).
+
+        On GPU the whole post-unroll pipeline (softmax, rho, both V-trace
+        scans, all three losses AND their backward) is the fused HIP kernel
+        pair in ops/hip/vtrace_loss.hip; the CPU path composes the same math
+        from algorithms/vtrace.py (the golden reference the GPU parity test
+        compares against).
         """
         clipped_r = clip_rewards(r, self.reward_clipping)
         discounts = (~d).float() * self.discount_factor
 
-        with self.autocast():
-            policy, value = self.model.unroll(s, pa, h0, c0)
+        if s.is_cuda:
+            from distributed_reinforcement_learning_amd.ops import (
+                fused_vtrace_loss,
+            )
+            logits, value = self.model.unroll_logits(s, pa, h0, c0)
+            pi_loss, baseline_loss, entropy = fused_vtrace_loss(
+                logits, value.float(), mu, a, clipped_r, discounts)
+            total = (pi_loss + baseline_loss * self.baseline_loss_coef
+                     + entropy * self.entropy_coef)
+            return pi_loss, baseline_loss, entropy, total
+
+        policy, value = self.model.unroll(s, pa, h0, c0)
         policy = policy.float()
         value = value.float()
 

@@ -101,11 +101,14 @@ class MLPHead(nn.Module):
         self.out = nn.Linear(prev, out_features)
         self.final_activation = final_activation
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def logits(self, x: torch.Tensor) -> torch.Tensor:
         x = x.to(self.out.weight.dtype)
         for layer in self.hidden:
             x = F.relu(layer(x))
-        x = self.out(x)
+        return self.out(x)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.logits(x)
         if self.final_activation == "softmax":
             # softmax/log run in fp32 regardless of the compute dtype
             x = F.softmax(x.float(), dim=-1)

@@ -52,19 +52,37 @@ class ImpalaActorCritic(nn.Module):
     def forward(self, state, prev_action, h, c):
         return self.single_step(state, prev_action, h, c)
 
+    def _unroll_features(self, traj_state, traj_prev_action, traj_h, traj_c):
+        B, T = traj_state.shape[:2]
+        flat_state = traj_state.reshape(B * T, *traj_state.shape[2:])
+        flat_pa = traj_prev_action.reshape(B * T)
+        flat_h = traj_h.reshape(B * T, -1)
+        flat_c = traj_c.reshape(B * T, -1)
+        feat = torch.cat(
+            [self.conv(flat_state), self.action_emb(flat_pa)], dim=1)
+        new_h, _ = self.lstm(feat, flat_h, flat_c)
+        return new_h, B, T
+
     def unroll(self, traj_state: torch.Tensor, traj_prev_action: torch.Tensor,
                traj_h: torch.Tensor, traj_c: torch.Tensor):
         """Batched evaluation over a whole trajectory.
 
         traj_state [B,T,84,84,C], traj_prev_action [B,T],
         traj_h/traj_c [B,T,H] (per-timestep stored actor states).
-        Returns (policy [B,T,A], value [B,T]).
+        Returns (policy softmax [B,T,A] f32, value [B,T]).
         """
-        B, T = traj_state.shape[:2]
-        flat_state = traj_state.reshape(B * T, *traj_state.shape[2:])
-        flat_pa = traj_prev_action.reshape(B * T)
-        flat_h = traj_h.reshape(B * T, -1)
-        flat_c = traj_c.reshape(B * T, -1)
-        policy, value, _, _ = self.single_step(flat_state, flat_pa,
-                                               flat_h, flat_c)
+        h, B, T = self._unroll_features(traj_state, traj_prev_action,
+                                        traj_h, traj_c)
+        policy = self.policy_head(h)
+        value = self.value_head(h).squeeze(-1)
         return policy.reshape(B, T, -1), value.reshape(B, T)
+
+    def unroll_logits(self, traj_state, traj_prev_action, traj_h, traj_c):
+        """Like unroll() but returns pre-softmax logits [B,T,A] (model
+        dtype) — the input of the fused V-trace loss kernel
+        (ops/vtrace_loss_op.py)."""
+        h, B, T = self._unroll_features(traj_state, traj_prev_action,
+                                        traj_h, traj_c)
+        logits = self.policy_head.logits(h)
+        value = self.value_head(h).squeeze(-1)
+        return logits.reshape(B, T, -1), value.reshape(B, T)

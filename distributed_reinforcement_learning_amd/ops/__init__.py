@@ -38,11 +38,12 @@ def require_ext() -> "object":
 
 
 from distributed_reinforcement_learning_amd.ops.vtrace_op import vtrace_scan
+from distributed_reinforcement_learning_amd.ops.vtrace_loss_op import fused_vtrace_loss
 from distributed_reinforcement_learning_amd.ops.preprocess import normalize_frames
 from distributed_reinforcement_learning_amd.ops.lstm_op import lstm_fused_step
 from distributed_reinforcement_learning_amd.ops.optim import FusedRMSProp, FusedAdam
 
 __all__ = [
-    "available", "require_ext", "vtrace_scan", "normalize_frames",
-    "lstm_fused_step", "FusedRMSProp", "FusedAdam",
+    "available", "require_ext", "vtrace_scan", "fused_vtrace_loss",
+    "normalize_frames", "lstm_fused_step", "FusedRMSProp", "FusedAdam",
 ]

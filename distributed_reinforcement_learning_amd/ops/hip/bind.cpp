@@ -25,6 +25,13 @@ extern "C" __global__ void drla_u8_normalize_bf16_tail(const unsigned char*,
                                                        long long, long long);
 extern "C" __global__ void drla_vtrace_scan(const float*, const float*,
                                             const float*, float*, int, int);
+extern "C" __global__ void drla_vtrace_loss_fwd(
+    const unsigned short*, const float*, const float*, const float*,
+    const int*, const float*, const float*, float*, float*, float*, float*,
+    int, int, int);
+extern "C" __global__ void drla_vtrace_loss_bwd(
+    const float*, const float*, const float*, const float*, const int*,
+    const float*, unsigned short*, float*, float*, int, int, int);
 extern "C" __global__ void drla_lstm_tail_fwd(const float*, const float*,
                                               float*, float*, float*, float,
                                               long long, int);
@@ -117,6 +124,56 @@ torch::Tensor vtrace_scan(torch::Tensor deltas, torch::Tensor discounts,
                      discounts.data_ptr<float>(), cs.data_ptr<float>(),
                      out.data_ptr<float>(), B, T);
   return out;
+}
+
+std::vector<torch::Tensor> vtrace_loss_fwd(
+    torch::Tensor logits, torch::Tensor value, torch::Tensor mu,
+    torch::Tensor actions, torch::Tensor rewards, torch::Tensor discounts) {
+  for (auto* t : {&logits, &value, &mu, &actions, &rewards, &discounts})
+    check_gpu_contig(*t, "vtrace_loss input");
+  TORCH_CHECK(actions.scalar_type() == torch::kInt, "actions must be int32");
+  const int B = logits.size(0), T = logits.size(1), A = logits.size(2);
+  TORCH_CHECK(A <= 64, "num_action cap is 64");
+  const bool bf16 = logits.scalar_type() == torch::kBFloat16;
+  auto fopt = value.options().dtype(torch::kFloat);
+  auto p_stash = torch::empty({B, T, A}, fopt);
+  auto vs_stash = torch::empty({B, T - 2}, fopt);
+  auto adv_stash = torch::empty({B, T - 2}, fopt);
+  auto losses = torch::empty({3}, fopt);
+  const int lds = (B * T + B * (T - 2) + 3 * (256 / DRLA_WAVE)) * 4 + 64;
+  hipLaunchKernelGGL(
+      drla_vtrace_loss_fwd, dim3(1), dim3(256), lds, cur_stream(),
+      bf16 ? reinterpret_cast<const unsigned short*>(logits.data_ptr())
+           : nullptr,
+      bf16 ? nullptr : logits.data_ptr<float>(), value.data_ptr<float>(),
+      mu.data_ptr<float>(), actions.data_ptr<int>(),
+      rewards.data_ptr<float>(), discounts.data_ptr<float>(),
+      p_stash.data_ptr<float>(), vs_stash.data_ptr<float>(),
+      adv_stash.data_ptr<float>(), losses.data_ptr<float>(), B, T, A);
+  return {losses, p_stash, vs_stash, adv_stash};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> vtrace_loss_bwd(
+    torch::Tensor p_stash, torch::Tensor vs_stash, torch::Tensor adv_stash,
+    torch::Tensor value, torch::Tensor actions, torch::Tensor grad3,
+    bool want_bf16) {
+  for (auto* t : {&p_stash, &vs_stash, &adv_stash, &value, &actions, &grad3})
+    check_gpu_contig(*t, "vtrace_loss bwd input");
+  const int B = p_stash.size(0), T = p_stash.size(1), A = p_stash.size(2);
+  auto dlogits = torch::empty(
+      {B, T, A},
+      value.options().dtype(want_bf16 ? torch::kBFloat16 : torch::kFloat));
+  auto dvalue = torch::empty({B, T}, value.options().dtype(torch::kFloat));
+  hipLaunchKernelGGL(
+      drla_vtrace_loss_bwd, dim3(drla_grid((long long)B * T)), dim3(256), 0,
+      cur_stream(), p_stash.data_ptr<float>(), vs_stash.data_ptr<float>(),
+      adv_stash.data_ptr<float>(), value.data_ptr<float>(),
+      actions.data_ptr<int>(), grad3.data_ptr<float>(),
+      want_bf16 ? reinterpret_cast<unsigned short*>(dlogits.data_ptr())
+                : nullptr,
+      want_bf16 ? nullptr : dlogits.data_ptr<float>(),
+      dvalue.data_ptr<float>(), B, T, A);
+  return {dlogits, dvalue};
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_tail_fwd(
@@ -291,6 +348,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("normalize_frames_bf16", &normalize_frames_bf16,
         "uint8 frames -> bf16/255 (K11)");
   m.def("vtrace_scan", &vtrace_scan, "fused V-trace reverse scan (K5)");
+  m.def("vtrace_loss_fwd", &vtrace_loss_fwd,
+        "fused IMPALA loss pipeline forward (K5+K7+K13)");
+  m.def("vtrace_loss_bwd", &vtrace_loss_bwd,
+        "fused IMPALA loss pipeline backward (closed form)");
   m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");
   m.def("lstm_tail_bwd", &lstm_tail_bwd, "fused LSTM gate tail bwd (K3)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");

@@ -1,0 +1,207 @@
+// K5+K7+K13 fused: the ENTIRE IMPALA loss pipeline in two
+// kernels.
+//
+// Forward (one workgroup — the whole problem is ~50 KB):
+//   softmax over A per (b,t) -> rho = pi[a]/mu[a] -> two V-trace reverse
+//   scans (first + middle windows) -> pg advantage -> the three losses
+//   (sum reductions, reference agent/impala.py:63-93 semantics).
+// Backward (closed form — everything but log pi(a) and the entropy is
+//   stop-gradient in the reference):
+//   dlogits[b,t,:] = gpi * -adv * sa/(sa+eps) * (onehot - s)
+//                  + ge  * s * (log s - sum_j s_j log s_j)   for t < T-2
+//   dvalue[b,t]    = gb * (v - vs)                           for t < T-2
+//
+// This replaces ~90 eager launches (~400 us/step measured,
+// profiles/impala_bench_r06_bf16_kernels.md) with two ~10 us kernels.
+// Replaces reference optimizer/vtrace.py:29-126 + the loss composition at
+// agent/impala.py:63-100.
+
+#include "drla_common.h"
+
+#define VT_MAX_A 64  // num_action cap (reference max is 18)
+
+typedef unsigned short bf16raw;
+
+__device__ __forceinline__ float vt_ld(const bf16raw* p, long long i) {
+  unsigned int x = ((unsigned int)p[i]) << 16;
+  return __uint_as_float(x);
+}
+
+// logits may be bf16 (GPU-native models) or f32; value f32, mu f32.
+// Layout: [B, T, A] row-major; scalars/scan state in LDS (single WG).
+extern "C" __global__ __launch_bounds__(256)
+void drla_vtrace_loss_fwd(
+    const bf16raw* __restrict__ logits_bf16,   // [B,T,A] (nullable)
+    const float* __restrict__ logits_f32,      // [B,T,A] (nullable)
+    const float* __restrict__ value,           // [B,T]
+    const float* __restrict__ mu,              // [B,T,A]
+    const int* __restrict__ actions,           // [B,T]
+    const float* __restrict__ rewards,         // [B,T] (already clipped)
+    const float* __restrict__ discounts,       // [B,T]
+    float* __restrict__ p_stash,               // [B,T,A] softmax out
+    float* __restrict__ vs_stash,              // [B,T-2]
+    float* __restrict__ adv_stash,             // [B,T-2]
+    float* __restrict__ losses,                // [3]: pi, baseline, entropy
+    int B, int T, int A) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* rho = reinterpret_cast<float*>(smem);          // [B*T]
+  float* vsm = rho + B * T;                             // [B*(T-2)] vs_plus_1
+  float* red = vsm + B * (T - 2);                       // [3 * nwaves]
+
+  const int tid = threadIdx.x;
+  const int nthreads = blockDim.x;
+  const int Tp = T - 2;
+
+  // phase A: softmax + rho per position (multi-pass over L2-resident rows —
+  // a local row array would be runtime-indexed and spill to scratch,
+  // guide §5.4 rule 20)
+  for (int i = tid; i < B * T; i += nthreads) {
+    const long long base = (long long)i * A;
+    float mx = -1e30f;
+    for (int k = 0; k < A; ++k) {
+      const float x = logits_bf16 ? vt_ld(logits_bf16, base + k)
+                                  : logits_f32[base + k];
+      mx = fmaxf(mx, x);
+    }
+    float denom = 0.0f;
+    for (int k = 0; k < A; ++k) {
+      const float x = logits_bf16 ? vt_ld(logits_bf16, base + k)
+                                  : logits_f32[base + k];
+      denom += __expf(x - mx);
+    }
+    const float inv = 1.0f / denom;
+    for (int k = 0; k < A; ++k) {
+      const float x = logits_bf16 ? vt_ld(logits_bf16, base + k)
+                                  : logits_f32[base + k];
+      p_stash[base + k] = __expf(x - mx) * inv;
+    }
+    const int a = actions[i];
+    rho[i] = p_stash[base + a] / mu[base + a];
+  }
+  __syncthreads();
+
+  // phase B: two reverse scans per batch row (lane-per-row, serial in T)
+  for (int b = tid; b < B; b += nthreads) {
+    const float* v = value + (long long)b * T;
+    const float* r = rewards + (long long)b * T;
+    const float* g = discounts + (long long)b * T;
+    const float* rh = rho + b * T;
+    // middle-window scan -> vs_plus_1 (bootstrap v[T-1])
+    float acc = 0.0f;
+    for (int t = Tp - 1; t >= 0; --t) {
+      const float c = fminf(1.0f, rh[t + 1]);
+      const float delta = c * (r[t + 1] + g[t + 1] * v[t + 2] - v[t + 1]);
+      acc = delta + g[t + 1] * c * acc;
+      vsm[b * Tp + t] = acc + v[t + 1];
+    }
+    // first-window scan -> vs (bootstrap v[T-2])
+    acc = 0.0f;
+    for (int t = Tp - 1; t >= 0; --t) {
+      const float c = fminf(1.0f, rh[t]);
+      const float delta = c * (r[t] + g[t] * v[t + 1] - v[t]);
+      acc = delta + g[t] * c * acc;
+      const float vs = acc + v[t];
+      vs_stash[b * Tp + t] = vs;
+      adv_stash[b * Tp + t] =
+          fminf(1.0f, rh[t]) * (r[t] + g[t] * vsm[b * Tp + t] - v[t]);
+    }
+  }
+  __syncthreads();
+
+  // phase C: loss sums over the first window
+  float pi_l = 0.0f, base_l = 0.0f, ent_l = 0.0f;
+  for (int i = tid; i < B * Tp; i += nthreads) {
+    const int b = i / Tp;
+    const int t = i - b * Tp;
+    const long long pos = (long long)b * T + t;
+    const long long pbase = pos * A;
+    const int a = actions[pos];
+    const float adv = adv_stash[i];
+    pi_l -= __logf(p_stash[pbase + a] + 1e-8f) * adv;
+    base_l += 0.5f * (vs_stash[i] - value[pos]) * (vs_stash[i] - value[pos]);
+    float e = 0.0f;
+    for (int k = 0; k < A; ++k) {
+      const float p = p_stash[pbase + k];
+      e += p * __logf(p);
+    }
+    ent_l += e;  // = negative entropy, matching reference vtrace.py:120-126
+  }
+  // block reduction
+  for (int off = DRLA_WAVE / 2; off > 0; off >>= 1) {
+    pi_l += __shfl_down(pi_l, off, DRLA_WAVE);
+    base_l += __shfl_down(base_l, off, DRLA_WAVE);
+    ent_l += __shfl_down(ent_l, off, DRLA_WAVE);
+  }
+  const int wave = tid / DRLA_WAVE;
+  const int lane = tid % DRLA_WAVE;
+  const int nwaves = nthreads / DRLA_WAVE;
+  if (lane == 0) {
+    red[wave] = pi_l;
+    red[nwaves + wave] = base_l;
+    red[2 * nwaves + wave] = ent_l;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    float s0 = 0, s1 = 0, s2 = 0;
+    for (int w = 0; w < nwaves; ++w) {
+      s0 += red[w];
+      s1 += red[nwaves + w];
+      s2 += red[2 * nwaves + w];
+    }
+    losses[0] = s0;
+    losses[1] = s1;
+    losses[2] = s2;
+  }
+}
+
+// one thread per (b,t): writes the full A-row of dlogits + dvalue.
+// gpi/gb/ge are the upstream gradients of the three loss outputs
+// (1, baseline_coef, entropy_coef when total.backward() is called).
+extern "C" __global__ void drla_vtrace_loss_bwd(
+    const float* __restrict__ p_stash, const float* __restrict__ vs_stash,
+    const float* __restrict__ adv_stash, const float* __restrict__ value,
+    const int* __restrict__ actions, const float* __restrict__ grad3,
+    bf16raw* __restrict__ dlogits_bf16, float* __restrict__ dlogits_f32,
+    float* __restrict__ dvalue, int B, int T, int A) {
+  const float gpi = grad3[0];
+  const float gb = grad3[1];
+  const float ge = grad3[2];
+  const int Tp = T - 2;
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long total = (long long)B * T;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < total; i += stride) {
+    const int b = i / T;
+    const int t = i - (long long)b * T;
+    const long long pbase = i * A;
+    if (t >= Tp) {
+      for (int k = 0; k < A; ++k) {
+        if (dlogits_bf16) dlogits_bf16[pbase + k] = 0;
+        else dlogits_f32[pbase + k] = 0.0f;
+      }
+      dvalue[i] = 0.0f;
+      continue;
+    }
+    const int wi = b * Tp + t;
+    const int a = actions[i];
+    const float adv = adv_stash[wi];
+    const float sa = p_stash[pbase + a];
+    const float w = sa / (sa + 1e-8f);
+    // row entropy term E = sum_j s_j log s_j
+    float E = 0.0f;
+    for (int k = 0; k < A; ++k) {
+      const float s = p_stash[pbase + k];
+      E += s * __logf(s);
+    }
+    for (int k = 0; k < A; ++k) {
+      const float s = p_stash[pbase + k];
+      const float onehot = (k == a) ? 1.0f : 0.0f;
+      const float d_pi = -adv * w * (onehot - s);
+      const float d_ent = s * (__logf(s) - E);
+      const float d = gpi * d_pi + ge * d_ent;
+      if (dlogits_bf16) dlogits_bf16[pbase + k] = drla_f32_to_bf16(d);
+      else dlogits_f32[pbase + k] = d;
+    }
+    dvalue[i] = gb * (value[i] - vs_stash[wi]);
+  }
+}
