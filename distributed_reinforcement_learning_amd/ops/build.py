@@ -27,6 +27,7 @@ SOURCES = [
     "bind.cpp",
     "elementwise.hip",
     "vtrace.hip",
+    "vtrace_loss.hip",
     "lstm_gates.hip",
     "optim.hip",
     "conv.hip",
@@ -75,8 +76,16 @@ def build(verbose: bool = True, force: bool = False) -> str:
     if verbose:
         print("[drla build]", " ".join(cmd))
     subprocess.run(cmd, check=True)
+    # import self-check in a FRESH interpreter: a kernel missing from SOURCES
+    # only surfaces as an undefined __device_stub__ symbol at import time
+    repo_root = os.path.dirname(os.path.dirname(HERE))
+    subprocess.run(
+        [sys.executable, "-c",
+         "import distributed_reinforcement_learning_amd.ops as o; "
+         "assert o.available(), o._IMPORT_ERROR"],
+        check=True, cwd=repo_root)
     if verbose:
-        print(f"[drla build] wrote {OUT_SO}")
+        print(f"[drla build] wrote {OUT_SO} (import check OK)")
     return OUT_SO
 
 
