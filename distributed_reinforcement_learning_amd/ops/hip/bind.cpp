@@ -23,6 +23,12 @@ extern "C" __global__ void drla_u8_normalize_f32_tail(const unsigned char*,
 extern "C" __global__ void drla_u8_normalize_bf16_tail(const unsigned char*,
                                                        unsigned short*,
                                                        long long, long long);
+extern "C" __global__ void drla_embed_bwd_scatter(
+    const long long*, const unsigned short*, const float*, float*,
+    long long, int);
+extern "C" __global__ void drla_f32_to_bf16_kernel(const float*,
+                                                   unsigned short*,
+                                                   long long);
 extern "C" __global__ void drla_vtrace_scan(const float*, const float*,
                                             const float*, float*, int, int);
 extern "C" __global__ void drla_vtrace_loss_fwd(
@@ -126,6 +132,34 @@ torch::Tensor vtrace_scan(torch::Tensor deltas, torch::Tensor discounts,
   return out;
 }
 
+torch::Tensor embed_bwd(torch::Tensor indices, torch::Tensor grad_out,
+                        int64_t num_rows, bool want_bf16) {
+  check_gpu_contig(indices, "indices");
+  check_gpu_contig(grad_out, "grad_out");
+  TORCH_CHECK(indices.scalar_type() == torch::kLong, "indices must be i64");
+  const long long N = grad_out.size(0);
+  const int H = grad_out.size(1);
+  const bool in16 = grad_out.scalar_type() == torch::kBFloat16;
+  auto scratch = torch::zeros({num_rows, H},
+                              grad_out.options().dtype(torch::kFloat));
+  hipLaunchKernelGGL(
+      drla_embed_bwd_scatter, dim3(drla_grid(N * H)), dim3(DRLA_BLOCK), 0,
+      cur_stream(), indices.data_ptr<int64_t>(),
+      in16 ? reinterpret_cast<const unsigned short*>(grad_out.data_ptr())
+           : nullptr,
+      in16 ? nullptr : grad_out.data_ptr<float>(), scratch.data_ptr<float>(),
+      N, H);
+  if (!want_bf16) return scratch;
+  auto out = torch::empty({num_rows, H},
+                          grad_out.options().dtype(torch::kBFloat16));
+  hipLaunchKernelGGL(drla_f32_to_bf16_kernel,
+                     dim3(drla_grid(num_rows * H)), dim3(DRLA_BLOCK), 0,
+                     cur_stream(), scratch.data_ptr<float>(),
+                     reinterpret_cast<unsigned short*>(out.data_ptr()),
+                     (long long)(num_rows * H));
+  return out;
+}
+
 std::vector<torch::Tensor> vtrace_loss_fwd(
     torch::Tensor logits, torch::Tensor value, torch::Tensor mu,
     torch::Tensor actions, torch::Tensor rewards, torch::Tensor discounts) {
@@ -136,13 +170,13 @@ std::vector<torch::Tensor> vtrace_loss_fwd(
   TORCH_CHECK(A <= 64, "num_action cap is 64");
   const bool bf16 = logits.scalar_type() == torch::kBFloat16;
   auto fopt = value.options().dtype(torch::kFloat);
+  TORCH_CHECK(T <= 128, "trajectory cap is 128 (VT_MAX_T)");
   auto p_stash = torch::empty({B, T, A}, fopt);
   auto vs_stash = torch::empty({B, T - 2}, fopt);
   auto adv_stash = torch::empty({B, T - 2}, fopt);
-  auto losses = torch::empty({3}, fopt);
-  const int lds = (B * T + B * (T - 2) + 3 * (256 / DRLA_WAVE)) * 4 + 64;
+  auto losses = torch::zeros({3}, fopt);  // blocks atomicAdd into it
   hipLaunchKernelGGL(
-      drla_vtrace_loss_fwd, dim3(1), dim3(256), lds, cur_stream(),
+      drla_vtrace_loss_fwd, dim3(B), dim3(256), 0, cur_stream(),
       bf16 ? reinterpret_cast<const unsigned short*>(logits.data_ptr())
            : nullptr,
       bf16 ? nullptr : logits.data_ptr<float>(), value.data_ptr<float>(),
@@ -352,6 +386,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused IMPALA loss pipeline forward (K5+K7+K13)");
   m.def("vtrace_loss_bwd", &vtrace_loss_bwd,
         "fused IMPALA loss pipeline backward (closed form)");
+  m.def("embed_bwd", &embed_bwd,
+        "action-embedding table gradient (K2 backward)");
   m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");
   m.def("lstm_tail_bwd", &lstm_tail_bwd, "fused LSTM gate tail bwd (K3)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");

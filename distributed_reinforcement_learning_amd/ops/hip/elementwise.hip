@@ -51,3 +51,38 @@ extern "C" __global__ void drla_u8_normalize_bf16_tail(
   long long i = start + blockIdx.x * (long long)blockDim.x + threadIdx.x;
   if (i < n) out[i] = drla_f32_to_bf16(in[i] * (1.0f / 255.0f));
 }
+
+// K2 backward: action-embedding table gradient.
+// torch's embedding_backward_feature_kernel costs ~38us/step on the tiny
+// [A<=64, 256] table; here: per-(n,h) atomicAdd into an f32 scratch
+// (A*H distinct addresses, deep L2 combining), then one cast kernel.
+extern "C" __global__ void drla_embed_bwd_scatter(
+    const long long* __restrict__ indices,      // [N]
+    const unsigned short* __restrict__ gout16,  // [N,H] bf16 (nullable)
+    const float* __restrict__ gout32,           // [N,H] f32 (nullable)
+    float* __restrict__ scratch,                // [A,H] f32, zeroed
+    long long N, int H) {
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long total = N * H;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < total; i += stride) {
+    const long long n = i / H;
+    const int h = i - n * H;
+    float g;
+    if (gout16) {
+      unsigned int x = ((unsigned int)gout16[i]) << 16;
+      g = __uint_as_float(x);
+    } else {
+      g = gout32[i];
+    }
+    atomicAdd(&scratch[indices[n] * H + h], g);
+  }
+}
+
+extern "C" __global__ void drla_f32_to_bf16_kernel(
+    const float* __restrict__ in, unsigned short* __restrict__ out,
+    long long n) {
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < n; i += stride) out[i] = drla_f32_to_bf16(in[i]);
+}

@@ -1,24 +1,22 @@
 // K5+K7+K13 fused: the ENTIRE IMPALA loss pipeline in two
 // kernels.
 //
-// Forward (one workgroup — the whole problem is ~50 KB):
-//   softmax over A per (b,t) -> rho = pi[a]/mu[a] -> two V-trace reverse
-//   scans (first + middle windows) -> pg advantage -> the three losses
-//   (sum reductions, reference agent/impala.py:63-93 semantics).
+// Forward: one workgroup per batch row (rows are fully independent —
+// softmax over A per (b,t) -> rho = pi[a]/mu[a] -> two V-trace reverse scans
+// (first + middle windows) -> pg advantage -> the three losses, accumulated
+// with one atomicAdd triple per block). Semantics: reference
+// agent/impala.py:63-100 + optimizer/vtrace.py:29-126 (sum reductions).
+//
 // Backward (closed form — everything but log pi(a) and the entropy is
-//   stop-gradient in the reference):
+// stop-gradient in the reference):
 //   dlogits[b,t,:] = gpi * -adv * sa/(sa+eps) * (onehot - s)
 //                  + ge  * s * (log s - sum_j s_j log s_j)   for t < T-2
 //   dvalue[b,t]    = gb * (v - vs)                           for t < T-2
 //
-// This replaces ~90 eager launches (~400 us/step measured,
-// profiles/impala_bench_r06_bf16_kernels.md) with two ~10 us kernels.
-// Replaces reference optimizer/vtrace.py:29-126 + the loss composition at
-// agent/impala.py:63-100.
+// Replaces ~90 eager launches (~400 us/step measured,
+// profiles/impala_bench_r06_bf16_kernels.md) with two small kernels.
 
 #include "drla_common.h"
-
-#define VT_MAX_A 64  // num_action cap (reference max is 18)
 
 typedef unsigned short bf16raw;
 
@@ -27,8 +25,9 @@ __device__ __forceinline__ float vt_ld(const bf16raw* p, long long i) {
   return __uint_as_float(x);
 }
 
-// logits may be bf16 (GPU-native models) or f32; value f32, mu f32.
-// Layout: [B, T, A] row-major; scalars/scan state in LDS (single WG).
+#define VT_MAX_T 128  // trajectory-length cap for the LDS scan buffers
+
+// grid = B blocks of 256; losses[3] must be ZERO on entry (atomicAdd).
 extern "C" __global__ __launch_bounds__(256)
 void drla_vtrace_loss_fwd(
     const bf16raw* __restrict__ logits_bf16,   // [B,T,A] (nullable)
@@ -43,20 +42,20 @@ void drla_vtrace_loss_fwd(
     float* __restrict__ adv_stash,             // [B,T-2]
     float* __restrict__ losses,                // [3]: pi, baseline, entropy
     int B, int T, int A) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* rho = reinterpret_cast<float*>(smem);          // [B*T]
-  float* vsm = rho + B * T;                             // [B*(T-2)] vs_plus_1
-  float* red = vsm + B * (T - 2);                       // [3 * nwaves]
+  __shared__ float rho[VT_MAX_T];
+  __shared__ float vsm[VT_MAX_T];
+  __shared__ float red[3 * 4];
 
+  const int b = blockIdx.x;
   const int tid = threadIdx.x;
-  const int nthreads = blockDim.x;
   const int Tp = T - 2;
+  const long long row = (long long)b * T;
 
-  // phase A: softmax + rho per position (multi-pass over L2-resident rows —
-  // a local row array would be runtime-indexed and spill to scratch,
+  // phase A: softmax + rho, one thread per t (multi-pass over the
+  // L2-resident row — a local A-array would be runtime-indexed and spill,
   // guide §5.4 rule 20)
-  for (int i = tid; i < B * T; i += nthreads) {
-    const long long base = (long long)i * A;
+  for (int t = tid; t < T; t += blockDim.x) {
+    const long long base = (row + t) * (long long)A;
     float mx = -1e30f;
     for (int k = 0; k < A; ++k) {
       const float x = logits_bf16 ? vt_ld(logits_bf16, base + k)
@@ -75,58 +74,53 @@ void drla_vtrace_loss_fwd(
                                   : logits_f32[base + k];
       p_stash[base + k] = __expf(x - mx) * inv;
     }
-    const int a = actions[i];
-    rho[i] = p_stash[base + a] / mu[base + a];
+    const int a = actions[row + t];
+    rho[t] = p_stash[base + a] / mu[base + a];
   }
   __syncthreads();
 
-  // phase B: two reverse scans per batch row (lane-per-row, serial in T)
-  for (int b = tid; b < B; b += nthreads) {
-    const float* v = value + (long long)b * T;
-    const float* r = rewards + (long long)b * T;
-    const float* g = discounts + (long long)b * T;
-    const float* rh = rho + b * T;
+  // phase B: the two reverse scans, serial in T (thread 0)
+  if (tid == 0) {
+    const float* v = value + row;
+    const float* r = rewards + row;
+    const float* g = discounts + row;
     // middle-window scan -> vs_plus_1 (bootstrap v[T-1])
     float acc = 0.0f;
     for (int t = Tp - 1; t >= 0; --t) {
-      const float c = fminf(1.0f, rh[t + 1]);
+      const float c = fminf(1.0f, rho[t + 1]);
       const float delta = c * (r[t + 1] + g[t + 1] * v[t + 2] - v[t + 1]);
       acc = delta + g[t + 1] * c * acc;
-      vsm[b * Tp + t] = acc + v[t + 1];
+      vsm[t] = acc + v[t + 1];
     }
     // first-window scan -> vs (bootstrap v[T-2])
     acc = 0.0f;
     for (int t = Tp - 1; t >= 0; --t) {
-      const float c = fminf(1.0f, rh[t]);
+      const float c = fminf(1.0f, rho[t]);
       const float delta = c * (r[t] + g[t] * v[t + 1] - v[t]);
       acc = delta + g[t] * c * acc;
       const float vs = acc + v[t];
       vs_stash[b * Tp + t] = vs;
-      adv_stash[b * Tp + t] =
-          fminf(1.0f, rh[t]) * (r[t] + g[t] * vsm[b * Tp + t] - v[t]);
+      adv_stash[b * Tp + t] = c * (r[t] + g[t] * vsm[t] - v[t]);
     }
   }
   __syncthreads();
 
-  // phase C: loss sums over the first window
+  // phase C: loss sums over this row's first window, (t, k) split across
+  // threads for the entropy inner loop
   float pi_l = 0.0f, base_l = 0.0f, ent_l = 0.0f;
-  for (int i = tid; i < B * Tp; i += nthreads) {
-    const int b = i / Tp;
-    const int t = i - b * Tp;
-    const long long pos = (long long)b * T + t;
-    const long long pbase = pos * A;
-    const int a = actions[pos];
-    const float adv = adv_stash[i];
-    pi_l -= __logf(p_stash[pbase + a] + 1e-8f) * adv;
-    base_l += 0.5f * (vs_stash[i] - value[pos]) * (vs_stash[i] - value[pos]);
-    float e = 0.0f;
-    for (int k = 0; k < A; ++k) {
-      const float p = p_stash[pbase + k];
-      e += p * __logf(p);
+  for (int i = tid; i < Tp * A; i += blockDim.x) {
+    const int t = i / A;
+    const int k = i - t * A;
+    const long long pbase = (row + t) * (long long)A;
+    const float p = p_stash[pbase + k];
+    ent_l += p * __logf(p);  // negative entropy (reference vtrace.py:120)
+    if (k == 0) {
+      const float adv = adv_stash[b * Tp + t];
+      pi_l -= __logf(p_stash[pbase + actions[row + t]] + 1e-8f) * adv;
+      const float diff = vs_stash[b * Tp + t] - value[row + t];
+      base_l += 0.5f * diff * diff;
     }
-    ent_l += e;  // = negative entropy, matching reference vtrace.py:120-126
   }
-  // block reduction
   for (int off = DRLA_WAVE / 2; off > 0; off >>= 1) {
     pi_l += __shfl_down(pi_l, off, DRLA_WAVE);
     base_l += __shfl_down(base_l, off, DRLA_WAVE);
@@ -134,23 +128,22 @@ void drla_vtrace_loss_fwd(
   }
   const int wave = tid / DRLA_WAVE;
   const int lane = tid % DRLA_WAVE;
-  const int nwaves = nthreads / DRLA_WAVE;
   if (lane == 0) {
     red[wave] = pi_l;
-    red[nwaves + wave] = base_l;
-    red[2 * nwaves + wave] = ent_l;
+    red[4 + wave] = base_l;
+    red[8 + wave] = ent_l;
   }
   __syncthreads();
   if (tid == 0) {
     float s0 = 0, s1 = 0, s2 = 0;
-    for (int w = 0; w < nwaves; ++w) {
+    for (int w = 0; w < (int)(blockDim.x / DRLA_WAVE); ++w) {
       s0 += red[w];
-      s1 += red[nwaves + w];
-      s2 += red[2 * nwaves + w];
+      s1 += red[4 + w];
+      s2 += red[8 + w];
     }
-    losses[0] = s0;
-    losses[1] = s1;
-    losses[2] = s2;
+    atomicAdd(&losses[0], s0);
+    atomicAdd(&losses[1], s1);
+    atomicAdd(&losses[2], s2);
   }
 }
 

@@ -191,6 +191,23 @@ def test_impala_gpu_matches_cpu_reference_forward():
     assert torch.allclose(V_cpu, V_gpu.float().cpu(), atol=0.15, rtol=0.1)
 
 
+def test_embed_backward_matches_torch(ext):
+    torch.manual_seed(9)
+    A, H, N = 18, 256, 640
+    table1 = torch.randn(A, H, device="cuda", requires_grad=True)
+    table2 = table1.detach().clone().requires_grad_(True)
+    idx = torch.randint(0, A, (N,), device="cuda")
+    go = torch.randn(N, H, device="cuda")
+    from distributed_reinforcement_learning_amd.ops.embed_op import embed_lookup
+    embed_lookup(table1, idx).backward(go)
+    torch.nn.functional.embedding(idx, table2).backward(go)
+    assert torch.allclose(table1.grad, table2.grad, atol=1e-4, rtol=1e-4)
+    # bf16 table path
+    t3 = torch.randn(A, H, device="cuda").bfloat16().requires_grad_(True)
+    embed_lookup(t3, idx).float().sum().backward()
+    assert t3.grad.dtype == torch.bfloat16
+
+
 def test_mixed_precision_rmsprop_master(ext):
     """bf16-model / fp32-master fused RMSProp vs manual fp32 math."""
     torch.manual_seed(7)
