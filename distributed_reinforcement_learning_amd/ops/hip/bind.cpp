@@ -144,7 +144,8 @@ torch::Tensor embed_bwd(torch::Tensor indices, torch::Tensor grad_out,
                               grad_out.options().dtype(torch::kFloat));
   hipLaunchKernelGGL(
       drla_embed_bwd_scatter, dim3(drla_grid(N * H)), dim3(DRLA_BLOCK), 0,
-      cur_stream(), indices.data_ptr<int64_t>(),
+      cur_stream(),
+      reinterpret_cast<const long long*>(indices.data_ptr<int64_t>()),
       in16 ? reinterpret_cast<const unsigned short*>(grad_out.data_ptr())
            : nullptr,
       in16 ? nullptr : grad_out.data_ptr<float>(), scratch.data_ptr<float>(),
