@@ -95,9 +95,21 @@ class AgentBase:
         t = torch.as_tensor(np.asarray(frames))
         if t.dtype == torch.uint8:
             t = t.to(self.device, non_blocking=True)
-            from distributed_reinforcement_learning_amd.ops import normalize_frames
-            return normalize_frames(t, out_dtype=self.model_dtype)
+            return self.prepare_frames(t)
         return t.to(self.device, dtype=torch.float32, non_blocking=True)
+
+    def prepare_frames(self, t_u8: torch.Tensor) -> torch.Tensor:
+        """Device-side frame prep. With the custom MFMA conv stack active
+        (bf16 GPU model, 84x84 u8 frames) the /255 normalize is fused into
+        conv layer 1, so frames stay uint8; otherwise normalize here."""
+        if (t_u8.is_cuda and self.model_dtype == torch.bfloat16
+                and t_u8.shape[-3:-1] == (84, 84)
+                and t_u8.shape[-1] in (1, 4)):
+            from distributed_reinforcement_learning_amd import ops as _o
+            if _o.available():
+                return t_u8
+        from distributed_reinforcement_learning_amd.ops import normalize_frames
+        return normalize_frames(t_u8, out_dtype=self.model_dtype)
 
     # -- model plumbing (subclasses set self.model / self.optimizer) ---------
 

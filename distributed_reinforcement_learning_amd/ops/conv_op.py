@@ -1,0 +1,77 @@
+"""Custom MFMA conv stack autograd wrapper (K1, ops/hip/conv.hip).
+
+Layer ids (conv.hip convcfg): 0 = 84x84x4 u8 -> 20x20x32 (IMPALA/A3C/Ape-X),
+1 = 84x84x1 u8 (R2D2 POMDP), 2 = 20x20x32 -> 9x9x64, 3 = 9x9x64 -> 7x7x64.
+
+The forward fuses /255-normalize (u8 layers), bias and ReLU; the backward is
+relu-mask -> custom wgrad (MFMA, split-M atomics) + dgrad (MFMA implicit
+GEMM) + bias column reduction. Weight layout is the channels_last flat view
+[CO][KH*KW*CI], built in-graph so grads flow back to the nn.Conv2d weight.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from distributed_reinforcement_learning_amd import ops as _ops
+
+_LAYER_CO = {0: 32, 1: 32, 2: 64, 3: 64}
+
+
+class _ConvLayer(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, w_flat: torch.Tensor,
+                bias_f32: torch.Tensor, layer: int):
+        ext = _ops.require_ext()
+        y = ext.conv_fwd(layer, x.contiguous(), w_flat.contiguous(),
+                         bias_f32.contiguous())
+        ctx.save_for_backward(x, w_flat, y)
+        ctx.layer = layer
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, w_flat, y = ctx.saved_tensors
+        ext = _ops.require_ext()
+        layer = ctx.layer
+        co = _LAYER_CO[layer]
+        dy_m = ext.relu_mask_bwd(dy.contiguous(), y)
+        dw = ext.conv_wgrad(layer, x, dy_m)
+        dbias = ext.bias_grad(dy_m, co)
+        dx = None
+        if layer >= 2 and ctx.needs_input_grad[0]:
+            dx = ext.conv_dgrad(layer, dy_m, w_flat)
+        return dx, dw, dbias, None
+
+
+def conv_layer(x: torch.Tensor, conv: torch.nn.Conv2d,
+               layer: int) -> torch.Tensor:
+    """x NHWC (u8 for layers 0/1, bf16 for 2/3); returns NHWC bf16 output.
+
+    conv.weight must be bf16 channels_last ([CO][KH][KW][CI] physically) —
+    the flat view below is then a no-copy reshape.
+    """
+    w = conv.weight
+    co = w.shape[0]
+    w_flat = w.permute(0, 2, 3, 1).reshape(co, -1)
+    return _ConvLayer.apply(x, w_flat, conv.bias.float(), layer)
+
+
+def atari_conv_stack(stack, x_u8_nhwc: torch.Tensor) -> torch.Tensor:
+    """Full custom stack: u8 [N,84,84,C] -> flat [N,3136] bf16 features,
+    flattened in NHWC order (identical to the torch path in
+    models/blocks.AtariConvStack)."""
+    ci = x_u8_nhwc.shape[-1]
+    l1 = 0 if ci == 4 else 1
+    y = conv_layer(x_u8_nhwc, stack.conv1, l1)
+    y = conv_layer(y, stack.conv2, 2)
+    y = conv_layer(y, stack.conv3, 3)
+    return y.reshape(y.shape[0], -1)
+
+
+def custom_stack_ok(stack, x: torch.Tensor) -> bool:
+    return (x.is_cuda and x.dtype == torch.uint8
+            and x.shape[1] == 84 and x.shape[2] == 84
+            and x.shape[3] in (1, 4)
+            and stack.conv1.weight.dtype == torch.bfloat16
+            and _ops.available())

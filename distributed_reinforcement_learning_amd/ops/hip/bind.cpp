@@ -23,6 +23,50 @@ extern "C" __global__ void drla_u8_normalize_f32_tail(const unsigned char*,
 extern "C" __global__ void drla_u8_normalize_bf16_tail(const unsigned char*,
                                                        unsigned short*,
                                                        long long, long long);
+// conv stack (conv.hip)
+extern "C" __global__ void drla_mfma_probe(const unsigned short*,
+                                           const unsigned short*, float*);
+extern "C" __global__ void drla_conv_fwd_l1(const unsigned char*,
+                                            const unsigned short*,
+                                            const float*, unsigned short*,
+                                            int);
+extern "C" __global__ void drla_conv_fwd_l1_c1(const unsigned char*,
+                                               const unsigned short*,
+                                               const float*, unsigned short*,
+                                               int);
+extern "C" __global__ void drla_conv_fwd_l2(const unsigned short*,
+                                            const unsigned short*,
+                                            const float*, unsigned short*,
+                                            int);
+extern "C" __global__ void drla_conv_fwd_l3(const unsigned short*,
+                                            const unsigned short*,
+                                            const float*, unsigned short*,
+                                            int);
+extern "C" __global__ void drla_relu_mask_bwd(const unsigned short*,
+                                              const unsigned short*,
+                                              unsigned short*, long long);
+extern "C" __global__ void drla_bias_grad(const unsigned short*, float*,
+                                          long long, int);
+extern "C" __global__ void drla_wgrad_finalize(const float*, unsigned short*,
+                                               int, int);
+extern "C" __global__ void drla_conv_wgrad_l1(const unsigned char*,
+                                              const unsigned short*, float*,
+                                              int);
+extern "C" __global__ void drla_conv_wgrad_l1_c1(const unsigned char*,
+                                                 const unsigned short*,
+                                                 float*, int);
+extern "C" __global__ void drla_conv_wgrad_l2(const unsigned short*,
+                                              const unsigned short*, float*,
+                                              int);
+extern "C" __global__ void drla_conv_wgrad_l3(const unsigned short*,
+                                              const unsigned short*, float*,
+                                              int);
+extern "C" __global__ void drla_conv_dgrad_l2(const unsigned short*,
+                                              const unsigned short*,
+                                              unsigned short*, int);
+extern "C" __global__ void drla_conv_dgrad_l3(const unsigned short*,
+                                              const unsigned short*,
+                                              unsigned short*, int);
 extern "C" __global__ void drla_embed_bwd_scatter(
     const long long*, const unsigned short*, const float*, float*,
     long long, int);
@@ -130,6 +174,156 @@ torch::Tensor vtrace_scan(torch::Tensor deltas, torch::Tensor discounts,
                      discounts.data_ptr<float>(), cs.data_ptr<float>(),
                      out.data_ptr<float>(), B, T);
   return out;
+}
+
+// ---- conv stack ----------------------------------------------------------
+namespace convcfg {
+struct Layer { int ci, co, kh, kw, stride, hi, wi, ho, wo; };
+static const Layer L[4] = {
+    {4, 32, 8, 8, 4, 84, 84, 20, 20},   // 0: l1 (u8 x4)
+    {1, 32, 8, 8, 4, 84, 84, 20, 20},   // 1: l1_c1 (u8 x1)
+    {32, 64, 4, 4, 2, 20, 20, 9, 9},    // 2: l2
+    {64, 64, 3, 3, 1, 9, 9, 7, 7},      // 3: l3
+};
+}  // namespace convcfg
+
+static const unsigned short* u16p(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+static unsigned short* u16pm(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  check_gpu_contig(A, "A");
+  check_gpu_contig(B, "B");
+  auto D = torch::empty({16, 16}, A.options().dtype(torch::kFloat));
+  hipLaunchKernelGGL(drla_mfma_probe, dim3(1), dim3(64), 0, cur_stream(),
+                     u16p(A), u16p(B), D.data_ptr<float>());
+  return D;
+}
+
+torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
+                       torch::Tensor bias) {
+  check_gpu_contig(in, "in");
+  check_gpu_contig(w, "w");
+  check_gpu_contig(bias, "bias");
+  const auto& cfg = convcfg::L[layer];
+  TORCH_CHECK(w.scalar_type() == torch::kBFloat16, "weights must be bf16");
+  TORCH_CHECK(bias.scalar_type() == torch::kFloat, "bias must be f32");
+  const int batch = in.size(0);
+  const int M = batch * cfg.ho * cfg.wo;
+  auto out = torch::empty({batch, cfg.ho, cfg.wo, cfg.co},
+                          in.options().dtype(torch::kBFloat16));
+  const int grid = (M + 127) / 128;
+  auto* outp = u16pm(out);
+  switch (layer) {
+    case 0:
+      hipLaunchKernelGGL(drla_conv_fwd_l1, dim3(grid), dim3(256), 0,
+                         cur_stream(), in.data_ptr<uint8_t>(), u16p(w),
+                         bias.data_ptr<float>(), outp, batch);
+      break;
+    case 1:
+      hipLaunchKernelGGL(drla_conv_fwd_l1_c1, dim3(grid), dim3(256), 0,
+                         cur_stream(), in.data_ptr<uint8_t>(), u16p(w),
+                         bias.data_ptr<float>(), outp, batch);
+      break;
+    case 2:
+      hipLaunchKernelGGL(drla_conv_fwd_l2, dim3(grid), dim3(256), 0,
+                         cur_stream(), u16p(in), u16p(w),
+                         bias.data_ptr<float>(), outp, batch);
+      break;
+    case 3:
+      hipLaunchKernelGGL(drla_conv_fwd_l3, dim3(grid), dim3(256), 0,
+                         cur_stream(), u16p(in), u16p(w),
+                         bias.data_ptr<float>(), outp, batch);
+      break;
+    default:
+      TORCH_CHECK(false, "bad layer");
+  }
+  return out;
+}
+
+torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y) {
+  check_gpu_contig(dy, "dy");
+  check_gpu_contig(y, "y");
+  auto out = torch::empty_like(dy);
+  const long long n = dy.numel();
+  hipLaunchKernelGGL(drla_relu_mask_bwd, dim3(drla_grid(n)), dim3(DRLA_BLOCK),
+                     0, cur_stream(), u16p(dy), u16p(y), u16pm(out), n);
+  return out;
+}
+
+torch::Tensor bias_grad(torch::Tensor dy_flat, int64_t CO) {
+  check_gpu_contig(dy_flat, "dy");
+  const long long M = dy_flat.numel() / CO;
+  auto out = torch::zeros({CO}, dy_flat.options().dtype(torch::kFloat));
+  hipLaunchKernelGGL(drla_bias_grad, dim3((CO + 63) / 64), dim3(256), 0,
+                     cur_stream(), u16p(dy_flat), out.data_ptr<float>(), M,
+                     (int)CO);
+  return out;
+}
+
+torch::Tensor conv_wgrad(int layer, torch::Tensor in, torch::Tensor dy) {
+  check_gpu_contig(in, "in");
+  check_gpu_contig(dy, "dy");
+  const auto& cfg = convcfg::L[layer];
+  const int batch = in.size(0);
+  const int K = cfg.kh * cfg.kw * cfg.ci;
+  auto scratch = torch::zeros({K, cfg.co},
+                              dy.options().dtype(torch::kFloat));
+  const int split = (layer <= 1) ? 64 : 32;
+  dim3 grid((K + 63) / 64, split);
+  switch (layer) {
+    case 0:
+      hipLaunchKernelGGL(drla_conv_wgrad_l1, grid, dim3(256), 0,
+                         cur_stream(), in.data_ptr<uint8_t>(), u16p(dy),
+                         scratch.data_ptr<float>(), batch);
+      break;
+    case 1:
+      hipLaunchKernelGGL(drla_conv_wgrad_l1_c1, grid, dim3(256), 0,
+                         cur_stream(), in.data_ptr<uint8_t>(), u16p(dy),
+                         scratch.data_ptr<float>(), batch);
+      break;
+    case 2:
+      hipLaunchKernelGGL(drla_conv_wgrad_l2, grid, dim3(256), 0,
+                         cur_stream(), u16p(in), u16p(dy),
+                         scratch.data_ptr<float>(), batch);
+      break;
+    case 3:
+      hipLaunchKernelGGL(drla_conv_wgrad_l3, grid, dim3(256), 0,
+                         cur_stream(), u16p(in), u16p(dy),
+                         scratch.data_ptr<float>(), batch);
+      break;
+    default:
+      TORCH_CHECK(false, "bad layer");
+  }
+  auto dw = torch::empty({cfg.co, K}, dy.options().dtype(torch::kBFloat16));
+  hipLaunchKernelGGL(drla_wgrad_finalize,
+                     dim3(drla_grid((long long)K * cfg.co)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     scratch.data_ptr<float>(), u16pm(dw), K, cfg.co);
+  return dw;
+}
+
+torch::Tensor conv_dgrad(int layer, torch::Tensor dy, torch::Tensor w) {
+  check_gpu_contig(dy, "dy");
+  check_gpu_contig(w, "w");
+  TORCH_CHECK(layer == 2 || layer == 3, "dgrad only for l2/l3");
+  const auto& cfg = convcfg::L[layer];
+  const int batch = dy.size(0);
+  const int M2 = batch * cfg.hi * cfg.wi;
+  auto dx = torch::empty({batch, cfg.hi, cfg.wi, cfg.ci},
+                         dy.options().dtype(torch::kBFloat16));
+  const int grid = (M2 + 127) / 128;
+  if (layer == 2) {
+    hipLaunchKernelGGL(drla_conv_dgrad_l2, dim3(grid), dim3(256), 0,
+                       cur_stream(), u16p(dy), u16p(w), u16pm(dx), batch);
+  } else {
+    hipLaunchKernelGGL(drla_conv_dgrad_l3, dim3(grid), dim3(256), 0,
+                       cur_stream(), u16p(dy), u16p(w), u16pm(dx), batch);
+  }
+  return dx;
 }
 
 torch::Tensor embed_bwd(torch::Tensor indices, torch::Tensor grad_out,
@@ -387,6 +581,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused IMPALA loss pipeline forward (K5+K7+K13)");
   m.def("vtrace_loss_bwd", &vtrace_loss_bwd,
         "fused IMPALA loss pipeline backward (closed form)");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("conv_fwd", &conv_fwd,
+        "MFMA implicit-GEMM conv fwd, fused normalize+bias+ReLU (K1)");
+  m.def("conv_wgrad", &conv_wgrad, "MFMA conv weight gradient (K1 bwd)");
+  m.def("conv_dgrad", &conv_dgrad, "MFMA conv data gradient (K1 bwd)");
+  m.def("relu_mask_bwd", &relu_mask_bwd, "fused-ReLU backward mask");
+  m.def("bias_grad", &bias_grad, "conv bias gradient column reduction");
   m.def("embed_bwd", &embed_bwd,
         "action-embedding table gradient (K2 backward)");
   m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");

@@ -1,0 +1,523 @@
+// K1: hand-written implicit-GEMM conv stack for the Atari
+// model on gfx950 — MFMA 16x16x32 bf16, LDS-tiled, NHWC, VALID padding.
+//
+// Forward kernels fuse: (layer 1) uint8 -> /255 normalize on load, and
+// (all layers) bias + ReLU epilogue. Weights are consumed exactly as torch
+// stores them channels_last: W[co][kh][kw][ci] == W[co][k] row-major with
+// k = (kh*KW + kw)*CI + ci — each LDS B-tile row is one contiguous read.
+//
+// GEMM view per layer (reference model/impala_actor_critic.py:5-10):
+//   L1: M = N*20*20, K = 8*8*C  (C=4: u8 input),  CO = 32
+//   L2: M = N*9*9,   K = 4*4*32,                  CO = 64
+//   L3: M = N*7*7,   K = 3*3*64,                  CO = 64
+//
+// Block: 256 threads (4 waves), BM = 128 rows x BN = CO cols, BK = 32.
+// Per wave: 32 rows x CO cols as 2 x (CO/16) mfma_f32_16x16x32_bf16
+// fragments. A-tile [128][32+8] and B-tile [CO][32+8] live in LDS with
+// +8 bf16 row padding (conflict-free ds_read_b128, guide §6 G4).
+//
+// Fragment maps (verified on-box by drla_mfma_probe + the parity tests):
+//   A: lane l holds A[l&15][(l>>4)*8 + e], e = 0..7   (8 bf16 = 4 VGPRs)
+//   B: lane l holds B[(l>>4)*8 + e][l&15]             (8 bf16)
+//   D: lane l, reg r -> row (l>>4)*4 + r, col l&15    (4 f32)
+
+#include "drla_common.h"
+
+typedef unsigned short bf16raw;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+
+__device__ __forceinline__ float cv_bf2f(bf16raw u) {
+  unsigned int x = ((unsigned int)u) << 16;
+  return __uint_as_float(x);
+}
+
+// ---------------------------------------------------------------------------
+// layout probe: D = A @ B for one 16x16x32 tile, used by tests to pin the
+// fragment maps empirically (guide §3: check with ASYMMETRIC operands).
+// A, B row-major [16][32] / [32][16] bf16; D [16][16] f32.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void drla_mfma_probe(const bf16raw* __restrict__ A,
+                                           const bf16raw* __restrict__ B,
+                                           float* __restrict__ D) {
+  const int l = threadIdx.x;
+  bf16x8 a_frag, b_frag;
+  for (int e = 0; e < 8; ++e) {
+    a_frag[e] = (short)A[(l & 15) * 32 + ((l >> 4) * 8 + e)];
+    b_frag[e] = (short)B[((l >> 4) * 8 + e) * 16 + (l & 15)];
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0, 0);
+  for (int r = 0; r < 4; ++r) {
+    D[((l >> 4) * 4 + r) * 16 + (l & 15)] = acc[r];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+
+template <typename IN_T, int CI, int CO, int KH, int KW, int STRIDE, int HI,
+          int WI, int HO, int WO>
+__device__ void conv_fwd_impl(const IN_T* __restrict__ in,
+                              const bf16raw* __restrict__ w,   // [CO][K]
+                              const float* __restrict__ bias,  // [CO]
+                              bf16raw* __restrict__ out,       // [M][CO]
+                              int batch) {
+  constexpr int K = KH * KW * CI;
+  constexpr int BM = 128;
+  constexpr int BK = 32;
+  constexpr int APAD = 8;
+  constexpr int NFRAG = CO / 16;
+  const int M = batch * HO * WO;
+
+  __shared__ bf16raw Abuf[BM][BK + APAD];
+  __shared__ bf16raw Bbuf[CO][BK + APAD];
+  __shared__ float BiasBuf[CO];
+
+  const int tid = threadIdx.x;
+  const int wave = tid / 64;
+  const int lane = tid % 64;
+  const int row0 = blockIdx.x * BM;
+
+  if (tid < CO) BiasBuf[tid] = bias[tid];
+
+  f32x4 acc[2][NFRAG];
+  for (int mi = 0; mi < 2; ++mi)
+    for (int ni = 0; ni < NFRAG; ++ni)
+      acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  // per-thread A staging assignment: thread t fills row (t>>1),
+  // k-halves (t&1)*16..+16 of the 128x32 tile
+  const int a_row = tid >> 1;
+  const int a_k0 = (tid & 1) * 16;
+  const int gm = row0 + a_row;
+  // decode output coordinate once
+  const int n_idx = gm / (HO * WO);
+  const int rem = gm - n_idx * (HO * WO);
+  const int ho = rem / WO;
+  const int wo = rem - ho * WO;
+  const long long in_base =
+      ((long long)n_idx * HI + ho * STRIDE) * WI + wo * STRIDE;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A[128][32] ----
+    if (gm < M) {
+      if constexpr (CI == 4) {
+        // u8 input, 4 channels per tap: 4 taps per k-half, one uchar4 each,
+        // /255 normalize fused into the load
+        for (int t = 0; t < 4; ++t) {
+          const int kk = k0 + a_k0 + t * 4;
+          const int kh = kk / (KW * CI);
+          const int kw = (kk - kh * KW * CI) / CI;
+          const uchar4 v = *reinterpret_cast<const uchar4*>(
+              in + (in_base + (long long)kh * WI + kw) * CI);
+          const float s = 1.0f / 255.0f;
+          Abuf[a_row][a_k0 + t * 4 + 0] = drla_f32_to_bf16(v.x * s);
+          Abuf[a_row][a_k0 + t * 4 + 1] = drla_f32_to_bf16(v.y * s);
+          Abuf[a_row][a_k0 + t * 4 + 2] = drla_f32_to_bf16(v.z * s);
+          Abuf[a_row][a_k0 + t * 4 + 3] = drla_f32_to_bf16(v.w * s);
+        }
+      } else if constexpr (CI == 1) {
+        // u8 single-channel (R2D2 POMDP): one tap per k element
+        for (int t = 0; t < 16; ++t) {
+          const int kk = k0 + a_k0 + t;
+          const int kh = kk / KW;
+          const int kw = kk - kh * KW;
+          const unsigned char v = in[in_base + (long long)kh * WI + kw];
+          Abuf[a_row][a_k0 + t] = drla_f32_to_bf16(v * (1.0f / 255.0f));
+        }
+      } else {
+        // bf16 input, CI >= 32: a 16-element k-half stays inside one tap
+        const int kk = k0 + a_k0;
+        const int kh = kk / (KW * CI);
+        const int kwci = kk - kh * KW * CI;
+        const int kw = kwci / CI;
+        const int ci = kwci - kw * CI;
+        const bf16raw* src = reinterpret_cast<const bf16raw*>(in) +
+                             (in_base + (long long)kh * WI + kw) * CI + ci;
+        *reinterpret_cast<uint4*>(&Abuf[a_row][a_k0]) =
+            *reinterpret_cast<const uint4*>(src);
+        *reinterpret_cast<uint4*>(&Abuf[a_row][a_k0 + 8]) =
+            *reinterpret_cast<const uint4*>(src + 8);
+      }
+    } else {
+      for (int t = 0; t < 16; t += 8) {
+        *reinterpret_cast<uint4*>(&Abuf[a_row][a_k0 + t]) = uint4{0, 0, 0, 0};
+      }
+    }
+    // ---- stage B[CO][32]: W[co][k0..k0+32) contiguous ----
+    {
+      // thread t stages co = t>>1? CO*32/256 = CO/8 elems per thread.
+      // assign: co = tid % CO, k-part = (tid / CO) * (32*CO/256/..)
+      constexpr int THREADS_PER_CO = 256 / CO;         // 8 (CO=32) or 4
+      constexpr int KCHUNK = BK / THREADS_PER_CO;      // 4 or 8
+      const int co = tid / THREADS_PER_CO;
+      const int kpart = (tid % THREADS_PER_CO) * KCHUNK;
+      const bf16raw* src = w + (long long)co * K + k0 + kpart;
+      if constexpr (KCHUNK == 8) {
+        *reinterpret_cast<uint4*>(&Bbuf[co][kpart]) =
+            *reinterpret_cast<const uint4*>(src);
+      } else {
+        *reinterpret_cast<uint2*>(&Bbuf[co][kpart]) =
+            *reinterpret_cast<const uint2*>(src);
+      }
+    }
+    __syncthreads();
+
+    // ---- MFMA: wave covers rows [wave*32, wave*32+32) x CO ----
+    for (int mi = 0; mi < 2; ++mi) {
+      const int arow = wave * 32 + mi * 16 + (lane & 15);
+      const bf16x8 a_frag =
+          *reinterpret_cast<const bf16x8*>(&Abuf[arow][(lane >> 4) * 8]);
+      for (int ni = 0; ni < NFRAG; ++ni) {
+        const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            &Bbuf[ni * 16 + (lane & 15)][(lane >> 4) * 8]);
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias + ReLU, bf16 NHWC store ----
+  for (int mi = 0; mi < 2; ++mi) {
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int col = ni * 16 + (lane & 15);
+      const float b = BiasBuf[col];
+      for (int r = 0; r < 4; ++r) {
+        const int grow = row0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + r;
+        if (grow < M) {
+          const float v = fmaxf(acc[mi][ni][r] + b, 0.0f);
+          out[(long long)grow * CO + col] = drla_f32_to_bf16(v);
+        }
+      }
+    }
+  }
+}
+
+// b_frag reads above: B fragment needs B[k][col] where the LDS image is
+// Bbuf[co][k] — i.e. we feed mfma(A, B^T-read) which computes A @ B with
+// B[k][col] = Bbuf[col][k]. The probe/parity tests pin this down.
+
+extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l1(
+    const unsigned char* in, const bf16raw* w, const float* bias,
+    bf16raw* out, int batch) {
+  conv_fwd_impl<unsigned char, 4, 32, 8, 8, 4, 84, 84, 20, 20>(
+      in, w, bias, out, batch);
+}
+
+extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l1_c1(
+    const unsigned char* in, const bf16raw* w, const float* bias,
+    bf16raw* out, int batch) {
+  // R2D2's single-channel POMDP frames: CI=1 -> K=64; stage per-tap scalars
+  conv_fwd_impl<unsigned char, 1, 32, 8, 8, 4, 84, 84, 20, 20>(
+      in, w, bias, out, batch);
+}
+
+extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l2(
+    const bf16raw* in, const bf16raw* w, const float* bias, bf16raw* out,
+    int batch) {
+  conv_fwd_impl<bf16raw, 32, 64, 4, 4, 2, 20, 20, 9, 9>(in, w, bias, out,
+                                                        batch);
+}
+
+extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l3(
+    const bf16raw* in, const bf16raw* w, const float* bias, bf16raw* out,
+    int batch) {
+  conv_fwd_impl<bf16raw, 64, 64, 3, 3, 1, 9, 9, 7, 7>(in, w, bias, out,
+                                                      batch);
+}
+
+// ---------------------------------------------------------------------------
+// backward helpers
+// ---------------------------------------------------------------------------
+
+// dY_masked = dY * (Y > 0): the fused-ReLU backward mask, one pass.
+extern "C" __global__ void drla_relu_mask_bwd(
+    const bf16raw* __restrict__ dy, const bf16raw* __restrict__ y,
+    bf16raw* __restrict__ out, long long n) {
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < n; i += stride) {
+    out[i] = (cv_bf2f(y[i]) > 0.0f) ? dy[i] : (bf16raw)0;
+  }
+}
+
+// dbias[co] = sum_m dY[m][co]  (f32 out)
+extern "C" __global__ void drla_bias_grad(
+    const bf16raw* __restrict__ dy, float* __restrict__ dbias, long long M,
+    int CO) {
+  // one block per co-chunk of 64; waves stride M (no early return: every
+  // thread must reach the barrier)
+  const int co = blockIdx.x * 64 + (threadIdx.x % 64);
+  const bool live = co < CO;
+  const int mslice = threadIdx.x / 64;     // 4 slices
+  float acc = 0.0f;
+  if (live) {
+    for (long long m = mslice; m < M; m += 4) {
+      acc += cv_bf2f(dy[m * CO + co]);
+    }
+  }
+  __shared__ float red[4][64];
+  red[mslice][threadIdx.x % 64] = acc;
+  __syncthreads();
+  if (mslice == 0 && live) {
+    atomicAdd(&dbias[co], red[0][threadIdx.x] + red[1][threadIdx.x] +
+                              red[2][threadIdx.x] + red[3][threadIdx.x]);
+  }
+}
+
+// cast + transpose the f32 [K][CO] wgrad scratch into bf16 [CO][K]
+// (the channels_last weight-grad layout)
+extern "C" __global__ void drla_wgrad_finalize(
+    const float* __restrict__ scratch, bf16raw* __restrict__ dw, int K,
+    int CO) {
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long total = (long long)K * CO;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; i < total; i += stride) {
+    const int co = i / K;
+    const int k = i - (long long)co * K;
+    dw[i] = drla_f32_to_bf16(scratch[(long long)k * CO + co]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad: dW^T[k][co] = sum_m A_im2col[m][k] * dY[m][co]
+// grid: (K/64, SPLIT_M); each block owns 64 k-rows x CO cols and a slice of
+// M; partials atomicAdd into the f32 [K][CO] scratch.
+// ---------------------------------------------------------------------------
+
+template <typename IN_T, int CI, int CO, int KH, int KW, int STRIDE, int HI,
+          int WI, int HO, int WO, int SPLIT_M>
+__device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
+                                const bf16raw* __restrict__ dy,  // [M][CO]
+                                float* __restrict__ scratch,     // [K][CO]
+                                int batch) {
+  constexpr int K = KH * KW * CI;
+  constexpr int BKM = 32;   // m-rows per MFMA step
+  constexpr int PAD = 8;
+  constexpr int NFRAG = CO / 16;
+  const int M = batch * HO * WO;
+
+  __shared__ bf16raw At[64][BKM + PAD];   // A^T image: [k][m]
+  __shared__ bf16raw Bt[CO][BKM + PAD];   // dY^T image: [co][m]
+
+  const int tid = threadIdx.x;
+  const int wave = tid / 64;
+  const int lane = tid % 64;
+  const int k_row0 = blockIdx.x * 64;
+
+  // wave w owns k-rows [w*16, w*16+16)
+  f32x4 acc[NFRAG];
+  for (int ni = 0; ni < NFRAG; ++ni) acc[ni] = {0.f, 0.f, 0.f, 0.f};
+
+  const int m_per_split = (M + SPLIT_M - 1) / SPLIT_M;
+  const int m_begin = blockIdx.y * m_per_split;
+  const int m_end = min(M, m_begin + m_per_split);
+
+  for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
+    // stage At[k][m] (transposed scatter: 64*32/256 = 8 elems/thread)
+    // thread t handles k = t>>2, m-quarter = (t&3)*8..+8
+    {
+      const int k = k_row0 + (tid >> 2);
+      const int mq = (tid & 3) * 8;
+      for (int e = 0; e < 8; ++e) {
+        const int m = m0 + mq + e;
+        bf16raw v = 0;
+        if (m < m_end && k < K) {
+          const int n_idx = m / (HO * WO);
+          const int rem = m - n_idx * (HO * WO);
+          const int ho = rem / WO;
+          const int wo = rem - ho * WO;
+          const int kh = k / (KW * CI);
+          const int kwci = k - kh * KW * CI;
+          const int kw = kwci / CI;
+          const int ci = kwci - kw * CI;
+          const long long src =
+              (((long long)n_idx * HI + ho * STRIDE + kh) * WI +
+               wo * STRIDE + kw) * CI + ci;
+          if constexpr (sizeof(IN_T) == 1) {
+            v = drla_f32_to_bf16(in[src] * (1.0f / 255.0f));
+          } else {
+            v = in[src];
+          }
+        }
+        At[tid >> 2][mq + e] = v;
+      }
+    }
+    // stage Bt[co][m] (transposed scatter: CO*32/256 elems/thread)
+    {
+      constexpr int CO_PER_T = (CO * BKM) / 256;  // 4 (CO=32) or 8
+      const int co = tid % CO;
+      const int mq = (tid / CO) * CO_PER_T;
+      for (int e = 0; e < CO_PER_T; ++e) {
+        const int m = m0 + mq + e;
+        Bt[co][mq + e] =
+            (m < m_end) ? dy[(long long)m * CO + co] : (bf16raw)0;
+      }
+    }
+    __syncthreads();
+
+    const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
+        &At[wave * 16 + (lane & 15)][(lane >> 4) * 8]);
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+          &Bt[ni * 16 + (lane & 15)][(lane >> 4) * 8]);
+      acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
+                                                        acc[ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  for (int ni = 0; ni < NFRAG; ++ni) {
+    const int co = ni * 16 + (lane & 15);
+    for (int r = 0; r < 4; ++r) {
+      const int k = k_row0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (k < K) atomicAdd(&scratch[(long long)k * CO + co], acc[ni][r]);
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l1(
+    const unsigned char* in, const bf16raw* dy, float* scratch, int batch) {
+  conv_wgrad_impl<unsigned char, 4, 32, 8, 8, 4, 84, 84, 20, 20, 64>(
+      in, dy, scratch, batch);
+}
+extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l1_c1(
+    const unsigned char* in, const bf16raw* dy, float* scratch, int batch) {
+  conv_wgrad_impl<unsigned char, 1, 32, 8, 8, 4, 84, 84, 20, 20, 64>(
+      in, dy, scratch, batch);
+}
+extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l2(
+    const bf16raw* in, const bf16raw* dy, float* scratch, int batch) {
+  conv_wgrad_impl<bf16raw, 32, 64, 4, 4, 2, 20, 20, 9, 9, 32>(in, dy,
+                                                              scratch, batch);
+}
+extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l3(
+    const bf16raw* in, const bf16raw* dy, float* scratch, int batch) {
+  conv_wgrad_impl<bf16raw, 64, 64, 3, 3, 1, 9, 9, 7, 7, 32>(in, dy, scratch,
+                                                            batch);
+}
+
+// ---------------------------------------------------------------------------
+// dgrad: dX[n,hi,wi,ci] = sum_{kh,kw,co valid} dY[n,ho',wo',co] *
+//        W[co][(kh*KW+kw)*CI+ci],  ho' = (hi-kh)/STRIDE (when integral).
+// Implicit GEMM: M2 = N*HI*WI rows, Kg = KH*KW*CO, CI cols.
+// ---------------------------------------------------------------------------
+
+template <int CI, int CO, int KH, int KW, int STRIDE, int HI, int WI, int HO,
+          int WO>
+__device__ void conv_dgrad_impl(const bf16raw* __restrict__ dy,  // [M][CO]
+                                const bf16raw* __restrict__ w,   // [CO][K]
+                                bf16raw* __restrict__ dx,        // [M2][CI]
+                                int batch) {
+  constexpr int K = KH * KW * CI;
+  constexpr int Kg = KH * KW * CO;
+  constexpr int BM = 128;
+  constexpr int BK = 32;   // kg per step: one tap x 32 co
+  constexpr int PAD = 8;
+  constexpr int NFRAG = CI / 16;
+  const int M2 = batch * HI * WI;
+
+  __shared__ bf16raw Ad[BM][BK + PAD];    // dY gather: [m2][kg_local]
+  __shared__ bf16raw Bd[CI][BK + PAD];    // W^T image: [ci][co_local]
+
+  const int tid = threadIdx.x;
+  const int wave = tid / 64;
+  const int lane = tid % 64;
+  const int row0 = blockIdx.x * BM;
+
+  f32x4 acc[2][NFRAG];
+  for (int mi = 0; mi < 2; ++mi)
+    for (int ni = 0; ni < NFRAG; ++ni)
+      acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  // per-thread A staging: row = tid>>1, kg-half = (tid&1)*16
+  const int a_row = tid >> 1;
+  const int a_k0 = (tid & 1) * 16;
+  const int m2 = row0 + a_row;
+  const int n_idx = m2 / (HI * WI);
+  const int rem = m2 - n_idx * (HI * WI);
+  const int hi = rem / WI;
+  const int wi = rem - hi * WI;
+
+  for (int kg0 = 0; kg0 < Kg; kg0 += BK) {
+    const int tap = kg0 / CO;
+    const int co0 = kg0 - tap * CO;
+    const int kh = tap / KW;
+    const int kw = tap - kh * KW;
+    // ---- stage Ad[m2][co within tap] with validity mask ----
+    {
+      bool valid = (m2 < M2);
+      int ho = 0, wo = 0;
+      if (valid) {
+        const int hh = hi - kh;
+        const int ww = wi - kw;
+        valid = hh >= 0 && ww >= 0 && (hh % STRIDE) == 0 &&
+                (ww % STRIDE) == 0;
+        if (valid) {
+          ho = hh / STRIDE;
+          wo = ww / STRIDE;
+          valid = ho < HO && wo < WO;
+        }
+      }
+      if (valid) {
+        const bf16raw* src =
+            dy + ((long long)(n_idx * HO + ho) * WO + wo) * CO + co0 + a_k0;
+        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) =
+            *reinterpret_cast<const uint4*>(src);
+        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) =
+            *reinterpret_cast<const uint4*>(src + 8);
+      } else {
+        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) = uint4{0, 0, 0, 0};
+        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) = uint4{0, 0, 0, 0};
+      }
+    }
+    // ---- stage Bd[ci][co_local]: W[co0+col][tap*CI + ci] transposed ----
+    {
+      constexpr int CI_PER_T = (CI * BK) / 256;  // 4 (CI=32) or 8 (CI=64)
+      const int col = tid % BK;                  // co_local
+      const int ci0 = (tid / BK) * CI_PER_T;
+      const bf16raw* src = w + (long long)(co0 + col) * K + tap * CI + ci0;
+      for (int e = 0; e < CI_PER_T; ++e) {
+        Bd[ci0 + e][col] = src[e];
+      }
+    }
+    __syncthreads();
+
+    for (int mi = 0; mi < 2; ++mi) {
+      const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
+          &Ad[wave * 32 + mi * 16 + (lane & 15)][(lane >> 4) * 8]);
+      for (int ni = 0; ni < NFRAG; ++ni) {
+        const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            &Bd[ni * 16 + (lane & 15)][(lane >> 4) * 8]);
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  for (int mi = 0; mi < 2; ++mi) {
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int ci = ni * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r) {
+        const int grow = row0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + r;
+        if (grow < M2) {
+          dx[(long long)grow * CI + ci] = drla_f32_to_bf16(acc[mi][ni][r]);
+        }
+      }
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256) void drla_conv_dgrad_l2(
+    const bf16raw* dy, const bf16raw* w, bf16raw* dx, int batch) {
+  conv_dgrad_impl<32, 64, 4, 4, 2, 20, 20, 9, 9>(dy, w, dx, batch);
+}
+extern "C" __global__ __launch_bounds__(256) void drla_conv_dgrad_l3(
+    const bf16raw* dy, const bf16raw* w, bf16raw* dx, int batch) {
+  conv_dgrad_impl<64, 64, 3, 3, 1, 9, 9, 7, 7>(dy, w, dx, batch);
+}

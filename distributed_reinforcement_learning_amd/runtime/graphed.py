@@ -101,11 +101,10 @@ class GraphedImpalaStep:
         self._distributed = is_distributed()
 
     def _fwd_bwd(self) -> Tuple[torch.Tensor, ...]:
-        from distributed_reinforcement_learning_amd.ops import normalize_frames
         agent = self.agent
         agent.optimizer.flat_grads.zero_()
         i = self.inputs
-        s = normalize_frames(i["state"], out_dtype=agent.model_dtype)
+        s = agent.prepare_frames(i["state"])
         pi_loss, baseline_loss, entropy, total = agent.compute_losses(
             s, i["reward"], i["action"], i["done"], i["behavior_policy"],
             i["previous_action"], i["initial_h"], i["initial_c"])
