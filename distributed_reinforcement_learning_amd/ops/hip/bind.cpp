@@ -258,9 +258,10 @@ torch::Tensor bias_grad(torch::Tensor dy_flat, int64_t CO) {
   check_gpu_contig(dy_flat, "dy");
   const long long M = dy_flat.numel() / CO;
   auto out = torch::zeros({CO}, dy_flat.options().dtype(torch::kFloat));
-  hipLaunchKernelGGL(drla_bias_grad, dim3((CO + 63) / 64), dim3(256), 0,
-                     cur_stream(), u16p(dy_flat), out.data_ptr<float>(), M,
-                     (int)CO);
+  const int msplit = (int)std::min<long long>(256, (M + 1023) / 1024) + 1;
+  hipLaunchKernelGGL(drla_bias_grad, dim3((CO + 63) / 64, msplit), dim3(256),
+                     0, cur_stream(), u16p(dy_flat), out.data_ptr<float>(),
+                     M, (int)CO);
   return out;
 }
 

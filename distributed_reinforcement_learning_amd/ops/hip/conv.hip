@@ -244,18 +244,21 @@ extern "C" __global__ void drla_relu_mask_bwd(
   }
 }
 
-// dbias[co] = sum_m dY[m][co]  (f32 out)
+// dbias[co] = sum_m dY[m][co]  (f32 out, zeroed by caller)
+// grid: (ceil(CO/64), M_SPLIT) — blockIdx.y slices M so hundreds of blocks
+// stream dY; partials atomicAdd per block.
 extern "C" __global__ void drla_bias_grad(
     const bf16raw* __restrict__ dy, float* __restrict__ dbias, long long M,
     int CO) {
-  // one block per co-chunk of 64; waves stride M (no early return: every
-  // thread must reach the barrier)
   const int co = blockIdx.x * 64 + (threadIdx.x % 64);
   const bool live = co < CO;
-  const int mslice = threadIdx.x / 64;     // 4 slices
+  const int mslice = threadIdx.x / 64;     // 4 slices within the block
+  const long long per = (M + gridDim.y - 1) / gridDim.y;
+  const long long m_begin = blockIdx.y * per;
+  const long long m_end = min(M, m_begin + per);
   float acc = 0.0f;
   if (live) {
-    for (long long m = mslice; m < M; m += 4) {
+    for (long long m = m_begin + mslice; m < m_end; m += 4) {
       acc += cv_bf2f(dy[m * CO + co]);
     }
   }
@@ -297,19 +300,23 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
                                 int batch) {
   constexpr int K = KH * KW * CI;
   constexpr int BKM = 32;   // m-rows per MFMA step
+  constexpr int BKK = 64;   // k-cols per block
   constexpr int PAD = 8;
   constexpr int NFRAG = CO / 16;
   const int M = batch * HO * WO;
 
-  __shared__ bf16raw At[64][BKM + PAD];   // A^T image: [k][m]
-  __shared__ bf16raw Bt[CO][BKM + PAD];   // dY^T image: [co][m]
+  // forward-style images (vector global loads, row-contiguous LDS writes);
+  // the MFMA fragments read them TRANSPOSED with scalar ds_read_u16 —
+  // cheaper than the division-heavy transposed-scatter staging this
+  // replaced (332 us -> see profiles).
+  __shared__ bf16raw Am[BKM][BKK + PAD];  // A image: [m][k]
+  __shared__ bf16raw Bm[BKM][CO + PAD];   // dY image: [m][co]
 
   const int tid = threadIdx.x;
   const int wave = tid / 64;
   const int lane = tid % 64;
-  const int k_row0 = blockIdx.x * 64;
+  const int k_row0 = blockIdx.x * BKK;
 
-  // wave w owns k-rows [w*16, w*16+16)
   f32x4 acc[NFRAG];
   for (int ni = 0; ni < NFRAG; ++ni) acc[ni] = {0.f, 0.f, 0.f, 0.f};
 
@@ -317,54 +324,94 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
   const int m_begin = blockIdx.y * m_per_split;
   const int m_end = min(M, m_begin + m_per_split);
 
+  // staging assignment (fwd-style): thread t -> m-row = t>>3,
+  // k-chunk = (t&7)*8 of the 32x64 A image; decode coords once per chunk
+  const int a_m = tid >> 3;
+  const int a_k = (tid & 7) * 8;
+
   for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
-    // stage At[k][m] (transposed scatter: 64*32/256 = 8 elems/thread)
-    // thread t handles k = t>>2, m-quarter = (t&3)*8..+8
+    // ---- stage Am[m][k0..k0+8) ----
     {
-      const int k = k_row0 + (tid >> 2);
-      const int mq = (tid & 3) * 8;
-      for (int e = 0; e < 8; ++e) {
-        const int m = m0 + mq + e;
-        bf16raw v = 0;
-        if (m < m_end && k < K) {
-          const int n_idx = m / (HO * WO);
-          const int rem = m - n_idx * (HO * WO);
-          const int ho = rem / WO;
-          const int wo = rem - ho * WO;
-          const int kh = k / (KW * CI);
-          const int kwci = k - kh * KW * CI;
+      const int m = m0 + a_m;
+      if (m < m_end) {
+        const int n_idx = m / (HO * WO);
+        const int rem = m - n_idx * (HO * WO);
+        const int ho = rem / WO;
+        const int wo = rem - ho * WO;
+        const long long base =
+            ((long long)n_idx * HI + ho * STRIDE) * WI + wo * STRIDE;
+        const int kk = k_row0 + a_k;
+        if constexpr (CI == 4) {
+          for (int t = 0; t < 2; ++t) {
+            const int k2 = kk + t * 4;
+            const int kh = k2 / (KW * CI);
+            const int kw = (k2 - kh * KW * CI) / CI;
+            const uchar4 v = *reinterpret_cast<const uchar4*>(
+                in + (base + (long long)kh * WI + kw) * CI);
+            const float s = 1.0f / 255.0f;
+            Am[a_m][a_k + t * 4 + 0] = drla_f32_to_bf16(v.x * s);
+            Am[a_m][a_k + t * 4 + 1] = drla_f32_to_bf16(v.y * s);
+            Am[a_m][a_k + t * 4 + 2] = drla_f32_to_bf16(v.z * s);
+            Am[a_m][a_k + t * 4 + 3] = drla_f32_to_bf16(v.w * s);
+          }
+        } else if constexpr (CI == 1) {
+          for (int t = 0; t < 8; ++t) {
+            const int k2 = kk + t;
+            const int kh = k2 / KW;
+            const int kw = k2 - kh * KW;
+            const unsigned char v =
+                (k2 < K) ? in[base + (long long)kh * WI + kw] : 0;
+            Am[a_m][a_k + t] = drla_f32_to_bf16(v * (1.0f / 255.0f));
+          }
+        } else {
+          const int kh = kk / (KW * CI);
+          const int kwci = kk - kh * KW * CI;
           const int kw = kwci / CI;
           const int ci = kwci - kw * CI;
-          const long long src =
-              (((long long)n_idx * HI + ho * STRIDE + kh) * WI +
-               wo * STRIDE + kw) * CI + ci;
-          if constexpr (sizeof(IN_T) == 1) {
-            v = drla_f32_to_bf16(in[src] * (1.0f / 255.0f));
-          } else {
-            v = in[src];
-          }
+          const bf16raw* src = reinterpret_cast<const bf16raw*>(in) +
+                               (base + (long long)kh * WI + kw) * CI + ci;
+          *reinterpret_cast<uint4*>(&Am[a_m][a_k]) =
+              (kk < K) ? *reinterpret_cast<const uint4*>(src)
+                       : uint4{0, 0, 0, 0};
         }
-        At[tid >> 2][mq + e] = v;
+      } else {
+        *reinterpret_cast<uint4*>(&Am[a_m][a_k]) = uint4{0, 0, 0, 0};
       }
     }
-    // stage Bt[co][m] (transposed scatter: CO*32/256 elems/thread)
+    // ---- stage Bm[m][co]: contiguous dY rows ----
     {
-      constexpr int CO_PER_T = (CO * BKM) / 256;  // 4 (CO=32) or 8
-      const int co = tid % CO;
-      const int mq = (tid / CO) * CO_PER_T;
-      for (int e = 0; e < CO_PER_T; ++e) {
-        const int m = m0 + mq + e;
-        Bt[co][mq + e] =
-            (m < m_end) ? dy[(long long)m * CO + co] : (bf16raw)0;
+      constexpr int CO_PER_T = (CO * BKM) / 256;  // 4 or 8
+      const int m = m0 + (tid / (CO / CO_PER_T));
+      const int co0 = (tid % (CO / CO_PER_T)) * CO_PER_T;
+      if constexpr (CO_PER_T == 8) {
+        *reinterpret_cast<uint4*>(&Bm[tid / (CO / CO_PER_T)][co0]) =
+            (m < m_end) ? *reinterpret_cast<const uint4*>(
+                              dy + (long long)m * CO + co0)
+                        : uint4{0, 0, 0, 0};
+      } else {
+        *reinterpret_cast<uint2*>(&Bm[tid / (CO / CO_PER_T)][co0]) =
+            (m < m_end) ? *reinterpret_cast<const uint2*>(
+                              dy + (long long)m * CO + co0)
+                        : uint2{0, 0};
       }
     }
     __syncthreads();
 
-    const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
-        &At[wave * 16 + (lane & 15)][(lane >> 4) * 8]);
+    // wave w owns k-rows [w*16, w*16+16): A_op[k][m] = Am[m][k] (transposed
+    // scalar reads), B_op[m][co] = Bm[m][co] (transposed scalar reads)
+    bf16x8 a_frag;
+    {
+      const int k = wave * 16 + (lane & 15);
+      for (int e = 0; e < 8; ++e) {
+        a_frag[e] = (short)Am[(lane >> 4) * 8 + e][k];
+      }
+    }
     for (int ni = 0; ni < NFRAG; ++ni) {
-      const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
-          &Bt[ni * 16 + (lane & 15)][(lane >> 4) * 8]);
+      bf16x8 b_frag;
+      const int co = ni * 16 + (lane & 15);
+      for (int e = 0; e < 8; ++e) {
+        b_frag[e] = (short)Bm[(lane >> 4) * 8 + e][co];
+      }
       acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
                                                         acc[ni], 0, 0, 0);
     }
