@@ -35,9 +35,8 @@ class _ConvLayer(torch.autograd.Function):
         ext = _ops.require_ext()
         layer = ctx.layer
         co = _LAYER_CO[layer]
-        dy_m = ext.relu_mask_bwd(dy.contiguous(), y)
+        dy_m, dbias = ext.relu_mask_bwd(dy.contiguous(), y, co)
         dw = ext.conv_wgrad(layer, x, dy_m)
-        dbias = ext.bias_grad(dy_m, co)
         dx = None
         if layer >= 2 and ctx.needs_input_grad[0]:
             dx = ext.conv_dgrad(layer, dy_m, w_flat)

@@ -44,9 +44,8 @@ extern "C" __global__ void drla_conv_fwd_l3(const unsigned short*,
                                             int);
 extern "C" __global__ void drla_relu_mask_bwd(const unsigned short*,
                                               const unsigned short*,
-                                              unsigned short*, long long);
-extern "C" __global__ void drla_bias_grad(const unsigned short*, float*,
-                                          long long, int);
+                                              unsigned short*, float*,
+                                              long long, int);
 extern "C" __global__ void drla_wgrad_finalize(const float*, unsigned short*,
                                                int, int);
 extern "C" __global__ void drla_conv_wgrad_l1(const unsigned char*,
@@ -244,25 +243,18 @@ torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
   return out;
 }
 
-torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y) {
+std::tuple<torch::Tensor, torch::Tensor> relu_mask_bwd(torch::Tensor dy,
+                                                       torch::Tensor y,
+                                                       int64_t CO) {
   check_gpu_contig(dy, "dy");
   check_gpu_contig(y, "y");
   auto out = torch::empty_like(dy);
+  auto dbias = torch::zeros({CO}, dy.options().dtype(torch::kFloat));
   const long long n = dy.numel();
   hipLaunchKernelGGL(drla_relu_mask_bwd, dim3(drla_grid(n)), dim3(DRLA_BLOCK),
-                     0, cur_stream(), u16p(dy), u16p(y), u16pm(out), n);
-  return out;
-}
-
-torch::Tensor bias_grad(torch::Tensor dy_flat, int64_t CO) {
-  check_gpu_contig(dy_flat, "dy");
-  const long long M = dy_flat.numel() / CO;
-  auto out = torch::zeros({CO}, dy_flat.options().dtype(torch::kFloat));
-  const int msplit = (int)std::min<long long>(256, (M + 1023) / 1024) + 1;
-  hipLaunchKernelGGL(drla_bias_grad, dim3((CO + 63) / 64, msplit), dim3(256),
-                     0, cur_stream(), u16p(dy_flat), out.data_ptr<float>(),
-                     M, (int)CO);
-  return out;
+                     0, cur_stream(), u16p(dy), u16p(y), u16pm(out),
+                     dbias.data_ptr<float>(), n, (int)CO);
+  return {out, dbias};
 }
 
 torch::Tensor conv_wgrad(int layer, torch::Tensor in, torch::Tensor dy) {
@@ -273,7 +265,7 @@ torch::Tensor conv_wgrad(int layer, torch::Tensor in, torch::Tensor dy) {
   const int K = cfg.kh * cfg.kw * cfg.ci;
   auto scratch = torch::zeros({K, cfg.co},
                               dy.options().dtype(torch::kFloat));
-  const int split = (layer <= 1) ? 64 : 32;
+  const int split = (layer <= 1) ? 256 : 128;
   dim3 grid((K + 63) / 64, split);
   switch (layer) {
     case 0:
@@ -587,8 +579,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA implicit-GEMM conv fwd, fused normalize+bias+ReLU (K1)");
   m.def("conv_wgrad", &conv_wgrad, "MFMA conv weight gradient (K1 bwd)");
   m.def("conv_dgrad", &conv_dgrad, "MFMA conv data gradient (K1 bwd)");
-  m.def("relu_mask_bwd", &relu_mask_bwd, "fused-ReLU backward mask");
-  m.def("bias_grad", &bias_grad, "conv bias gradient column reduction");
+  m.def("relu_mask_bwd", &relu_mask_bwd,
+        "fused-ReLU backward mask + bias gradient");
   m.def("embed_bwd", &embed_bwd,
         "action-embedding table gradient (K2 backward)");
   m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");

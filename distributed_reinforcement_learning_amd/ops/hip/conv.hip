@@ -233,14 +233,32 @@ extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l3(
 // backward helpers
 // ---------------------------------------------------------------------------
 
-// dY_masked = dY * (Y > 0): the fused-ReLU backward mask, one pass.
+// dY_masked = dY * (Y > 0) with dbias[co] = sum_m dY_masked[m][co] FUSED
+// into the same pass. Thread-to-column pinning: with blockDim (256) and the
+// grid both multiples of CO (CO in {32,64}), a grid-stride walk keeps every
+// thread on ONE co column — local accumulate, LDS-reduce per co, one
+// atomicAdd per (block, co). dbias must be zeroed by the caller.
 extern "C" __global__ void drla_relu_mask_bwd(
     const bf16raw* __restrict__ dy, const bf16raw* __restrict__ y,
-    bf16raw* __restrict__ out, long long n) {
-  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+    bf16raw* __restrict__ out, float* __restrict__ dbias, long long n,
+    int CO) {
   const long long stride = gridDim.x * (long long)blockDim.x;
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const int my_co = (int)(i % CO);
+  float acc = 0.0f;
   for (; i < n; i += stride) {
-    out[i] = (cv_bf2f(y[i]) > 0.0f) ? dy[i] : (bf16raw)0;
+    const bf16raw v = (cv_bf2f(y[i]) > 0.0f) ? dy[i] : (bf16raw)0;
+    out[i] = v;
+    acc += cv_bf2f(v);
+  }
+  __shared__ float red[DRLA_BLOCK];
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  // threads {co, co+CO, co+2CO, ...} within the block share a column
+  if (threadIdx.x < (unsigned)CO) {
+    float s = 0.0f;
+    for (int t = threadIdx.x; t < DRLA_BLOCK; t += CO) s += red[t];
+    atomicAdd(&dbias[my_co], s);
   }
 }
 
@@ -429,22 +447,22 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
 
 extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l1(
     const unsigned char* in, const bf16raw* dy, float* scratch, int batch) {
-  conv_wgrad_impl<unsigned char, 4, 32, 8, 8, 4, 84, 84, 20, 20, 64>(
+  conv_wgrad_impl<unsigned char, 4, 32, 8, 8, 4, 84, 84, 20, 20, 256>(
       in, dy, scratch, batch);
 }
 extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l1_c1(
     const unsigned char* in, const bf16raw* dy, float* scratch, int batch) {
-  conv_wgrad_impl<unsigned char, 1, 32, 8, 8, 4, 84, 84, 20, 20, 64>(
+  conv_wgrad_impl<unsigned char, 1, 32, 8, 8, 4, 84, 84, 20, 20, 256>(
       in, dy, scratch, batch);
 }
 extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l2(
     const bf16raw* in, const bf16raw* dy, float* scratch, int batch) {
-  conv_wgrad_impl<bf16raw, 32, 64, 4, 4, 2, 20, 20, 9, 9, 32>(in, dy,
+  conv_wgrad_impl<bf16raw, 32, 64, 4, 4, 2, 20, 20, 9, 9, 128>(in, dy,
                                                               scratch, batch);
 }
 extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l3(
     const bf16raw* in, const bf16raw* dy, float* scratch, int batch) {
-  conv_wgrad_impl<bf16raw, 64, 64, 3, 3, 1, 9, 9, 7, 7, 32>(in, dy, scratch,
+  conv_wgrad_impl<bf16raw, 64, 64, 3, 3, 1, 9, 9, 7, 7, 128>(in, dy, scratch,
                                                             batch);
 }
 
