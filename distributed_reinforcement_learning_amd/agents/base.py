@@ -80,7 +80,8 @@ class AgentBase:
     # -- data movement -------------------------------------------------------
 
     def to_device(self, arr, dtype=None) -> torch.Tensor:
-        t = torch.as_tensor(np.asarray(arr))
+        t = arr if isinstance(arr, torch.Tensor) \
+            else torch.as_tensor(np.asarray(arr))
         if dtype is not None:
             t = t.to(dtype)
         return t.to(self.device, non_blocking=True)
@@ -89,10 +90,13 @@ class AgentBase:
         """uint8 (or float) frames -> normalized float32 on device.
 
         uint8 stays uint8 across the H2D copy (1 byte/pixel on the bus); the
-        /255 cast runs on-device (ops/preprocess.py, HIP kernel on gfx950).
-        Float inputs (vector obs) pass through unscaled.
+        /255 cast runs on-device (ops/preprocess.py, HIP kernel on gfx950) or
+        fused into conv layer 1 (custom MFMA stack). Device tensors (e.g.
+        from the GPU replay shard) pass straight through. Float inputs
+        (vector obs) pass through unscaled.
         """
-        t = torch.as_tensor(np.asarray(frames))
+        t = frames if isinstance(frames, torch.Tensor) \
+            else torch.as_tensor(np.asarray(frames))
         if t.dtype == torch.uint8:
             t = t.to(self.device, non_blocking=True)
             return self.prepare_frames(t)

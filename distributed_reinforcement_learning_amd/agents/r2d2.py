@@ -160,13 +160,24 @@ class Agent(AgentBase):
             with_grad=False)
         return float((target_value - sav).mean().abs())
 
+    @torch.no_grad()
+    def get_td_error_batch(self, state, previous_action, action, h0, c0,
+                           reward, done, as_tensor: bool = False):
+        """Per-sequence |mean TD| for a whole [B, L] batch in one forward
+        (device tensors ok; feeds the GPU replay shard directly)."""
+        _, target_value, sav = self._sequence_losses(
+            state, previous_action, action, h0, c0, reward, done,
+            with_grad=False)
+        td = (target_value - sav).mean(dim=1).abs()
+        return td if as_tensor else td.cpu().numpy()
+
     def train(self, state, previous_action, action, h, c, reward, done,
-              weight) -> Tuple[float, np.ndarray]:
+              weight, as_tensor: bool = False) -> Tuple[float, np.ndarray]:
         """IS-weighted batch update; h/c [B, L, H] stored per step — the
         sequence-start state is index 0 (reference train_r2d2.py:135-136).
         Returns (loss, per-sequence |mean TD|)."""
-        h0 = np.asarray(h)[:, 0]
-        c0 = np.asarray(c)[:, 0]
+        h0 = h[:, 0] if isinstance(h, torch.Tensor) else np.asarray(h)[:, 0]
+        c0 = c[:, 0] if isinstance(c, torch.Tensor) else np.asarray(c)[:, 0]
         unweighted, target_value, sav = self._sequence_losses(
             state, previous_action, action, h0, c0, reward, done,
             with_grad=True)
@@ -180,7 +191,9 @@ class Agent(AgentBase):
         self.optimizer.step(lr=lr)
         self.global_step += 1
         self.num_env_frames += int(np.prod(np.shape(reward)))
-        td = (target_value - sav).mean(dim=1).abs().detach().cpu().numpy()
+        td = (target_value - sav).mean(dim=1).abs().detach()
+        if not as_tensor:
+            td = td.cpu().numpy()
         return float(loss.detach()), td
 
     @torch.no_grad()

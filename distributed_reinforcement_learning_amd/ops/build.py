@@ -32,6 +32,7 @@ SOURCES = [
     "optim.hip",
     "conv.hip",
     "per_tree.hip",
+    "dqn_loss.hip",
 ]
 
 

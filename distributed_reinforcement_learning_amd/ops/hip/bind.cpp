@@ -66,6 +66,18 @@ extern "C" __global__ void drla_conv_dgrad_l2(const unsigned short*,
 extern "C" __global__ void drla_conv_dgrad_l3(const unsigned short*,
                                               const unsigned short*,
                                               unsigned short*, int);
+extern "C" __global__ void drla_dqn_loss_fwd(
+    const unsigned short*, const float*, const float*, const float*,
+    const int*, const float*, const float*, const float*, float*, float*,
+    int, int);
+extern "C" __global__ void drla_dqn_loss_bwd(
+    const float*, const int*, const float*, const float*, unsigned short*,
+    float*, int, int);
+extern "C" __global__ void drla_per_update(float*, const long long*,
+                                           const float*, int, long long);
+extern "C" __global__ void drla_per_sample(const float*, const float*,
+                                           long long*, float*, int,
+                                           long long);
 extern "C" __global__ void drla_embed_bwd_scatter(
     const long long*, const unsigned short*, const float*, float*,
     long long, int);
@@ -317,6 +329,76 @@ torch::Tensor conv_dgrad(int layer, torch::Tensor dy, torch::Tensor w) {
                        cur_stream(), u16p(dy), u16p(w), u16pm(dx), batch);
   }
   return dx;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> dqn_loss_fwd(
+    torch::Tensor main_q, torch::Tensor next_main, torch::Tensor next_tgt,
+    torch::Tensor actions, torch::Tensor rewards, torch::Tensor discounts,
+    torch::Tensor weights) {
+  for (auto* t : {&main_q, &next_main, &next_tgt, &actions, &rewards,
+                  &discounts, &weights})
+    check_gpu_contig(*t, "dqn_loss input");
+  TORCH_CHECK(actions.scalar_type() == torch::kInt, "actions must be i32");
+  const int B = main_q.size(0), A = main_q.size(1);
+  const bool bf16 = main_q.scalar_type() == torch::kBFloat16;
+  auto fopt = rewards.options().dtype(torch::kFloat);
+  auto loss = torch::zeros({1}, fopt);
+  auto td = torch::empty({B}, fopt);
+  hipLaunchKernelGGL(
+      drla_dqn_loss_fwd, dim3((B + 255) / 256), dim3(256), 0, cur_stream(),
+      bf16 ? u16p(main_q) : nullptr,
+      bf16 ? nullptr : main_q.data_ptr<float>(),
+      next_main.data_ptr<float>(), next_tgt.data_ptr<float>(),
+      actions.data_ptr<int>(), rewards.data_ptr<float>(),
+      discounts.data_ptr<float>(), weights.data_ptr<float>(),
+      loss.data_ptr<float>(), td.data_ptr<float>(), B, A);
+  return {loss, td};
+}
+
+torch::Tensor dqn_loss_bwd(torch::Tensor td, torch::Tensor actions,
+                           torch::Tensor weights, torch::Tensor gloss,
+                           int64_t A, bool want_bf16) {
+  for (auto* t : {&td, &actions, &weights, &gloss})
+    check_gpu_contig(*t, "dqn_loss bwd input");
+  const int B = td.numel();
+  auto dmq = torch::empty(
+      {B, A},
+      td.options().dtype(want_bf16 ? torch::kBFloat16 : torch::kFloat));
+  hipLaunchKernelGGL(
+      drla_dqn_loss_bwd, dim3(((long long)B * A + 255) / 256), dim3(256), 0,
+      cur_stream(), td.data_ptr<float>(), actions.data_ptr<int>(),
+      weights.data_ptr<float>(), gloss.data_ptr<float>(),
+      want_bf16 ? u16pm(dmq) : nullptr,
+      want_bf16 ? nullptr : dmq.data_ptr<float>(), B, (int)A);
+  return dmq;
+}
+
+void per_update(torch::Tensor tree, torch::Tensor idxs,
+                torch::Tensor prios, int64_t cap) {
+  for (auto* t : {&tree, &idxs, &prios}) check_gpu_contig(*t, "per tensor");
+  TORCH_CHECK(idxs.scalar_type() == torch::kLong, "idxs must be i64");
+  const int n = idxs.numel();
+  hipLaunchKernelGGL(drla_per_update, dim3((n + 255) / 256), dim3(256), 0,
+                     cur_stream(), tree.data_ptr<float>(),
+                     reinterpret_cast<const long long*>(
+                         idxs.data_ptr<int64_t>()),
+                     prios.data_ptr<float>(), n, cap);
+}
+
+std::tuple<torch::Tensor, torch::Tensor> per_sample(torch::Tensor tree,
+                                                    torch::Tensor s,
+                                                    int64_t cap) {
+  check_gpu_contig(tree, "tree");
+  check_gpu_contig(s, "s");
+  const int n = s.numel();
+  auto idx = torch::empty({n}, s.options().dtype(torch::kLong));
+  auto prio = torch::empty({n}, s.options().dtype(torch::kFloat));
+  hipLaunchKernelGGL(drla_per_sample, dim3((n + 255) / 256), dim3(256), 0,
+                     cur_stream(), tree.data_ptr<float>(),
+                     s.data_ptr<float>(),
+                     reinterpret_cast<long long*>(idx.data_ptr<int64_t>()),
+                     prio.data_ptr<float>(), n, cap);
+  return {idx, prio};
 }
 
 torch::Tensor embed_bwd(torch::Tensor indices, torch::Tensor grad_out,
@@ -581,6 +663,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_dgrad", &conv_dgrad, "MFMA conv data gradient (K1 bwd)");
   m.def("relu_mask_bwd", &relu_mask_bwd,
         "fused-ReLU backward mask + bias gradient");
+  m.def("dqn_loss_fwd", &dqn_loss_fwd,
+        "fused double-DQN target + IS-weighted TD loss (K8)");
+  m.def("dqn_loss_bwd", &dqn_loss_bwd, "closed-form K8 backward");
+  m.def("per_update", &per_update, "GPU PER segment-tree batched update");
+  m.def("per_sample", &per_sample, "GPU PER stratified sample descent");
   m.def("embed_bwd", &embed_bwd,
         "action-embedding table gradient (K2 backward)");
   m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");

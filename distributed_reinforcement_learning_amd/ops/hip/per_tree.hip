@@ -1,0 +1,53 @@
+// K10: GPU prioritized-replay segment tree.
+//
+// Same array layout as the CPU replay (replay/sum_tree.py, mirroring
+// reference buffer_queue.py:326-371): tree[0 .. 2*cap-2] f32, leaves at
+// cap-1. Payloads live in preallocated HBM tensors (replay/gpu_memory.py),
+// so learner-side priorities NEVER round-trip to the host — TD errors from
+// the loss kernels feed update_batch directly (SURVEY §7 build plan item 7).
+//
+// update: one thread per index — atomicExch the leaf (duplicate updates
+// serialize into a consistent delta chain), then atomicAdd the delta up to
+// the root. O(log cap) atomics per update, batched.
+// sample: one thread per query — root-to-leaf descent.
+
+#include "drla_common.h"
+
+extern "C" __global__ void drla_per_update(
+    float* __restrict__ tree, const long long* __restrict__ idxs,
+    const float* __restrict__ prios, int n, long long cap) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  long long idx = idxs[i];
+  const float p = prios[i];
+  const float old = atomicExch(&tree[idx], p);
+  float delta = p - old;
+  while (idx != 0) {
+    idx = (idx - 1) >> 1;
+    atomicAdd(&tree[idx], delta);
+  }
+}
+
+extern "C" __global__ void drla_per_sample(
+    const float* __restrict__ tree, const float* __restrict__ s,
+    long long* __restrict__ out_idx, float* __restrict__ out_prio, int n,
+    long long cap) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const long long tree_len = 2 * cap - 1;
+  float rem = s[i];
+  long long idx = 0;
+  while (true) {
+    const long long left = 2 * idx + 1;
+    if (left >= tree_len) break;
+    const float ls = tree[left];
+    if (rem <= ls) {
+      idx = left;
+    } else {
+      rem -= ls;
+      idx = left + 1;
+    }
+  }
+  out_idx[i] = idx;
+  out_prio[i] = tree[idx];
+}

@@ -72,7 +72,25 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
     monitor = common.learner_supervision(ctx, queue, supervisor)
     writer = SummaryWriter(ctx.learner_logdir())
     timer = StageTimer()
-    memory = Memory(MEMORY_CAPACITY, seed=args.seed)
+    import torch
+    use_gpu_replay = torch.cuda.is_available()
+    if use_gpu_replay:
+        # GPU-resident PER shard (replay/gpu_memory.py): segment tree,
+        # priorities and payloads all on-device
+        from distributed_reinforcement_learning_amd.replay.gpu_memory import (
+            GpuMemory,
+        )
+        H, W, C = cfg.model_input
+        memory = GpuMemory(MEMORY_CAPACITY, fields={
+            "state": ((H, W, C), torch.uint8),
+            "next_state": ((H, W, C), torch.uint8),
+            "previous_action": ((), torch.int32),
+            "action": ((), torch.int32),
+            "reward": ((), torch.float32),
+            "done": ((), torch.bool),
+        }, device=ctx.device, seed=args.seed)
+    else:
+        memory = Memory(MEMORY_CAPACITY, seed=args.seed)
     train_step, buffer_steps = 0, 0
     try:
         while args.max_steps <= 0 or train_step < args.max_steps:
@@ -85,31 +103,51 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
                 with timer.track("ingest"):
                     u = queue.sample_batch(1)
                     T = u["state"].shape[1]
-                    td = agent.get_td_error(
-                        u["state"][0], u["next_state"][0],
-                        u["previous_action"][0], u["action"][0],
-                        u["reward"][0], u["done"][0])
-                    samples = [
-                        (u["state"][0, t], u["next_state"][0, t],
-                         u["previous_action"][0, t], u["action"][0, t],
-                         u["reward"][0, t], u["done"][0, t])
-                        for t in range(T)
-                    ]
-                    memory.add_batch(td, samples)
+                    if use_gpu_replay:
+                        dev = {k: agent.to_device(v[0]) for k, v in u.items()}
+                        td = agent.get_td_error(
+                            dev["state"], dev["next_state"],
+                            dev["previous_action"], dev["action"],
+                            dev["reward"], dev["done"], as_tensor=True)
+                        memory.add_batch(td, dev)
+                    else:
+                        td = agent.get_td_error(
+                            u["state"][0], u["next_state"][0],
+                            u["previous_action"][0], u["action"][0],
+                            u["reward"][0], u["done"][0])
+                        samples = [
+                            (u["state"][0, t], u["next_state"][0, t],
+                             u["previous_action"][0, t], u["action"][0, t],
+                             u["reward"][0, t], u["done"][0, t])
+                            for t in range(T)
+                        ]
+                        memory.add_batch(td, samples)
                     buffer_steps += 1
             if buffer_steps <= TRAIN_AFTER:
                 continue
             # Phase B: PER train
             t0 = time.time()
-            with timer.track("sample"):
-                batch, idxs, is_weight = memory.sample(cfg.batch_size)
-                stacked = [np.stack([b[i] for b in batch])
-                           for i in range(6)]
-            with timer.track("train"):
-                loss, td_error = agent.distributed_train(
-                    *stacked, is_weight)
-            with timer.track("per_update"):
-                memory.update_batch(idxs, td_error)
+            if use_gpu_replay:
+                with timer.track("sample"):
+                    rows, idxs, is_weight = memory.sample(cfg.batch_size)
+                    b = memory.gather(rows)
+                with timer.track("train"):
+                    loss, td_error = agent.distributed_train(
+                        b["state"], b["next_state"], b["previous_action"],
+                        b["action"], b["reward"], b["done"], is_weight,
+                        as_tensor=True)
+                with timer.track("per_update"):
+                    memory.update_batch(idxs, td_error)
+            else:
+                with timer.track("sample"):
+                    batch, idxs, is_weight = memory.sample(cfg.batch_size)
+                    stacked = [np.stack([b[i] for b in batch])
+                               for i in range(6)]
+                with timer.track("train"):
+                    loss, td_error = agent.distributed_train(
+                        *stacked, is_weight)
+                with timer.track("per_update"):
+                    memory.update_batch(idxs, td_error)
             train_step += 1
             if train_step % TARGET_SYNC_EVERY == 0:
                 agent.target_to_main()

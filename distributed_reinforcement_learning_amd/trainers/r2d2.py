@@ -69,7 +69,25 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
     monitor = common.learner_supervision(ctx, queue, supervisor)
     writer = SummaryWriter(ctx.learner_logdir())
     timer = StageTimer()
-    memory = Memory(MEMORY_CAPACITY, seed=args.seed)
+    import torch
+    use_gpu_replay = torch.cuda.is_available()
+    if use_gpu_replay:
+        from distributed_reinforcement_learning_amd.replay.gpu_memory import (
+            GpuMemory,
+        )
+        H, W, C = cfg.model_input
+        L, Hs = cfg.seq_len, cfg.lstm_size
+        memory = GpuMemory(MEMORY_CAPACITY, fields={
+            "state": ((L, H, W, C), torch.uint8),
+            "previous_action": ((L,), torch.int32),
+            "action": ((L,), torch.int32),
+            "reward": ((L,), torch.float32),
+            "done": ((L,), torch.bool),
+            "initial_h": ((L, Hs), torch.float32),
+            "initial_c": ((L, Hs), torch.float32),
+        }, device=ctx.device, seed=args.seed)
+    else:
+        memory = Memory(MEMORY_CAPACITY, seed=args.seed)
     train_step, buffer_steps = 0, 0
     min_warm = 2 * cfg.batch_size  # reference :122
     try:
@@ -79,28 +97,51 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
             if queue.get_size() > 0 or need_data:
                 with timer.track("ingest"):
                     u = queue.sample_batch(1)
-                    td = agent.get_td_error(
-                        u["state"][0], u["previous_action"][0],
-                        u["action"][0], u["initial_h"][0],
-                        u["initial_c"][0], u["reward"][0], u["done"][0])
-                    memory.add(td, {k: v[0] for k, v in u.items()})
+                    if use_gpu_replay:
+                        dev = {k: agent.to_device(v) for k, v in u.items()}
+                        td = agent.get_td_error_batch(
+                            dev["state"], dev["previous_action"],
+                            dev["action"], dev["initial_h"][:, 0],
+                            dev["initial_c"][:, 0], dev["reward"],
+                            dev["done"], as_tensor=True)
+                        memory.add_batch(td, dev)
+                    else:
+                        td = agent.get_td_error(
+                            u["state"][0], u["previous_action"][0],
+                            u["action"][0], u["initial_h"][0],
+                            u["initial_c"][0], u["reward"][0], u["done"][0])
+                        memory.add(td, {k: v[0] for k, v in u.items()})
                     buffer_steps += 1
             if buffer_steps < min_warm:
                 continue
             t0 = time.time()
-            with timer.track("sample"):
-                batch, idxs, is_weight = memory.sample(cfg.batch_size)
-                stacked = {k: np.stack([b[k] for b in batch])
-                           for k in batch[0]}
-            with timer.track("train"):
-                loss, td_error = agent.train(
-                    state=stacked["state"],
-                    previous_action=stacked["previous_action"],
-                    action=stacked["action"], h=stacked["initial_h"],
-                    c=stacked["initial_c"], reward=stacked["reward"],
-                    done=stacked["done"], weight=is_weight)
-            with timer.track("per_update"):
-                memory.update_batch(idxs, td_error)
+            if use_gpu_replay:
+                with timer.track("sample"):
+                    rows, idxs, is_weight = memory.sample(cfg.batch_size)
+                    b = memory.gather(rows)
+                with timer.track("train"):
+                    loss, td_error = agent.train(
+                        state=b["state"],
+                        previous_action=b["previous_action"],
+                        action=b["action"], h=b["initial_h"],
+                        c=b["initial_c"], reward=b["reward"],
+                        done=b["done"], weight=is_weight, as_tensor=True)
+                with timer.track("per_update"):
+                    memory.update_batch(idxs, td_error)
+            else:
+                with timer.track("sample"):
+                    batch, idxs, is_weight = memory.sample(cfg.batch_size)
+                    stacked = {k: np.stack([b[k] for b in batch])
+                               for k in batch[0]}
+                with timer.track("train"):
+                    loss, td_error = agent.train(
+                        state=stacked["state"],
+                        previous_action=stacked["previous_action"],
+                        action=stacked["action"], h=stacked["initial_h"],
+                        c=stacked["initial_c"], reward=stacked["reward"],
+                        done=stacked["done"], weight=is_weight)
+                with timer.track("per_update"):
+                    memory.update_batch(idxs, td_error)
             train_step += 1
             if train_step % TARGET_SYNC_EVERY == 0:
                 agent.main_to_target()
