@@ -133,7 +133,7 @@ class Agent(AgentBase):
             self.optimizer.step(lr=lr)
             self.global_step += 1
             self.num_env_frames += len(r)
-            td_error = td_signed.abs()
+            td_error = td_signed.detach().abs()
             if not as_tensor:
                 td_error = td_error.cpu().numpy()
             return float(loss.detach()), td_error

@@ -138,11 +138,16 @@ class TrajectoryRing:
     # -- consumer ------------------------------------------------------------
 
     def try_pop_into(self, out: Dict[str, np.ndarray], row: int) -> bool:
+        """Pop one slot into row ``row`` of ``out``. Fields absent from
+        ``out`` are skipped (e.g. the IMPALA learner's pinned staging has no
+        next_state buffer — the batched unroll never reads it)."""
         if self.size() == 0:
             return False
         views = self._slot_views(self.head % self.capacity)
         for name, view in views.items():
-            out[name][row] = view
+            dst = out.get(name)
+            if dst is not None:
+                dst[row] = view
         self._ctr[1] = np.uint64(self.head + 1)
         return True
 

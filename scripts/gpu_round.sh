@@ -5,13 +5,11 @@ mkdir -p gpurun_out
 export PYTORCH_ROCM_ARCH=gfx950
 timeout 900 python -m distributed_reinforcement_learning_amd.ops.build > gpurun_out/build.log 2>&1
 echo "build rc=$?"
-timeout 600 python -m pytest tests/test_gpu_conv.py -q -m gpu -x > gpurun_out/conv_tests.log 2>&1
-echo "conv tests rc=$?"; tail -20 gpurun_out/conv_tests.log
-timeout 1200 python -m pytest tests/ -q -m gpu > gpurun_out/gpu_tests.log 2>&1
-echo "gpu tests rc=$?"; tail -4 gpurun_out/gpu_tests.log
+timeout 1500 python -m pytest tests/ -q -m gpu > gpurun_out/gpu_tests.log 2>&1
+echo "gpu tests rc=$?"; tail -6 gpurun_out/gpu_tests.log
+# end-to-end spawn-mode runs ON GPU: learner on cuda:0 with GPU replay,
+# CPU actors, shm transport
+timeout 1500 python -m pytest tests/test_end_to_end.py -q > gpurun_out/e2e_gpu.log 2>&1
+echo "e2e-on-gpu rc=$?"; tail -4 gpurun_out/e2e_gpu.log
 timeout 900 python bench.py --steps 150 --warmup 30 > gpurun_out/bench_graph.log 2>&1
 echo "bench rc=$?"; cat gpurun_out/bench_graph.log
-cd /tmp && export TMPDIR=/tmp && cd /root/repo
-timeout 900 rocprofv3 --kernel-trace --stats -d gpurun_out/prof5 -o conv \
-  -- python bench.py --steps 30 --warmup 10 > gpurun_out/rocprof5.log 2>&1
-echo "rocprof rc=$?"

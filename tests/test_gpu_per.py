@@ -33,7 +33,7 @@ def test_ring_overwrite_and_update():
                                    dtype=torch.float32).reshape(12, 3)})
     assert len(m) == 8
     # ring: rows hold samples 4..11
-    assert m.data["x"][m.write % 8, 0] == pytest.approx(4 * 3.0)
+    assert float(m.data["x"][m.write % 8, 0]) == pytest.approx(4 * 3.0)
     # update one leaf to dominate
     idxs = torch.tensor([8 - 1 + 0], device="cuda")  # leaf 0
     m.update_batch(idxs, torch.full((1,), 1000.0, device="cuda"))
@@ -51,7 +51,7 @@ def test_sampling_tracks_priorities():
         rows, idxs, w = m.sample(8)
         for r in rows.cpu():
             counts[r] += 1
-        assert w.max() == pytest.approx(1.0)
+        assert float(w.max()) == pytest.approx(1.0)
         assert (rows >= 0).all() and (rows < 64).all()
     assert counts[7] > counts.sum() * 0.5  # dominant priority wins
 
