@@ -106,7 +106,7 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
             ctx.maybe_checkpoint(agent)
             if monitor is not None and train_step % 50 == 0:
                 supervisor.check()
-            if ctx.rank == 0 and (graphed is None
+            if ctx.rank == 0 and (graphed is None or train_step == 1
                                   or train_step % log_every == 0):
                 if graphed is not None:
                     pi_loss, v_loss, entropy, lr = graphed.last_losses()
