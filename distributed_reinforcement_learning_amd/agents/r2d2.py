@@ -80,24 +80,22 @@ class Agent(AgentBase):
     # -- core sequence evaluation -------------------------------------------
 
     def _unroll_q(self, model, s, pa, h0, c0, d, with_grad: bool):
-        """Unroll ``model`` over the full sequence; optionally run the first
-        burn_in steps detached (hidden-state recompute only)."""
+        """Unroll ``model`` over the full sequence; by default the first
+        burn_in steps run as a detached hidden-state recompute
+        (model.burn_in_states — one fused kernel on GPU), and only the
+        trained window unrolls with gradient."""
         if with_grad and not self.burn_in_gradient and self.burn_in > 0:
             b = self.burn_in
-            with torch.no_grad():
-                feat = model.features(
-                    s[:, :b].reshape(-1, *s.shape[2:]),
-                    pa[:, :b].reshape(-1)).reshape(s.shape[0], b, -1)
-                h, c = h0, c0
-                q_burn = []
-                for i in range(b):
-                    h, c = model.lstm(feat[:, i], h, c)
-                    q_burn.append(model._head(h))
-                    keep = (~d[:, i]).to(h.dtype).unsqueeze(1)
-                    h, c = h * keep, c * keep
+            h, c = self.model.burn_in_states(s[:, :b], pa[:, :b], h0, c0,
+                                             d[:, :b])
             q_rest = model.unroll_sequence(s[:, b:], pa[:, b:], h.detach(),
                                            c.detach(), d[:, b:])
-            return torch.cat([torch.stack(q_burn, 1), q_rest], dim=1)
+            # burn-in Q values never enter the loss (it slices [:, b:]);
+            # pad with detached zeros to keep the [B, L] interface
+            B = s.shape[0]
+            pad = torch.zeros(B, b, q_rest.shape[-1], dtype=q_rest.dtype,
+                              device=q_rest.device)
+            return torch.cat([pad, q_rest], dim=1)
         ctx = torch.enable_grad() if with_grad else torch.no_grad()
         with ctx:
             return model.unroll_sequence(s, pa, h0, c0, d)

@@ -63,12 +63,35 @@ class R2D2LstmQ(nn.Module):
         seq_state [B,L,84,84,C], seq_prev_action [B,L], h0/c0 [B,H],
         seq_done [B,L] bool. Returns q_stack [B,L,A].
         After step i: (h, c) *= (~done_i) — reference r2d2_lstm.py:80-82.
+
+        When no gradient is required on GPU (burn-in recompute, target-net
+        unroll, TD scoring), the whole recurrence runs as ONE kernel
+        (ops/hip drla_lstm_seq_fwd): the x-projection is a single batched
+        GEMM and only the tiny h @ Wh chain loops inside the kernel.
         """
         B, L = seq_state.shape[:2]
         feat = self.features(
             seq_state.reshape(B * L, *seq_state.shape[2:]),
             seq_prev_action.reshape(B * L),
         ).reshape(B, L, -1)
+
+        use_seq_kernel = (feat.is_cuda and not torch.is_grad_enabled()
+                          and self.lstm.weight.dtype == torch.bfloat16
+                          and 4 * self.lstm_size <= 1024)
+        if use_seq_kernel:
+            from distributed_reinforcement_learning_amd import ops as _o
+            ext = _o.require_ext()
+            F = feat.shape[-1]
+            w = self.lstm.weight
+            xg = feat.reshape(B * L, F) @ w[:F] + self.lstm.bias
+            h_all, _, _ = ext.lstm_seq_fwd(
+                xg.reshape(B, L, -1).contiguous(),
+                w[F:].contiguous(), h0.float().contiguous(),
+                c0.float().contiguous(), seq_done.contiguous(),
+                self.lstm.forget_bias)
+            q = self._head(h_all.reshape(B * L, -1))
+            return q.reshape(B, L, -1)
+
         h, c = h0, c0
         qs = []
         for i in range(L):
@@ -78,3 +101,35 @@ class R2D2LstmQ(nn.Module):
             h = h * keep
             c = c * keep
         return torch.stack(qs, dim=1)
+
+    @torch.no_grad()
+    def burn_in_states(self, seq_state: torch.Tensor,
+                       seq_prev_action: torch.Tensor, h0: torch.Tensor,
+                       c0: torch.Tensor, seq_done: torch.Tensor):
+        """Hidden-state recompute over the burn-in window (no gradient, no
+        Q heads) — BASELINE's "burn_in hidden-state recompute" hot-path
+        item. Returns (h, c) to start the trained window from. One fused
+        kernel on GPU (drla_lstm_seq_fwd), torch loop on CPU."""
+        B, L = seq_state.shape[:2]
+        feat = self.features(
+            seq_state.reshape(B * L, *seq_state.shape[2:]),
+            seq_prev_action.reshape(B * L)).reshape(B, L, -1)
+        if (feat.is_cuda and self.lstm.weight.dtype == torch.bfloat16
+                and 4 * self.lstm_size <= 1024):
+            from distributed_reinforcement_learning_amd import ops as _o
+            ext = _o.require_ext()
+            F = feat.shape[-1]
+            w = self.lstm.weight
+            xg = feat.reshape(B * L, F) @ w[:F] + self.lstm.bias
+            _, h_fin, c_fin = ext.lstm_seq_fwd(
+                xg.reshape(B, L, -1).contiguous(), w[F:].contiguous(),
+                h0.float().contiguous(), c0.float().contiguous(),
+                seq_done.contiguous(), self.lstm.forget_bias)
+            return h_fin, c_fin
+        h, c = h0, c0
+        for i in range(L):
+            h, c = self.lstm(feat[:, i], h, c)
+            keep = (~seq_done[:, i]).to(h.dtype).unsqueeze(1)
+            h = h * keep
+            c = c * keep
+        return h, c

@@ -100,6 +100,10 @@ extern "C" __global__ void drla_lstm_tail_bwd(const float*, const float*,
                                               const float*, const float*,
                                               const float*, float*, float*,
                                               long long, int);
+extern "C" __global__ void drla_lstm_seq_fwd(
+    const unsigned short*, const float*, const unsigned short*, const float*,
+    const float*, const unsigned char*, float*, float*, float*, float, int,
+    int, int);
 extern "C" __global__ void drla_sq_norm(const float*, float*, long long);
 extern "C" __global__ void drla_sq_norm_bf16(const unsigned short*, float*,
                                              long long);
@@ -521,6 +525,34 @@ std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
   return {grad_gates, grad_c_prev};
 }
 
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_seq_fwd(
+    torch::Tensor xgates, torch::Tensor Wh, torch::Tensor h0,
+    torch::Tensor c0, torch::Tensor done, double forget_bias) {
+  for (auto* t : {&xgates, &Wh, &h0, &c0, &done})
+    check_gpu_contig(*t, "lstm_seq input");
+  TORCH_CHECK(Wh.scalar_type() == torch::kBFloat16, "Wh must be bf16");
+  TORCH_CHECK(done.scalar_type() == torch::kBool, "done must be bool");
+  const int B = xgates.size(0), L = xgates.size(1);
+  const int H = xgates.size(2) / 4;
+  TORCH_CHECK(4 * H <= 1024, "lstm hidden cap is 256");
+  const bool bf16 = xgates.scalar_type() == torch::kBFloat16;
+  auto fopt = h0.options().dtype(torch::kFloat);
+  auto h_out = torch::empty({B, L, H}, fopt);
+  auto h_fin = torch::empty({B, H}, fopt);
+  auto c_fin = torch::empty({B, H}, fopt);
+  const int lds = H * 4 * H * 2 + 2 * H * 4 + 16;
+  hipLaunchKernelGGL(
+      drla_lstm_seq_fwd, dim3(B), dim3(4 * H), lds, cur_stream(),
+      bf16 ? u16p(xgates) : nullptr,
+      bf16 ? nullptr : xgates.data_ptr<float>(), u16p(Wh),
+      h0.data_ptr<float>(), c0.data_ptr<float>(),
+      reinterpret_cast<const unsigned char*>(done.data_ptr<bool>()),
+      h_out.data_ptr<float>(),
+      h_fin.data_ptr<float>(), c_fin.data_ptr<float>(),
+      static_cast<float>(forget_bias), B, L, H);
+  return {h_out, h_fin, c_fin};
+}
+
 torch::Tensor sq_norm(torch::Tensor x) {
   check_gpu_contig(x, "x");
   TORCH_CHECK(x.scalar_type() == torch::kFloat, "sq_norm wants float32");
@@ -672,6 +704,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "action-embedding table gradient (K2 backward)");
   m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");
   m.def("lstm_tail_bwd", &lstm_tail_bwd, "fused LSTM gate tail bwd (K3)");
+  m.def("lstm_seq_fwd", &lstm_seq_fwd,
+        "whole no-grad LSTM unroll in one kernel (K3 seq / burn-in)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");
   m.def("rmsprop_step", &rmsprop_step,
         "fused global-norm-clip + TF-RMSProp update (K12)");

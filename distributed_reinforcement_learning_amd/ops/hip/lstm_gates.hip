@@ -67,3 +67,82 @@ extern "C" __global__ void drla_lstm_tail_bwd(
     grad_gates[g0 + 3 * H] = do_ * o_s * (1.0f - o_s);
   }
 }
+
+// K3 sequence form (SURVEY §5.7a + BASELINE "burn_in hidden-state
+// recompute"): the WHOLE no-grad LSTM unroll in ONE kernel.
+//
+// The x-projection (feat @ Wx + b) has no recurrence and runs as one
+// batched MFMA GEMM outside; only the tiny h @ Wh chain is sequential.
+// One block per batch row: Wh [H][4H] staged in LDS once, then L steps of
+// {gates_h = h @ Wh + xgates[t]; cell tail; done-mask reset} with h/c in
+// LDS. blockDim = 4H (H=64 -> 256 threads); consecutive lanes read
+// consecutive Wh bytes (conflict-free).
+//
+// Replaces the reference's per-timestep replica chain
+// (r2d2_lstm.py:67-114): ~5 launches/step -> 1 launch per sequence.
+
+typedef unsigned short lstm_bf16;
+
+__device__ __forceinline__ float lstm_b2f(lstm_bf16 u) {
+  unsigned int x = ((unsigned int)u) << 16;
+  return __uint_as_float(x);
+}
+
+extern "C" __global__ void drla_lstm_seq_fwd(
+    const lstm_bf16* __restrict__ xg16,  // [B,L,4H] (nullable)
+    const float* __restrict__ xg32,      // [B,L,4H] (nullable)
+    const lstm_bf16* __restrict__ Wh,    // [H][4H] bf16
+    const float* __restrict__ h0,        // [B,H]
+    const float* __restrict__ c0,        // [B,H]
+    const unsigned char* __restrict__ done,  // [B,L]
+    float* __restrict__ h_out,           // [B,L,H]
+    float* __restrict__ h_fin,           // [B,H]
+    float* __restrict__ c_fin,           // [B,H]
+    float forget_bias, int B, int L, int H) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  lstm_bf16* wh = reinterpret_cast<lstm_bf16*>(smem);        // [H*4H]
+  float* hbuf = reinterpret_cast<float*>(wh + H * 4 * H);    // [H]
+  float* cbuf = hbuf + H;                                    // [H]
+
+  const int b = blockIdx.x;
+  const int j = threadIdx.x;          // gate index in [0, 4H)
+  const int G = 4 * H;
+
+  for (int i = j; i < H * G; i += blockDim.x) wh[i] = Wh[i];
+  if (j < H) {
+    hbuf[j] = h0[(long long)b * H + j];
+    cbuf[j] = c0[(long long)b * H + j];
+  }
+  __syncthreads();
+
+  for (int t = 0; t < L; ++t) {
+    // gates_h[j] = sum_k h[k] * Wh[k][j]
+    float g = 0.0f;
+    for (int k = 0; k < H; ++k) {
+      g = fmaf(hbuf[k], lstm_b2f(wh[k * G + j]), g);
+    }
+    const long long xbase = ((long long)b * L + t) * G + j;
+    g += xg16 ? lstm_b2f(xg16[xbase]) : xg32[xbase];
+    __shared__ float gates[1024];
+    gates[j] = g;
+    __syncthreads();
+    if (j < H) {
+      const float i_s = drla_sigmoid(gates[j]);
+      const float g_t = tanhf(gates[H + j]);
+      const float f_s = drla_sigmoid(gates[2 * H + j] + forget_bias);
+      const float o_s = drla_sigmoid(gates[3 * H + j]);
+      const float c_new = f_s * cbuf[j] + i_s * g_t;
+      const float h_new = o_s * tanhf(c_new);
+      h_out[((long long)b * L + t) * H + j] = h_new;
+      // done-mask reset applies to the CARRY, not the emitted h
+      const float keep = done[(long long)b * L + t] ? 0.0f : 1.0f;
+      hbuf[j] = h_new * keep;
+      cbuf[j] = c_new * keep;
+    }
+    __syncthreads();
+  }
+  if (j < H) {
+    h_fin[(long long)b * H + j] = hbuf[j];
+    c_fin[(long long)b * H + j] = cbuf[j];
+  }
+}
