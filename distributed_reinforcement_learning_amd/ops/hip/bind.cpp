@@ -267,7 +267,11 @@ std::tuple<torch::Tensor, torch::Tensor> relu_mask_bwd(torch::Tensor dy,
   auto out = torch::empty_like(dy);
   auto dbias = torch::zeros({CO}, dy.options().dtype(torch::kFloat));
   const long long n = dy.numel();
-  hipLaunchKernelGGL(drla_relu_mask_bwd, dim3(drla_grid(n)), dim3(DRLA_BLOCK),
+  TORCH_CHECK(n % 8 == 0, "relu_mask_bwd wants numel % 8 == 0");
+  // cap blocks: each extra block costs CO atomicAdds on dbias
+  int grid = drla_grid(n / 8);
+  if (grid > 640) grid = 640;
+  hipLaunchKernelGGL(drla_relu_mask_bwd, dim3(grid), dim3(DRLA_BLOCK),
                      0, cur_stream(), u16p(dy), u16p(y), u16pm(out),
                      dbias.data_ptr<float>(), n, (int)CO);
   return {out, dbias};

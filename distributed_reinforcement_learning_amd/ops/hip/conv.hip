@@ -242,23 +242,50 @@ extern "C" __global__ void drla_relu_mask_bwd(
     const bf16raw* __restrict__ dy, const bf16raw* __restrict__ y,
     bf16raw* __restrict__ out, float* __restrict__ dbias, long long n,
     int CO) {
+  // vectorized: each thread moves 8 bf16 per iteration (uint4), which pins
+  // it to ONE 8-column group because stride*8 % CO == 0 (CO in {32,64},
+  // blockDim 256). Per-thread acc[8] -> LDS -> one atomicAdd per (block,
+  // column).
+  const long long n8 = n / 8;
   const long long stride = gridDim.x * (long long)blockDim.x;
   long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-  const int my_co = (int)(i % CO);
-  float acc = 0.0f;
-  for (; i < n; i += stride) {
-    const bf16raw v = (cv_bf2f(y[i]) > 0.0f) ? dy[i] : (bf16raw)0;
-    out[i] = v;
-    acc += cv_bf2f(v);
+  const int tid = threadIdx.x;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  const uint4* dy4 = reinterpret_cast<const uint4*>(dy);
+  const uint4* y4 = reinterpret_cast<const uint4*>(y);
+  uint4* out4 = reinterpret_cast<uint4*>(out);
+  for (; i < n8; i += stride) {
+    const uint4 dv = dy4[i];
+    const uint4 yv = y4[i];
+    uint4 ov;
+    const unsigned int* dp = reinterpret_cast<const unsigned int*>(&dv);
+    const unsigned int* yp = reinterpret_cast<const unsigned int*>(&yv);
+    unsigned int* op = reinterpret_cast<unsigned int*>(&ov);
+    for (int w = 0; w < 4; ++w) {
+      const bf16raw d0 = (bf16raw)(dp[w] & 0xFFFF);
+      const bf16raw d1 = (bf16raw)(dp[w] >> 16);
+      const bf16raw y0 = (bf16raw)(yp[w] & 0xFFFF);
+      const bf16raw y1 = (bf16raw)(yp[w] >> 16);
+      const bf16raw m0 = (cv_bf2f(y0) > 0.0f) ? d0 : (bf16raw)0;
+      const bf16raw m1 = (cv_bf2f(y1) > 0.0f) ? d1 : (bf16raw)0;
+      op[w] = ((unsigned int)m1 << 16) | m0;
+      acc[2 * w] += cv_bf2f(m0);
+      acc[2 * w + 1] += cv_bf2f(m1);
+    }
+    out4[i] = ov;
   }
-  __shared__ float red[DRLA_BLOCK];
-  red[threadIdx.x] = acc;
+  __shared__ float red[DRLA_BLOCK * 8];
+  for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc[e];
   __syncthreads();
-  // threads {co, co+CO, co+2CO, ...} within the block share a column
-  if (threadIdx.x < (unsigned)CO) {
+  // column j gets acc[j%8] of every thread whose group (tid*8)%CO == j-j%8
+  if (tid < CO) {
+    const int g = tid / 8;
+    const int e = tid % 8;
+    // threads with in-block index t where (t*8)%CO == g*8, i.e.
+    // t = g + k*(CO/8)
     float s = 0.0f;
-    for (int t = threadIdx.x; t < DRLA_BLOCK; t += CO) s += red[t];
-    atomicAdd(&dbias[my_co], s);
+    for (int t = g; t < DRLA_BLOCK; t += CO / 8) s += red[t * 8 + e];
+    atomicAdd(&dbias[tid], s);
   }
 }
 
