@@ -160,6 +160,18 @@ class Agent(AgentBase):
                                       action, reward, done, ones)
 
     @torch.no_grad()
+    def target_main_test(self, state, previous_action):
+        """Debug helper (reference agent/apex.py:109-117): print the
+        next-state main/target Q values for one state."""
+        s = self.frames_to_device(np.asarray(state)[None])
+        a = self.to_device(np.asarray([previous_action]), torch.int64)
+        next_main = self.model(s, a).float().cpu().numpy()
+        target = self.target_model(s, a).float().cpu().numpy()
+        print(next_main)
+        print(target)
+        return next_main, target
+
+    @torch.no_grad()
     def get_policy_and_action(self, state, previous_action, epsilon: float
                               ) -> Tuple[int, np.ndarray, float]:
         s = self.frames_to_device(np.asarray(state)[None])

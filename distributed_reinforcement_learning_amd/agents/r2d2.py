@@ -195,6 +195,30 @@ class Agent(AgentBase):
         return float(loss.detach()), td
 
     @torch.no_grad()
+    def main_q_value_test(self, state, h, c, done, previous_action):
+        """Debug helper (reference agent/r2d2.py:188-199): main-net Q
+        values over one stored sequence."""
+        s = self.frames_to_device(np.asarray(state)[None])
+        pa = self.to_device(np.asarray(previous_action)[None], torch.int64)
+        d = self.to_device(np.asarray(done)[None], torch.bool)
+        h0 = self.to_device(np.asarray(h)[0][None], torch.float32)
+        c0 = self.to_device(np.asarray(c)[0][None], torch.float32)
+        q = self.model.unroll_sequence(s, pa, h0, c0, d)
+        return q[0].float().cpu().numpy()
+
+    @torch.no_grad()
+    def target_q_value_test(self, state, h, c, done, previous_action):
+        """Debug helper (reference agent/r2d2.py:201-211): target-net Q
+        values over one stored sequence."""
+        s = self.frames_to_device(np.asarray(state)[None])
+        pa = self.to_device(np.asarray(previous_action)[None], torch.int64)
+        d = self.to_device(np.asarray(done)[None], torch.bool)
+        h0 = self.to_device(np.asarray(h)[0][None], torch.float32)
+        c0 = self.to_device(np.asarray(c)[0][None], torch.float32)
+        q = self.target_model.unroll_sequence(s, pa, h0, c0, d)
+        return q[0].float().cpu().numpy()
+
+    @torch.no_grad()
     def get_action(self, state, h, c, previous_action, epsilon: float):
         """Epsilon-greedy single-step acting (reference agent/r2d2.py:166-186).
         Returns (action, q[action], h', c')."""
