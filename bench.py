@@ -48,6 +48,10 @@ def main(argv=None) -> None:
                    help="distinct synthetic batches cycled through")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph capture (eager launches)")
+    p.add_argument("--model", default="deep_conv",
+                   choices=["deep_conv", "resnet"],
+                   help="flagship (deep_conv) or IMPALA ResNet-large "
+                        "(BASELINE config #5, stresses the all-reduce)")
     args = p.parse_args(argv)
 
     from distributed_reinforcement_learning_amd.agents import impala
@@ -65,7 +69,8 @@ def main(argv=None) -> None:
         lstm_hidden_size=H, discount_factor=0.99, start_learning_rate=6e-4,
         end_learning_rate=0.0, learning_frame=10 ** 9,
         baseline_loss_coef=1.0, entropy_coef=0.05, gradient_clip_norm=40.0,
-        reward_clipping="abs_one", device=device, seed=1234 + rank)
+        reward_clipping="abs_one", device=device, seed=1234 + rank,
+        model_arch=args.model)
     if world > 1:
         from distributed_reinforcement_learning_amd.parallel.dist import (
             broadcast_module,
@@ -157,7 +162,10 @@ def main(argv=None) -> None:
             "dtype": "bf16" if have_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "impala deep-conv 84x84x4 + LSTM256 V-trace",
+                "model": ("impala deep-conv 84x84x4 + LSTM256 V-trace"
+                          if args.model == "deep_conv"
+                          else "impala resnet-large 84x84x4 + LSTM256 "
+                               "V-trace"),
                 "global_batch": B * world,
                 "seq_len": T,
                 "parallelism": f"dp{world}",

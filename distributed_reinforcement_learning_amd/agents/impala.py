@@ -34,7 +34,8 @@ class Agent(AgentBase):
                  entropy_coef: float, gradient_clip_norm: float,
                  reward_clipping: str, device: str = "cpu",
                  compute_dtype: torch.dtype = torch.bfloat16,
-                 build_optimizer: bool = True, seed: Optional[int] = None):
+                 build_optimizer: bool = True, seed: Optional[int] = None,
+                 model_arch: str = "deep_conv"):
         super().__init__(device=device, compute_dtype=compute_dtype)
         self.trajectory = trajectory
         self.input_shape = tuple(input_shape)
@@ -52,9 +53,18 @@ class Agent(AgentBase):
             torch.manual_seed(seed)
         self.rng = np.random.default_rng(seed)
 
-        self.model = self.finalize_model(
-            ImpalaActorCritic(self.input_shape, num_action,
-                              lstm_hidden_size))
+        if model_arch == "resnet":
+            # BASELINE config #5: IMPALA ResNet-large (xGMI scaling curve)
+            from distributed_reinforcement_learning_amd.models import (
+                ImpalaResNetActorCritic,
+            )
+            net = ImpalaResNetActorCritic(self.input_shape, num_action,
+                                          lstm_hidden_size)
+        else:
+            net = ImpalaActorCritic(self.input_shape, num_action,
+                                    lstm_hidden_size)
+        self.model_arch = model_arch
+        self.model = self.finalize_model(net)
         self.optimizer = None
         if build_optimizer:
             # TF RMSProp(decay=.99, momentum=0, eps=.1) + clip_by_global_norm

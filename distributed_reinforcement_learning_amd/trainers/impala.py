@@ -44,7 +44,8 @@ def build_agent(ctx, device: str, build_optimizer: bool,
         entropy_coef=cfg.entropy_coef,
         gradient_clip_norm=cfg.gradient_clip_norm,
         reward_clipping=cfg.reward_clipping, device=device,
-        build_optimizer=build_optimizer, seed=seed)
+        build_optimizer=build_optimizer, seed=seed,
+        model_arch=cfg.get("model_arch", "deep_conv"))
 
 
 def learner(ctx: common.TrainerContext, supervisor=None) -> None:

@@ -91,3 +91,34 @@ def test_param_count_impala_close_to_reference_scale():
     m = ImpalaActorCritic([84, 84, 4], 18, 256)
     n = sum(p.numel() for p in m.parameters())
     assert 3_500_000 < n < 5_000_000
+
+
+def test_impala_resnet_model_and_agent():
+    """BASELINE config #5 model: ResNet torso, same agent interface."""
+    from distributed_reinforcement_learning_amd.models import (
+        ImpalaResNetActorCritic,
+    )
+    from distributed_reinforcement_learning_amd.agents import impala
+    m = ImpalaResNetActorCritic([84, 84, 4], 6, 32)
+    p, v, h, c = m.single_step(torch.rand(2, 84, 84, 4),
+                               torch.tensor([0, 1]),
+                               torch.zeros(2, 32), torch.zeros(2, 32))
+    assert p.shape == (2, 6) and torch.allclose(p.sum(1), torch.ones(2),
+                                                atol=1e-5)
+    agent = impala.Agent(
+        trajectory=4, input_shape=[84, 84, 4], num_action=6,
+        lstm_hidden_size=32, discount_factor=0.99, start_learning_rate=1e-3,
+        end_learning_rate=0.0, learning_frame=10 ** 9,
+        baseline_loss_coef=1.0, entropy_coef=0.05, gradient_clip_norm=40.0,
+        reward_clipping="abs_one", seed=0, model_arch="resnet")
+    rng = np.random.default_rng(0)
+    out = agent.train(
+        state=rng.integers(0, 255, (2, 4, 84, 84, 4), dtype=np.uint8),
+        reward=rng.normal(size=(2, 4)).astype(np.float32),
+        action=rng.integers(0, 6, (2, 4)).astype(np.int32),
+        done=np.zeros((2, 4), dtype=bool),
+        behavior_policy=np.full((2, 4, 6), 1 / 6, dtype=np.float32),
+        previous_action=rng.integers(0, 6, (2, 4)).astype(np.int32),
+        initial_h=np.zeros((2, 4, 32), dtype=np.float32),
+        initial_c=np.zeros((2, 4, 32), dtype=np.float32))
+    assert np.isfinite(out).all()
