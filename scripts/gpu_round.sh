@@ -15,3 +15,5 @@ cd /tmp && export TMPDIR=/tmp && cd /root/repo
 timeout 900 rocprofv3 --kernel-trace --stats -d gpurun_out/proflatest -o latest \
   -- python bench.py --steps 30 --warmup 10 > gpurun_out/rocproflatest.log 2>&1
 echo "rocprof rc=$?"
+timeout 900 python bench.py --steps 60 --warmup 15 --model resnet > gpurun_out/bench_resnet.log 2>&1
+echo "bench-resnet rc=$?"; cat gpurun_out/bench_resnet.log
