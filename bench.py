@@ -30,6 +30,9 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+from distributed_reinforcement_learning_amd.utils import tunableop
+tunableop.enable()  # shipped hipBLASLt tuning table (+3.5%; must precede torch GEMMs)
+
 import numpy as np
 import torch
 

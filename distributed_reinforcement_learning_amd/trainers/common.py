@@ -20,6 +20,9 @@ import os
 import time
 from typing import Callable, Optional
 
+from distributed_reinforcement_learning_amd.utils import tunableop
+tunableop.enable()  # before the first GEMM (learner hot path)
+
 import torch
 
 from distributed_reinforcement_learning_amd.config import (
