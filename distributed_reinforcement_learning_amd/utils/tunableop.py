@@ -24,7 +24,6 @@ def enable() -> bool:
         return False
     os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
     os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"  # use the table, no search
-    # %d expands to the device ordinal; we ship one table (single-arch node)
     # %d expands to the device ordinal; identical GPUs share one tuning,
     # shipped as tunableop_gfx950_{0..7}.csv
     os.environ["PYTORCH_TUNABLEOP_FILENAME"] = table.replace(
