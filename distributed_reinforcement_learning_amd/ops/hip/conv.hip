@@ -344,7 +344,7 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
                                 float* __restrict__ scratch,     // [K][CO]
                                 int batch) {
   constexpr int K = KH * KW * CI;
-  constexpr int BKM = 32;   // m-rows per MFMA step
+  constexpr int BKM = 64;   // m-rows per chunk (2 MFMA k-steps; fewer barriers)
   constexpr int BKK = 64;   // k-cols per block
   constexpr int PAD = 8;
   constexpr int NFRAG = CO / 16;
@@ -369,15 +369,16 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
   const int m_begin = blockIdx.y * m_per_split;
   const int m_end = min(M, m_begin + m_per_split);
 
-  // staging assignment (fwd-style): thread t -> m-row = t>>3,
-  // k-chunk = (t&7)*8 of the 32x64 A image; decode coords once per chunk
-  const int a_m = tid >> 3;
-  const int a_k = (tid & 7) * 8;
+  // staging assignment (fwd-style): thread t -> m-rows {t>>2, 32+(t>>2)},
+  // k-chunk = (t&3)*16 of the 64x64 A image; decode coords once per chunk
+  const int a_m = tid >> 2;
+  const int a_k = (tid & 3) * 16;
 
   for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
-    // ---- stage Am[m][k0..k0+8) ----
+    // ---- stage Am[m][a_k..a_k+16): one row, two 8-wide segments ----
     {
-      const int m = m0 + a_m;
+      const int lm = a_m;
+      const int m = m0 + lm;
       if (m < m_end) {
         const int n_idx = m / (HO * WO);
         const int rem = m - n_idx * (HO * WO);
@@ -385,80 +386,90 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
         const int wo = rem - ho * WO;
         const long long base =
             ((long long)n_idx * HI + ho * STRIDE) * WI + wo * STRIDE;
-        const int kk = k_row0 + a_k;
-        if constexpr (CI == 4) {
-          for (int t = 0; t < 2; ++t) {
-            const int k2 = kk + t * 4;
-            const int kh = k2 / (KW * CI);
-            const int kw = (k2 - kh * KW * CI) / CI;
-            const uchar4 v = *reinterpret_cast<const uchar4*>(
-                in + (base + (long long)kh * WI + kw) * CI);
-            const float s = 1.0f / 255.0f;
-            Am[a_m][a_k + t * 4 + 0] = drla_f32_to_bf16(v.x * s);
-            Am[a_m][a_k + t * 4 + 1] = drla_f32_to_bf16(v.y * s);
-            Am[a_m][a_k + t * 4 + 2] = drla_f32_to_bf16(v.z * s);
-            Am[a_m][a_k + t * 4 + 3] = drla_f32_to_bf16(v.w * s);
+        for (int seg = 0; seg < 2; ++seg) {
+          const int kk = k_row0 + a_k + seg * 8;
+          if constexpr (CI == 4) {
+            for (int t = 0; t < 2; ++t) {
+              const int k2 = kk + t * 4;
+              const int kh = k2 / (KW * CI);
+              const int kw = (k2 - kh * KW * CI) / CI;
+              const uchar4 v = *reinterpret_cast<const uchar4*>(
+                  in + (base + (long long)kh * WI + kw) * CI);
+              const float s = 1.0f / 255.0f;
+              const int o = a_k + seg * 8 + t * 4;
+              Am[lm][o + 0] = drla_f32_to_bf16(v.x * s);
+              Am[lm][o + 1] = drla_f32_to_bf16(v.y * s);
+              Am[lm][o + 2] = drla_f32_to_bf16(v.z * s);
+              Am[lm][o + 3] = drla_f32_to_bf16(v.w * s);
+            }
+          } else if constexpr (CI == 1) {
+            for (int t = 0; t < 8; ++t) {
+              const int k2 = kk + t;
+              const int kh = k2 / KW;
+              const int kw = k2 - kh * KW;
+              const unsigned char v =
+                  (k2 < K) ? in[base + (long long)kh * WI + kw] : 0;
+              Am[lm][a_k + seg * 8 + t] =
+                  drla_f32_to_bf16(v * (1.0f / 255.0f));
+            }
+          } else {
+            const int kh = kk / (KW * CI);
+            const int kwci = kk - kh * KW * CI;
+            const int kw = kwci / CI;
+            const int ci = kwci - kw * CI;
+            const bf16raw* src = reinterpret_cast<const bf16raw*>(in) +
+                                 (base + (long long)kh * WI + kw) * CI + ci;
+            *reinterpret_cast<uint4*>(&Am[lm][a_k + seg * 8]) =
+                (kk < K) ? *reinterpret_cast<const uint4*>(src)
+                         : uint4{0, 0, 0, 0};
           }
-        } else if constexpr (CI == 1) {
-          for (int t = 0; t < 8; ++t) {
-            const int k2 = kk + t;
-            const int kh = k2 / KW;
-            const int kw = k2 - kh * KW;
-            const unsigned char v =
-                (k2 < K) ? in[base + (long long)kh * WI + kw] : 0;
-            Am[a_m][a_k + t] = drla_f32_to_bf16(v * (1.0f / 255.0f));
-          }
-        } else {
-          const int kh = kk / (KW * CI);
-          const int kwci = kk - kh * KW * CI;
-          const int kw = kwci / CI;
-          const int ci = kwci - kw * CI;
-          const bf16raw* src = reinterpret_cast<const bf16raw*>(in) +
-                               (base + (long long)kh * WI + kw) * CI + ci;
-          *reinterpret_cast<uint4*>(&Am[a_m][a_k]) =
-              (kk < K) ? *reinterpret_cast<const uint4*>(src)
-                       : uint4{0, 0, 0, 0};
         }
       } else {
-        *reinterpret_cast<uint4*>(&Am[a_m][a_k]) = uint4{0, 0, 0, 0};
+        *reinterpret_cast<uint4*>(&Am[lm][a_k]) = uint4{0, 0, 0, 0};
+        *reinterpret_cast<uint4*>(&Am[lm][a_k + 8]) = uint4{0, 0, 0, 0};
       }
     }
-    // ---- stage Bm[m][co]: contiguous dY rows ----
+    // ---- stage Bm[m][co]: contiguous dY rows, both m-halves ----
     {
-      constexpr int CO_PER_T = (CO * BKM) / 256;  // 4 or 8
-      const int m = m0 + (tid / (CO / CO_PER_T));
-      const int co0 = (tid % (CO / CO_PER_T)) * CO_PER_T;
-      if constexpr (CO_PER_T == 8) {
-        *reinterpret_cast<uint4*>(&Bm[tid / (CO / CO_PER_T)][co0]) =
-            (m < m_end) ? *reinterpret_cast<const uint4*>(
-                              dy + (long long)m * CO + co0)
-                        : uint4{0, 0, 0, 0};
-      } else {
-        *reinterpret_cast<uint2*>(&Bm[tid / (CO / CO_PER_T)][co0]) =
-            (m < m_end) ? *reinterpret_cast<const uint2*>(
-                              dy + (long long)m * CO + co0)
-                        : uint2{0, 0};
+      constexpr int CO_PER_T = (CO * 32) / 256;  // 4 or 8 per half
+      for (int half = 0; half < 2; ++half) {
+        const int lm = (tid / (CO / CO_PER_T)) + half * 32;
+        const int m = m0 + lm;
+        const int co0 = (tid % (CO / CO_PER_T)) * CO_PER_T;
+        if constexpr (CO_PER_T == 8) {
+          *reinterpret_cast<uint4*>(&Bm[lm][co0]) =
+              (m < m_end) ? *reinterpret_cast<const uint4*>(
+                                dy + (long long)m * CO + co0)
+                          : uint4{0, 0, 0, 0};
+        } else {
+          *reinterpret_cast<uint2*>(&Bm[lm][co0]) =
+              (m < m_end) ? *reinterpret_cast<const uint2*>(
+                                dy + (long long)m * CO + co0)
+                          : uint2{0, 0};
+        }
       }
     }
     __syncthreads();
 
-    // wave w owns k-rows [w*16, w*16+16): A_op[k][m] = Am[m][k] (transposed
-    // scalar reads), B_op[m][co] = Bm[m][co] (transposed scalar reads)
-    bf16x8 a_frag;
-    {
-      const int k = wave * 16 + (lane & 15);
-      for (int e = 0; e < 8; ++e) {
-        a_frag[e] = (short)Am[(lane >> 4) * 8 + e][k];
+    // wave w owns k-rows [w*16, w*16+16); two MFMA k-steps per chunk.
+    // A_op[k][m] = Am[m][k], B_op[m][co] = Bm[m][co] (transposed reads)
+    for (int kk = 0; kk < BKM; kk += 32) {
+      bf16x8 a_frag;
+      {
+        const int k = wave * 16 + (lane & 15);
+        for (int e = 0; e < 8; ++e) {
+          a_frag[e] = (short)Am[kk + (lane >> 4) * 8 + e][k];
+        }
       }
-    }
-    for (int ni = 0; ni < NFRAG; ++ni) {
-      bf16x8 b_frag;
-      const int co = ni * 16 + (lane & 15);
-      for (int e = 0; e < 8; ++e) {
-        b_frag[e] = (short)Bm[(lane >> 4) * 8 + e][co];
+      for (int ni = 0; ni < NFRAG; ++ni) {
+        bf16x8 b_frag;
+        const int co = ni * 16 + (lane & 15);
+        for (int e = 0; e < 8; ++e) {
+          b_frag[e] = (short)Bm[kk + (lane >> 4) * 8 + e][co];
+        }
+        acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
+                                                          acc[ni], 0, 0, 0);
       }
-      acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
-                                                        acc[ni], 0, 0, 0);
     }
     __syncthreads();
   }

@@ -5,16 +5,9 @@ mkdir -p gpurun_out
 export PYTORCH_ROCM_ARCH=gfx950
 timeout 900 python -m distributed_reinforcement_learning_amd.ops.build > gpurun_out/build.log 2>&1
 echo "build rc=$?"
-# A: baseline
-timeout 900 python bench.py --steps 100 --warmup 20 > gpurun_out/bench_a.log 2>&1
-grep -o '"ms_per_step": [0-9.]*' gpurun_out/bench_a.log
-# B: TunableOp (tune during warmup, then measure)
-export PYTORCH_TUNABLEOP_ENABLED=1
-export PYTORCH_TUNABLEOP_FILENAME=gpurun_out/tunableop_%d.csv
-timeout 1200 python bench.py --steps 100 --warmup 20 > gpurun_out/bench_tuned.log 2>&1
-echo "tuned rc=$?"
-grep -o '"ms_per_step": [0-9.]*' gpurun_out/bench_tuned.log
-# C: tuned again (reads the CSV, no tuning overhead)
-timeout 900 python bench.py --steps 100 --warmup 20 > gpurun_out/bench_tuned2.log 2>&1
-grep -o '"ms_per_step": [0-9.]*' gpurun_out/bench_tuned2.log
-unset PYTORCH_TUNABLEOP_ENABLED
+timeout 1500 python -m pytest tests/ -q -m gpu > gpurun_out/gpu_tests.log 2>&1
+echo "gpu tests rc=$?"; tail -3 gpurun_out/gpu_tests.log
+timeout 900 python bench.py --steps 150 --warmup 30 > gpurun_out/bench_graph.log 2>&1
+echo "bench rc=$?"; grep -o '"value": [0-9.]*, "unit[^,]*, "n_gpus": [0-9]*' gpurun_out/bench_graph.log; grep -o '"ms_per_step": [0-9.]*' gpurun_out/bench_graph.log
+timeout 600 python -c "import __graft_entry__ as g; g.build(); g.smoke(); print('graft clean')" > gpurun_out/smoke.log 2>&1
+echo "smoke rc=$?"; tail -2 gpurun_out/smoke.log
