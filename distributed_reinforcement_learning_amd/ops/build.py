@@ -33,6 +33,7 @@ SOURCES = [
     "conv.hip",
     "per_tree.hip",
     "dqn_loss.hip",
+    "mlp_heads.hip",
 ]
 
 

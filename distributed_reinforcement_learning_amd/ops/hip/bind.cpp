@@ -100,6 +100,17 @@ extern "C" __global__ void drla_lstm_tail_bwd(const float*, const float*,
                                               const float*, const float*,
                                               const float*, float*, float*,
                                               long long, int);
+extern "C" __global__ void drla_mlp_heads_fwd(
+    const float*, const unsigned short*, const float*, const unsigned short*,
+    const float*, const unsigned short*, const float*, const unsigned short*,
+    const float*, const unsigned short*, const float*, const unsigned short*,
+    const float*, unsigned short*, float*, unsigned short*, int, int);
+extern "C" __global__ void drla_mlp_heads_bwd(
+    const unsigned short*, const float*, const unsigned short*,
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    unsigned short*, unsigned short*, unsigned short*, unsigned short*,
+    float*, float*, float*, float*, float*, float*, float*, int, int);
 extern "C" __global__ void drla_lstm_seq_fwd(
     const unsigned short*, const float*, const unsigned short*, const float*,
     const float*, const unsigned char*, float*, float*, float*, float, int,
@@ -529,6 +540,65 @@ std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
   return {grad_gates, grad_c_prev};
 }
 
+std::vector<torch::Tensor> mlp_heads_fwd(
+    torch::Tensor h, std::vector<torch::Tensor> weights,
+    std::vector<torch::Tensor> biases_f32, int64_t A) {
+  check_gpu_contig(h, "h");
+  TORCH_CHECK(weights.size() == 6 && biases_f32.size() == 6);
+  for (auto& t : weights) check_gpu_contig(t, "W");
+  for (auto& t : biases_f32) check_gpu_contig(t, "b");
+  const int N = h.size(0);
+  TORCH_CHECK(h.size(1) == 256 && A <= 32);
+  auto bopt = h.options().dtype(torch::kBFloat16);
+  auto logits = torch::empty({N, A}, bopt);
+  auto value = torch::empty({N}, h.options().dtype(torch::kFloat));
+  auto stash = torch::empty({N, 5 * 256}, bopt);
+  hipLaunchKernelGGL(
+      drla_mlp_heads_fwd, dim3((N + 31) / 32), dim3(256), 0, cur_stream(),
+      h.data_ptr<float>(), u16p(weights[0]),
+      biases_f32[0].data_ptr<float>(), u16p(weights[1]),
+      biases_f32[1].data_ptr<float>(), u16p(weights[2]),
+      biases_f32[2].data_ptr<float>(), u16p(weights[3]),
+      biases_f32[3].data_ptr<float>(), u16p(weights[4]),
+      biases_f32[4].data_ptr<float>(), u16p(weights[5]),
+      biases_f32[5].data_ptr<float>(), u16pm(logits),
+      value.data_ptr<float>(), u16pm(stash), N, (int)A);
+  return {logits, value, stash};
+}
+
+std::vector<torch::Tensor> mlp_heads_bwd(
+    torch::Tensor dlogits, torch::Tensor dvalue, torch::Tensor stash,
+    std::vector<torch::Tensor> weights, int64_t A) {
+  check_gpu_contig(dlogits, "dlogits");
+  check_gpu_contig(dvalue, "dvalue");
+  check_gpu_contig(stash, "stash");
+  TORCH_CHECK(weights.size() == 6);
+  const int N = dvalue.numel();
+  auto bopt = dlogits.options().dtype(torch::kBFloat16);
+  auto fopt = dvalue.options().dtype(torch::kFloat);
+  auto dz1p = torch::empty({N, 256}, bopt);
+  auto dz2p = torch::empty({N, 256}, bopt);
+  auto dz1v = torch::empty({N, 256}, bopt);
+  auto dz2v = torch::empty({N, 256}, bopt);
+  auto dh = torch::empty({N, 256}, fopt);
+  auto db1p = torch::zeros({256}, fopt);
+  auto db2p = torch::zeros({256}, fopt);
+  auto db3p = torch::zeros({A}, fopt);
+  auto db1v = torch::zeros({256}, fopt);
+  auto db2v = torch::zeros({256}, fopt);
+  auto db3v = torch::zeros({1}, fopt);
+  hipLaunchKernelGGL(
+      drla_mlp_heads_bwd, dim3((N + 31) / 32), dim3(256), 0, cur_stream(),
+      u16p(dlogits), dvalue.data_ptr<float>(), u16p(stash),
+      u16p(weights[0]), u16p(weights[1]), u16p(weights[2]),
+      u16p(weights[3]), u16p(weights[4]), u16p(weights[5]), u16pm(dz1p),
+      u16pm(dz2p), u16pm(dz1v), u16pm(dz2v), dh.data_ptr<float>(),
+      db1p.data_ptr<float>(), db2p.data_ptr<float>(),
+      db3p.data_ptr<float>(), db1v.data_ptr<float>(),
+      db2v.data_ptr<float>(), db3v.data_ptr<float>(), N, (int)A);
+  return {dz1p, dz2p, dz1v, dz2v, dh, db1p, db2p, db3p, db1v, db2v, db3v};
+}
+
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_seq_fwd(
     torch::Tensor xgates, torch::Tensor Wh, torch::Tensor h0,
     torch::Tensor c0, torch::Tensor done, double forget_bias) {
@@ -708,6 +778,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "action-embedding table gradient (K2 backward)");
   m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");
   m.def("lstm_tail_bwd", &lstm_tail_bwd, "fused LSTM gate tail bwd (K3)");
+  m.def("mlp_heads_fwd", &mlp_heads_fwd,
+        "fused policy+value MLP heads forward (K4)");
+  m.def("mlp_heads_bwd", &mlp_heads_bwd,
+        "fused heads dgrad chain + ReLU masks + bias grads (K4 bwd)");
   m.def("lstm_seq_fwd", &lstm_seq_fwd,
         "whole no-grad LSTM unroll in one kernel (K3 seq / burn-in)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");
