@@ -321,8 +321,11 @@ extern "C" __global__ __launch_bounds__(256) void drla_mlp_heads_bwd(
     atomicAdd(&db3p[tid], s);
   }
   __syncthreads();
+  // kdim = A (not 32): rows >= A of W3p do not exist — the stage must
+  // zero-fill them, or garbage (possibly NaN bits) meets the zero-padded
+  // dz and 0 * NaN = NaN
   mh_dgrad_layer<true, false>(dza, dzb, W3p, wbuf, a2p, dz2p, nullptr,
-                              colsum, row0, N, 32);
+                              colsum, row0, N, A);
   // flush db2p
   if (tid < MH_HID) {
     atomicAdd(&db2p[tid], colsum[tid]);
