@@ -44,9 +44,15 @@ class _FusedMlpHeads(torch.autograd.Function):
             dvalue = torch.zeros(N, dtype=torch.float32, device=dev)
         dlogits = dlogits.to(torch.bfloat16).contiguous()
         dvalue = dvalue.float().contiguous()
+        # the dgrad kernel consumes PRE-TRANSPOSED weights (its A-operand
+        # streams matrix rows); layer-3 policy is zero-padded to K=32
+        wT3p = torch.zeros(256, 32, dtype=torch.bfloat16, device=dev)
+        wT3p[:, :A] = w3p.t()
+        wT = [w1p.t().contiguous(), w2p.t().contiguous(), wT3p,
+              w1v.t().contiguous(), w2v.t().contiguous(),
+              w3v.contiguous()]
         (dz1p, dz2p, dz1v, dz2v, dh, db1p, db2p, db3p, db1v, db2v,
-         db3v) = ext.mlp_heads_bwd(dlogits, dvalue, stash,
-                                   [w1p, w2p, w3p, w1v, w2v, w3v], A)
+         db3v) = ext.mlp_heads_bwd(dlogits, dvalue, stash, wT, A)
         soff = N * 256
         flat = stash.reshape(-1)
         a1p = flat[0:soff].reshape(N, 256)
