@@ -554,7 +554,8 @@ std::vector<torch::Tensor> mlp_heads_fwd(
   auto value = torch::empty({N}, h.options().dtype(torch::kFloat));
   auto stash = torch::empty({N, 5 * 256}, bopt);
   hipLaunchKernelGGL(
-      drla_mlp_heads_fwd, dim3((N + 31) / 32), dim3(256), 0, cur_stream(),
+      drla_mlp_heads_fwd, dim3((N + 15) / 16, 2), dim3(256), 0,
+      cur_stream(),
       h.data_ptr<float>(), u16p(weights[0]),
       biases_f32[0].data_ptr<float>(), u16p(weights[1]),
       biases_f32[1].data_ptr<float>(), u16p(weights[2]),
@@ -580,7 +581,7 @@ std::vector<torch::Tensor> mlp_heads_bwd(
   auto dz2p = torch::empty({N, 256}, bopt);
   auto dz1v = torch::empty({N, 256}, bopt);
   auto dz2v = torch::empty({N, 256}, bopt);
-  auto dh = torch::empty({N, 256}, fopt);
+  auto dh = torch::zeros({N, 256}, fopt);  // both head chains atomicAdd
   auto db1p = torch::zeros({256}, fopt);
   auto db2p = torch::zeros({256}, fopt);
   auto db3p = torch::zeros({A}, fopt);
@@ -588,7 +589,8 @@ std::vector<torch::Tensor> mlp_heads_bwd(
   auto db2v = torch::zeros({256}, fopt);
   auto db3v = torch::zeros({1}, fopt);
   hipLaunchKernelGGL(
-      drla_mlp_heads_bwd, dim3((N + 31) / 32), dim3(256), 0, cur_stream(),
+      drla_mlp_heads_bwd, dim3((N + 15) / 16, 2), dim3(256), 0,
+      cur_stream(),
       u16p(dlogits), dvalue.data_ptr<float>(), u16p(stash),
       u16p(weights[0]), u16p(weights[1]), u16p(weights[2]),
       u16p(weights[3]), u16p(weights[4]), u16p(weights[5]), u16pm(dz1p),

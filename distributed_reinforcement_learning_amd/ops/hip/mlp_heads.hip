@@ -26,7 +26,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 
 #define MH_HID 256
-#define MH_BM 32
+#define MH_BM 16
 #define MH_PAD 8
 #define MH_LD (MH_HID + MH_PAD)
 
@@ -60,10 +60,8 @@ __device__ void mh_pass(const bf16raw (*act_in)[MH_LD],
   const int out0 = wave * 64;          // this wave's 64 output columns
   const bool live = out0 < nout;
 
-  f32x4 acc[4][2];
-  for (int ni = 0; ni < 4; ++ni)
-    for (int rj = 0; rj < 2; ++rj)
-      acc[ni][rj] = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc[4];
+  for (int ni = 0; ni < 4; ++ni) acc[ni] = {0.f, 0.f, 0.f, 0.f};
 
   const bf16x8 zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
   if (live) {
@@ -78,12 +76,10 @@ __device__ void mh_pass(const bf16raw (*act_in)[MH_LD],
                 ? *reinterpret_cast<const bf16x8*>(
                       M + (long long)orow * kdim + k0 + (lane >> 4) * 8)
                 : zero8;
-        for (int rj = 0; rj < 2; ++rj) {
-          const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
-              &act_in[rj * 16 + (lane & 15)][k0 + (lane >> 4) * 8]);
-          acc[ni][rj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag, b_frag, acc[ni][rj], 0, 0, 0);
-        }
+        const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            &act_in[lane & 15][k0 + (lane >> 4) * 8]);
+        acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag, acc[ni], 0, 0, 0);
       }
     }
   }
@@ -95,36 +91,35 @@ __device__ void mh_pass(const bf16raw (*act_in)[MH_LD],
     for (int ni = 0; ni < 4; ++ni) {
       for (int r = 0; r < 4; ++r) {
         const int oc = out0 + ni * 16 + (lane >> 4) * 4 + r;
-        float csum = 0.0f;
-        for (int rj = 0; rj < 2; ++rj) {
-          const int lrow = rj * 16 + (lane & 15);
-          const int grow = row0 + lrow;
-          float v = acc[ni][rj][r];
-          if (bias) v += (oc < nout) ? bias[oc] : 0.0f;
-          if (RELU) v = fmaxf(v, 0.0f);
-          if (mask_stash) {
-            const float a = (grow < N)
-                ? mh_b2f(mask_stash[(long long)grow * MH_HID + oc]) : 0.0f;
-            v = (a > 0.0f) ? v : 0.0f;
+        const int lrow = lane & 15;
+        const int grow = row0 + lrow;
+        float v = acc[ni][r];
+        if (bias) v += (oc < nout) ? bias[oc] : 0.0f;
+        if (RELU) v = fmaxf(v, 0.0f);
+        if (mask_stash) {
+          const float a = (grow < N)
+              ? mh_b2f(mask_stash[(long long)grow * MH_HID + oc]) : 0.0f;
+          v = (a > 0.0f) ? v : 0.0f;
+        }
+        if (dh_global) {
+          // both head chains land here concurrently (blockIdx.y split):
+          // accumulate atomically into the zeroed dh buffer
+          if (grow < N && oc < nout) {
+            atomicAdd(&dh_global[(long long)grow * MH_HID + oc], v);
           }
-          if (dh_global) {
-            if (grow < N && oc < nout) {
-              if (ACCUM) dh_global[(long long)grow * MH_HID + oc] += v;
-              else dh_global[(long long)grow * MH_HID + oc] = v;
-            }
-          } else {
-            const bf16raw bv = drla_f32_to_bf16(v);
-            act_out[lrow][oc] = bv;
-            if (stash && grow < N && oc < nout) {
-              stash[(long long)grow * nout + oc] = bv;
-            }
-            if (dz_global && grow < N && oc < nout) {
-              dz_global[(long long)grow * MH_HID + oc] = bv;
-            }
-            if (grow < N) csum += v;
+        } else {
+          const bf16raw bv = drla_f32_to_bf16(v);
+          act_out[lrow][oc] = bv;
+          if (stash && grow < N && oc < nout) {
+            stash[(long long)grow * nout + oc] = bv;
+          }
+          if (dz_global && grow < N && oc < nout) {
+            dz_global[(long long)grow * MH_HID + oc] = bv;
+          }
+          if (colsum && oc < nout && grow < N) {
+            atomicAdd(&colsum[oc], v);
           }
         }
-        if (colsum && oc < nout) atomicAdd(&colsum[oc], csum);
       }
     }
   }
@@ -144,57 +139,61 @@ extern "C" __global__ __launch_bounds__(256) void drla_mlp_heads_fwd(
     float* __restrict__ value,          // [N]
     bf16raw* __restrict__ stash,    // [N,5*256]: a1p,a2p,a1v,a2v,h_bf16
     int N, int A) {
+  // grid: (ceil(N/16), 2) — blockIdx.y 0 = policy chain, 1 = value chain
   __shared__ bf16raw hb[MH_BM][MH_LD];
   __shared__ bf16raw acta[MH_BM][MH_LD];
   __shared__ bf16raw actb[MH_BM][MH_LD];
 
   const int tid = threadIdx.x;
   const int row0 = blockIdx.x * MH_BM;
+  const bool policy = blockIdx.y == 0;
   const long long soff = (long long)N * MH_HID;
 
-  // load h (f32 -> bf16) + stash the bf16 copy (wgrad input)
+  // load h (f32 -> bf16); the policy block stashes the bf16 copy
   {
-    const int r = tid >> 3;
-    const int c0 = (tid & 7) * 32;
+    const int r = tid >> 4;
+    const int c0 = (tid & 15) * 16;
     const int grow = row0 + r;
-    for (int c = 0; c < 32; ++c) {
+    for (int c = 0; c < 16; ++c) {
       const float v = (grow < N) ? h[(long long)grow * MH_HID + c0 + c]
                                  : 0.0f;
       const bf16raw bv = drla_f32_to_bf16(v);
       hb[r][c0 + c] = bv;
-      if (grow < N) {
+      if (policy && grow < N) {
         stash[4 * soff + (long long)grow * MH_HID + c0 + c] = bv;
       }
     }
   }
   __syncthreads();
 
-  // policy head
-  mh_pass<true, false>(hb, acta, W1p, b1p, nullptr, nullptr, nullptr,
-                       stash, nullptr, row0, N, MH_HID, MH_HID);
-  mh_pass<true, false>(acta, actb, W2p, b2p, nullptr, nullptr, nullptr,
-                       stash + soff, nullptr, row0, N, MH_HID, MH_HID);
-  mh_pass<false, false>(actb, acta, W3p, b3p, nullptr, nullptr, nullptr,
-                        nullptr, nullptr, row0, N, A, MH_HID);
-  {
+  if (policy) {
+    mh_pass<true, false>(hb, acta, W1p, b1p, nullptr, nullptr, nullptr,
+                         stash, nullptr, row0, N, MH_HID, MH_HID);
+    mh_pass<true, false>(acta, actb, W2p, b2p, nullptr, nullptr, nullptr,
+                         stash + soff, nullptr, row0, N, MH_HID, MH_HID);
+    mh_pass<false, false>(actb, acta, W3p, b3p, nullptr, nullptr, nullptr,
+                          nullptr, nullptr, row0, N, A, MH_HID);
     const int r = tid >> 3;
     const int c = tid & 7;
-    for (int cc = c; cc < A; cc += 8) {
-      if (row0 + r < N) {
-        logits[(long long)(row0 + r) * A + cc] = acta[r][cc];
+    if (r < MH_BM) {
+      for (int cc = c; cc < A; cc += 8) {
+        if (row0 + r < N) {
+          logits[(long long)(row0 + r) * A + cc] = acta[r][cc];
+        }
       }
     }
-  }
-  __syncthreads();
-  // value head
-  mh_pass<true, false>(hb, acta, W1v, b1v, nullptr, nullptr, nullptr,
-                       stash + 2 * soff, nullptr, row0, N, MH_HID, MH_HID);
-  mh_pass<true, false>(acta, actb, W2v, b2v, nullptr, nullptr, nullptr,
-                       stash + 3 * soff, nullptr, row0, N, MH_HID, MH_HID);
-  mh_pass<false, false>(actb, acta, W3v, b3v, nullptr, nullptr, nullptr,
-                        nullptr, nullptr, row0, N, 1, MH_HID);
-  if (tid < MH_BM && row0 + tid < N) {
-    value[row0 + tid] = mh_b2f(acta[tid][0]);
+  } else {
+    mh_pass<true, false>(hb, acta, W1v, b1v, nullptr, nullptr, nullptr,
+                         stash + 2 * soff, nullptr, row0, N, MH_HID,
+                         MH_HID);
+    mh_pass<true, false>(acta, actb, W2v, b2v, nullptr, nullptr, nullptr,
+                         stash + 3 * soff, nullptr, row0, N, MH_HID,
+                         MH_HID);
+    mh_pass<false, false>(actb, acta, W3v, b3v, nullptr, nullptr, nullptr,
+                          nullptr, nullptr, row0, N, 1, MH_HID);
+    if (tid < MH_BM && row0 + tid < N) {
+      value[row0 + tid] = mh_b2f(acta[tid][0]);
+    }
   }
 }
 
@@ -211,12 +210,14 @@ extern "C" __global__ __launch_bounds__(256) void drla_mlp_heads_bwd(
     const bf16raw* __restrict__ W3v,       // [1,256]
     bf16raw* __restrict__ dz1p, bf16raw* __restrict__ dz2p,  // [N,256]
     bf16raw* __restrict__ dz1v, bf16raw* __restrict__ dz2v,  // [N,256]
-    float* __restrict__ dh,                // [N,256]
+    float* __restrict__ dh,                // [N,256], ZEROED (atomics)
     float* __restrict__ db1p, float* __restrict__ db2p,
     float* __restrict__ db3p,              // [A]
     float* __restrict__ db1v, float* __restrict__ db2v,
     float* __restrict__ db3v,              // [1]
     int N, int A) {
+  // grid: (ceil(N/16), 2) — blockIdx.y 0 = policy chain, 1 = value chain;
+  // both accumulate dh atomically.
   __shared__ bf16raw dza[MH_BM][MH_LD];
   __shared__ bf16raw dzb[MH_BM][MH_LD];
   __shared__ float colsum[MH_HID];
@@ -229,82 +230,81 @@ extern "C" __global__ __launch_bounds__(256) void drla_mlp_heads_bwd(
   const bf16raw* a1v = stash + 2 * soff;
   const bf16raw* a2v = stash + 3 * soff;
 
-  // stage dz3p zero-padded to 32 cols
-  {
-    const int r = tid >> 3;
-    const int c = tid & 7;
-    for (int cc = c; cc < 32; cc += 8) {
-      const int grow = row0 + r;
-      bf16raw v = 0;
-      if (cc < A && grow < N) v = dlogits[(long long)grow * A + cc];
-      dza[r][cc] = v;
-    }
-  }
   if (tid < MH_HID) colsum[tid] = 0.0f;
-  __syncthreads();
-  if (tid < A) {
-    float s = 0.0f;
-    for (int r = 0; r < MH_BM; ++r) s += mh_b2f(dza[r][tid]);
-    atomicAdd(&db3p[tid], s);
-  }
-  __syncthreads();
-  // da2p = dz3 @ W3p = dz3(padded) x wT3p^T... in transposed-output form:
-  // out[r][i] with M = wT3p [256 i-rows][32], kdim = 32
-  mh_pass<false, false>(dza, dzb, wT3p, nullptr, a2p, dz2p, colsum, nullptr,
-                        nullptr, row0, N, MH_HID, 32);
-  if (tid < MH_HID) {
-    atomicAdd(&db2p[tid], colsum[tid]);
-    colsum[tid] = 0.0f;
-  }
-  __syncthreads();
-  mh_pass<false, false>(dzb, dza, wT2p, nullptr, a1p, dz1p, colsum, nullptr,
-                        nullptr, row0, N, MH_HID, MH_HID);
-  if (tid < MH_HID) {
-    atomicAdd(&db1p[tid], colsum[tid]);
-    colsum[tid] = 0.0f;
-  }
-  __syncthreads();
-  mh_pass<false, false>(dza, dzb, wT1p, nullptr, nullptr, nullptr, nullptr,
-                        nullptr, dh, row0, N, MH_HID, MH_HID);
 
-  // value chain: da2v[r][i] = dvalue[r] * W3v[i], masked by a2v
-  {
-    const int r = tid >> 3;
-    const int c0 = (tid & 7) * 32;
-    const int grow = row0 + r;
-    const float dv = (grow < N) ? dvalue[grow] : 0.0f;
-    for (int c = 0; c < 32; ++c) {
-      const int ic = c0 + c;
-      float v = dv * mh_b2f(W3v[ic]);
-      if (grow < N) {
-        const float a = mh_b2f(a2v[(long long)grow * MH_HID + ic]);
-        v = (a > 0.0f) ? v : 0.0f;
-      } else {
-        v = 0.0f;
+  if (blockIdx.y == 0) {
+    // ---- policy chain ----
+    {
+      const int r = tid >> 4;
+      const int c = tid & 15;
+      for (int cc = c; cc < 32; cc += 16) {
+        const int grow = row0 + r;
+        bf16raw v = 0;
+        if (cc < A && grow < N) v = dlogits[(long long)grow * A + cc];
+        dza[r][cc] = v;
       }
-      const bf16raw bv = drla_f32_to_bf16(v);
-      dza[r][ic] = bv;
-      if (grow < N) dz2v[(long long)grow * MH_HID + ic] = bv;
-      atomicAdd(&colsum[ic], v);
     }
+    __syncthreads();
+    if (tid < A) {
+      float s = 0.0f;
+      for (int r = 0; r < MH_BM; ++r) {
+        if (row0 + r < N) s += mh_b2f(dza[r][tid]);
+      }
+      atomicAdd(&db3p[tid], s);
+    }
+    __syncthreads();
+    mh_pass<false, false>(dza, dzb, wT3p, nullptr, a2p, dz2p, colsum,
+                          nullptr, nullptr, row0, N, MH_HID, 32);
+    if (tid < MH_HID) {
+      atomicAdd(&db2p[tid], colsum[tid]);
+      colsum[tid] = 0.0f;
+    }
+    __syncthreads();
+    mh_pass<false, false>(dzb, dza, wT2p, nullptr, a1p, dz1p, colsum,
+                          nullptr, nullptr, row0, N, MH_HID, MH_HID);
+    if (tid < MH_HID) atomicAdd(&db1p[tid], colsum[tid]);
+    __syncthreads();
+    mh_pass<false, false>(dza, dzb, wT1p, nullptr, nullptr, nullptr,
+                          nullptr, nullptr, dh, row0, N, MH_HID, MH_HID);
+  } else {
+    // ---- value chain ----
+    // da2v[r][i] = dvalue[r] * W3v[i], masked by a2v
+    {
+      const int r = tid >> 4;
+      const int c0 = (tid & 15) * 16;
+      const int grow = row0 + r;
+      const float dv = (grow < N) ? dvalue[grow] : 0.0f;
+      for (int c = 0; c < 16; ++c) {
+        const int ic = c0 + c;
+        float v = dv * mh_b2f(W3v[ic]);
+        if (grow < N) {
+          const float a = mh_b2f(a2v[(long long)grow * MH_HID + ic]);
+          v = (a > 0.0f) ? v : 0.0f;
+        } else {
+          v = 0.0f;
+        }
+        const bf16raw bv = drla_f32_to_bf16(v);
+        dza[r][ic] = bv;
+        if (grow < N) dz2v[(long long)grow * MH_HID + ic] = bv;
+        atomicAdd(&colsum[ic], v);
+      }
+    }
+    __syncthreads();
+    if (tid < MH_HID) {
+      atomicAdd(&db2v[tid], colsum[tid]);
+      colsum[tid] = 0.0f;
+    }
+    if (tid == 0) {
+      float s = 0.0f;
+      for (int r = 0; r < MH_BM && row0 + r < N; ++r) s += dvalue[row0 + r];
+      atomicAdd(&db3v[0], s);
+    }
+    __syncthreads();
+    mh_pass<false, false>(dza, dzb, wT2v, nullptr, a1v, dz1v, colsum,
+                          nullptr, nullptr, row0, N, MH_HID, MH_HID);
+    if (tid < MH_HID) atomicAdd(&db1v[tid], colsum[tid]);
+    __syncthreads();
+    mh_pass<false, true>(dzb, dza, wT1v, nullptr, nullptr, nullptr,
+                         nullptr, nullptr, dh, row0, N, MH_HID, MH_HID);
   }
-  __syncthreads();
-  if (tid < MH_HID) {
-    atomicAdd(&db2v[tid], colsum[tid]);
-    colsum[tid] = 0.0f;
-  }
-  if (tid == 0) {
-    float s = 0.0f;
-    for (int r = 0; r < MH_BM && row0 + r < N; ++r) s += dvalue[row0 + r];
-    atomicAdd(&db3v[0], s);
-  }
-  __syncthreads();
-  mh_pass<false, false>(dza, dzb, wT2v, nullptr, a1v, dz1v, colsum, nullptr,
-                        nullptr, row0, N, MH_HID, MH_HID);
-  if (tid < MH_HID) {
-    atomicAdd(&db1v[tid], colsum[tid]);
-  }
-  __syncthreads();
-  mh_pass<false, true>(dzb, dza, wT1v, nullptr, nullptr, nullptr, nullptr,
-                       nullptr, dh, row0, N, MH_HID, MH_HID);
 }
