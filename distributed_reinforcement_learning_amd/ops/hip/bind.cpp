@@ -581,13 +581,19 @@ std::vector<torch::Tensor> mlp_heads_bwd(
   auto dz2p = torch::empty({N, 256}, bopt);
   auto dz1v = torch::empty({N, 256}, bopt);
   auto dz2v = torch::empty({N, 256}, bopt);
-  auto dh = torch::zeros({N, 256}, fopt);  // both head chains atomicAdd
-  auto db1p = torch::zeros({256}, fopt);
-  auto db2p = torch::zeros({256}, fopt);
-  auto db3p = torch::zeros({A}, fopt);
-  auto db1v = torch::zeros({256}, fopt);
-  auto db2v = torch::zeros({256}, fopt);
-  auto db3v = torch::zeros({1}, fopt);
+  // dh + all six bias grads live in ONE zeroed workspace (atomics targets;
+  // one fill kernel instead of seven)
+  const long long ws_n = (long long)N * 256 + 4 * 256 + A + 1;
+  auto ws = torch::zeros({ws_n}, fopt);
+  long long off = 0;
+  auto dh = ws.narrow(0, off, (long long)N * 256).view({N, 256});
+  off += (long long)N * 256;
+  auto db1p = ws.narrow(0, off, 256); off += 256;
+  auto db2p = ws.narrow(0, off, 256); off += 256;
+  auto db3p = ws.narrow(0, off, A); off += A;
+  auto db1v = ws.narrow(0, off, 256); off += 256;
+  auto db2v = ws.narrow(0, off, 256); off += 256;
+  auto db3v = ws.narrow(0, off, 1);
   hipLaunchKernelGGL(
       drla_mlp_heads_bwd, dim3((N + 15) / 16, 2), dim3(256), 0,
       cur_stream(),

@@ -46,8 +46,8 @@ class _FusedMlpHeads(torch.autograd.Function):
         dvalue = dvalue.float().contiguous()
         # the dgrad kernel consumes PRE-TRANSPOSED weights (its A-operand
         # streams matrix rows); layer-3 policy is zero-padded to K=32
-        wT3p = torch.zeros(256, 32, dtype=torch.bfloat16, device=dev)
-        wT3p[:, :A] = w3p.t()
+        wT3p = torch.nn.functional.pad(
+            w3p.t().contiguous(), (0, 32 - A))
         wT = [w1p.t().contiguous(), w2p.t().contiguous(), wT3p,
               w1v.t().contiguous(), w2v.t().contiguous(),
               w3v.contiguous()]
