@@ -90,12 +90,23 @@ extern "C" __global__ void drla_sq_norm_bf16(
   const uint4* x8 = reinterpret_cast<const uint4*>(x);
   for (long long k = i; k < n8; k += stride) {
     const uint4 v = x8[k];
-    const unsigned int* w = reinterpret_cast<const unsigned int*>(&v);
-    for (int j = 0; j < 4; ++j) {
-      const float lo = drla_bf16_to_f32((unsigned short)(w[j] & 0xFFFF));
-      const float hi = drla_bf16_to_f32((unsigned short)(w[j] >> 16));
-      acc += lo * lo + hi * hi;
-    }
+    // named components (no address-of: a pointer into a local can force
+    // the vector to scratch — guide §5.4 rule 20)
+
+    float a0 = 0, a1 = 0, a2 = 0, a3 = 0;
+    a0 = drla_bf16_to_f32((unsigned short)(v.x & 0xFFFF));
+    a1 = drla_bf16_to_f32((unsigned short)(v.x >> 16));
+    acc += a0 * a0 + a1 * a1;
+    a2 = drla_bf16_to_f32((unsigned short)(v.y & 0xFFFF));
+    a3 = drla_bf16_to_f32((unsigned short)(v.y >> 16));
+    acc += a2 * a2 + a3 * a3;
+    a0 = drla_bf16_to_f32((unsigned short)(v.z & 0xFFFF));
+    a1 = drla_bf16_to_f32((unsigned short)(v.z >> 16));
+    acc += a0 * a0 + a1 * a1;
+    a2 = drla_bf16_to_f32((unsigned short)(v.w & 0xFFFF));
+    a3 = drla_bf16_to_f32((unsigned short)(v.w >> 16));
+    acc += a2 * a2 + a3 * a3;
+
   }
   for (long long k = n8 * 8 + i; k < n; k += stride) {
     float a = drla_bf16_to_f32(x[k]);
