@@ -111,6 +111,15 @@ extern "C" __global__ void drla_mlp_heads_bwd(
     const unsigned short*, const unsigned short*, const unsigned short*,
     unsigned short*, unsigned short*, unsigned short*, unsigned short*,
     float*, float*, float*, float*, float*, float*, float*, int, int);
+extern "C" __global__ void drla_heads_wt_pack(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    unsigned short*, int);
+extern "C" __global__ void drla_heads_wgrad(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, const float*,
+    const unsigned short*, unsigned short*, unsigned short*, unsigned short*,
+    unsigned short*, unsigned short*, unsigned short*, int, int);
 extern "C" __global__ void drla_lstm_seq_fwd(
     const unsigned short*, const float*, const unsigned short*, const float*,
     const float*, const unsigned char*, float*, float*, float*, float, int,
@@ -604,7 +613,47 @@ std::vector<torch::Tensor> mlp_heads_bwd(
       db1p.data_ptr<float>(), db2p.data_ptr<float>(),
       db3p.data_ptr<float>(), db1v.data_ptr<float>(),
       db2v.data_ptr<float>(), db3v.data_ptr<float>(), N, (int)A);
-  return {dz1p, dz2p, dz1v, dz2v, dh, db1p, db2p, db3p, db1v, db2v, db3v};
+  // ws comes back too so the wrapper can cast the contiguous bias-grad
+  // tail (db1p..db3v) to bf16 in ONE kernel instead of six
+  return {dz1p, dz2p, dz1v, dz2v, dh,   db1p, db2p, db3p, db1v, db2v,
+          db3v, ws};
+}
+
+torch::Tensor mlp_heads_pack_wt(std::vector<torch::Tensor> weights,
+                                int64_t A) {
+  TORCH_CHECK(weights.size() == 6);
+  for (auto& w : weights) check_gpu_contig(w, "pack weight");
+  auto out = torch::empty({4 * 65536 + 256 * 32 + 256},
+                          weights[0].options());
+  hipLaunchKernelGGL(drla_heads_wt_pack, dim3(drla_grid(out.numel())),
+                     dim3(DRLA_BLOCK), 0, cur_stream(), u16p(weights[0]),
+                     u16p(weights[1]), u16p(weights[2]), u16p(weights[3]),
+                     u16p(weights[4]), u16p(weights[5]), u16pm(out),
+                     (int)A);
+  return out;
+}
+
+std::vector<torch::Tensor> mlp_heads_wgrad(
+    torch::Tensor dz1p, torch::Tensor dz2p, torch::Tensor dz1v,
+    torch::Tensor dz2v, torch::Tensor dlogits, torch::Tensor dvalue,
+    torch::Tensor stash, int64_t A) {
+  for (auto* t : {&dz1p, &dz2p, &dz1v, &dz2v, &dlogits, &stash})
+    check_gpu_contig(*t, "wgrad input");
+  check_gpu_contig(dvalue, "dvalue");
+  const int N = dvalue.numel();
+  auto bopt = dz1p.options();
+  auto dw1p = torch::empty({256, 256}, bopt);
+  auto dw2p = torch::empty({256, 256}, bopt);
+  auto dw3p = torch::empty({A, 256}, bopt);
+  auto dw1v = torch::empty({256, 256}, bopt);
+  auto dw2v = torch::empty({256, 256}, bopt);
+  auto dw3v = torch::empty({1, 256}, bopt);
+  hipLaunchKernelGGL(drla_heads_wgrad, dim3(72), dim3(256), 0, cur_stream(),
+                     u16p(dz1p), u16p(dz2p), u16p(dz1v), u16p(dz2v),
+                     u16p(dlogits), dvalue.data_ptr<float>(), u16p(stash),
+                     u16pm(dw1p), u16pm(dw2p), u16pm(dw3p), u16pm(dw1v),
+                     u16pm(dw2v), u16pm(dw3v), N, (int)A);
+  return {dw1p, dw2p, dw3p, dw1v, dw2v, dw3v};
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_seq_fwd(
@@ -638,7 +687,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_seq_fwd(
 torch::Tensor sq_norm(torch::Tensor x) {
   check_gpu_contig(x, "x");
   TORCH_CHECK(x.scalar_type() == torch::kFloat, "sq_norm wants float32");
-  auto out = torch::zeros({1}, x.options());
+  // DRLA_NORM_SLOTS cache-line-spread partials (total = .sum())
+  auto out = torch::zeros({DRLA_NORM_SLOTS * 16}, x.options());
   hipLaunchKernelGGL(drla_sq_norm, dim3(drla_grid(x.numel() / 4 + 1)),
                      dim3(DRLA_BLOCK), 0, cur_stream(), x.data_ptr<float>(),
                      out.data_ptr<float>(), (long long)x.numel());
@@ -648,7 +698,8 @@ torch::Tensor sq_norm(torch::Tensor x) {
 torch::Tensor sq_norm_bf16(torch::Tensor x) {
   check_gpu_contig(x, "x");
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "wants bf16");
-  auto out = torch::zeros({1}, x.options().dtype(torch::kFloat));
+  auto out = torch::zeros({DRLA_NORM_SLOTS * 16},
+                          x.options().dtype(torch::kFloat));
   hipLaunchKernelGGL(drla_sq_norm_bf16, dim3(drla_grid(x.numel() / 4 + 1)),
                      dim3(DRLA_BLOCK), 0, cur_stream(),
                      reinterpret_cast<const unsigned short*>(x.data_ptr()),
@@ -790,6 +841,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused policy+value MLP heads forward (K4)");
   m.def("mlp_heads_bwd", &mlp_heads_bwd,
         "fused heads dgrad chain + ReLU masks + bias grads (K4 bwd)");
+  m.def("mlp_heads_pack_wt", &mlp_heads_pack_wt,
+        "one-kernel transposed-weight pack for the heads dgrad");
+  m.def("mlp_heads_wgrad", &mlp_heads_wgrad,
+        "all six head wgrads (dW = dz^T @ act) in one MFMA launch");
   m.def("lstm_seq_fwd", &lstm_seq_fwd,
         "whole no-grad LSTM unroll in one kernel (K3 seq / burn-in)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");

@@ -7,6 +7,9 @@
 #include <hip/hip_bf16.h>
 
 #define DRLA_WAVE 64
+// squared-norm partials: 16 slots, each on its own 64 B cache line
+// (buffer length = 16*16 floats; slot s lives at [s*16])
+#define DRLA_NORM_SLOTS 16
 #define DRLA_BLOCK 256
 // cap grids at ~8 blocks/CU x 256 CUs and grid-stride the rest
 #define DRLA_MAX_BLOCKS 2048

@@ -308,3 +308,168 @@ extern "C" __global__ __launch_bounds__(256) void drla_mlp_heads_bwd(
                          nullptr, nullptr, dh, row0, N, MH_HID, MH_HID);
   }
 }
+
+// ---------------------------------------------------------------------------
+// Weight-transpose pack: the dgrad kernel above wants W^T operands, which
+// the wrapper used to build with 5 .t().contiguous() copies + an F.pad —
+// ~7 eager launches per step at the ~4.5 us small-kernel floor (profile
+// r23). One kernel writes the whole packed buffer instead.
+// Layout (bf16 elements): wT1p@0, wT2p@65536, wT3p@131072 ([256,32],
+// cols >= A zero), wT1v@139264, wT2v@204800, W3v copy @270336; total 270592.
+extern "C" __global__ void drla_heads_wt_pack(
+    const bf16raw* __restrict__ w1p, const bf16raw* __restrict__ w2p,
+    const bf16raw* __restrict__ w3p, const bf16raw* __restrict__ w1v,
+    const bf16raw* __restrict__ w2v, const bf16raw* __restrict__ w3v,
+    bf16raw* __restrict__ out, int A) {
+  const int total = 4 * 65536 + 256 * 32 + 256;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    bf16raw v;
+    if (i < 65536) {
+      v = w1p[(i & 255) * 256 + (i >> 8)];
+    } else if (i < 131072) {
+      const int j = i - 65536;
+      v = w2p[(j & 255) * 256 + (j >> 8)];
+    } else if (i < 139264) {
+      const int j = i - 131072;
+      const int c = j & 31;
+      v = (c < A) ? w3p[c * 256 + (j >> 5)] : (bf16raw)0;
+    } else if (i < 204800) {
+      const int j = i - 139264;
+      v = w1v[(j & 255) * 256 + (j >> 8)];
+    } else if (i < 270336) {
+      const int j = i - 204800;
+      v = w2v[(j & 255) * 256 + (j >> 8)];
+    } else {
+      v = w3v[i - 270336];
+    }
+    out[i] = v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// All six head wgrads in ONE launch: dW = dz^T @ act with K = N rows.
+// Replaces 6 hipBLASLt GEMM dispatches (~5/step at ~13 us each, r23).
+//
+// Jobs (all outputs [M,256] bf16):
+//   0: dW1p = dz1p^T @ hb     M=256      3: dW2v = dz2v^T @ a1v  M=256
+//   1: dW2p = dz2p^T @ a1p    M=256      4: dW3p = dlogits^T @ a2p  M=A
+//   2: dW1v = dz1v^T @ hb     M=256      5: dW3v = dvalue^T @ a2v   M=1
+//
+// Both MFMA operands are k-major in global (dz[k][m], act[k][n]), so each
+// 32k-chunk is staged TRANSPOSED into padded LDS (vec8 global row loads,
+// scalar LDS writes) and the fragments read back as vec8 ds_read_b128
+// (row stride 40 elems = 80 B keeps 16 B alignment).
+#define HW_LDK 40  // 32 + 8 pad
+
+extern "C" __global__ __launch_bounds__(256) void drla_heads_wgrad(
+    const bf16raw* __restrict__ dz1p, const bf16raw* __restrict__ dz2p,
+    const bf16raw* __restrict__ dz1v, const bf16raw* __restrict__ dz2v,
+    const bf16raw* __restrict__ dlogits,  // [N,A]
+    const float* __restrict__ dvalue,     // [N]
+    const bf16raw* __restrict__ stash,    // [N,5*256]
+    bf16raw* __restrict__ dw1p, bf16raw* __restrict__ dw2p,
+    bf16raw* __restrict__ dw3p,           // [A,256]
+    bf16raw* __restrict__ dw1v, bf16raw* __restrict__ dw2v,
+    bf16raw* __restrict__ dw3v,           // [1,256]
+    int N, int A) {
+  // grid.x = 72: jobs 0-3 are 4x4 tiles of 64x64 (blocks 0..63), job 4 is
+  // blocks 64..67 (one m-tile, A <= 32), job 5 blocks 68..71.
+  const int bid = blockIdx.x;
+  int job, mt, nt;
+  if (bid < 64) {
+    job = bid >> 4;
+    mt = (bid >> 2) & 3;
+    nt = bid & 3;
+  } else if (bid < 68) {
+    job = 4; mt = 0; nt = bid - 64;
+  } else {
+    job = 5; mt = 0; nt = bid - 68;
+  }
+  const long long soff = (long long)N * MH_HID;
+  const bf16raw* dz;
+  const bf16raw* act;
+  bf16raw* out;
+  int M;
+  switch (job) {
+    case 0: dz = dz1p; act = stash + 4 * soff; out = dw1p; M = 256; break;
+    case 1: dz = dz2p; act = stash;            out = dw2p; M = 256; break;
+    case 2: dz = dz1v; act = stash + 4 * soff; out = dw1v; M = 256; break;
+    case 3: dz = dz2v; act = stash + 2 * soff; out = dw2v; M = 256; break;
+    case 4: dz = dlogits; act = stash + soff;  out = dw3p; M = A;   break;
+    default: dz = nullptr; act = stash + 3 * soff; out = dw3v; M = 1;
+  }
+  const int m0 = mt * 64;
+  const int n0 = nt * 64;
+
+  __shared__ bf16raw dzT[64][HW_LDK];   // [m_local][k_local]
+  __shared__ bf16raw actT[64][HW_LDK];  // [n_local][k_local]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+
+  f32x4 acc[4];
+  for (int ni = 0; ni < 4; ++ni) acc[ni] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < N; k0 += 32) {
+    // stage: thread t owns global row k = k0 + t/8, columns (t%8)*8..+8
+    const int kk = tid >> 3;
+    const int c0 = (tid & 7) * 8;
+    const int k = k0 + kk;
+    // act is always [N,256] bf16, vec8
+    {
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (k < N) {
+        v = *reinterpret_cast<const bf16x8*>(
+            act + (long long)k * MH_HID + n0 + c0);
+      }
+      actT[c0 + 0][kk] = v[0]; actT[c0 + 1][kk] = v[1];
+      actT[c0 + 2][kk] = v[2]; actT[c0 + 3][kk] = v[3];
+      actT[c0 + 4][kk] = v[4]; actT[c0 + 5][kk] = v[5];
+      actT[c0 + 6][kk] = v[6]; actT[c0 + 7][kk] = v[7];
+    }
+    if (job < 4) {           // dz [N,256] bf16, vec8
+      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (k < N) {
+        v = *reinterpret_cast<const bf16x8*>(
+            dz + (long long)k * MH_HID + m0 + c0);
+      }
+      dzT[c0 + 0][kk] = v[0]; dzT[c0 + 1][kk] = v[1];
+      dzT[c0 + 2][kk] = v[2]; dzT[c0 + 3][kk] = v[3];
+      dzT[c0 + 4][kk] = v[4]; dzT[c0 + 5][kk] = v[5];
+      dzT[c0 + 6][kk] = v[6]; dzT[c0 + 7][kk] = v[7];
+    } else if (job == 4) {   // dlogits [N,A], scalar predicated
+      for (int e = 0; e < 8; ++e) {
+        const int m = c0 + e;
+        dzT[m][kk] = (k < N && m < A) ? dlogits[(long long)k * A + m]
+                                      : (bf16raw)0;
+      }
+    } else {                 // dvalue [N] f32, M = 1
+      for (int e = 0; e < 8; ++e) {
+        const int m = c0 + e;
+        dzT[m][kk] = (k < N && m == 0) ? drla_f32_to_bf16(dvalue[k])
+                                       : (bf16raw)0;
+      }
+    }
+    __syncthreads();
+    // wave w: m rows [w*16, w*16+16); frags vec8 from LDS
+    const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
+        &dzT[wave * 16 + (lane & 15)][(lane >> 4) * 8]);
+    for (int ni = 0; ni < 4; ++ni) {
+      const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+          &actT[ni * 16 + (lane & 15)][(lane >> 4) * 8]);
+      acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_frag, b_frag, acc[ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  // D: lane l reg r -> row (l>>4)*4+r (m), col l&15 (n)
+  for (int ni = 0; ni < 4; ++ni) {
+    for (int r = 0; r < 4; ++r) {
+      const int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      const int n = n0 + ni * 16 + (lane & 15);
+      if (m < M) out[(long long)m * MH_HID + n] = drla_f32_to_bf16(acc[ni][r]);
+    }
+  }
+}

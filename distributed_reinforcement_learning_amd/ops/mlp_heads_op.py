@@ -45,32 +45,32 @@ class _FusedMlpHeads(torch.autograd.Function):
         dlogits = dlogits.to(torch.bfloat16).contiguous()
         dvalue = dvalue.float().contiguous()
         # the dgrad kernel consumes PRE-TRANSPOSED weights (its A-operand
-        # streams matrix rows); layer-3 policy is zero-padded to K=32
-        wT3p = torch.nn.functional.pad(
-            w3p.t().contiguous(), (0, 32 - A))
-        wT = [w1p.t().contiguous(), w2p.t().contiguous(), wT3p,
-              w1v.t().contiguous(), w2v.t().contiguous(),
-              w3v.contiguous()]
-        (dz1p, dz2p, dz1v, dz2v, dh, db1p, db2p, db3p, db1v, db2v,
-         db3v) = ext.mlp_heads_bwd(dlogits, dvalue, stash, wT, A)
-        soff = N * 256
-        flat = stash.reshape(-1)
-        a1p = flat[0:soff].reshape(N, 256)
-        a2p = flat[soff:2 * soff].reshape(N, 256)
-        a1v = flat[2 * soff:3 * soff].reshape(N, 256)
-        a2v = flat[3 * soff:4 * soff].reshape(N, 256)
-        hb = flat[4 * soff:5 * soff].reshape(N, 256)
-        # wgrads: dW = dz^T @ a_prev (torch Linear convention W [out,in])
-        dw1p = dz1p.t().mm(hb)
-        dw2p = dz2p.t().mm(a1p)
-        dw3p = dlogits.t().mm(a2p)
-        dw1v = dz1v.t().mm(hb)
-        dw2v = dz2v.t().mm(a1v)
-        dw3v = dvalue.to(torch.bfloat16).unsqueeze(0).mm(a2v)
-        bf = torch.bfloat16
-        return (dh, dw1p, db1p.to(bf), dw2p, db2p.to(bf), dw3p,
-                db3p.to(bf), dw1v, db1v.to(bf), dw2v, db2v.to(bf), dw3v,
-                db3v.to(bf))
+        # streams matrix rows); ONE pack kernel builds all of W^T + the
+        # layer-3 zero-pad to K=32 (was 5 .t().contiguous() + F.pad)
+        packed = ext.mlp_heads_pack_wt(
+            [w1p, w2p, w3p, w1v, w2v, w3v], A)
+        wT = [packed.narrow(0, 0, 65536).view(256, 256),
+              packed.narrow(0, 65536, 65536).view(256, 256),
+              packed.narrow(0, 131072, 8192).view(256, 32),
+              packed.narrow(0, 139264, 65536).view(256, 256),
+              packed.narrow(0, 204800, 65536).view(256, 256),
+              packed.narrow(0, 270336, 256).view(1, 256)]
+        (dz1p, dz2p, dz1v, dz2v, dh, _db1p, _db2p, _db3p, _db1v, _db2v,
+         _db3v, ws) = ext.mlp_heads_bwd(dlogits, dvalue, stash, wT, A)
+        # all six wgrads (dW = dz^T @ act) in one MFMA launch
+        dw1p, dw2p, dw3p, dw1v, dw2v, dw3v = ext.mlp_heads_wgrad(
+            dz1p, dz2p, dz1v, dz2v, dlogits, dvalue, stash, A)
+        # bias grads: cast the contiguous workspace tail once (one kernel),
+        # then slice — not six tiny .to(bf16) launches
+        tail = ws.narrow(0, N * 256, 4 * 256 + A + 1).to(torch.bfloat16)
+        db1p = tail.narrow(0, 0, 256)
+        db2p = tail.narrow(0, 256, 256)
+        db3p = tail.narrow(0, 512, A)
+        db1v = tail.narrow(0, 512 + A, 256)
+        db2v = tail.narrow(0, 768 + A, 256)
+        db3v = tail.narrow(0, 1024 + A, 1)
+        return (dh, dw1p, db1p, dw2p, db2p, dw3p, db3p,
+                dw1v, db1v, dw2v, db2v, dw3v, db3v)
 
 
 def fused_mlp_heads(h: torch.Tensor, policy_head, value_head

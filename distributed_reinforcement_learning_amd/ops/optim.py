@@ -79,8 +79,8 @@ class _FlatOptimizerBase:
         if self.flat_grads.is_cuda:
             ext = _ops.require_ext()
             if self.flat_grads.dtype == torch.bfloat16:
-                return ext.sq_norm_bf16(self.flat_grads).sqrt()[0]
-            return ext.sq_norm(self.flat_grads).sqrt()[0]
+                return ext.sq_norm_bf16(self.flat_grads).sum().sqrt()
+            return ext.sq_norm(self.flat_grads).sum().sqrt()
         return self.flat_grads.float().norm()
 
     def _clip_scale(self) -> torch.Tensor:

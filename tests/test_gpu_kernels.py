@@ -79,7 +79,7 @@ def test_lstm_tail_forward_backward_parity(ext):
 def test_sq_norm(ext):
     x = torch.randn(1_000_003, device="cuda")
     out = ext.sq_norm(x)
-    assert torch.allclose(out[0], (x * x).sum(), rtol=1e-4)
+    assert torch.allclose(out.sum(), (x * x).sum(), rtol=1e-4)
 
 
 def _cpu_rmsprop_golden(p, g, ms, clip, lr, rho, eps):
