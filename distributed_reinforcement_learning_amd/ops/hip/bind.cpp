@@ -119,7 +119,12 @@ extern "C" __global__ void drla_heads_wgrad(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const unsigned short*, const unsigned short*, const float*,
     const unsigned short*, unsigned short*, unsigned short*, unsigned short*,
-    unsigned short*, unsigned short*, unsigned short*, int, int);
+    unsigned short*, unsigned short*, unsigned short*, const float*,
+    unsigned short*, unsigned short*, unsigned short*, unsigned short*,
+    unsigned short*, unsigned short*, int, int);
+extern "C" __global__ void drla_grad_gather(
+    const unsigned long long*, const long long*, const long long*,
+    unsigned short*, int, long long);
 extern "C" __global__ void drla_lstm_seq_fwd(
     const unsigned short*, const float*, const unsigned short*, const float*,
     const float*, const unsigned char*, float*, float*, float*, float, int,
@@ -636,8 +641,8 @@ torch::Tensor mlp_heads_pack_wt(std::vector<torch::Tensor> weights,
 std::vector<torch::Tensor> mlp_heads_wgrad(
     torch::Tensor dz1p, torch::Tensor dz2p, torch::Tensor dz1v,
     torch::Tensor dz2v, torch::Tensor dlogits, torch::Tensor dvalue,
-    torch::Tensor stash, int64_t A) {
-  for (auto* t : {&dz1p, &dz2p, &dz1v, &dz2v, &dlogits, &stash})
+    torch::Tensor stash, torch::Tensor ws, int64_t A) {
+  for (auto* t : {&dz1p, &dz2p, &dz1v, &dz2v, &dlogits, &stash, &ws})
     check_gpu_contig(*t, "wgrad input");
   check_gpu_contig(dvalue, "dvalue");
   const int N = dvalue.numel();
@@ -648,12 +653,23 @@ std::vector<torch::Tensor> mlp_heads_wgrad(
   auto dw1v = torch::empty({256, 256}, bopt);
   auto dw2v = torch::empty({256, 256}, bopt);
   auto dw3v = torch::empty({1, 256}, bopt);
-  hipLaunchKernelGGL(drla_heads_wgrad, dim3(72), dim3(256), 0, cur_stream(),
+  auto db1p = torch::empty({256}, bopt);
+  auto db2p = torch::empty({256}, bopt);
+  auto db3p = torch::empty({A}, bopt);
+  auto db1v = torch::empty({256}, bopt);
+  auto db2v = torch::empty({256}, bopt);
+  auto db3v = torch::empty({1}, bopt);
+  // ws tail = the f32 bias-grad partials written by mlp_heads_bwd
+  const float* ws_tail = ws.data_ptr<float>() + (long long)N * 256;
+  hipLaunchKernelGGL(drla_heads_wgrad, dim3(73), dim3(256), 0, cur_stream(),
                      u16p(dz1p), u16p(dz2p), u16p(dz1v), u16p(dz2v),
                      u16p(dlogits), dvalue.data_ptr<float>(), u16p(stash),
                      u16pm(dw1p), u16pm(dw2p), u16pm(dw3p), u16pm(dw1v),
-                     u16pm(dw2v), u16pm(dw3v), N, (int)A);
-  return {dw1p, dw2p, dw3p, dw1v, dw2v, dw3v};
+                     u16pm(dw2v), u16pm(dw3v), ws_tail, u16pm(db1p),
+                     u16pm(db2p), u16pm(db3p), u16pm(db1v), u16pm(db2v),
+                     u16pm(db3v), N, (int)A);
+  return {dw1p, dw2p, dw3p, dw1v, dw2v, dw3v,
+          db1p, db2p, db3p, db1v, db2v, db3v};
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_seq_fwd(
@@ -682,6 +698,25 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_seq_fwd(
       h_fin.data_ptr<float>(), c_fin.data_ptr<float>(),
       static_cast<float>(forget_bias), B, L, H);
   return {h_out, h_fin, c_fin};
+}
+
+void grad_gather(torch::Tensor srcs, torch::Tensor offs,
+                 torch::Tensor sizes, torch::Tensor dst) {
+  check_gpu_contig(dst, "dst");
+  TORCH_CHECK(dst.scalar_type() == torch::kBFloat16,
+              "grad_gather wants a bf16 flat bucket");
+  TORCH_CHECK(dst.numel() % 8 == 0, "flat bucket must be 8-aligned");
+  for (auto* t : {&srcs, &offs, &sizes}) {
+    check_gpu_contig(*t, "gather table");
+    TORCH_CHECK(t->scalar_type() == torch::kLong, "table must be int64");
+  }
+  const long long chunks = dst.numel() / 8;
+  hipLaunchKernelGGL(
+      drla_grad_gather, dim3(drla_grid(chunks)), dim3(DRLA_BLOCK), 0,
+      cur_stream(),
+      reinterpret_cast<const unsigned long long*>(srcs.data_ptr<int64_t>()),
+      offs.data_ptr<int64_t>(), sizes.data_ptr<int64_t>(), u16pm(dst),
+      (int)srcs.numel(), chunks);
 }
 
 torch::Tensor sq_norm(torch::Tensor x) {
@@ -848,6 +883,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_seq_fwd", &lstm_seq_fwd,
         "whole no-grad LSTM unroll in one kernel (K3 seq / burn-in)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");
+  m.def("grad_gather", &grad_gather,
+        "one-kernel scattered-grad -> flat bucket pack (K12b)");
   m.def("rmsprop_step", &rmsprop_step,
         "fused global-norm-clip + TF-RMSProp update (K12)");
   m.def("rmsprop_step_t", &rmsprop_step_t,

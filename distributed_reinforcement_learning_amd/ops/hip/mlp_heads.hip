@@ -372,10 +372,30 @@ extern "C" __global__ __launch_bounds__(256) void drla_heads_wgrad(
     bf16raw* __restrict__ dw3p,           // [A,256]
     bf16raw* __restrict__ dw1v, bf16raw* __restrict__ dw2v,
     bf16raw* __restrict__ dw3v,           // [1,256]
+    const float* __restrict__ ws_tail,    // f32 bias partials from heads_bwd
+    bf16raw* __restrict__ db1p, bf16raw* __restrict__ db2p,
+    bf16raw* __restrict__ db3p, bf16raw* __restrict__ db1v,
+    bf16raw* __restrict__ db2v, bf16raw* __restrict__ db3v,
     int N, int A) {
-  // grid.x = 72: jobs 0-3 are 4x4 tiles of 64x64 (blocks 0..63), job 4 is
-  // blocks 64..67 (one m-tile, A <= 32), job 5 blocks 68..71.
+  // grid.x = 73: jobs 0-3 are 4x4 tiles of 64x64 (blocks 0..63), job 4 is
+  // blocks 64..67 (one m-tile, A <= 32), job 5 blocks 68..71; block 72
+  // converts the f32 bias-grad partials to six contiguous bf16 tensors
+  // (contiguous outputs avoid AccumulateGrad's clone-a-view kernels).
   const int bid = blockIdx.x;
+  if (bid == 72) {
+    const int tid = threadIdx.x;
+    const int total = 4 * 256 + A + 1;
+    for (int i = tid; i < total; i += 256) {
+      const bf16raw v = drla_f32_to_bf16(ws_tail[i]);
+      if (i < 256)            db1p[i] = v;
+      else if (i < 512)       db2p[i - 256] = v;
+      else if (i < 512 + A)   db3p[i - 512] = v;
+      else if (i < 768 + A)   db1v[i - 512 - A] = v;
+      else if (i < 1024 + A)  db2v[i - 768 - A] = v;
+      else                    db3v[0] = v;
+    }
+    return;
+  }
   int job, mt, nt;
   if (bid < 64) {
     job = bid >> 4;

@@ -192,3 +192,37 @@ extern "C" __global__ void drla_adam_step(
     p[i] -= lr_t * mi / (sqrtf(vi) + eps);
   }
 }
+
+// K12b: scatter-mode grad gather — pack the autograd-owned grad tensors
+// (stable hipGraph-pool addresses) into the flat bucket in ONE launch,
+// replacing ~20 per-param AccumulateGrad add kernels. Slots are 8-element
+// aligned (optim.py flatten_dense_params), so a vec8 chunk never crosses a
+// slot; source pool allocations are >=256 B aligned.
+typedef __attribute__((ext_vector_type(4))) unsigned int drla_u32x4;
+
+extern "C" __global__ void drla_grad_gather(
+    const unsigned long long* __restrict__ srcs,  // [nseg] device pointers
+    const long long* __restrict__ offs,           // [nseg] padded starts
+    const long long* __restrict__ sizes,          // [nseg] true numels
+    unsigned short* __restrict__ dst, int nseg, long long chunks) {
+  for (long long c = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       c < chunks; c += gridDim.x * (long long)blockDim.x) {
+    const long long i = c * 8;
+    int lo = 0, hi = nseg - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (offs[mid] <= i) lo = mid; else hi = mid - 1;
+    }
+    const long long local = i - offs[lo];
+    const long long n = sizes[lo];
+    if (local >= n) continue;  // alignment hole: flat stays zero
+    const unsigned short* src =
+        reinterpret_cast<const unsigned short*>(srcs[lo]);
+    if (local + 8 <= n) {
+      *reinterpret_cast<drla_u32x4*>(dst + i) =
+          *reinterpret_cast<const drla_u32x4*>(src + local);
+    } else {
+      for (int e = 0; e < 8 && local + e < n; ++e) dst[i + e] = src[local + e];
+    }
+  }
+}

@@ -2,8 +2,10 @@
 
 Forward: one kernel runs both three-layer heads for a block of rows with
 weights streamed through LDS (fused bias+ReLU, activations stashed bf16).
-Backward: one kernel fuses the dgrad chains, ReLU masks and bias grads; the
-six wgrad GEMMs ([256 x N x 256], MFMA-efficient) stay on hipBLASLt.
+Backward: one kernel fuses the dgrad chains, ReLU masks and f32 bias-grad
+partials; a second MFMA kernel computes all six wgrads (dW = dz^T @ act)
+plus the bf16 bias-grad conversion — so the whole bwd is 3 launches
+(W^T pack, dgrad chain, wgrad) instead of ~20 torch ops.
 """
 
 from __future__ import annotations
@@ -57,18 +59,11 @@ class _FusedMlpHeads(torch.autograd.Function):
               packed.narrow(0, 270336, 256).view(1, 256)]
         (dz1p, dz2p, dz1v, dz2v, dh, _db1p, _db2p, _db3p, _db1v, _db2v,
          _db3v, ws) = ext.mlp_heads_bwd(dlogits, dvalue, stash, wT, A)
-        # all six wgrads (dW = dz^T @ act) in one MFMA launch
-        dw1p, dw2p, dw3p, dw1v, dw2v, dw3v = ext.mlp_heads_wgrad(
-            dz1p, dz2p, dz1v, dz2v, dlogits, dvalue, stash, A)
-        # bias grads: cast the contiguous workspace tail once (one kernel),
-        # then slice — not six tiny .to(bf16) launches
-        tail = ws.narrow(0, N * 256, 4 * 256 + A + 1).to(torch.bfloat16)
-        db1p = tail.narrow(0, 0, 256)
-        db2p = tail.narrow(0, 256, 256)
-        db3p = tail.narrow(0, 512, A)
-        db1v = tail.narrow(0, 512 + A, 256)
-        db2v = tail.narrow(0, 768 + A, 256)
-        db3v = tail.narrow(0, 1024 + A, 1)
+        # all six wgrads (dW = dz^T @ act) in one MFMA launch; its block 72
+        # also converts the f32 bias-grad partials to contiguous bf16
+        (dw1p, dw2p, dw3p, dw1v, dw2v, dw3v, db1p, db2p, db3p, db1v,
+         db2v, db3v) = ext.mlp_heads_wgrad(
+            dz1p, dz2p, dz1v, dz2v, dlogits, dvalue, stash, ws, A)
         return (dh, dw1p, db1p, dw2p, db2p, dw3p, db3p,
                 dw1v, db1v, dw2v, db2v, dw3v, db3v)
 

@@ -31,19 +31,23 @@ import torch
 from distributed_reinforcement_learning_amd import ops as _ops
 
 
-def flatten_dense_params(params: List[torch.Tensor]) -> torch.Tensor:
-    """Re-home ``params`` as views into one new contiguous flat buffer."""
-    total = sum(p.numel() for p in params)
-    if total == 0:
-        return torch.empty(0)
-    flat = torch.empty(total, dtype=params[0].dtype, device=params[0].device)
+def flatten_dense_params(params: List[torch.Tensor]):
+    """Re-home ``params`` as views into one new contiguous flat buffer.
+
+    Slots are 8-element aligned (small zero holes between params) so the
+    scatter-mode gather kernel can use vec8 loads/stores; returns
+    (flat, slots) with slots = [(offset, numel), ...]."""
+    slots = []
     offset = 0
     for p in params:
-        n = p.numel()
-        flat[offset:offset + n].copy_(p.detach().reshape(-1))
-        p.data = flat[offset:offset + n].view_as(p)
-        offset += n
-    return flat
+        slots.append((offset, p.numel()))
+        offset += (p.numel() + 7) & ~7
+    flat = torch.zeros(offset, dtype=params[0].dtype,
+                       device=params[0].device)
+    for p, (off, n) in zip(params, slots):
+        flat[off:off + n].copy_(p.detach().reshape(-1))
+        p.data = flat[off:off + n].view_as(p)
+    return flat, slots
 
 
 class _FlatOptimizerBase:
@@ -57,23 +61,69 @@ class _FlatOptimizerBase:
                  clip_norm: Optional[float]):
         self.params = [p for p in params if p.requires_grad]
         assert len(self.params) > 0
-        self.flat_params = flatten_dense_params(self.params)
+        self.flat_params, self.slots = flatten_dense_params(self.params)
         self.flat_grads = torch.zeros_like(self.flat_params)
         self.mixed = self.flat_params.dtype != torch.float32
         self.master = (self.flat_params.detach().float()
                        if self.mixed else self.flat_params)
+        self.scatter = False
+        self._gather_table = None
         # route autograd into the flat grad buffer
-        offset = 0
-        for p in self.params:
-            n = p.numel()
-            p.grad = self.flat_grads[offset:offset + n].view_as(p)
-            offset += n
+        for p, (off, n) in zip(self.params, self.slots):
+            p.grad = self.flat_grads[off:off + n].view_as(p)
         self.lr = lr
         self.clip_norm = clip_norm
         self.step_count = 0
 
     def zero_grad(self) -> None:
         self.flat_grads.zero_()
+
+    # -- scatter-grad mode (graphed GPU learner) ---------------------------
+    # With .grad routed into flat views, every param costs one AccumulateGrad
+    # add kernel per backward (~20 x 4.5 us/step at the small-kernel floor,
+    # profile r23). Scatter mode leaves .grad = None so backward ASSIGNS the
+    # grad tensors (no kernels); under hipGraph capture those tensors live at
+    # stable pool addresses, and ONE gather kernel packs them into
+    # flat_grads (which the DP all-reduce and the fused update consume).
+
+    def enable_scatter_grads(self) -> None:
+        self.scatter = True
+        self._gather_table = None
+        for p in self.params:
+            p.grad = None
+
+    def gather_grads_eager(self) -> None:
+        """Eager/warmup-path gather: per-param copies, then detach again."""
+        with torch.no_grad():
+            for p, (off, n) in zip(self.params, self.slots):
+                self.flat_grads[off:off + n].copy_(p.grad.reshape(-1))
+                p.grad = None
+
+    def build_gather_table(self) -> None:
+        """Call once right after hipGraph capture of backward: .grad now
+        holds capture-pool tensors whose addresses are replay-stable."""
+        dev = self.flat_grads.device
+        ptrs, offs, sizes = [], [], []
+        for p, (off, n) in zip(self.params, self.slots):
+            g = p.grad
+            if g is None or not g.is_contiguous() or g.numel() != n \
+                    or g.dtype != self.flat_grads.dtype:
+                raise RuntimeError(
+                    f"scatter-grad table: bad grad for slot {off} "
+                    f"(shape {None if g is None else tuple(g.shape)})")
+            ptrs.append(g.data_ptr())
+            offs.append(off)
+            sizes.append(n)
+        self._gather_table = (
+            torch.tensor(ptrs, dtype=torch.int64, device=dev),
+            torch.tensor(offs, dtype=torch.int64, device=dev),
+            torch.tensor(sizes, dtype=torch.int64, device=dev))
+
+    def gather_grads(self) -> None:
+        """One-kernel scattered-grad -> flat_grads pack."""
+        ext = _ops.require_ext()
+        srcs, offs, sizes = self._gather_table
+        ext.grad_gather(srcs, offs, sizes, self.flat_grads)
 
     def grad_global_norm(self) -> torch.Tensor:
         if self.flat_grads.is_cuda:

@@ -70,6 +70,9 @@ class GraphedImpalaStep:
 
         # ---- warmup (eager, side stream), with state snapshot/restore -----
         opt = agent.optimizer
+        # scatter-grad mode: backward ASSIGNS grads (no per-param
+        # AccumulateGrad adds); one gather kernel packs them (optim.py)
+        opt.enable_scatter_grads()
         snap_params = opt.flat_params.detach().clone()
         snap_state = {k: v.detach().clone()
                       for k, v in opt._state_tensors().items()}
@@ -78,6 +81,7 @@ class GraphedImpalaStep:
         with torch.cuda.stream(side):
             for _ in range(warmup_iters):
                 self._fwd_bwd()
+                opt.gather_grads_eager()
                 opt.step_tensor_lr(self.lr_buf)
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
@@ -91,6 +95,8 @@ class GraphedImpalaStep:
         self.g_fwd_bwd = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_fwd_bwd):
             self.losses = self._fwd_bwd()
+        # .grad now holds capture-pool tensors at replay-stable addresses
+        opt.build_gather_table()
         self.g_opt = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_opt):
             opt.step_tensor_lr(self.lr_buf)
@@ -102,7 +108,8 @@ class GraphedImpalaStep:
 
     def _fwd_bwd(self) -> Tuple[torch.Tensor, ...]:
         agent = self.agent
-        agent.optimizer.flat_grads.zero_()
+        if not agent.optimizer.scatter:
+            agent.optimizer.flat_grads.zero_()
         i = self.inputs
         s = agent.prepare_frames(i["state"])
         pi_loss, baseline_loss, entropy, total = agent.compute_losses(
@@ -152,6 +159,7 @@ class GraphedImpalaStep:
         lr = agent.lr_at(agent.global_step)
         self.lr_buf.fill_(lr)
         self.g_fwd_bwd.replay()
+        agent.optimizer.gather_grads()
         if self._distributed:
             agent.reduce_gradients()
         self.g_opt.replay()
