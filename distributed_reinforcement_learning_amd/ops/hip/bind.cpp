@@ -715,8 +715,9 @@ void grad_gather(torch::Tensor srcs, torch::Tensor offs,
       drla_grad_gather, dim3(drla_grid(chunks)), dim3(DRLA_BLOCK), 0,
       cur_stream(),
       reinterpret_cast<const unsigned long long*>(srcs.data_ptr<int64_t>()),
-      offs.data_ptr<int64_t>(), sizes.data_ptr<int64_t>(), u16pm(dst),
-      (int)srcs.numel(), chunks);
+      reinterpret_cast<const long long*>(offs.data_ptr<int64_t>()),
+      reinterpret_cast<const long long*>(sizes.data_ptr<int64_t>()),
+      u16pm(dst), (int)srcs.numel(), chunks);
 }
 
 torch::Tensor sq_norm(torch::Tensor x) {
