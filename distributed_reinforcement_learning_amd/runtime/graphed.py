@@ -97,14 +97,18 @@ class GraphedImpalaStep:
             self.losses = self._fwd_bwd()
         # .grad now holds capture-pool tensors at replay-stable addresses
         opt.build_gather_table()
-        self.g_opt = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_opt):
-            opt.step_tensor_lr(self.lr_buf)
-
         from distributed_reinforcement_learning_amd.parallel.dist import (
             is_distributed,
         )
         self._distributed = is_distributed()
+        self.g_opt = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_opt):
+            if not self._distributed:
+                # single-GPU: the gather rides inside the optimizer graph;
+                # distributed keeps it eager so the RCCL all-reduce sees
+                # the packed bucket between the two replays
+                opt.gather_grads()
+            opt.step_tensor_lr(self.lr_buf)
 
     def _fwd_bwd(self) -> Tuple[torch.Tensor, ...]:
         agent = self.agent
@@ -159,8 +163,8 @@ class GraphedImpalaStep:
         lr = agent.lr_at(agent.global_step)
         self.lr_buf.fill_(lr)
         self.g_fwd_bwd.replay()
-        agent.optimizer.gather_grads()
         if self._distributed:
+            agent.optimizer.gather_grads()
             agent.reduce_gradients()
         self.g_opt.replay()
         self._consumed.record()
