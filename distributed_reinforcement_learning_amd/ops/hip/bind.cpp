@@ -355,7 +355,10 @@ torch::Tensor conv_dgrad(int layer, torch::Tensor dy, torch::Tensor w) {
                          dy.options().dtype(torch::kBFloat16));
   const int grid = (M2 + 127) / 128;
   if (layer == 2) {
-    hipLaunchKernelGGL(drla_conv_dgrad_l2, dim3(grid), dim3(256), 0,
+    // parity-class kernel: blockIdx.y = (hi%2, wi%2) class, each class
+    // covers batch * (hi/2) * (wi/2) rows
+    const int grid_c = (batch * (cfg.hi / 2) * (cfg.wi / 2) + 127) / 128;
+    hipLaunchKernelGGL(drla_conv_dgrad_l2, dim3(grid_c, 4), dim3(256), 0,
                        cur_stream(), u16p(dy), u16p(w), u16pm(dx), batch);
   } else {
     hipLaunchKernelGGL(drla_conv_dgrad_l3, dim3(grid), dim3(256), 0,

@@ -254,26 +254,31 @@ extern "C" __global__ void drla_relu_mask_bwd(
   const uint4* dy4 = reinterpret_cast<const uint4*>(dy);
   const uint4* y4 = reinterpret_cast<const uint4*>(y);
   uint4* out4 = reinterpret_cast<uint4*>(out);
+  // named .x/.y/.z/.w only — a pointer into a local vector can force it
+  // to scratch (guide §5.4 rule 20; cost measured on sq_norm_bf16)
+#define DRLA_RM_LANE(dw, yw, ow, a0, a1)                              \
+  {                                                                   \
+    const bf16raw d0 = (bf16raw)((dw) & 0xFFFF);                      \
+    const bf16raw d1 = (bf16raw)((dw) >> 16);                         \
+    const bf16raw m0 =                                                \
+        (cv_bf2f((bf16raw)((yw) & 0xFFFF)) > 0.0f) ? d0 : (bf16raw)0; \
+    const bf16raw m1 =                                                \
+        (cv_bf2f((bf16raw)((yw) >> 16)) > 0.0f) ? d1 : (bf16raw)0;    \
+    (ow) = ((unsigned int)m1 << 16) | m0;                             \
+    (a0) += cv_bf2f(m0);                                              \
+    (a1) += cv_bf2f(m1);                                              \
+  }
   for (; i < n8; i += stride) {
     const uint4 dv = dy4[i];
     const uint4 yv = y4[i];
     uint4 ov;
-    const unsigned int* dp = reinterpret_cast<const unsigned int*>(&dv);
-    const unsigned int* yp = reinterpret_cast<const unsigned int*>(&yv);
-    unsigned int* op = reinterpret_cast<unsigned int*>(&ov);
-    for (int w = 0; w < 4; ++w) {
-      const bf16raw d0 = (bf16raw)(dp[w] & 0xFFFF);
-      const bf16raw d1 = (bf16raw)(dp[w] >> 16);
-      const bf16raw y0 = (bf16raw)(yp[w] & 0xFFFF);
-      const bf16raw y1 = (bf16raw)(yp[w] >> 16);
-      const bf16raw m0 = (cv_bf2f(y0) > 0.0f) ? d0 : (bf16raw)0;
-      const bf16raw m1 = (cv_bf2f(y1) > 0.0f) ? d1 : (bf16raw)0;
-      op[w] = ((unsigned int)m1 << 16) | m0;
-      acc[2 * w] += cv_bf2f(m0);
-      acc[2 * w + 1] += cv_bf2f(m1);
-    }
+    DRLA_RM_LANE(dv.x, yv.x, ov.x, acc[0], acc[1]);
+    DRLA_RM_LANE(dv.y, yv.y, ov.y, acc[2], acc[3]);
+    DRLA_RM_LANE(dv.z, yv.z, ov.z, acc[4], acc[5]);
+    DRLA_RM_LANE(dv.w, yv.w, ov.w, acc[6], acc[7]);
     out4[i] = ov;
   }
+#undef DRLA_RM_LANE
   __shared__ float red[DRLA_BLOCK * 8];
   for (int e = 0; e < 8; ++e) red[tid * 8 + e] = acc[e];
   __syncthreads();
@@ -616,9 +621,120 @@ __device__ void conv_dgrad_impl(const bf16raw* __restrict__ dy,  // [M][CO]
   }
 }
 
+// stride-2 dgrad, parity-class decomposition: for fixed (hi%2, wi%2) only
+// taps with kh≡hi, kw≡wi (mod 2) can contribute, so 4 of the 16 taps are
+// live — the generic kernel burns 3/4 of its k-loop staging zeros. Blocks
+// are grouped by parity class (blockIdx.y) so the whole block walks just
+// the 4 live taps: 8 BK=32 iterations instead of 32.
+template <int CI, int CO, int KH, int KW, int HI, int WI, int HO, int WO>
+__device__ void conv_dgrad_s2_impl(const bf16raw* __restrict__ dy,
+                                   const bf16raw* __restrict__ w,
+                                   bf16raw* __restrict__ dx, int batch) {
+  constexpr int K = KH * KW * CI;
+  constexpr int BM = 128;
+  constexpr int BK = 32;
+  constexpr int PAD = 8;
+  constexpr int NFRAG = CI / 16;
+  constexpr int H2 = HI / 2, W2 = WI / 2;   // per-class extent
+  const int ph = blockIdx.y >> 1, pw = blockIdx.y & 1;
+  const int Mc = batch * H2 * W2;           // rows in this class
+
+  __shared__ bf16raw Ad[BM][BK + PAD];
+  __shared__ bf16raw Bd[CI][BK + PAD];
+
+  const int tid = threadIdx.x;
+  const int wave = tid / 64;
+  const int lane = tid % 64;
+  const int row0 = blockIdx.x * BM;
+
+  f32x4 acc[2][NFRAG];
+  for (int mi = 0; mi < 2; ++mi)
+    for (int ni = 0; ni < NFRAG; ++ni)
+      acc[mi][ni] = {0.f, 0.f, 0.f, 0.f};
+
+  const int a_row = tid >> 1;
+  const int a_k0 = (tid & 1) * 16;
+  const int mc = row0 + a_row;
+  const int n_idx = mc / (H2 * W2);
+  const int rem = mc - n_idx * (H2 * W2);
+  const int hi = (rem / W2) * 2 + ph;
+  const int wi = (rem - (rem / W2) * W2) * 2 + pw;
+
+  for (int it = 0; it < 4 * (CO / BK); ++it) {
+    const int t = it / (CO / BK);           // live tap 0..3
+    const int co0 = (it - t * (CO / BK)) * BK;
+    const int kh = ph + 2 * (t >> 1);
+    const int kw = pw + 2 * (t & 1);
+    {
+      bool valid = (mc < Mc);
+      int ho = 0, wo = 0;
+      if (valid) {
+        const int hh = hi - kh;
+        const int ww = wi - kw;
+        valid = hh >= 0 && ww >= 0;
+        if (valid) {
+          ho = hh >> 1;
+          wo = ww >> 1;
+          valid = ho < HO && wo < WO;
+        }
+      }
+      if (valid) {
+        const bf16raw* src =
+            dy + ((long long)(n_idx * HO + ho) * WO + wo) * CO + co0 + a_k0;
+        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) =
+            *reinterpret_cast<const uint4*>(src);
+        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) =
+            *reinterpret_cast<const uint4*>(src + 8);
+      } else {
+        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) = uint4{0, 0, 0, 0};
+        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) = uint4{0, 0, 0, 0};
+      }
+    }
+    {
+      constexpr int CI_PER_T = (CI * BK) / 256;
+      const int col = tid % BK;
+      const int ci0 = (tid / BK) * CI_PER_T;
+      const int tap = kh * KW + kw;
+      const bf16raw* src = w + (long long)(co0 + col) * K + tap * CI + ci0;
+      for (int e = 0; e < CI_PER_T; ++e) {
+        Bd[ci0 + e][col] = src[e];
+      }
+    }
+    __syncthreads();
+    for (int mi = 0; mi < 2; ++mi) {
+      const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
+          &Ad[wave * 32 + mi * 16 + (lane & 15)][(lane >> 4) * 8]);
+      for (int ni = 0; ni < NFRAG; ++ni) {
+        const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            &Bd[ni * 16 + (lane & 15)][(lane >> 4) * 8]);
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  for (int mi = 0; mi < 2; ++mi) {
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int ci = ni * 16 + (lane & 15);
+      for (int r = 0; r < 4; ++r) {
+        const int growc = row0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + r;
+        if (growc < Mc) {
+          const int gn = growc / (H2 * W2);
+          const int grem = growc - gn * (H2 * W2);
+          const int ghi = (grem / W2) * 2 + ph;
+          const int gwi = (grem - (grem / W2) * W2) * 2 + pw;
+          dx[(((long long)gn * HI + ghi) * WI + gwi) * CI + ci] =
+              drla_f32_to_bf16(acc[mi][ni][r]);
+        }
+      }
+    }
+  }
+}
+
 extern "C" __global__ __launch_bounds__(256) void drla_conv_dgrad_l2(
     const bf16raw* dy, const bf16raw* w, bf16raw* dx, int batch) {
-  conv_dgrad_impl<32, 64, 4, 4, 2, 20, 20, 9, 9>(dy, w, dx, batch);
+  conv_dgrad_s2_impl<32, 64, 4, 4, 20, 20, 9, 9>(dy, w, dx, batch);
 }
 extern "C" __global__ __launch_bounds__(256) void drla_conv_dgrad_l3(
     const bf16raw* dy, const bf16raw* w, bf16raw* dx, int batch) {
