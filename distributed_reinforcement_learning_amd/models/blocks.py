@@ -159,4 +159,7 @@ class LSTMCellTF(nn.Module):
         # elementwise tail: fused HIP kernel on gfx950 (fp32 states), eager
         # math on CPU — ops/lstm_op.py
         from distributed_reinforcement_learning_amd.ops import lstm_fused_step
+        if g.is_cuda:
+            # the fused tail consumes bf16 gates natively
+            return lstm_fused_step(g, c.float(), self.forget_bias)
         return lstm_fused_step(g.float(), c.float(), self.forget_bias)

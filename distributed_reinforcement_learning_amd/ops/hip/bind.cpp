@@ -46,7 +46,7 @@ extern "C" __global__ void drla_relu_mask_bwd(const unsigned short*,
                                               const unsigned short*,
                                               unsigned short*, float*,
                                               long long, int);
-extern "C" __global__ void drla_wgrad_finalize(const float*, unsigned short*,
+extern "C" __global__ void drla_wgrad_finalize(float*, unsigned short*,
                                                int, int);
 extern "C" __global__ void drla_conv_wgrad_l1(const unsigned char*,
                                               const unsigned short*, float*,
@@ -84,6 +84,9 @@ extern "C" __global__ void drla_embed_bwd_scatter(
 extern "C" __global__ void drla_f32_to_bf16_kernel(const float*,
                                                    unsigned short*,
                                                    long long);
+extern "C" __global__ void drla_f32_to_bf16_zero_kernel(float*,
+                                                        unsigned short*,
+                                                        long long);
 extern "C" __global__ void drla_vtrace_scan(const float*, const float*,
                                             const float*, float*, int, int);
 extern "C" __global__ void drla_vtrace_loss_fwd(
@@ -100,6 +103,12 @@ extern "C" __global__ void drla_lstm_tail_bwd(const float*, const float*,
                                               const float*, const float*,
                                               const float*, float*, float*,
                                               long long, int);
+extern "C" __global__ void drla_lstm_tail_fwd_bf16(
+    const unsigned short*, const float*, float*, float*, float*, float,
+    long long, int);
+extern "C" __global__ void drla_lstm_tail_bwd_bf16(
+    const float*, const float*, const float*, const float*, const float*,
+    unsigned short*, float*, long long, int);
 extern "C" __global__ void drla_mlp_heads_fwd(
     const float*, const unsigned short*, const float*, const unsigned short*,
     const float*, const unsigned short*, const float*, const unsigned short*,
@@ -124,7 +133,7 @@ extern "C" __global__ void drla_heads_wgrad(
     unsigned short*, unsigned short*, int, int);
 extern "C" __global__ void drla_grad_gather(
     const unsigned long long*, const long long*, const long long*,
-    unsigned short*, int, long long);
+    unsigned short*, int, long long, float*, int);
 extern "C" __global__ void drla_lstm_seq_fwd(
     const unsigned short*, const float*, const unsigned short*, const float*,
     const float*, const unsigned char*, float*, float*, float*, float, int,
@@ -308,8 +317,14 @@ torch::Tensor conv_wgrad(int layer, torch::Tensor in, torch::Tensor dy) {
   const auto& cfg = convcfg::L[layer];
   const int batch = in.size(0);
   const int K = cfg.kh * cfg.kw * cfg.ci;
-  auto scratch = torch::zeros({K, cfg.co},
-                              dy.options().dtype(torch::kFloat));
+  // persistent zero-between-calls scratch per layer: drla_wgrad_finalize
+  // re-zeroes it as it reads, so no per-step fill kernel
+  static torch::Tensor scratch_cache[4];
+  if (!scratch_cache[layer].defined()) {
+    scratch_cache[layer] =
+        torch::zeros({K, cfg.co}, dy.options().dtype(torch::kFloat));
+  }
+  auto scratch = scratch_cache[layer];
   const int split = (layer <= 1) ? 256 : 128;
   dim3 grid((K + 63) / 64, split);
   switch (layer) {
@@ -445,8 +460,22 @@ torch::Tensor embed_bwd(torch::Tensor indices, torch::Tensor grad_out,
   const long long N = grad_out.size(0);
   const int H = grad_out.size(1);
   const bool in16 = grad_out.scalar_type() == torch::kBFloat16;
-  auto scratch = torch::zeros({num_rows, H},
-                              grad_out.options().dtype(torch::kFloat));
+  torch::Tensor scratch;
+  if (want_bf16) {
+    // persistent zero-between-calls scratch (the zeroing cast below
+    // restores the invariant); f32 callers own the returned tensor, so
+    // they keep the per-call allocation
+    static torch::Tensor emb_scratch;
+    if (!emb_scratch.defined() || emb_scratch.size(0) != num_rows ||
+        emb_scratch.size(1) != H) {
+      emb_scratch = torch::zeros({num_rows, H},
+                                 grad_out.options().dtype(torch::kFloat));
+    }
+    scratch = emb_scratch;
+  } else {
+    scratch = torch::zeros({num_rows, H},
+                           grad_out.options().dtype(torch::kFloat));
+  }
   hipLaunchKernelGGL(
       drla_embed_bwd_scatter, dim3(drla_grid(N * H)), dim3(DRLA_BLOCK), 0,
       cur_stream(),
@@ -458,7 +487,7 @@ torch::Tensor embed_bwd(torch::Tensor indices, torch::Tensor grad_out,
   if (!want_bf16) return scratch;
   auto out = torch::empty({num_rows, H},
                           grad_out.options().dtype(torch::kBFloat16));
-  hipLaunchKernelGGL(drla_f32_to_bf16_kernel,
+  hipLaunchKernelGGL(drla_f32_to_bf16_zero_kernel,
                      dim3(drla_grid(num_rows * H)), dim3(DRLA_BLOCK), 0,
                      cur_stream(), scratch.data_ptr<float>(),
                      reinterpret_cast<unsigned short*>(out.data_ptr()),
@@ -520,7 +549,9 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_tail_fwd(
     torch::Tensor gates, torch::Tensor c_prev, double forget_bias) {
   check_gpu_contig(gates, "gates");
   check_gpu_contig(c_prev, "c_prev");
-  TORCH_CHECK(gates.scalar_type() == torch::kFloat, "gates must be float32");
+  const bool bf16 = gates.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || gates.scalar_type() == torch::kFloat,
+              "gates must be bf16 or float32");
   TORCH_CHECK(c_prev.scalar_type() == torch::kFloat, "c_prev must be float32");
   const long long N = gates.size(0);
   const int H4 = gates.size(1);
@@ -529,25 +560,46 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_tail_fwd(
   TORCH_CHECK(c_prev.size(1) == H, "c_prev width mismatch");
   auto new_h = torch::empty_like(c_prev);
   auto new_c = torch::empty_like(c_prev);
-  auto stash = torch::empty_like(gates);
-  hipLaunchKernelGGL(drla_lstm_tail_fwd, dim3(drla_grid(N * H)),
-                     dim3(DRLA_BLOCK), 0, cur_stream(),
-                     gates.data_ptr<float>(), c_prev.data_ptr<float>(),
-                     new_h.data_ptr<float>(), new_c.data_ptr<float>(),
-                     stash.data_ptr<float>(),
-                     static_cast<float>(forget_bias), N, H);
+  auto stash = torch::empty({N, (long long)H4},
+                            c_prev.options().dtype(torch::kFloat));
+  if (bf16) {
+    hipLaunchKernelGGL(drla_lstm_tail_fwd_bf16, dim3(drla_grid(N * H)),
+                       dim3(DRLA_BLOCK), 0, cur_stream(), u16p(gates),
+                       c_prev.data_ptr<float>(), new_h.data_ptr<float>(),
+                       new_c.data_ptr<float>(), stash.data_ptr<float>(),
+                       static_cast<float>(forget_bias), N, H);
+  } else {
+    hipLaunchKernelGGL(drla_lstm_tail_fwd, dim3(drla_grid(N * H)),
+                       dim3(DRLA_BLOCK), 0, cur_stream(),
+                       gates.data_ptr<float>(), c_prev.data_ptr<float>(),
+                       new_h.data_ptr<float>(), new_c.data_ptr<float>(),
+                       stash.data_ptr<float>(),
+                       static_cast<float>(forget_bias), N, H);
+  }
   return {new_h, new_c, stash};
 }
 
 std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
     torch::Tensor grad_h, torch::Tensor grad_c, torch::Tensor stash,
-    torch::Tensor c_prev, torch::Tensor new_c) {
+    torch::Tensor c_prev, torch::Tensor new_c, bool bf16_gates) {
   for (auto* t : {&grad_h, &grad_c, &stash, &c_prev, &new_c})
     check_gpu_contig(*t, "lstm bwd input");
   const long long N = stash.size(0);
   const int H = stash.size(1) / 4;
-  auto grad_gates = torch::empty_like(stash);
   auto grad_c_prev = torch::empty_like(c_prev);
+  if (bf16_gates) {
+    auto grad_gates = torch::empty(
+        {N, (long long)stash.size(1)},
+        stash.options().dtype(torch::kBFloat16));
+    hipLaunchKernelGGL(drla_lstm_tail_bwd_bf16, dim3(drla_grid(N * H)),
+                       dim3(DRLA_BLOCK), 0, cur_stream(),
+                       grad_h.data_ptr<float>(), grad_c.data_ptr<float>(),
+                       stash.data_ptr<float>(), c_prev.data_ptr<float>(),
+                       new_c.data_ptr<float>(), u16pm(grad_gates),
+                       grad_c_prev.data_ptr<float>(), N, H);
+    return {grad_gates, grad_c_prev};
+  }
+  auto grad_gates = torch::empty_like(stash);
   hipLaunchKernelGGL(drla_lstm_tail_bwd, dim3(drla_grid(N * H)),
                      dim3(DRLA_BLOCK), 0, cur_stream(),
                      grad_h.data_ptr<float>(), grad_c.data_ptr<float>(),
@@ -704,7 +756,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_seq_fwd(
 }
 
 void grad_gather(torch::Tensor srcs, torch::Tensor offs,
-                 torch::Tensor sizes, torch::Tensor dst) {
+                 torch::Tensor sizes, torch::Tensor dst,
+                 c10::optional<torch::Tensor> norm_ws) {
   check_gpu_contig(dst, "dst");
   TORCH_CHECK(dst.scalar_type() == torch::kBFloat16,
               "grad_gather wants a bf16 flat bucket");
@@ -714,13 +767,20 @@ void grad_gather(torch::Tensor srcs, torch::Tensor offs,
     TORCH_CHECK(t->scalar_type() == torch::kLong, "table must be int64");
   }
   const long long chunks = dst.numel() / 8;
+  float* nw = nullptr;
+  int nw_n = 0;
+  if (norm_ws.has_value()) {
+    check_gpu_contig(*norm_ws, "norm_ws");
+    nw = norm_ws->data_ptr<float>();
+    nw_n = (int)norm_ws->numel();
+  }
   hipLaunchKernelGGL(
       drla_grad_gather, dim3(drla_grid(chunks)), dim3(DRLA_BLOCK), 0,
       cur_stream(),
       reinterpret_cast<const unsigned long long*>(srcs.data_ptr<int64_t>()),
       reinterpret_cast<const long long*>(offs.data_ptr<int64_t>()),
       reinterpret_cast<const long long*>(sizes.data_ptr<int64_t>()),
-      u16pm(dst), (int)srcs.numel(), chunks);
+      u16pm(dst), (int)srcs.numel(), chunks, nw, nw_n);
 }
 
 torch::Tensor sq_norm(torch::Tensor x) {
@@ -748,7 +808,8 @@ torch::Tensor sq_norm_bf16(torch::Tensor x) {
 
 void rmsprop_step_bf16_t(torch::Tensor p, torch::Tensor g,
                          torch::Tensor master, torch::Tensor ms, double clip,
-                         torch::Tensor lr_buf, double rho, double eps) {
+                         torch::Tensor lr_buf, double rho, double eps,
+                         c10::optional<torch::Tensor> norm_ws) {
   for (auto* t : {&p, &g, &master, &ms, &lr_buf})
     check_gpu_contig(*t, "rmsprop bf16 tensor");
   TORCH_CHECK(p.scalar_type() == torch::kBFloat16 &&
@@ -756,10 +817,18 @@ void rmsprop_step_bf16_t(torch::Tensor p, torch::Tensor g,
               "params/grads must be bf16");
   const long long n = p.numel();
   torch::Tensor norm_buf;
-  if (clip > 0) {
+  if (norm_ws.has_value() && clip > 0) {
+    // persistent buffer, zeroed by grad_gather earlier on this stream
+    norm_buf = *norm_ws;
+    hipLaunchKernelGGL(drla_sq_norm_bf16,
+                       dim3(drla_grid(g.numel() / 4 + 1)), dim3(DRLA_BLOCK),
+                       0, cur_stream(),
+                       reinterpret_cast<const unsigned short*>(g.data_ptr()),
+                       norm_buf.data_ptr<float>(), (long long)g.numel());
+  } else if (clip > 0) {
     norm_buf = sq_norm_bf16(g);
   } else {
-    norm_buf = torch::zeros({1}, master.options());
+    norm_buf = torch::zeros({DRLA_NORM_SLOTS * 16}, master.options());
   }
   hipLaunchKernelGGL(drla_rmsprop_step_bf16, dim3(drla_grid(n)),
                      dim3(DRLA_BLOCK), 0, cur_stream(),

@@ -322,9 +322,11 @@ extern "C" __global__ void drla_bias_grad(
 }
 
 // cast + transpose the f32 [K][CO] wgrad scratch into bf16 [CO][K]
-// (the channels_last weight-grad layout)
+// (the channels_last weight-grad layout). Zeroes the scratch as it reads:
+// the binding keeps ONE persistent scratch per layer whose invariant is
+// "zero between calls", so the per-step torch::zeros fill disappears.
 extern "C" __global__ void drla_wgrad_finalize(
-    const float* __restrict__ scratch, bf16raw* __restrict__ dw, int K,
+    float* __restrict__ scratch, bf16raw* __restrict__ dw, int K,
     int CO) {
   long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
   const long long total = (long long)K * CO;
@@ -332,7 +334,9 @@ extern "C" __global__ void drla_wgrad_finalize(
   for (; i < total; i += stride) {
     const int co = i / K;
     const int k = i - (long long)co * K;
-    dw[i] = drla_f32_to_bf16(scratch[(long long)k * CO + co]);
+    const long long si = (long long)k * CO + co;
+    dw[i] = drla_f32_to_bf16(scratch[si]);
+    scratch[si] = 0.0f;
   }
 }
 

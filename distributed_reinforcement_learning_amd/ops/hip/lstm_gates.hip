@@ -11,6 +11,14 @@
 
 #include "drla_common.h"
 
+typedef unsigned short lstm_bf16;
+
+__device__ __forceinline__ float lstm_b2f(lstm_bf16 u) {
+  unsigned int x = ((unsigned int)u) << 16;
+  return __uint_as_float(x);
+}
+
+
 extern "C" __global__ void drla_lstm_tail_fwd(
     const float* __restrict__ gates, const float* __restrict__ c_prev,
     float* __restrict__ new_h, float* __restrict__ new_c,
@@ -68,6 +76,68 @@ extern "C" __global__ void drla_lstm_tail_bwd(
   }
 }
 
+
+// bf16-gates variants: the gate GEMM (addmm) emits bf16; reading it
+// directly (and emitting bf16 grad_gates) removes the [N,4H] f32 cast
+// kernel on each side of the tail (~2 x 4.7 us/step).
+extern "C" __global__ void drla_lstm_tail_fwd_bf16(
+    const unsigned short* __restrict__ gates,
+    const float* __restrict__ c_prev, float* __restrict__ new_h,
+    float* __restrict__ new_c, float* __restrict__ stash, float forget_bias,
+    long long N, int H) {
+  long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long total = N * H;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; idx < total; idx += stride) {
+    const long long n = idx / H;
+    const int h = idx - n * H;
+    const long long g0 = n * 4LL * H + h;
+    const float i_s = drla_sigmoid(lstm_b2f(gates[g0]));
+    const float g_t = tanhf(lstm_b2f(gates[g0 + H]));
+    const float f_s = drla_sigmoid(lstm_b2f(gates[g0 + 2 * H])
+                                   + forget_bias);
+    const float o_s = drla_sigmoid(lstm_b2f(gates[g0 + 3 * H]));
+    const float c_new = f_s * c_prev[idx] + i_s * g_t;
+    new_c[idx] = c_new;
+    new_h[idx] = o_s * tanhf(c_new);
+    stash[g0] = i_s;
+    stash[g0 + H] = g_t;
+    stash[g0 + 2 * H] = f_s;
+    stash[g0 + 3 * H] = o_s;
+  }
+}
+
+extern "C" __global__ void drla_lstm_tail_bwd_bf16(
+    const float* __restrict__ grad_h, const float* __restrict__ grad_c,
+    const float* __restrict__ stash, const float* __restrict__ c_prev,
+    const float* __restrict__ new_c, unsigned short* __restrict__ grad_gates,
+    float* __restrict__ grad_c_prev, long long N, int H) {
+  long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  const long long total = N * H;
+  const long long stride = gridDim.x * (long long)blockDim.x;
+  for (; idx < total; idx += stride) {
+    const long long n = idx / H;
+    const int h = idx - n * H;
+    const long long g0 = n * 4LL * H + h;
+    const float i_s = stash[g0];
+    const float g_t = stash[g0 + H];
+    const float f_s = stash[g0 + 2 * H];
+    const float o_s = stash[g0 + 3 * H];
+    const float tc = tanhf(new_c[idx]);
+    const float dh = grad_h[idx];
+    const float d_tc = dh * o_s * (1.0f - tc * tc) + grad_c[idx];
+    grad_c_prev[idx] = d_tc * f_s;
+    const float di = d_tc * g_t;
+    const float dg = d_tc * i_s;
+    const float df = d_tc * c_prev[idx];
+    const float do_ = dh * tc;
+    grad_gates[g0] = drla_f32_to_bf16(di * i_s * (1.0f - i_s));
+    grad_gates[g0 + H] = drla_f32_to_bf16(dg * (1.0f - g_t * g_t));
+    grad_gates[g0 + 2 * H] = drla_f32_to_bf16(df * f_s * (1.0f - f_s));
+    grad_gates[g0 + 3 * H] = drla_f32_to_bf16(do_ * o_s * (1.0f - o_s));
+  }
+}
+
 // K3 sequence form (SURVEY §5.7a + BASELINE "burn_in hidden-state
 // recompute"): the WHOLE no-grad LSTM unroll in ONE kernel.
 //
@@ -80,13 +150,6 @@ extern "C" __global__ void drla_lstm_tail_bwd(
 //
 // Replaces the reference's per-timestep replica chain
 // (r2d2_lstm.py:67-114): ~5 launches/step -> 1 launch per sequence.
-
-typedef unsigned short lstm_bf16;
-
-__device__ __forceinline__ float lstm_b2f(lstm_bf16 u) {
-  unsigned int x = ((unsigned int)u) << 16;
-  return __uint_as_float(x);
-}
 
 extern "C" __global__ void drla_lstm_seq_fwd(
     const lstm_bf16* __restrict__ xg16,  // [B,L,4H] (nullable)

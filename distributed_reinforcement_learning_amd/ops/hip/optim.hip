@@ -204,7 +204,13 @@ extern "C" __global__ void drla_grad_gather(
     const unsigned long long* __restrict__ srcs,  // [nseg] device pointers
     const long long* __restrict__ offs,           // [nseg] padded starts
     const long long* __restrict__ sizes,          // [nseg] true numels
-    unsigned short* __restrict__ dst, int nseg, long long chunks) {
+    unsigned short* __restrict__ dst, int nseg, long long chunks,
+    float* __restrict__ norm_ws, int norm_n) {
+  // piggyback: zero the persistent sq-norm partial buffer (consumed by
+  // the sq_norm launch that follows on this stream) — saves a fill kernel
+  if (norm_ws && blockIdx.x == 0 && (int)threadIdx.x < norm_n) {
+    norm_ws[threadIdx.x] = 0.0f;
+  }
   for (long long c = blockIdx.x * (long long)blockDim.x + threadIdx.x;
        c < chunks; c += gridDim.x * (long long)blockDim.x) {
     const long long i = c * 8;

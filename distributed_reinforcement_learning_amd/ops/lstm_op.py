@@ -22,10 +22,13 @@ class _FusedLstmTail(torch.autograd.Function):
     def forward(ctx, gates: torch.Tensor, c_prev: torch.Tensor,
                 forget_bias: float):
         ext = _ops.require_ext()
+        # bf16 gates (straight from the addmm) are consumed natively —
+        # no [N,4H] f32 cast on either side of the tail
         new_h, new_c, stash = ext.lstm_tail_fwd(gates.contiguous(),
                                                 c_prev.contiguous(),
                                                 forget_bias)
         ctx.save_for_backward(stash, c_prev, new_c)
+        ctx.bf16_gates = gates.dtype == torch.bfloat16
         return new_h, new_c
 
     @staticmethod
@@ -34,17 +37,18 @@ class _FusedLstmTail(torch.autograd.Function):
         ext = _ops.require_ext()
         grad_gates, grad_c_prev = ext.lstm_tail_bwd(
             grad_h.contiguous(), grad_c.contiguous(), stash,
-            c_prev.contiguous(), new_c)
+            c_prev.contiguous(), new_c, ctx.bf16_gates)
         return grad_gates, grad_c_prev, None
 
 
 def lstm_fused_step(gates: torch.Tensor, c_prev: torch.Tensor,
                     forget_bias: float = 1.0
                     ) -> Tuple[torch.Tensor, torch.Tensor]:
-    """gates [N,4H] (pre-activation, order i,g,f,o), c_prev [N,H] ->
-    (new_h, new_c)."""
+    """gates [N,4H] (pre-activation, order i,g,f,o; f32 or bf16 on GPU),
+    c_prev [N,H] -> (new_h, new_c) f32."""
     if gates.is_cuda:
         return _FusedLstmTail.apply(gates, c_prev, forget_bias)
+    gates = gates.float()
     i, g, f, o = gates.chunk(4, dim=1)
     new_c = torch.sigmoid(f + forget_bias) * c_prev \
         + torch.sigmoid(i) * torch.tanh(g)

@@ -118,12 +118,16 @@ class _FlatOptimizerBase:
             torch.tensor(ptrs, dtype=torch.int64, device=dev),
             torch.tensor(offs, dtype=torch.int64, device=dev),
             torch.tensor(sizes, dtype=torch.int64, device=dev))
+        # persistent sq-norm partials (16 slots x 16-float cache lines),
+        # zeroed by the gather kernel each step
+        self._norm_ws = torch.zeros(256, dtype=torch.float32, device=dev)
 
     def gather_grads(self) -> None:
-        """One-kernel scattered-grad -> flat_grads pack."""
+        """One-kernel scattered-grad -> flat_grads pack (also re-zeroes
+        the persistent sq-norm buffer for the update that follows)."""
         ext = _ops.require_ext()
         srcs, offs, sizes = self._gather_table
-        ext.grad_gather(srcs, offs, sizes, self.flat_grads)
+        ext.grad_gather(srcs, offs, sizes, self.flat_grads, self._norm_ws)
 
     def grad_global_norm(self) -> torch.Tensor:
         if self.flat_grads.is_cuda:
@@ -207,7 +211,8 @@ class FusedRMSProp(_FlatOptimizerBase):
         if self.mixed:
             ext.rmsprop_step_bf16_t(self.flat_params, self.flat_grads,
                                     self.master, self.ms, clip, lr_buf,
-                                    self.rho, self.eps)
+                                    self.rho, self.eps,
+                                    getattr(self, "_norm_ws", None))
         else:
             ext.rmsprop_step_t(self.flat_params, self.flat_grads, self.ms,
                                clip, lr_buf, self.rho, self.eps)

@@ -219,7 +219,8 @@ def test_mixed_precision_rmsprop_master(ext):
     ms = torch.rand(n).cuda()
     lr_buf = torch.full((1,), 1e-3, device="cuda")
     ms_ref = ms.cpu().clone()
-    ext.rmsprop_step_bf16_t(p, g, master, ms, 40.0, lr_buf, 0.99, 0.1)
+    ext.rmsprop_step_bf16_t(p, g, master, ms, 40.0, lr_buf, 0.99, 0.1,
+                            None)
     # golden: clip on bf16 grads, update fp32 master, round to bf16
     gf = g.float().cpu()
     norm = gf.norm()
