@@ -110,10 +110,12 @@ extern "C" __global__ void drla_lstm_tail_bwd_bf16(
     const float*, const float*, const float*, const float*, const float*,
     unsigned short*, float*, long long, int);
 extern "C" __global__ void drla_mlp_heads_fwd(
-    const float*, const unsigned short*, const float*, const unsigned short*,
-    const float*, const unsigned short*, const float*, const unsigned short*,
-    const float*, const unsigned short*, const float*, const unsigned short*,
-    const float*, unsigned short*, float*, unsigned short*, int, int);
+    const float*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const unsigned short*, unsigned short*, float*, unsigned short*, int,
+    int);
 extern "C" __global__ void drla_mlp_heads_bwd(
     const unsigned short*, const float*, const unsigned short*,
     const unsigned short*, const unsigned short*, const unsigned short*,
@@ -611,11 +613,11 @@ std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
 
 std::vector<torch::Tensor> mlp_heads_fwd(
     torch::Tensor h, std::vector<torch::Tensor> weights,
-    std::vector<torch::Tensor> biases_f32, int64_t A) {
+    std::vector<torch::Tensor> biases, int64_t A) {
   check_gpu_contig(h, "h");
-  TORCH_CHECK(weights.size() == 6 && biases_f32.size() == 6);
+  TORCH_CHECK(weights.size() == 6 && biases.size() == 6);
   for (auto& t : weights) check_gpu_contig(t, "W");
-  for (auto& t : biases_f32) check_gpu_contig(t, "b");
+  for (auto& t : biases) check_gpu_contig(t, "b");
   const int N = h.size(0);
   TORCH_CHECK(h.size(1) == 256 && A <= 32);
   auto bopt = h.options().dtype(torch::kBFloat16);
@@ -626,12 +628,12 @@ std::vector<torch::Tensor> mlp_heads_fwd(
       drla_mlp_heads_fwd, dim3((N + 15) / 16, 2), dim3(256), 0,
       cur_stream(),
       h.data_ptr<float>(), u16p(weights[0]),
-      biases_f32[0].data_ptr<float>(), u16p(weights[1]),
-      biases_f32[1].data_ptr<float>(), u16p(weights[2]),
-      biases_f32[2].data_ptr<float>(), u16p(weights[3]),
-      biases_f32[3].data_ptr<float>(), u16p(weights[4]),
-      biases_f32[4].data_ptr<float>(), u16p(weights[5]),
-      biases_f32[5].data_ptr<float>(), u16pm(logits),
+      u16p(biases[0]), u16p(weights[1]),
+      u16p(biases[1]), u16p(weights[2]),
+      u16p(biases[2]), u16p(weights[3]),
+      u16p(biases[3]), u16p(weights[4]),
+      u16p(biases[4]), u16p(weights[5]),
+      u16p(biases[5]), u16pm(logits),
       value.data_ptr<float>(), u16pm(stash), N, (int)A);
   return {logits, value, stash};
 }

@@ -47,7 +47,7 @@ template <bool RELU, bool ACCUM>
 __device__ void mh_pass(const bf16raw (*act_in)[MH_LD],
                         bf16raw (*act_out)[MH_LD],
                         const bf16raw* __restrict__ M,   // [nout][kdim]
-                        const float* __restrict__ bias,  // [nout] or null
+                        const bf16raw* __restrict__ bias,  // [nout] or null
                         const bf16raw* __restrict__ mask_stash,  // [N,256]
                         bf16raw* __restrict__ dz_global,         // [N,256]
                         float* __restrict__ colsum,              // [256]
@@ -94,7 +94,7 @@ __device__ void mh_pass(const bf16raw (*act_in)[MH_LD],
         const int lrow = lane & 15;
         const int grow = row0 + lrow;
         float v = acc[ni][r];
-        if (bias) v += (oc < nout) ? bias[oc] : 0.0f;
+        if (bias) v += (oc < nout) ? mh_b2f(bias[oc]) : 0.0f;
         if (RELU) v = fmaxf(v, 0.0f);
         if (mask_stash) {
           const float a = (grow < N)
@@ -129,12 +129,12 @@ __device__ void mh_pass(const bf16raw (*act_in)[MH_LD],
 
 extern "C" __global__ __launch_bounds__(256) void drla_mlp_heads_fwd(
     const float* __restrict__ h,        // [N,256]
-    const bf16raw* __restrict__ W1p, const float* __restrict__ b1p,
-    const bf16raw* __restrict__ W2p, const float* __restrict__ b2p,
-    const bf16raw* __restrict__ W3p, const float* __restrict__ b3p,
-    const bf16raw* __restrict__ W1v, const float* __restrict__ b1v,
-    const bf16raw* __restrict__ W2v, const float* __restrict__ b2v,
-    const bf16raw* __restrict__ W3v, const float* __restrict__ b3v,
+    const bf16raw* __restrict__ W1p, const bf16raw* __restrict__ b1p,
+    const bf16raw* __restrict__ W2p, const bf16raw* __restrict__ b2p,
+    const bf16raw* __restrict__ W3p, const bf16raw* __restrict__ b3p,
+    const bf16raw* __restrict__ W1v, const bf16raw* __restrict__ b1v,
+    const bf16raw* __restrict__ W2v, const bf16raw* __restrict__ b2v,
+    const bf16raw* __restrict__ W3v, const bf16raw* __restrict__ b3v,
     bf16raw* __restrict__ logits,       // [N,A]
     float* __restrict__ value,          // [N]
     bf16raw* __restrict__ stash,    // [N,5*256]: a1p,a2p,a1v,a2v,h_bf16

@@ -25,8 +25,8 @@ class _FusedMlpHeads(torch.autograd.Function):
         ext = _ops.require_ext()
         A = w3p.shape[0]
         weights = [w1p, w2p, w3p, w1v, w2v, w3v]
-        biases = [b.float().contiguous() for b in (b1p, b2p, b3p,
-                                                   b1v, b2v, b3v)]
+        # biases are consumed bf16 in-kernel (no per-step f32 casts)
+        biases = [b.contiguous() for b in (b1p, b2p, b3p, b1v, b2v, b3v)]
         logits, value, stash = ext.mlp_heads_fwd(
             h.contiguous(), [w.contiguous() for w in weights], biases, A)
         ctx.save_for_backward(stash, *weights)
