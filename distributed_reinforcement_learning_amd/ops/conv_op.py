@@ -35,8 +35,10 @@ class _ConvLayer(torch.autograd.Function):
         ext = _ops.require_ext()
         layer = ctx.layer
         co = _LAYER_CO[layer]
-        dy_m, dbias = ext.relu_mask_bwd(dy.contiguous(), y, co)
-        dw = ext.conv_wgrad(layer, x, dy_m)
+        # mask kernel writes bias-grad partials into a persistent slot
+        # buffer; the wgrad finalize that follows sums them -> bf16 dbias
+        dy_m = ext.relu_mask_bwd(dy.contiguous(), y, co)
+        dw, dbias = ext.conv_wgrad(layer, x, dy_m)
         dx = None
         if layer >= 2 and ctx.needs_input_grad[0]:
             dx = ext.conv_dgrad(layer, dy_m, w_flat)
@@ -53,7 +55,7 @@ def conv_layer(x: torch.Tensor, conv: torch.nn.Conv2d,
     w = conv.weight
     co = w.shape[0]
     w_flat = w.permute(0, 2, 3, 1).reshape(co, -1)
-    return _ConvLayer.apply(x, w_flat, conv.bias.float(), layer)
+    return _ConvLayer.apply(x, w_flat, conv.bias, layer)
 
 
 def atari_conv_stack(stack, x_u8_nhwc: torch.Tensor) -> torch.Tensor:

@@ -47,7 +47,8 @@ extern "C" __global__ void drla_relu_mask_bwd(const unsigned short*,
                                               unsigned short*, float*,
                                               long long, int);
 extern "C" __global__ void drla_wgrad_finalize(float*, unsigned short*,
-                                               int, int);
+                                               int, int, float*,
+                                               unsigned short*);
 extern "C" __global__ void drla_conv_wgrad_l1(const unsigned char*,
                                               const unsigned short*, float*,
                                               int);
@@ -261,7 +262,7 @@ torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
   check_gpu_contig(bias, "bias");
   const auto& cfg = convcfg::L[layer];
   TORCH_CHECK(w.scalar_type() == torch::kBFloat16, "weights must be bf16");
-  TORCH_CHECK(bias.scalar_type() == torch::kFloat, "bias must be f32");
+  TORCH_CHECK(bias.scalar_type() == torch::kBFloat16, "bias must be bf16");
   const int batch = in.size(0);
   const int M = batch * cfg.ho * cfg.wo;
   auto out = torch::empty({batch, cfg.ho, cfg.wo, cfg.co},
@@ -272,22 +273,22 @@ torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
     case 0:
       hipLaunchKernelGGL(drla_conv_fwd_l1, dim3(grid), dim3(256), 0,
                          cur_stream(), in.data_ptr<uint8_t>(), u16p(w),
-                         bias.data_ptr<float>(), outp, batch);
+                         u16p(bias), outp, batch);
       break;
     case 1:
       hipLaunchKernelGGL(drla_conv_fwd_l1_c1, dim3(grid), dim3(256), 0,
                          cur_stream(), in.data_ptr<uint8_t>(), u16p(w),
-                         bias.data_ptr<float>(), outp, batch);
+                         u16p(bias), outp, batch);
       break;
     case 2:
       hipLaunchKernelGGL(drla_conv_fwd_l2, dim3(grid), dim3(256), 0,
                          cur_stream(), u16p(in), u16p(w),
-                         bias.data_ptr<float>(), outp, batch);
+                         u16p(bias), outp, batch);
       break;
     case 3:
       hipLaunchKernelGGL(drla_conv_fwd_l3, dim3(grid), dim3(256), 0,
                          cur_stream(), u16p(in), u16p(w),
-                         bias.data_ptr<float>(), outp, batch);
+                         u16p(bias), outp, batch);
       break;
     default:
       TORCH_CHECK(false, "bad layer");
@@ -309,11 +310,13 @@ std::tuple<torch::Tensor, torch::Tensor> relu_mask_bwd(torch::Tensor dy,
   if (grid > 640) grid = 640;
   hipLaunchKernelGGL(drla_relu_mask_bwd, dim3(grid), dim3(DRLA_BLOCK),
                      0, cur_stream(), u16p(dy), u16p(y), u16pm(out),
-                     dbias.data_ptr<float>(), n, (int)CO);
+                     du16p(bias), n, (int)CO);
   return {out, dbias};
 }
 
-torch::Tensor conv_wgrad(int layer, torch::Tensor in, torch::Tensor dy) {
+std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
+                                                    torch::Tensor in,
+                                                    torch::Tensor dy) {
   check_gpu_contig(in, "in");
   check_gpu_contig(dy, "dy");
   const auto& cfg = convcfg::L[layer];
@@ -354,11 +357,14 @@ torch::Tensor conv_wgrad(int layer, torch::Tensor in, torch::Tensor dy) {
       TORCH_CHECK(false, "bad layer");
   }
   auto dw = torch::empty({cfg.co, K}, dy.options().dtype(torch::kBFloat16));
+  auto dbias = torch::empty({cfg.co}, dy.options().dtype(torch::kBFloat16));
+  auto& slots = dbias_slot_buf(dy);
   hipLaunchKernelGGL(drla_wgrad_finalize,
                      dim3(drla_grid((long long)K * cfg.co)),
                      dim3(DRLA_BLOCK), 0, cur_stream(),
-                     scratch.data_ptr<float>(), u16pm(dw), K, cfg.co);
-  return dw;
+                     scratch.data_ptr<float>(), u16pm(dw), K, cfg.co,
+                     slots.data_ptr<float>(), u16pm(dbias));
+  return {dw, dbias};
 }
 
 torch::Tensor conv_dgrad(int layer, torch::Tensor dy, torch::Tensor w) {

@@ -61,7 +61,7 @@ template <typename IN_T, int CI, int CO, int KH, int KW, int STRIDE, int HI,
           int WI, int HO, int WO>
 __device__ void conv_fwd_impl(const IN_T* __restrict__ in,
                               const bf16raw* __restrict__ w,   // [CO][K]
-                              const float* __restrict__ bias,  // [CO]
+                              const bf16raw* __restrict__ bias,  // [CO]
                               bf16raw* __restrict__ out,       // [M][CO]
                               int batch) {
   constexpr int K = KH * KW * CI;
@@ -80,7 +80,7 @@ __device__ void conv_fwd_impl(const IN_T* __restrict__ in,
   const int lane = tid % 64;
   const int row0 = blockIdx.x * BM;
 
-  if (tid < CO) BiasBuf[tid] = bias[tid];
+  if (tid < CO) BiasBuf[tid] = cv_bf2f(bias[tid]);
 
   f32x4 acc[2][NFRAG];
   for (int mi = 0; mi < 2; ++mi)
@@ -201,14 +201,14 @@ __device__ void conv_fwd_impl(const IN_T* __restrict__ in,
 // B[k][col] = Bbuf[col][k]. The probe/parity tests pin this down.
 
 extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l1(
-    const unsigned char* in, const bf16raw* w, const float* bias,
+    const unsigned char* in, const bf16raw* w, const bf16raw* bias,
     bf16raw* out, int batch) {
   conv_fwd_impl<unsigned char, 4, 32, 8, 8, 4, 84, 84, 20, 20>(
       in, w, bias, out, batch);
 }
 
 extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l1_c1(
-    const unsigned char* in, const bf16raw* w, const float* bias,
+    const unsigned char* in, const bf16raw* w, const bf16raw* bias,
     bf16raw* out, int batch) {
   // R2D2's single-channel POMDP frames: CI=1 -> K=64; stage per-tap scalars
   conv_fwd_impl<unsigned char, 1, 32, 8, 8, 4, 84, 84, 20, 20>(
@@ -216,14 +216,14 @@ extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l1_c1(
 }
 
 extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l2(
-    const bf16raw* in, const bf16raw* w, const float* bias, bf16raw* out,
+    const bf16raw* in, const bf16raw* w, const bf16raw* bias, bf16raw* out,
     int batch) {
   conv_fwd_impl<bf16raw, 32, 64, 4, 4, 2, 20, 20, 9, 9>(in, w, bias, out,
                                                         batch);
 }
 
 extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l3(
-    const bf16raw* in, const bf16raw* w, const float* bias, bf16raw* out,
+    const bf16raw* in, const bf16raw* w, const bf16raw* bias, bf16raw* out,
     int batch) {
   conv_fwd_impl<bf16raw, 64, 64, 3, 3, 1, 9, 9, 7, 7>(in, w, bias, out,
                                                       batch);
@@ -233,15 +233,19 @@ extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l3(
 // backward helpers
 // ---------------------------------------------------------------------------
 
-// dY_masked = dY * (Y > 0) with dbias[co] = sum_m dY_masked[m][co] FUSED
-// into the same pass. Thread-to-column pinning: with blockDim (256) and the
-// grid both multiples of CO (CO in {32,64}), a grid-stride walk keeps every
-// thread on ONE co column — local accumulate, LDS-reduce per co, one
-// atomicAdd per (block, co). dbias must be zeroed by the caller.
+// dY_masked = dY * (Y > 0) with dbias partials FUSED into the same pass.
+// Thread-to-column pinning: with blockDim (256) and the grid both multiples
+// of CO (CO in {32,64}), a grid-stride walk keeps every thread on ONE co
+// column — local accumulate, LDS-reduce per co, one atomicAdd per (block,
+// co). 640 blocks adding the SAME [CO] addresses serialize (~10 us,
+// r24/r25 — same pathology as sq_norm), so partials spread over 16 slots
+// of a persistent [16][64] buffer; drla_wgrad_finalize (which always runs
+// next in the conv backward) does the 16-way sum, emits bf16 dbias, and
+// re-zeroes the slots.
 extern "C" __global__ void drla_relu_mask_bwd(
     const bf16raw* __restrict__ dy, const bf16raw* __restrict__ y,
-    bf16raw* __restrict__ out, float* __restrict__ dbias, long long n,
-    int CO) {
+    bf16raw* __restrict__ out, float* __restrict__ dbias_slots,
+    long long n, int CO) {
   // vectorized: each thread moves 8 bf16 per iteration (uint4), which pins
   // it to ONE 8-column group because stride*8 % CO == 0 (CO in {32,64},
   // blockDim 256). Per-thread acc[8] -> LDS -> one atomicAdd per (block,
@@ -290,7 +294,7 @@ extern "C" __global__ void drla_relu_mask_bwd(
     // t = g + k*(CO/8)
     float s = 0.0f;
     for (int t = g; t < DRLA_BLOCK; t += CO / 8) s += red[t * 8 + e];
-    atomicAdd(&dbias[tid], s);
+    atomicAdd(&dbias_slots[(blockIdx.x & 15) * 64 + tid], s);
   }
 }
 
@@ -327,7 +331,18 @@ extern "C" __global__ void drla_bias_grad(
 // "zero between calls", so the per-step torch::zeros fill disappears.
 extern "C" __global__ void drla_wgrad_finalize(
     float* __restrict__ scratch, bf16raw* __restrict__ dw, int K,
-    int CO) {
+    int CO, float* __restrict__ dbias_slots,
+    bf16raw* __restrict__ dbias) {
+  // bias-grad epilogue: 16-way slot sum from relu_mask_bwd, bf16 out,
+  // slots re-zeroed (persistent zero-between-calls buffer)
+  if (dbias_slots && blockIdx.x == 0 && (int)threadIdx.x < CO) {
+    float s = 0.0f;
+    for (int t = 0; t < 16; ++t) {
+      s += dbias_slots[t * 64 + threadIdx.x];
+      dbias_slots[t * 64 + threadIdx.x] = 0.0f;
+    }
+    dbias[threadIdx.x] = drla_f32_to_bf16(s);
+  }
   long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
   const long long total = (long long)K * CO;
   const long long stride = gridDim.x * (long long)blockDim.x;

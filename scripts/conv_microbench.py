@@ -57,7 +57,7 @@ def main():
                 memory_format=torch.channels_last)
         w_flat = conv_bf.weight.permute(0, 2, 3, 1).reshape(
             co, -1).contiguous()
-        bias = conv_bf.bias.float().contiguous()
+        bias = conv_bf.bias.contiguous()
         xc = x.contiguous()
 
         t_custom = t_ms(lambda: ext.conv_fwd(layer, xc, w_flat, bias))
@@ -73,7 +73,7 @@ def main():
         dy = torch.randn(N, ho, ho, co, device="cuda").bfloat16().contiguous()
         y = ext.conv_fwd(layer, xc, w_flat, bias)
         def custom_bwd():
-            dy_m, dbias = ext.relu_mask_bwd(dy, y, co)
+            dy_m = ext.relu_mask_bwd(dy, y, co)
             ext.conv_wgrad(layer, xc, dy_m)
             if layer >= 2:
                 ext.conv_dgrad(layer, dy_m, w_flat)
