@@ -28,20 +28,20 @@ extern "C" __global__ void drla_mfma_probe(const unsigned short*,
                                            const unsigned short*, float*);
 extern "C" __global__ void drla_conv_fwd_l1(const unsigned char*,
                                             const unsigned short*,
-                                            const float*, unsigned short*,
-                                            int);
+                                            const unsigned short*,
+                                            unsigned short*, int);
 extern "C" __global__ void drla_conv_fwd_l1_c1(const unsigned char*,
                                                const unsigned short*,
-                                               const float*, unsigned short*,
-                                               int);
+                                               const unsigned short*,
+                                               unsigned short*, int);
 extern "C" __global__ void drla_conv_fwd_l2(const unsigned short*,
                                             const unsigned short*,
-                                            const float*, unsigned short*,
-                                            int);
+                                            const unsigned short*,
+                                            unsigned short*, int);
 extern "C" __global__ void drla_conv_fwd_l3(const unsigned short*,
                                             const unsigned short*,
-                                            const float*, unsigned short*,
-                                            int);
+                                            const unsigned short*,
+                                            unsigned short*, int);
 extern "C" __global__ void drla_relu_mask_bwd(const unsigned short*,
                                               const unsigned short*,
                                               unsigned short*, float*,
@@ -296,22 +296,30 @@ torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
   return out;
 }
 
-std::tuple<torch::Tensor, torch::Tensor> relu_mask_bwd(torch::Tensor dy,
-                                                       torch::Tensor y,
-                                                       int64_t CO) {
+// persistent [16][64] bias-grad partial slots shared by relu_mask_bwd
+// (atomic producers) and wgrad_finalize (sum + re-zero consumer); the conv
+// backward always runs them as a mask -> wgrad pair on one stream.
+static torch::Tensor& dbias_slot_buf(const torch::Tensor& like) {
+  static torch::Tensor buf;
+  if (!buf.defined()) {
+    buf = torch::zeros({16 * 64}, like.options().dtype(torch::kFloat));
+  }
+  return buf;
+}
+
+torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y, int64_t CO) {
   check_gpu_contig(dy, "dy");
   check_gpu_contig(y, "y");
   auto out = torch::empty_like(dy);
-  auto dbias = torch::zeros({CO}, dy.options().dtype(torch::kFloat));
+  auto& slots = dbias_slot_buf(dy);
   const long long n = dy.numel();
   TORCH_CHECK(n % 8 == 0, "relu_mask_bwd wants numel % 8 == 0");
-  // cap blocks: each extra block costs CO atomicAdds on dbias
   int grid = drla_grid(n / 8);
   if (grid > 640) grid = 640;
   hipLaunchKernelGGL(drla_relu_mask_bwd, dim3(grid), dim3(DRLA_BLOCK),
                      0, cur_stream(), u16p(dy), u16p(y), u16pm(out),
-                     du16p(bias), n, (int)CO);
-  return {out, dbias};
+                     slots.data_ptr<float>(), n, (int)CO);
+  return out;
 }
 
 std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
