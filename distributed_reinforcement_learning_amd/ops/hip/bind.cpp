@@ -88,15 +88,19 @@ extern "C" __global__ void drla_f32_to_bf16_kernel(const float*,
 extern "C" __global__ void drla_f32_to_bf16_zero_kernel(float*,
                                                         unsigned short*,
                                                         long long);
+extern "C" __global__ void drla_colsum_bf16(const unsigned short*,
+                                            unsigned short*, long long,
+                                            int);
 extern "C" __global__ void drla_vtrace_scan(const float*, const float*,
                                             const float*, float*, int, int);
 extern "C" __global__ void drla_vtrace_loss_fwd(
     const unsigned short*, const float*, const float*, const float*,
-    const int*, const float*, const float*, float*, float*, float*, float*,
-    int, int, int);
+    const int*, const float*, const unsigned char*, float, int, float,
+    float, float*, float*, float*, float*, int, int, int);
 extern "C" __global__ void drla_vtrace_loss_bwd(
     const float*, const float*, const float*, const float*, const int*,
-    const float*, unsigned short*, float*, float*, int, int, int);
+    const float*, int, float, float, unsigned short*, float*, float*, int,
+    int, int);
 extern "C" __global__ void drla_lstm_tail_fwd(const float*, const float*,
                                               float*, float*, float*, float,
                                               long long, int);
@@ -513,10 +517,12 @@ torch::Tensor embed_bwd(torch::Tensor indices, torch::Tensor grad_out,
 
 std::vector<torch::Tensor> vtrace_loss_fwd(
     torch::Tensor logits, torch::Tensor value, torch::Tensor mu,
-    torch::Tensor actions, torch::Tensor rewards, torch::Tensor discounts) {
-  for (auto* t : {&logits, &value, &mu, &actions, &rewards, &discounts})
+    torch::Tensor actions, torch::Tensor rewards, torch::Tensor done,
+    double gamma, int64_t clip_mode, double c_bl, double c_ent) {
+  for (auto* t : {&logits, &value, &mu, &actions, &rewards, &done})
     check_gpu_contig(*t, "vtrace_loss input");
   TORCH_CHECK(actions.scalar_type() == torch::kInt, "actions must be int32");
+  TORCH_CHECK(done.scalar_type() == torch::kBool, "done must be bool");
   const int B = logits.size(0), T = logits.size(1), A = logits.size(2);
   TORCH_CHECK(A <= 64, "num_action cap is 64");
   const bool bf16 = logits.scalar_type() == torch::kBFloat16;
@@ -525,23 +531,26 @@ std::vector<torch::Tensor> vtrace_loss_fwd(
   auto p_stash = torch::empty({B, T, A}, fopt);
   auto vs_stash = torch::empty({B, T - 2}, fopt);
   auto adv_stash = torch::empty({B, T - 2}, fopt);
-  auto losses = torch::zeros({3}, fopt);  // blocks atomicAdd into it
+  auto losses = torch::zeros({4}, fopt);  // blocks atomicAdd into it
   hipLaunchKernelGGL(
       drla_vtrace_loss_fwd, dim3(B), dim3(256), 0, cur_stream(),
       bf16 ? reinterpret_cast<const unsigned short*>(logits.data_ptr())
            : nullptr,
       bf16 ? nullptr : logits.data_ptr<float>(), value.data_ptr<float>(),
       mu.data_ptr<float>(), actions.data_ptr<int>(),
-      rewards.data_ptr<float>(), discounts.data_ptr<float>(),
-      p_stash.data_ptr<float>(), vs_stash.data_ptr<float>(),
-      adv_stash.data_ptr<float>(), losses.data_ptr<float>(), B, T, A);
+      rewards.data_ptr<float>(),
+      reinterpret_cast<const unsigned char*>(done.data_ptr<bool>()),
+      static_cast<float>(gamma), (int)clip_mode, static_cast<float>(c_bl),
+      static_cast<float>(c_ent), p_stash.data_ptr<float>(),
+      vs_stash.data_ptr<float>(), adv_stash.data_ptr<float>(),
+      losses.data_ptr<float>(), B, T, A);
   return {losses, p_stash, vs_stash, adv_stash};
 }
 
 std::tuple<torch::Tensor, torch::Tensor> vtrace_loss_bwd(
     torch::Tensor p_stash, torch::Tensor vs_stash, torch::Tensor adv_stash,
     torch::Tensor value, torch::Tensor actions, torch::Tensor grad3,
-    bool want_bf16) {
+    bool from_total, double c_bl, double c_ent, bool want_bf16) {
   for (auto* t : {&p_stash, &vs_stash, &adv_stash, &value, &actions, &grad3})
     check_gpu_contig(*t, "vtrace_loss bwd input");
   const int B = p_stash.size(0), T = p_stash.size(1), A = p_stash.size(2);
@@ -554,6 +563,8 @@ std::tuple<torch::Tensor, torch::Tensor> vtrace_loss_bwd(
       cur_stream(), p_stash.data_ptr<float>(), vs_stash.data_ptr<float>(),
       adv_stash.data_ptr<float>(), value.data_ptr<float>(),
       actions.data_ptr<int>(), grad3.data_ptr<float>(),
+      from_total ? 1 : 0, static_cast<float>(c_bl),
+      static_cast<float>(c_ent),
       want_bf16 ? reinterpret_cast<unsigned short*>(dlogits.data_ptr())
                 : nullptr,
       want_bf16 ? nullptr : dlogits.data_ptr<float>(),
@@ -799,6 +810,17 @@ void grad_gather(torch::Tensor srcs, torch::Tensor offs,
       u16pm(dst), (int)srcs.numel(), chunks, nw, nw_n);
 }
 
+torch::Tensor colsum_bf16(torch::Tensor x) {
+  check_gpu_contig(x, "x");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.dim() == 2);
+  const long long N = x.size(0);
+  const int C = x.size(1);
+  auto out = torch::empty({C}, x.options());
+  hipLaunchKernelGGL(drla_colsum_bf16, dim3((C + 63) / 64), dim3(256), 0,
+                     cur_stream(), u16p(x), u16pm(out), N, C);
+  return out;
+}
+
 torch::Tensor sq_norm(torch::Tensor x) {
   check_gpu_contig(x, "x");
   TORCH_CHECK(x.scalar_type() == torch::kFloat, "sq_norm wants float32");
@@ -972,6 +994,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_seq_fwd", &lstm_seq_fwd,
         "whole no-grad LSTM unroll in one kernel (K3 seq / burn-in)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");
+  m.def("colsum_bf16", &colsum_bf16,
+        "bf16 column sum (bias grads) without torch's reduce kernel");
   m.def("grad_gather", &grad_gather,
         "one-kernel scattered-grad -> flat bucket pack (K12b)");
   m.def("rmsprop_step", &rmsprop_step,

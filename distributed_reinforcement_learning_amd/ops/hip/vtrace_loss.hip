@@ -27,7 +27,13 @@ __device__ __forceinline__ float vt_ld(const bf16raw* p, long long i) {
 
 #define VT_MAX_T 128  // trajectory-length cap for the LDS scan buffers
 
-// grid = B blocks of 256; losses[3] must be ZERO on entry (atomicAdd).
+// grid = B blocks of 256; losses[4] must be ZERO on entry (atomicAdd).
+// Reward clipping (clip_mode: 0 = abs_one, 1 = soft_asymmetric, 2 = none,
+// reference agent/impala.py:45-49) and discounts = (1-done)*gamma are
+// computed IN-KERNEL from the raw reward/done rows — the python-side
+// clamp/not/cast/mul prep kernels disappear. losses[3] accumulates the
+// combined total = pi + c_bl*baseline + c_ent*entropy so the loss-combine
+// arithmetic costs nothing host-side either.
 extern "C" __global__ __launch_bounds__(256)
 void drla_vtrace_loss_fwd(
     const bf16raw* __restrict__ logits_bf16,   // [B,T,A] (nullable)
@@ -35,15 +41,18 @@ void drla_vtrace_loss_fwd(
     const float* __restrict__ value,           // [B,T]
     const float* __restrict__ mu,              // [B,T,A]
     const int* __restrict__ actions,           // [B,T]
-    const float* __restrict__ rewards,         // [B,T] (already clipped)
-    const float* __restrict__ discounts,       // [B,T]
+    const float* __restrict__ rewards,         // [B,T] RAW
+    const unsigned char* __restrict__ done,    // [B,T] bool
+    float gamma, int clip_mode, float c_bl, float c_ent,
     float* __restrict__ p_stash,               // [B,T,A] softmax out
     float* __restrict__ vs_stash,              // [B,T-2]
     float* __restrict__ adv_stash,             // [B,T-2]
-    float* __restrict__ losses,                // [3]: pi, baseline, entropy
+    float* __restrict__ losses,                // [4]: pi, base, ent, total
     int B, int T, int A) {
   __shared__ float rho[VT_MAX_T];
   __shared__ float vsm[VT_MAX_T];
+  __shared__ float rc[VT_MAX_T];
+  __shared__ float gc[VT_MAX_T];
   __shared__ float red[3 * 4];
 
   const int b = blockIdx.x;
@@ -53,8 +62,22 @@ void drla_vtrace_loss_fwd(
 
   // phase A: softmax + rho, one thread per t (multi-pass over the
   // L2-resident row — a local A-array would be runtime-indexed and spill,
-  // guide §5.4 rule 20)
+  // guide §5.4 rule 20); reward clip + discount computed alongside
   for (int t = tid; t < T; t += blockDim.x) {
+    {
+      const float rr = rewards[row + t];
+      float rcl;
+      if (clip_mode == 0) {
+        rcl = fminf(1.0f, fmaxf(-1.0f, rr));
+      } else if (clip_mode == 1) {
+        const float sq = tanhf(rr * 0.2f);
+        rcl = ((rr < 0.0f) ? 0.3f * sq : sq) * 5.0f;
+      } else {
+        rcl = rr;
+      }
+      rc[t] = rcl;
+      gc[t] = done[row + t] ? 0.0f : gamma;
+    }
     const long long base = (row + t) * (long long)A;
     float mx = -1e30f;
     for (int k = 0; k < A; ++k) {
@@ -82,8 +105,8 @@ void drla_vtrace_loss_fwd(
   // phase B: the two reverse scans, serial in T (thread 0)
   if (tid == 0) {
     const float* v = value + row;
-    const float* r = rewards + row;
-    const float* g = discounts + row;
+    const float* r = rc;
+    const float* g = gc;
     // middle-window scan -> vs_plus_1 (bootstrap v[T-1])
     float acc = 0.0f;
     for (int t = Tp - 1; t >= 0; --t) {
@@ -144,21 +167,26 @@ void drla_vtrace_loss_fwd(
     atomicAdd(&losses[0], s0);
     atomicAdd(&losses[1], s1);
     atomicAdd(&losses[2], s2);
+    atomicAdd(&losses[3], s0 + c_bl * s1 + c_ent * s2);
   }
 }
 
 // one thread per (b,t): writes the full A-row of dlogits + dvalue.
 // gpi/gb/ge are the upstream gradients of the three loss outputs
 // (1, baseline_coef, entropy_coef when total.backward() is called).
+// from_total != 0: grad3 is the 1-element upstream grad of the COMBINED
+// total and the per-loss grads are (gt, gt*c_bl, gt*c_ent) — no host-side
+// stack/scale kernels for the standard total.backward() path.
 extern "C" __global__ void drla_vtrace_loss_bwd(
     const float* __restrict__ p_stash, const float* __restrict__ vs_stash,
     const float* __restrict__ adv_stash, const float* __restrict__ value,
     const int* __restrict__ actions, const float* __restrict__ grad3,
+    int from_total, float c_bl, float c_ent,
     bf16raw* __restrict__ dlogits_bf16, float* __restrict__ dlogits_f32,
     float* __restrict__ dvalue, int B, int T, int A) {
   const float gpi = grad3[0];
-  const float gb = grad3[1];
-  const float ge = grad3[2];
+  const float gb = from_total ? gpi * c_bl : grad3[1];
+  const float ge = from_total ? gpi * c_ent : grad3[2];
   const int Tp = T - 2;
   long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
   const long long total = (long long)B * T;

@@ -43,18 +43,24 @@ def test_fused_vtrace_loss_matches_torch(dtype):
     value_g = value.detach().clone().requires_grad_(True)
     mu = torch.softmax(torch.randn(B, T, A, device="cuda"), -1)
     actions = torch.randint(0, A, (B, T), device="cuda")
-    rewards = torch.randn(B, T, device="cuda").clamp(-1, 1)
-    discounts = (torch.rand(B, T, device="cuda") > 0.1).float() * 0.99
+    rewards = torch.randn(B, T, device="cuda") * 2.0
+    done = torch.rand(B, T, device="cuda") < 0.1
+    discounts = (~done).float() * 0.99
 
-    pi_g, bl_g, ent_g = fused_vtrace_loss(logits_g, value_g, mu, actions,
-                                          rewards, discounts)
-    total_g = pi_g + c_b * bl_g + c_e * ent_g
+    # the fused path clips in-kernel (abs_one) and folds the coef-combined
+    # total; backward through total exercises the from_total fast path
+    pi_g, bl_g, ent_g, total_g = fused_vtrace_loss(
+        logits_g, value_g, mu, actions, rewards, done, 0.99, "abs_one",
+        c_b, c_e)
     total_g.backward()
+    rewards = rewards.clamp(-1, 1)
 
     pi_f, bl_f, ent_f = _torch_composed(logits_f, value_f, mu, actions,
                                         rewards, discounts, c_b, c_e)
     total_f = pi_f + c_b * bl_f + c_e * ent_f
     total_f.backward()
+    assert float(total_g) == pytest.approx(float(total_f), rel=5e-3,
+                                           abs=2e-2)
 
     rtol = 2e-4 if dtype == torch.float32 else 2e-3
     assert float(pi_g) == pytest.approx(float(pi_f), rel=rtol, abs=1e-2)
