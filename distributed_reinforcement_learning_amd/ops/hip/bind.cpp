@@ -88,9 +88,7 @@ extern "C" __global__ void drla_f32_to_bf16_kernel(const float*,
 extern "C" __global__ void drla_f32_to_bf16_zero_kernel(float*,
                                                         unsigned short*,
                                                         long long);
-extern "C" __global__ void drla_colsum_bf16(const unsigned short*,
-                                            unsigned short*, long long,
-                                            int);
+
 extern "C" __global__ void drla_vtrace_scan(const float*, const float*,
                                             const float*, float*, int, int);
 extern "C" __global__ void drla_vtrace_loss_fwd(
@@ -810,17 +808,6 @@ void grad_gather(torch::Tensor srcs, torch::Tensor offs,
       u16pm(dst), (int)srcs.numel(), chunks, nw, nw_n);
 }
 
-torch::Tensor colsum_bf16(torch::Tensor x) {
-  check_gpu_contig(x, "x");
-  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.dim() == 2);
-  const long long N = x.size(0);
-  const int C = x.size(1);
-  auto out = torch::empty({C}, x.options());
-  hipLaunchKernelGGL(drla_colsum_bf16, dim3((C + 63) / 64), dim3(256), 0,
-                     cur_stream(), u16p(x), u16pm(out), N, C);
-  return out;
-}
-
 torch::Tensor sq_norm(torch::Tensor x) {
   check_gpu_contig(x, "x");
   TORCH_CHECK(x.scalar_type() == torch::kFloat, "sq_norm wants float32");
@@ -994,8 +981,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_seq_fwd", &lstm_seq_fwd,
         "whole no-grad LSTM unroll in one kernel (K3 seq / burn-in)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");
-  m.def("colsum_bf16", &colsum_bf16,
-        "bf16 column sum (bias grads) without torch's reduce kernel");
   m.def("grad_gather", &grad_gather,
         "one-kernel scattered-grad -> flat bucket pack (K12b)");
   m.def("rmsprop_step", &rmsprop_step,

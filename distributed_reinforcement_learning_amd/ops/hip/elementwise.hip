@@ -100,29 +100,3 @@ extern "C" __global__ void drla_f32_to_bf16_zero_kernel(
     src[i] = 0.0f;
   }
 }
-
-// column sum of a bf16 [N,C] row-major matrix -> bf16 [C] (bias grads for
-// the LSTM gate GEMM and the action-embedding layers; torch's bf16 column
-// reduce_kernel takes ~9 us at [640,1024], this is load-bound ~3 us).
-// blockDim 256 = 64 cols x 4 row-groups; grid.x = ceil(C/64).
-extern "C" __global__ void drla_colsum_bf16(
-    const unsigned short* __restrict__ x, unsigned short* __restrict__ out,
-    long long N, int C) {
-  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  const int rg = threadIdx.x >> 6;
-  float acc = 0.0f;
-  if (c < C) {
-    for (long long i = rg; i < N; i += 4) {
-      const unsigned short u = x[i * C + c];
-      acc += __uint_as_float(((unsigned int)u) << 16);
-    }
-  }
-  __shared__ float red[4][64];
-  red[rg][threadIdx.x & 63] = acc;
-  __syncthreads();
-  if (rg == 0 && c < C) {
-    const float s = red[0][threadIdx.x] + red[1][threadIdx.x] +
-                    red[2][threadIdx.x] + red[3][threadIdx.x];
-    out[c] = drla_f32_to_bf16(s);
-  }
-}
