@@ -378,8 +378,14 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
   // the MFMA fragments read them TRANSPOSED with scalar ds_read_u16 —
   // cheaper than the division-heavy transposed-scatter staging this
   // replaced (332 us -> see profiles).
-  __shared__ bf16raw Am[BKM][BKK + PAD];  // A image: [m][k]
-  __shared__ bf16raw Bm[BKM][CO + PAD];   // dY image: [m][co]
+  // A image also TRANSPOSED ([k][m]): the u8 staging is per-element
+  // anyway (same write count), and a_frag turns into one ds_read_b128
+  __shared__ bf16raw Am[BKK][BKM + PAD];  // A image: [k][m]
+  // dY staged TRANSPOSED: b_frag wants B_op[m][co] with m varying inside
+  // the fragment, and every element is read by ALL 4 waves — row-major Bm
+  // cost 32 scalar ds_read_u16 per 8 elements (PMC r16: wgrads VALU-bound
+  // at ~31% MFMA util). BmT makes that one ds_read_b128 per wave.
+  __shared__ bf16raw BmT[CO][BKM + PAD];  // dY image: [co][m]
 
   const int tid = threadIdx.x;
   const int wave = tid / 64;
@@ -421,10 +427,10 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
                   in + (base + (long long)kh * WI + kw) * CI);
               const float s = 1.0f / 255.0f;
               const int o = a_k + seg * 8 + t * 4;
-              Am[lm][o + 0] = drla_f32_to_bf16(v.x * s);
-              Am[lm][o + 1] = drla_f32_to_bf16(v.y * s);
-              Am[lm][o + 2] = drla_f32_to_bf16(v.z * s);
-              Am[lm][o + 3] = drla_f32_to_bf16(v.w * s);
+              Am[o + 0][lm] = drla_f32_to_bf16(v.x * s);
+              Am[o + 1][lm] = drla_f32_to_bf16(v.y * s);
+              Am[o + 2][lm] = drla_f32_to_bf16(v.z * s);
+              Am[o + 3][lm] = drla_f32_to_bf16(v.w * s);
             }
           } else if constexpr (CI == 1) {
             for (int t = 0; t < 8; ++t) {
@@ -433,7 +439,7 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
               const int kw = k2 - kh * KW;
               const unsigned char v =
                   (k2 < K) ? in[base + (long long)kh * WI + kw] : 0;
-              Am[lm][a_k + seg * 8 + t] =
+              Am[a_k + seg * 8 + t][lm] =
                   drla_f32_to_bf16(v * (1.0f / 255.0f));
             }
           } else {
@@ -441,19 +447,24 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
             const int kwci = kk - kh * KW * CI;
             const int kw = kwci / CI;
             const int ci = kwci - kw * CI;
-            const bf16raw* src = reinterpret_cast<const bf16raw*>(in) +
-                                 (base + (long long)kh * WI + kw) * CI + ci;
-            *reinterpret_cast<uint4*>(&Am[lm][a_k + seg * 8]) =
-                (kk < K) ? *reinterpret_cast<const uint4*>(src)
-                         : uint4{0, 0, 0, 0};
+            bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+            if (kk < K) {
+              v = *reinterpret_cast<const bf16x8*>(
+                  reinterpret_cast<const bf16raw*>(in) +
+                  (base + (long long)kh * WI + kw) * CI + ci);
+            }
+            const int o = a_k + seg * 8;
+            Am[o + 0][lm] = v[0]; Am[o + 1][lm] = v[1];
+            Am[o + 2][lm] = v[2]; Am[o + 3][lm] = v[3];
+            Am[o + 4][lm] = v[4]; Am[o + 5][lm] = v[5];
+            Am[o + 6][lm] = v[6]; Am[o + 7][lm] = v[7];
           }
         }
       } else {
-        *reinterpret_cast<uint4*>(&Am[lm][a_k]) = uint4{0, 0, 0, 0};
-        *reinterpret_cast<uint4*>(&Am[lm][a_k + 8]) = uint4{0, 0, 0, 0};
+        for (int t = 0; t < 16; ++t) Am[a_k + t][lm] = 0;
       }
     }
-    // ---- stage Bm[m][co]: contiguous dY rows, both m-halves ----
+    // ---- stage BmT[co][m]: vec global dY rows, scalar transposed writes
     {
       constexpr int CO_PER_T = (CO * 32) / 256;  // 4 or 8 per half
       for (int half = 0; half < 2; ++half) {
@@ -461,15 +472,25 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
         const int m = m0 + lm;
         const int co0 = (tid % (CO / CO_PER_T)) * CO_PER_T;
         if constexpr (CO_PER_T == 8) {
-          *reinterpret_cast<uint4*>(&Bm[lm][co0]) =
-              (m < m_end) ? *reinterpret_cast<const uint4*>(
-                                dy + (long long)m * CO + co0)
-                          : uint4{0, 0, 0, 0};
+          bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+          if (m < m_end) {
+            v = *reinterpret_cast<const bf16x8*>(dy + (long long)m * CO +
+                                                 co0);
+          }
+          BmT[co0 + 0][lm] = v[0]; BmT[co0 + 1][lm] = v[1];
+          BmT[co0 + 2][lm] = v[2]; BmT[co0 + 3][lm] = v[3];
+          BmT[co0 + 4][lm] = v[4]; BmT[co0 + 5][lm] = v[5];
+          BmT[co0 + 6][lm] = v[6]; BmT[co0 + 7][lm] = v[7];
         } else {
-          *reinterpret_cast<uint2*>(&Bm[lm][co0]) =
-              (m < m_end) ? *reinterpret_cast<const uint2*>(
-                                dy + (long long)m * CO + co0)
-                          : uint2{0, 0};
+          uint2 v = {0, 0};
+          if (m < m_end) {
+            v = *reinterpret_cast<const uint2*>(dy + (long long)m * CO +
+                                                co0);
+          }
+          BmT[co0 + 0][lm] = (bf16raw)(v.x & 0xFFFF);
+          BmT[co0 + 1][lm] = (bf16raw)(v.x >> 16);
+          BmT[co0 + 2][lm] = (bf16raw)(v.y & 0xFFFF);
+          BmT[co0 + 3][lm] = (bf16raw)(v.y >> 16);
         }
       }
     }
@@ -478,19 +499,12 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
     // wave w owns k-rows [w*16, w*16+16); two MFMA k-steps per chunk.
     // A_op[k][m] = Am[m][k], B_op[m][co] = Bm[m][co] (transposed reads)
     for (int kk = 0; kk < BKM; kk += 32) {
-      bf16x8 a_frag;
-      {
-        const int k = wave * 16 + (lane & 15);
-        for (int e = 0; e < 8; ++e) {
-          a_frag[e] = (short)Am[kk + (lane >> 4) * 8 + e][k];
-        }
-      }
+      const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
+          &Am[wave * 16 + (lane & 15)][kk + (lane >> 4) * 8]);
       for (int ni = 0; ni < NFRAG; ++ni) {
-        bf16x8 b_frag;
         const int co = ni * 16 + (lane & 15);
-        for (int e = 0; e < 8; ++e) {
-          b_frag[e] = (short)Bm[kk + (lane >> 4) * 8 + e][co];
-        }
+        const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            &BmT[co][kk + (lane >> 4) * 8]);
         acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
                                                           acc[ni], 0, 0, 0);
       }
