@@ -340,7 +340,13 @@ std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
         torch::zeros({K, cfg.co}, dy.options().dtype(torch::kFloat));
   }
   auto scratch = scratch_cache[layer];
-  const int split = (layer <= 1) ? 256 : 128;
+  // runtime-tunable M-split (DRLA_WGRAD_SPLIT sweeps it; defaults from the
+  // r28 on-box sweep)
+  static int split_env = [] {
+    const char* e = getenv("DRLA_WGRAD_SPLIT");
+    return e ? atoi(e) : 0;
+  }();
+  const int split = split_env ? split_env : ((layer <= 1) ? 256 : 128);
   dim3 grid((K + 63) / 64, split);
   switch (layer) {
     case 0:

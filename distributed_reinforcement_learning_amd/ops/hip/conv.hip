@@ -362,7 +362,7 @@ extern "C" __global__ void drla_wgrad_finalize(
 // ---------------------------------------------------------------------------
 
 template <typename IN_T, int CI, int CO, int KH, int KW, int STRIDE, int HI,
-          int WI, int HO, int WO, int SPLIT_M>
+          int WI, int HO, int WO>
 __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
                                 const bf16raw* __restrict__ dy,  // [M][CO]
                                 float* __restrict__ scratch,     // [K][CO]
@@ -395,7 +395,9 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
   f32x4 acc[NFRAG];
   for (int ni = 0; ni < NFRAG; ++ni) acc[ni] = {0.f, 0.f, 0.f, 0.f};
 
-  const int m_per_split = (M + SPLIT_M - 1) / SPLIT_M;
+  // M-split comes from gridDim.y (runtime-tunable: more splits hide the
+  // per-chunk latency chain, fewer splits cut the atomic partial count)
+  const int m_per_split = (M + gridDim.y - 1) / gridDim.y;
   const int m_begin = blockIdx.y * m_per_split;
   const int m_end = min(M, m_begin + m_per_split);
 
@@ -523,22 +525,22 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
 
 extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l1(
     const unsigned char* in, const bf16raw* dy, float* scratch, int batch) {
-  conv_wgrad_impl<unsigned char, 4, 32, 8, 8, 4, 84, 84, 20, 20, 256>(
+  conv_wgrad_impl<unsigned char, 4, 32, 8, 8, 4, 84, 84, 20, 20>(
       in, dy, scratch, batch);
 }
 extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l1_c1(
     const unsigned char* in, const bf16raw* dy, float* scratch, int batch) {
-  conv_wgrad_impl<unsigned char, 1, 32, 8, 8, 4, 84, 84, 20, 20, 256>(
+  conv_wgrad_impl<unsigned char, 1, 32, 8, 8, 4, 84, 84, 20, 20>(
       in, dy, scratch, batch);
 }
 extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l2(
     const bf16raw* in, const bf16raw* dy, float* scratch, int batch) {
-  conv_wgrad_impl<bf16raw, 32, 64, 4, 4, 2, 20, 20, 9, 9, 128>(in, dy,
+  conv_wgrad_impl<bf16raw, 32, 64, 4, 4, 2, 20, 20, 9, 9>(in, dy,
                                                               scratch, batch);
 }
 extern "C" __global__ __launch_bounds__(256) void drla_conv_wgrad_l3(
     const bf16raw* in, const bf16raw* dy, float* scratch, int batch) {
-  conv_wgrad_impl<bf16raw, 64, 64, 3, 3, 1, 9, 9, 7, 7, 128>(in, dy, scratch,
+  conv_wgrad_impl<bf16raw, 64, 64, 3, 3, 1, 9, 9, 7, 7>(in, dy, scratch,
                                                             batch);
 }
 
