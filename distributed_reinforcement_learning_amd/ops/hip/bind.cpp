@@ -46,8 +46,9 @@ extern "C" __global__ void drla_relu_mask_bwd(const unsigned short*,
                                               const unsigned short*,
                                               unsigned short*, float*,
                                               long long, int);
-extern "C" __global__ void drla_wgrad_finalize(float*, unsigned short*,
-                                               int, int, float*,
+extern "C" __global__ void drla_wgrad_finalize(const float*,
+                                               unsigned short*, int, int,
+                                               int, float*,
                                                unsigned short*);
 extern "C" __global__ void drla_conv_wgrad_l1(const unsigned char*,
                                               const unsigned short*, float*,
@@ -332,21 +333,22 @@ std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
   const auto& cfg = convcfg::L[layer];
   const int batch = in.size(0);
   const int K = cfg.kh * cfg.kw * cfg.ci;
-  // persistent zero-between-calls scratch per layer: drla_wgrad_finalize
-  // re-zeroes it as it reads, so no per-step fill kernel
-  static torch::Tensor scratch_cache[4];
-  if (!scratch_cache[layer].defined()) {
-    scratch_cache[layer] =
-        torch::zeros({K, cfg.co}, dy.options().dtype(torch::kFloat));
-  }
-  auto scratch = scratch_cache[layer];
   // runtime-tunable M-split (DRLA_WGRAD_SPLIT sweeps it; defaults from the
-  // r28 on-box sweep)
+  // on-box sweeps — atomic-free slab stores make high splits cheap)
   static int split_env = [] {
     const char* e = getenv("DRLA_WGRAD_SPLIT");
     return e ? atoi(e) : 0;
   }();
-  const int split = split_env ? split_env : ((layer <= 1) ? 256 : 128);
+  const int split = split_env ? split_env : ((layer <= 1) ? 512 : 256);
+  // persistent per-layer slab scratch [split][K][CO] (fully rewritten
+  // every call — no zeroing)
+  static torch::Tensor scratch_cache[4];
+  if (!scratch_cache[layer].defined() ||
+      scratch_cache[layer].numel() < (long long)split * K * cfg.co) {
+    scratch_cache[layer] = torch::empty(
+        {(long long)split * K * cfg.co}, dy.options().dtype(torch::kFloat));
+  }
+  auto scratch = scratch_cache[layer];
   dim3 grid((K + 63) / 64, split);
   switch (layer) {
     case 0:
@@ -379,7 +381,7 @@ std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
                      dim3(drla_grid((long long)K * cfg.co)),
                      dim3(DRLA_BLOCK), 0, cur_stream(),
                      scratch.data_ptr<float>(), u16pm(dw), K, cfg.co,
-                     slots.data_ptr<float>(), u16pm(dbias));
+                     split, slots.data_ptr<float>(), u16pm(dbias));
   return {dw, dbias};
 }
 
