@@ -46,9 +46,8 @@ extern "C" __global__ void drla_relu_mask_bwd(const unsigned short*,
                                               const unsigned short*,
                                               unsigned short*, float*,
                                               long long, int);
-extern "C" __global__ void drla_wgrad_finalize(const float*,
-                                               unsigned short*, int, int,
-                                               int, float*,
+extern "C" __global__ void drla_wgrad_finalize(float*, unsigned short*,
+                                               int, int, float*,
                                                unsigned short*);
 extern "C" __global__ void drla_conv_wgrad_l1(const unsigned char*,
                                               const unsigned short*, float*,
@@ -339,14 +338,12 @@ std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
     const char* e = getenv("DRLA_WGRAD_SPLIT");
     return e ? atoi(e) : 0;
   }();
-  const int split = split_env ? split_env : ((layer <= 1) ? 512 : 256);
-  // persistent per-layer slab scratch [split][K][CO] (fully rewritten
-  // every call — no zeroing)
+  const int split = split_env ? split_env : ((layer <= 1) ? 512 : 128);
+  // persistent zero-between-calls scratch (finalize re-zeroes on read)
   static torch::Tensor scratch_cache[4];
-  if (!scratch_cache[layer].defined() ||
-      scratch_cache[layer].numel() < (long long)split * K * cfg.co) {
-    scratch_cache[layer] = torch::empty(
-        {(long long)split * K * cfg.co}, dy.options().dtype(torch::kFloat));
+  if (!scratch_cache[layer].defined()) {
+    scratch_cache[layer] =
+        torch::zeros({K, cfg.co}, dy.options().dtype(torch::kFloat));
   }
   auto scratch = scratch_cache[layer];
   dim3 grid((K + 63) / 64, split);
@@ -381,7 +378,7 @@ std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
                      dim3(drla_grid((long long)K * cfg.co)),
                      dim3(DRLA_BLOCK), 0, cur_stream(),
                      scratch.data_ptr<float>(), u16pm(dw), K, cfg.co,
-                     split, slots.data_ptr<float>(), u16pm(dbias));
+                     slots.data_ptr<float>(), u16pm(dbias));
   return {dw, dbias};
 }
 

@@ -325,12 +325,12 @@ extern "C" __global__ void drla_bias_grad(
   }
 }
 
-// S-deep sum + cast + transpose of the f32 [S][K][CO] wgrad slab scratch
-// into bf16 [CO][K] (the channels_last weight-grad layout). Every slab
-// element is freshly written each call, so no zeroing anywhere.
+// cast + transpose the f32 [K][CO] wgrad scratch into bf16 [CO][K]
+// (the channels_last weight-grad layout), re-zeroing the persistent
+// scratch as it reads (zero-between-calls invariant, no fill kernel).
 extern "C" __global__ void drla_wgrad_finalize(
-    const float* __restrict__ scratch, bf16raw* __restrict__ dw, int K,
-    int CO, int S, float* __restrict__ dbias_slots,
+    float* __restrict__ scratch, bf16raw* __restrict__ dw, int K,
+    int CO, float* __restrict__ dbias_slots,
     bf16raw* __restrict__ dbias) {
   // bias-grad epilogue: 16-way slot sum from relu_mask_bwd, bf16 out,
   // slots re-zeroed (persistent zero-between-calls buffer)
@@ -349,19 +349,18 @@ extern "C" __global__ void drla_wgrad_finalize(
     const int co = i / K;
     const int k = i - (long long)co * K;
     const long long si = (long long)k * CO + co;
-    float acc = 0.0f;
-    for (int sl = 0; sl < S; ++sl) acc += scratch[(long long)sl * total + si];
-    dw[i] = drla_f32_to_bf16(acc);
+    dw[i] = drla_f32_to_bf16(scratch[si]);
+    scratch[si] = 0.0f;
   }
 }
 
 // ---------------------------------------------------------------------------
 // wgrad: dW^T[k][co] = sum_m A_im2col[m][k] * dY[m][co]
-// grid: (K/64, S); each block owns 64 k-rows x CO cols and a slice of M,
-// and STORES its partial tile into its own slab of the f32 [S][K][CO]
-// scratch (no atomics: at the sweep-optimal high M-splits, K*CO*S atomic
-// adds serialized ~25 us/layer — plain slab stores are bandwidth-priced
-// and drla_wgrad_finalize does the S-deep sum).
+// grid: (K/64, S); each block owns 64 k-rows x CO cols and a slice of
+// M; partials atomicAdd into the f32 [K][CO] scratch. (A no-atomic
+// [S][K][CO] slab variant measured WORSE across the board — the S-deep
+// finalize reduce is the new bottleneck — so atomics stay; the real cost
+// is the per-chunk global-load latency chain, hence the prefetch below.)
 // ---------------------------------------------------------------------------
 
 template <typename IN_T, int CI, int CO, int KH, int KW, int STRIDE, int HI,
@@ -517,12 +516,11 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
     __syncthreads();
   }
 
-  float* slab = scratch + (long long)blockIdx.y * K * CO;
   for (int ni = 0; ni < NFRAG; ++ni) {
     const int co = ni * 16 + (lane & 15);
     for (int r = 0; r < 4; ++r) {
       const int k = k_row0 + wave * 16 + (lane >> 4) * 4 + r;
-      if (k < K) slab[(long long)k * CO + co] = acc[ni][r];
+      if (k < K) atomicAdd(&scratch[(long long)k * CO + co], acc[ni][r]);
     }
   }
 }
