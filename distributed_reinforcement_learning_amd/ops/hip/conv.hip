@@ -370,23 +370,19 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
                                 float* __restrict__ scratch,     // [K][CO]
                                 int batch) {
   constexpr int K = KH * KW * CI;
-  constexpr int BKM = 64;   // m-rows per chunk (2 MFMA k-steps; fewer barriers)
+  constexpr int BKM = 64;   // m-rows per chunk (2 MFMA k-steps)
   constexpr int BKK = 64;   // k-cols per block
   constexpr int PAD = 8;
   constexpr int NFRAG = CO / 16;
+  constexpr int CO_PER_T = (CO * 32) / 256;  // B staging: co per thread/half
   const int M = batch * HO * WO;
 
-  // forward-style images (vector global loads, row-contiguous LDS writes);
-  // the MFMA fragments read them TRANSPOSED with scalar ds_read_u16 —
-  // cheaper than the division-heavy transposed-scatter staging this
-  // replaced (332 us -> see profiles).
-  // A image also TRANSPOSED ([k][m]): the u8 staging is per-element
-  // anyway (same write count), and a_frag turns into one ds_read_b128
+  // Both images staged TRANSPOSED so the MFMA fragments are single
+  // ds_read_b128s, and SOFTWARE-PIPELINED: chunk m+1's global loads issue
+  // into registers while chunk m's MFMA runs — the per-chunk load-latency
+  // chain (the measured bottleneck: time scaled with M/split, not with
+  // atomics or LDS ops) overlaps compute.
   __shared__ bf16raw Am[BKK][BKM + PAD];  // A image: [k][m]
-  // dY staged TRANSPOSED: b_frag wants B_op[m][co] with m varying inside
-  // the fragment, and every element is read by ALL 4 waves — row-major Bm
-  // cost 32 scalar ds_read_u16 per 8 elements (PMC r16: wgrads VALU-bound
-  // at ~31% MFMA util). BmT makes that one ds_read_b128 per wave.
   __shared__ bf16raw BmT[CO][BKM + PAD];  // dY image: [co][m]
 
   const int tid = threadIdx.x;
@@ -397,111 +393,181 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
   f32x4 acc[NFRAG];
   for (int ni = 0; ni < NFRAG; ++ni) acc[ni] = {0.f, 0.f, 0.f, 0.f};
 
-  // M-split comes from gridDim.y (runtime-tunable: more splits hide the
-  // per-chunk latency chain, fewer splits cut the atomic partial count)
   const int m_per_split = (M + gridDim.y - 1) / gridDim.y;
   const int m_begin = blockIdx.y * m_per_split;
   const int m_end = min(M, m_begin + m_per_split);
 
-  // staging assignment (fwd-style): thread t -> m-rows {t>>2, 32+(t>>2)},
-  // k-chunk = (t&3)*16 of the 64x64 A image; decode coords once per chunk
+  // staging assignment: thread t -> A row t>>2 with k-chunk (t&3)*16; the
+  // k-position decode is m-invariant, so tap offsets precompute once
   const int a_m = tid >> 2;
   const int a_k = (tid & 3) * 16;
+  long long a_off0 = 0, a_off1 = 0, a_off2 = 0, a_off3 = 0;
+  long long a_offc[16];
+  if constexpr (CI == 4) {
+    const int kb = k_row0 + a_k;
+#define DRLA_WG_TAP(st, dst)                                   \
+    {                                                          \
+      const int k2 = kb + ((st) >> 1) * 8 + ((st) & 1) * 4;    \
+      const int kh = k2 / (KW * CI);                           \
+      const int kw = (k2 - kh * KW * CI) / CI;                 \
+      dst = (long long)kh * WI + kw;                           \
+    }
+    DRLA_WG_TAP(0, a_off0); DRLA_WG_TAP(1, a_off1);
+    DRLA_WG_TAP(2, a_off2); DRLA_WG_TAP(3, a_off3);
+#undef DRLA_WG_TAP
+  } else if constexpr (CI == 1) {
+#pragma unroll
+    for (int t = 0; t < 16; ++t) {
+      const int k2 = k_row0 + a_k + t;
+      const int kh = k2 / KW;
+      a_offc[t] = (k2 < K) ? (long long)kh * WI + (k2 - kh * KW) : -1;
+    }
+  } else {
+    const int kk0 = k_row0 + a_k;
+    const int kh0 = kk0 / (KW * CI);
+    const int kwci0 = kk0 - kh0 * KW * CI;
+    a_off0 = (kk0 < K)
+        ? ((long long)kh0 * WI + kwci0 / CI) * CI + (kwci0 % CI) : -1;
+    const int kk1 = kk0 + 8;
+    const int kh1 = kk1 / (KW * CI);
+    const int kwci1 = kk1 - kh1 * KW * CI;
+    a_off1 = (kk1 < K)
+        ? ((long long)kh1 * WI + kwci1 / CI) * CI + (kwci1 % CI) : -1;
+  }
+  const int b_lm0 = tid / (CO / CO_PER_T);
+  const int b_co0 = (tid % (CO / CO_PER_T)) * CO_PER_T;
 
-  for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
-    // ---- stage Am[m][a_k..a_k+16): one row, two 8-wide segments ----
-    {
-      const int lm = a_m;
-      const int m = m0 + lm;
-      if (m < m_end) {
-        const int n_idx = m / (HO * WO);
-        const int rem = m - n_idx * (HO * WO);
-        const int ho = rem / WO;
-        const int wo = rem - ho * WO;
-        const long long base =
-            ((long long)n_idx * HI + ho * STRIDE) * WI + wo * STRIDE;
-        for (int seg = 0; seg < 2; ++seg) {
-          const int kk = k_row0 + a_k + seg * 8;
-          if constexpr (CI == 4) {
-            for (int t = 0; t < 2; ++t) {
-              const int k2 = kk + t * 4;
-              const int kh = k2 / (KW * CI);
-              const int kw = (k2 - kh * KW * CI) / CI;
-              const uchar4 v = *reinterpret_cast<const uchar4*>(
-                  in + (base + (long long)kh * WI + kw) * CI);
-              const float s = 1.0f / 255.0f;
-              const int o = a_k + seg * 8 + t * 4;
-              Am[o + 0][lm] = drla_f32_to_bf16(v.x * s);
-              Am[o + 1][lm] = drla_f32_to_bf16(v.y * s);
-              Am[o + 2][lm] = drla_f32_to_bf16(v.z * s);
-              Am[o + 3][lm] = drla_f32_to_bf16(v.w * s);
-            }
-          } else if constexpr (CI == 1) {
-            for (int t = 0; t < 8; ++t) {
-              const int k2 = kk + t;
-              const int kh = k2 / KW;
-              const int kw = k2 - kh * KW;
-              const unsigned char v =
-                  (k2 < K) ? in[base + (long long)kh * WI + kw] : 0;
-              Am[a_k + seg * 8 + t][lm] =
-                  drla_f32_to_bf16(v * (1.0f / 255.0f));
-            }
-          } else {
-            const int kh = kk / (KW * CI);
-            const int kwci = kk - kh * KW * CI;
-            const int kw = kwci / CI;
-            const int ci = kwci - kw * CI;
-            bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-            if (kk < K) {
-              v = *reinterpret_cast<const bf16x8*>(
-                  reinterpret_cast<const bf16raw*>(in) +
-                  (base + (long long)kh * WI + kw) * CI + ci);
-            }
-            const int o = a_k + seg * 8;
-            Am[o + 0][lm] = v[0]; Am[o + 1][lm] = v[1];
-            Am[o + 2][lm] = v[2]; Am[o + 3][lm] = v[3];
-            Am[o + 4][lm] = v[4]; Am[o + 5][lm] = v[5];
-            Am[o + 6][lm] = v[6]; Am[o + 7][lm] = v[7];
-          }
+  // register prefetch state
+  uchar4 ra4_0, ra4_1, ra4_2, ra4_3;
+  unsigned char ra1[16];
+  bf16x8 rab0 = {0, 0, 0, 0, 0, 0, 0, 0}, rab1 = rab0;
+  bf16x8 rb8_0 = rab0, rb8_1 = rab0;
+  uint2 rb4_0 = {0, 0}, rb4_1 = {0, 0};
+  bool a_live = false, b_live0 = false, b_live1 = false;
+
+  auto load_chunk = [&](int m0) {
+    const int m = m0 + a_m;
+    a_live = (m < m_end);
+    if (a_live) {
+      const int n_idx = m / (HO * WO);
+      const int rem = m - n_idx * (HO * WO);
+      const int ho = rem / WO;
+      const int wo = rem - ho * WO;
+      const long long base =
+          ((long long)n_idx * HI + ho * STRIDE) * WI + wo * STRIDE;
+      if constexpr (CI == 4) {
+        ra4_0 = *reinterpret_cast<const uchar4*>(in + (base + a_off0) * CI);
+        ra4_1 = *reinterpret_cast<const uchar4*>(in + (base + a_off1) * CI);
+        ra4_2 = *reinterpret_cast<const uchar4*>(in + (base + a_off2) * CI);
+        ra4_3 = *reinterpret_cast<const uchar4*>(in + (base + a_off3) * CI);
+      } else if constexpr (CI == 1) {
+#pragma unroll
+        for (int t = 0; t < 16; ++t) {
+          ra1[t] = (a_offc[t] >= 0) ? in[base + a_offc[t]] : 0;
         }
       } else {
-        for (int t = 0; t < 16; ++t) Am[a_k + t][lm] = 0;
-      }
-    }
-    // ---- stage BmT[co][m]: vec global dY rows, scalar transposed writes
-    {
-      constexpr int CO_PER_T = (CO * 32) / 256;  // 4 or 8 per half
-      for (int half = 0; half < 2; ++half) {
-        const int lm = (tid / (CO / CO_PER_T)) + half * 32;
-        const int m = m0 + lm;
-        const int co0 = (tid % (CO / CO_PER_T)) * CO_PER_T;
-        if constexpr (CO_PER_T == 8) {
-          bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-          if (m < m_end) {
-            v = *reinterpret_cast<const bf16x8*>(dy + (long long)m * CO +
-                                                 co0);
-          }
-          BmT[co0 + 0][lm] = v[0]; BmT[co0 + 1][lm] = v[1];
-          BmT[co0 + 2][lm] = v[2]; BmT[co0 + 3][lm] = v[3];
-          BmT[co0 + 4][lm] = v[4]; BmT[co0 + 5][lm] = v[5];
-          BmT[co0 + 6][lm] = v[6]; BmT[co0 + 7][lm] = v[7];
-        } else {
-          uint2 v = {0, 0};
-          if (m < m_end) {
-            v = *reinterpret_cast<const uint2*>(dy + (long long)m * CO +
-                                                co0);
-          }
-          BmT[co0 + 0][lm] = (bf16raw)(v.x & 0xFFFF);
-          BmT[co0 + 1][lm] = (bf16raw)(v.x >> 16);
-          BmT[co0 + 2][lm] = (bf16raw)(v.y & 0xFFFF);
-          BmT[co0 + 3][lm] = (bf16raw)(v.y >> 16);
+        const bf16raw* inb = reinterpret_cast<const bf16raw*>(in);
+        if (a_off0 >= 0) {
+          rab0 = *reinterpret_cast<const bf16x8*>(inb + base * CI + a_off0);
+        }
+        if (a_off1 >= 0) {
+          rab1 = *reinterpret_cast<const bf16x8*>(inb + base * CI + a_off1);
         }
       }
     }
-    __syncthreads();
+    const int mb0 = m0 + b_lm0;
+    b_live0 = (mb0 < m_end);
+    if (b_live0) {
+      if constexpr (CO_PER_T == 8) {
+        rb8_0 = *reinterpret_cast<const bf16x8*>(dy + (long long)mb0 * CO +
+                                                 b_co0);
+      } else {
+        rb4_0 = *reinterpret_cast<const uint2*>(dy + (long long)mb0 * CO +
+                                                b_co0);
+      }
+    }
+    const int mb1 = m0 + b_lm0 + 32;
+    b_live1 = (mb1 < m_end);
+    if (b_live1) {
+      if constexpr (CO_PER_T == 8) {
+        rb8_1 = *reinterpret_cast<const bf16x8*>(dy + (long long)mb1 * CO +
+                                                 b_co0);
+      } else {
+        rb4_1 = *reinterpret_cast<const uint2*>(dy + (long long)mb1 * CO +
+                                                b_co0);
+      }
+    }
+  };
 
-    // wave w owns k-rows [w*16, w*16+16); two MFMA k-steps per chunk.
-    // A_op[k][m] = Am[m][k], B_op[m][co] = Bm[m][co] (transposed reads)
+  auto store_chunk = [&]() {
+    const int lm = a_m;
+    if (a_live) {
+      if constexpr (CI == 4) {
+        const float sc = 1.0f / 255.0f;
+#define DRLA_WG_PUT(st, v)                                   \
+        {                                                    \
+          const int o = a_k + ((st) >> 1) * 8 + ((st) & 1) * 4; \
+          Am[o + 0][lm] = drla_f32_to_bf16(v.x * sc);        \
+          Am[o + 1][lm] = drla_f32_to_bf16(v.y * sc);        \
+          Am[o + 2][lm] = drla_f32_to_bf16(v.z * sc);        \
+          Am[o + 3][lm] = drla_f32_to_bf16(v.w * sc);        \
+        }
+        DRLA_WG_PUT(0, ra4_0); DRLA_WG_PUT(1, ra4_1);
+        DRLA_WG_PUT(2, ra4_2); DRLA_WG_PUT(3, ra4_3);
+#undef DRLA_WG_PUT
+      } else if constexpr (CI == 1) {
+#pragma unroll
+        for (int t = 0; t < 16; ++t) {
+          Am[a_k + t][lm] = drla_f32_to_bf16(ra1[t] * (1.0f / 255.0f));
+        }
+      } else {
+        Am[a_k + 0][lm] = rab0[0]; Am[a_k + 1][lm] = rab0[1];
+        Am[a_k + 2][lm] = rab0[2]; Am[a_k + 3][lm] = rab0[3];
+        Am[a_k + 4][lm] = rab0[4]; Am[a_k + 5][lm] = rab0[5];
+        Am[a_k + 6][lm] = rab0[6]; Am[a_k + 7][lm] = rab0[7];
+        Am[a_k + 8][lm] = rab1[0]; Am[a_k + 9][lm] = rab1[1];
+        Am[a_k + 10][lm] = rab1[2]; Am[a_k + 11][lm] = rab1[3];
+        Am[a_k + 12][lm] = rab1[4]; Am[a_k + 13][lm] = rab1[5];
+        Am[a_k + 14][lm] = rab1[6]; Am[a_k + 15][lm] = rab1[7];
+      }
+    } else {
+#pragma unroll
+      for (int t = 0; t < 16; ++t) Am[a_k + t][lm] = 0;
+    }
+    if constexpr (CO_PER_T == 8) {
+      const bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+      const bf16x8 v0 = b_live0 ? rb8_0 : z;
+      BmT[b_co0 + 0][b_lm0] = v0[0]; BmT[b_co0 + 1][b_lm0] = v0[1];
+      BmT[b_co0 + 2][b_lm0] = v0[2]; BmT[b_co0 + 3][b_lm0] = v0[3];
+      BmT[b_co0 + 4][b_lm0] = v0[4]; BmT[b_co0 + 5][b_lm0] = v0[5];
+      BmT[b_co0 + 6][b_lm0] = v0[6]; BmT[b_co0 + 7][b_lm0] = v0[7];
+      const bf16x8 v1 = b_live1 ? rb8_1 : z;
+      const int lm1 = b_lm0 + 32;
+      BmT[b_co0 + 0][lm1] = v1[0]; BmT[b_co0 + 1][lm1] = v1[1];
+      BmT[b_co0 + 2][lm1] = v1[2]; BmT[b_co0 + 3][lm1] = v1[3];
+      BmT[b_co0 + 4][lm1] = v1[4]; BmT[b_co0 + 5][lm1] = v1[5];
+      BmT[b_co0 + 6][lm1] = v1[6]; BmT[b_co0 + 7][lm1] = v1[7];
+    } else {
+      const uint2 z = {0, 0};
+      const uint2 v0 = b_live0 ? rb4_0 : z;
+      BmT[b_co0 + 0][b_lm0] = (bf16raw)(v0.x & 0xFFFF);
+      BmT[b_co0 + 1][b_lm0] = (bf16raw)(v0.x >> 16);
+      BmT[b_co0 + 2][b_lm0] = (bf16raw)(v0.y & 0xFFFF);
+      BmT[b_co0 + 3][b_lm0] = (bf16raw)(v0.y >> 16);
+      const uint2 v1 = b_live1 ? rb4_1 : z;
+      const int lm1 = b_lm0 + 32;
+      BmT[b_co0 + 0][lm1] = (bf16raw)(v1.x & 0xFFFF);
+      BmT[b_co0 + 1][lm1] = (bf16raw)(v1.x >> 16);
+      BmT[b_co0 + 2][lm1] = (bf16raw)(v1.y & 0xFFFF);
+      BmT[b_co0 + 3][lm1] = (bf16raw)(v1.y >> 16);
+    }
+  };
+
+  load_chunk(m_begin);
+  for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
+    store_chunk();
+    __syncthreads();
+    if (m0 + BKM < m_end) load_chunk(m0 + BKM);  // prefetch next chunk
     for (int kk = 0; kk < BKM; kk += 32) {
       const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
           &Am[wave * 16 + (lane & 15)][kk + (lane >> 4) * 8]);
