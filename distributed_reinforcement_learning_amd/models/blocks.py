@@ -149,6 +149,22 @@ class LSTMCellTF(nn.Module):
         self.bias = nn.Parameter(torch.zeros(4 * hidden_size))
         nn.init.xavier_uniform_(self.weight)
 
+    def gates_from_xh(self, xh: torch.Tensor) -> torch.Tensor:
+        """Pre-concatenated [x, h] input (the batched unroll builds ONE
+        cat of conv-features + action-embedding + h instead of two)."""
+        if xh.is_cuda and xh.dtype == torch.bfloat16:
+            return torch.addmm(self.bias, xh, self.weight)
+        return torch.addmm(self.bias.to(xh.dtype), xh, self.weight)
+
+    def forward_xh(self, xh: torch.Tensor, c: torch.Tensor):
+        from distributed_reinforcement_learning_amd.ops.lstm_op import (
+            lstm_fused_step,
+        )
+        g = self.gates_from_xh(xh)
+        if g.is_cuda:
+            return lstm_fused_step(g, c.float(), self.forget_bias)
+        return lstm_fused_step(g.float(), c.float(), self.forget_bias)
+
     def gates(self, x: torch.Tensor, h: torch.Tensor) -> torch.Tensor:
         xh = torch.cat([x, h.to(x.dtype)], dim=1)
         # addmm fuses the bias into the GEMM epilogue (one fewer launch);

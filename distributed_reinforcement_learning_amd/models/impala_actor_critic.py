@@ -58,9 +58,13 @@ class ImpalaActorCritic(nn.Module):
         flat_pa = traj_prev_action.reshape(B * T)
         flat_h = traj_h.reshape(B * T, -1)
         flat_c = traj_c.reshape(B * T, -1)
-        feat = torch.cat(
-            [self.conv(flat_state), self.action_emb(flat_pa)], dim=1)
-        new_h, _ = self.lstm(feat, flat_h, flat_c)
+        conv_out = self.conv(flat_state)
+        emb = self.action_emb(flat_pa)
+        # ONE cat builds the gate-GEMM input directly (conv || emb || h);
+        # backward slices it in place — the strided-view support in the
+        # conv/LSTM backward kernels means no .contiguous() copies
+        xh = torch.cat([conv_out, emb, flat_h.to(conv_out.dtype)], dim=1)
+        new_h, _ = self.lstm.forward_xh(xh, flat_c)
         return new_h, B, T
 
     def unroll(self, traj_state: torch.Tensor, traj_prev_action: torch.Tensor,

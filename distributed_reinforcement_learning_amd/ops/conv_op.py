@@ -37,7 +37,10 @@ class _ConvLayer(torch.autograd.Function):
         co = _LAYER_CO[layer]
         # mask kernel writes bias-grad partials into a persistent slot
         # buffer; the wgrad finalize that follows sums them -> bf16 dbias
-        dy_m = ext.relu_mask_bwd(dy.contiguous(), y, co)
+        if not (dy.is_contiguous() or (dy.stride(-1) == 1
+                                        and dy.dim() >= 2)):
+            dy = dy.contiguous()
+        dy_m = ext.relu_mask_bwd(dy, y, co)
         dw, dbias = ext.conv_wgrad(layer, x, dy_m)
         dx = None
         if layer >= 2 and ctx.needs_input_grad[0]:

@@ -45,7 +45,8 @@ extern "C" __global__ void drla_conv_fwd_l3(const unsigned short*,
 extern "C" __global__ void drla_relu_mask_bwd(const unsigned short*,
                                               const unsigned short*,
                                               unsigned short*, float*,
-                                              long long, int);
+                                              long long, int, long long,
+                                              long long);
 extern "C" __global__ void drla_wgrad_finalize(float*, unsigned short*,
                                                int, int, float*,
                                                unsigned short*);
@@ -105,13 +106,13 @@ extern "C" __global__ void drla_lstm_tail_fwd(const float*, const float*,
 extern "C" __global__ void drla_lstm_tail_bwd(const float*, const float*,
                                               const float*, const float*,
                                               const float*, float*, float*,
-                                              long long, int);
+                                              long long, int, long long);
 extern "C" __global__ void drla_lstm_tail_fwd_bf16(
     const unsigned short*, const float*, float*, float*, float*, float,
     long long, int);
 extern "C" __global__ void drla_lstm_tail_bwd_bf16(
     const float*, const float*, const float*, const float*, const float*,
-    unsigned short*, float*, long long, int);
+    unsigned short*, float*, long long, int, long long);
 extern "C" __global__ void drla_mlp_heads_fwd(
     const float*, const unsigned short*, const unsigned short*,
     const unsigned short*, const unsigned short*, const unsigned short*,
@@ -310,17 +311,39 @@ static torch::Tensor& dbias_slot_buf(const torch::Tensor& like) {
 }
 
 torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y, int64_t CO) {
-  check_gpu_contig(dy, "dy");
   check_gpu_contig(y, "y");
-  auto out = torch::empty_like(dy);
+  TORCH_CHECK(dy.is_cuda(), "dy must be on GPU");
+  // dy may be an N-strided view (slice of the fused xh grad): inner dims
+  // contiguous, only stride(0) loose
+  long long n_stride, row_elems;
+  if (dy.is_contiguous()) {
+    n_stride = row_elems = dy.numel() / dy.size(0);
+  } else {
+    row_elems = 1;
+    for (int d = 1; d < dy.dim(); ++d) {
+      TORCH_CHECK(dy.stride(d - 1) >= dy.stride(d), "dy must be row-major");
+      row_elems *= dy.size(d);
+    }
+    TORCH_CHECK(dy.stride(dy.dim() - 1) == 1 &&
+                    dy.stride(0) >= row_elems && row_elems % 8 == 0,
+                "dy view must be inner-contiguous");
+    for (int d = 1; d + 1 < dy.dim(); ++d) {
+      TORCH_CHECK(dy.stride(d) == dy.size(d + 1) * dy.stride(d + 1),
+                  "dy inner dims must be contiguous");
+    }
+    n_stride = dy.stride(0);
+  }
+  auto out = torch::empty(dy.sizes(), dy.options());
   auto& slots = dbias_slot_buf(dy);
   const long long n = dy.numel();
   TORCH_CHECK(n % 8 == 0, "relu_mask_bwd wants numel % 8 == 0");
   int grid = drla_grid(n / 8);
   if (grid > 640) grid = 640;
   hipLaunchKernelGGL(drla_relu_mask_bwd, dim3(grid), dim3(DRLA_BLOCK),
-                     0, cur_stream(), u16p(dy), u16p(y), u16pm(out),
-                     slots.data_ptr<float>(), n, (int)CO);
+                     0, cur_stream(),
+                     reinterpret_cast<const unsigned short*>(dy.data_ptr()),
+                     u16p(y), u16pm(out), slots.data_ptr<float>(), n,
+                     (int)CO, n_stride, row_elems);
   return out;
 }
 
@@ -338,7 +361,7 @@ std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
     const char* e = getenv("DRLA_WGRAD_SPLIT");
     return e ? atoi(e) : 0;
   }();
-  const int split = split_env ? split_env : ((layer <= 1) ? 512 : 128);
+  const int split = split_env ? split_env : ((layer <= 1) ? 512 : 64);
   // persistent zero-between-calls scratch (finalize re-zeroes on read)
   static torch::Tensor scratch_cache[4];
   if (!scratch_cache[layer].defined()) {
@@ -612,8 +635,12 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_tail_fwd(
 std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
     torch::Tensor grad_h, torch::Tensor grad_c, torch::Tensor stash,
     torch::Tensor c_prev, torch::Tensor new_c, bool bf16_gates) {
-  for (auto* t : {&grad_h, &grad_c, &stash, &c_prev, &new_c})
+  for (auto* t : {&grad_c, &stash, &c_prev, &new_c})
     check_gpu_contig(*t, "lstm bwd input");
+  // grad_h may be an N-strided [N,H] view (slice of the xh gradient)
+  TORCH_CHECK(grad_h.is_cuda() && grad_h.dim() == 2 &&
+              grad_h.stride(1) == 1, "grad_h must be row-contiguous");
+  const long long gh_stride = grad_h.stride(0);
   const long long N = stash.size(0);
   const int H = stash.size(1) / 4;
   auto grad_c_prev = torch::empty_like(c_prev);
@@ -626,7 +653,7 @@ std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
                        grad_h.data_ptr<float>(), grad_c.data_ptr<float>(),
                        stash.data_ptr<float>(), c_prev.data_ptr<float>(),
                        new_c.data_ptr<float>(), u16pm(grad_gates),
-                       grad_c_prev.data_ptr<float>(), N, H);
+                       grad_c_prev.data_ptr<float>(), N, H, gh_stride);
     return {grad_gates, grad_c_prev};
   }
   auto grad_gates = torch::empty_like(stash);
@@ -635,7 +662,7 @@ std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
                      grad_h.data_ptr<float>(), grad_c.data_ptr<float>(),
                      stash.data_ptr<float>(), c_prev.data_ptr<float>(),
                      new_c.data_ptr<float>(), grad_gates.data_ptr<float>(),
-                     grad_c_prev.data_ptr<float>(), N, H);
+                     grad_c_prev.data_ptr<float>(), N, H, gh_stride);
   return {grad_gates, grad_c_prev};
 }
 

@@ -242,10 +242,14 @@ extern "C" __global__ __launch_bounds__(256) void drla_conv_fwd_l3(
 // of a persistent [16][64] buffer; drla_wgrad_finalize (which always runs
 // next in the conv backward) does the 16-way sum, emits bf16 dbias, and
 // re-zeroes the slots.
+// dy may be an N-strided view (a slice of the fused xh gradient):
+// row_elems = elements per logical row, n_stride = elements between row
+// starts; row_elems == n_stride means contiguous. Chunks never straddle
+// rows (row_elems % 8 == 0).
 extern "C" __global__ void drla_relu_mask_bwd(
     const bf16raw* __restrict__ dy, const bf16raw* __restrict__ y,
     bf16raw* __restrict__ out, float* __restrict__ dbias_slots,
-    long long n, int CO) {
+    long long n, int CO, long long n_stride, long long row_elems) {
   // vectorized: each thread moves 8 bf16 per iteration (uint4), which pins
   // it to ONE 8-column group because stride*8 % CO == 0 (CO in {32,64},
   // blockDim 256). Per-thread acc[8] -> LDS -> one atomicAdd per (block,
@@ -272,8 +276,15 @@ extern "C" __global__ void drla_relu_mask_bwd(
     (a0) += cv_bf2f(m0);                                              \
     (a1) += cv_bf2f(m1);                                              \
   }
+  const long long row8 = row_elems / 8;
+  const bool strided = (n_stride != row_elems);
   for (; i < n8; i += stride) {
-    const uint4 dv = dy4[i];
+    long long si = i;
+    if (strided) {
+      const long long r = i / row8;
+      si = (r * n_stride + (i - r * row8) * 8) / 8;
+    }
+    const uint4 dv = dy4[si];
     const uint4 yv = y4[i];
     uint4 ov;
     DRLA_RM_LANE(dv.x, yv.x, ov.x, acc[0], acc[1]);

@@ -49,7 +49,8 @@ extern "C" __global__ void drla_lstm_tail_bwd(
     const float* __restrict__ grad_h, const float* __restrict__ grad_c,
     const float* __restrict__ stash, const float* __restrict__ c_prev,
     const float* __restrict__ new_c, float* __restrict__ grad_gates,
-    float* __restrict__ grad_c_prev, long long N, int H) {
+    float* __restrict__ grad_c_prev, long long N, int H,
+    long long gh_stride) {
   long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
   const long long total = N * H;
   const long long stride = gridDim.x * (long long)blockDim.x;
@@ -62,7 +63,7 @@ extern "C" __global__ void drla_lstm_tail_bwd(
     const float f_s = stash[g0 + 2 * H];
     const float o_s = stash[g0 + 3 * H];
     const float tc = tanhf(new_c[idx]);
-    const float dh = grad_h[idx];
+    const float dh = grad_h[n * gh_stride + h];
     const float d_tc = dh * o_s * (1.0f - tc * tc) + grad_c[idx];
     grad_c_prev[idx] = d_tc * f_s;
     const float di = d_tc * g_t;
@@ -111,7 +112,8 @@ extern "C" __global__ void drla_lstm_tail_bwd_bf16(
     const float* __restrict__ grad_h, const float* __restrict__ grad_c,
     const float* __restrict__ stash, const float* __restrict__ c_prev,
     const float* __restrict__ new_c, unsigned short* __restrict__ grad_gates,
-    float* __restrict__ grad_c_prev, long long N, int H) {
+    float* __restrict__ grad_c_prev, long long N, int H,
+    long long gh_stride) {
   long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
   const long long total = N * H;
   const long long stride = gridDim.x * (long long)blockDim.x;
@@ -124,7 +126,7 @@ extern "C" __global__ void drla_lstm_tail_bwd_bf16(
     const float f_s = stash[g0 + 2 * H];
     const float o_s = stash[g0 + 3 * H];
     const float tc = tanhf(new_c[idx]);
-    const float dh = grad_h[idx];
+    const float dh = grad_h[n * gh_stride + h];
     const float d_tc = dh * o_s * (1.0f - tc * tc) + grad_c[idx];
     grad_c_prev[idx] = d_tc * f_s;
     const float di = d_tc * g_t;

@@ -35,8 +35,10 @@ class _FusedLstmTail(torch.autograd.Function):
     def backward(ctx, grad_h: torch.Tensor, grad_c: torch.Tensor):
         stash, c_prev, new_c = ctx.saved_tensors
         ext = _ops.require_ext()
+        if not (grad_h.dim() == 2 and grad_h.stride(1) == 1):
+            grad_h = grad_h.contiguous()
         grad_gates, grad_c_prev = ext.lstm_tail_bwd(
-            grad_h.contiguous(), grad_c.contiguous(), stash,
+            grad_h, grad_c.contiguous(), stash,
             c_prev.contiguous(), new_c, ctx.bf16_gates)
         return grad_gates, grad_c_prev, None
 
