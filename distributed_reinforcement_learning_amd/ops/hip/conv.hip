@@ -100,35 +100,41 @@ __device__ void conv_fwd_impl(const IN_T* __restrict__ in,
   const long long in_base =
       ((long long)n_idx * HI + ho * STRIDE) * WI + wo * STRIDE;
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // ---- stage A[128][32] ----
+  // software-pipelined K-loop: chunk k0+BK's global loads issue into
+  // registers while chunk k0's MFMA runs (same structure as the wgrad —
+  // the per-chunk load-latency chain is what sets kernel time here)
+  uchar4 ra4_0, ra4_1, ra4_2, ra4_3;
+  unsigned char ra1[16];
+  bf16x8 rab0 = {0, 0, 0, 0, 0, 0, 0, 0}, rab1 = rab0;
+  uint4 rbw4;
+  uint2 rbw2;
+  constexpr int THREADS_PER_CO = 256 / CO;       // 8 (CO=32) or 4
+  constexpr int KCHUNK = BK / THREADS_PER_CO;    // 4 or 8
+  const int b_co = tid / THREADS_PER_CO;
+  const int b_kpart = (tid % THREADS_PER_CO) * KCHUNK;
+
+  auto load_chunk = [&](int k0) {
     if (gm < M) {
       if constexpr (CI == 4) {
-        // u8 input, 4 channels per tap: 4 taps per k-half, one uchar4 each,
-        // /255 normalize fused into the load
-        for (int t = 0; t < 4; ++t) {
-          const int kk = k0 + a_k0 + t * 4;
-          const int kh = kk / (KW * CI);
-          const int kw = (kk - kh * KW * CI) / CI;
-          const uchar4 v = *reinterpret_cast<const uchar4*>(
-              in + (in_base + (long long)kh * WI + kw) * CI);
-          const float s = 1.0f / 255.0f;
-          Abuf[a_row][a_k0 + t * 4 + 0] = drla_f32_to_bf16(v.x * s);
-          Abuf[a_row][a_k0 + t * 4 + 1] = drla_f32_to_bf16(v.y * s);
-          Abuf[a_row][a_k0 + t * 4 + 2] = drla_f32_to_bf16(v.z * s);
-          Abuf[a_row][a_k0 + t * 4 + 3] = drla_f32_to_bf16(v.w * s);
+#define DRLA_CF_TAP(t, dst)                                       \
+        {                                                         \
+          const int kk = k0 + a_k0 + (t) * 4;                     \
+          const int kh = kk / (KW * CI);                          \
+          const int kw = (kk - kh * KW * CI) / CI;                \
+          dst = *reinterpret_cast<const uchar4*>(                 \
+              in + (in_base + (long long)kh * WI + kw) * CI);     \
         }
+        DRLA_CF_TAP(0, ra4_0); DRLA_CF_TAP(1, ra4_1);
+        DRLA_CF_TAP(2, ra4_2); DRLA_CF_TAP(3, ra4_3);
+#undef DRLA_CF_TAP
       } else if constexpr (CI == 1) {
-        // u8 single-channel (R2D2 POMDP): one tap per k element
+#pragma unroll
         for (int t = 0; t < 16; ++t) {
           const int kk = k0 + a_k0 + t;
           const int kh = kk / KW;
-          const int kw = kk - kh * KW;
-          const unsigned char v = in[in_base + (long long)kh * WI + kw];
-          Abuf[a_row][a_k0 + t] = drla_f32_to_bf16(v * (1.0f / 255.0f));
+          ra1[t] = in[in_base + (long long)kh * WI + (kk - kh * KW)];
         }
       } else {
-        // bf16 input, CI >= 32: a 16-element k-half stays inside one tap
         const int kk = k0 + a_k0;
         const int kh = kk / (KW * CI);
         const int kwci = kk - kh * KW * CI;
@@ -136,35 +142,59 @@ __device__ void conv_fwd_impl(const IN_T* __restrict__ in,
         const int ci = kwci - kw * CI;
         const bf16raw* src = reinterpret_cast<const bf16raw*>(in) +
                              (in_base + (long long)kh * WI + kw) * CI + ci;
-        *reinterpret_cast<uint4*>(&Abuf[a_row][a_k0]) =
-            *reinterpret_cast<const uint4*>(src);
-        *reinterpret_cast<uint4*>(&Abuf[a_row][a_k0 + 8]) =
-            *reinterpret_cast<const uint4*>(src + 8);
+        rab0 = *reinterpret_cast<const bf16x8*>(src);
+        rab1 = *reinterpret_cast<const bf16x8*>(src + 8);
+      }
+    }
+    const bf16raw* src = w + (long long)b_co * K + k0 + b_kpart;
+    if constexpr (KCHUNK == 8) {
+      rbw4 = *reinterpret_cast<const uint4*>(src);
+    } else {
+      rbw2 = *reinterpret_cast<const uint2*>(src);
+    }
+  };
+
+  auto store_chunk = [&]() {
+    if (gm < M) {
+      if constexpr (CI == 4) {
+        const float sc = 1.0f / 255.0f;
+#define DRLA_CF_PUT(t, v)                                        \
+        {                                                        \
+          const int o = a_k0 + (t) * 4;                          \
+          Abuf[a_row][o + 0] = drla_f32_to_bf16(v.x * sc);       \
+          Abuf[a_row][o + 1] = drla_f32_to_bf16(v.y * sc);       \
+          Abuf[a_row][o + 2] = drla_f32_to_bf16(v.z * sc);       \
+          Abuf[a_row][o + 3] = drla_f32_to_bf16(v.w * sc);       \
+        }
+        DRLA_CF_PUT(0, ra4_0); DRLA_CF_PUT(1, ra4_1);
+        DRLA_CF_PUT(2, ra4_2); DRLA_CF_PUT(3, ra4_3);
+#undef DRLA_CF_PUT
+      } else if constexpr (CI == 1) {
+#pragma unroll
+        for (int t = 0; t < 16; ++t) {
+          Abuf[a_row][a_k0 + t] = drla_f32_to_bf16(ra1[t] * (1.0f / 255.0f));
+        }
+      } else {
+        *reinterpret_cast<bf16x8*>(&Abuf[a_row][a_k0]) = rab0;
+        *reinterpret_cast<bf16x8*>(&Abuf[a_row][a_k0 + 8]) = rab1;
       }
     } else {
       for (int t = 0; t < 16; t += 8) {
         *reinterpret_cast<uint4*>(&Abuf[a_row][a_k0 + t]) = uint4{0, 0, 0, 0};
       }
     }
-    // ---- stage B[CO][32]: W[co][k0..k0+32) contiguous ----
-    {
-      // thread t stages co = t>>1? CO*32/256 = CO/8 elems per thread.
-      // assign: co = tid % CO, k-part = (tid / CO) * (32*CO/256/..)
-      constexpr int THREADS_PER_CO = 256 / CO;         // 8 (CO=32) or 4
-      constexpr int KCHUNK = BK / THREADS_PER_CO;      // 4 or 8
-      const int co = tid / THREADS_PER_CO;
-      const int kpart = (tid % THREADS_PER_CO) * KCHUNK;
-      const bf16raw* src = w + (long long)co * K + k0 + kpart;
-      if constexpr (KCHUNK == 8) {
-        *reinterpret_cast<uint4*>(&Bbuf[co][kpart]) =
-            *reinterpret_cast<const uint4*>(src);
-      } else {
-        *reinterpret_cast<uint2*>(&Bbuf[co][kpart]) =
-            *reinterpret_cast<const uint2*>(src);
-      }
+    if constexpr (KCHUNK == 8) {
+      *reinterpret_cast<uint4*>(&Bbuf[b_co][b_kpart]) = rbw4;
+    } else {
+      *reinterpret_cast<uint2*>(&Bbuf[b_co][b_kpart]) = rbw2;
     }
-    __syncthreads();
+  };
 
+  load_chunk(0);
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    store_chunk();
+    __syncthreads();
+    if (k0 + BK < K) load_chunk(k0 + BK);
     // ---- MFMA: wave covers rows [wave*32, wave*32+32) x CO ----
     for (int mi = 0; mi < 2; ++mi) {
       const int arow = wave * 32 + mi * 16 + (lane & 15);
@@ -665,49 +695,57 @@ __device__ void conv_dgrad_impl(const bf16raw* __restrict__ dy,  // [M][CO]
   const int hi = rem / WI;
   const int wi = rem - hi * WI;
 
-  for (int kg0 = 0; kg0 < Kg; kg0 += BK) {
+  // software-pipelined: next tap-chunk's globals prefetch into registers
+  // during this chunk's MFMA (see wgrad note)
+  uint4 rda0, rda1;
+  bf16raw rbw[8];
+  constexpr int CI_PER_T = (CI * BK) / 256;  // 4 (CI=32) or 8 (CI=64)
+  const int b_col = tid % BK;
+  const int b_ci0 = (tid / BK) * CI_PER_T;
+
+  auto load_chunk = [&](int kg0) {
     const int tap = kg0 / CO;
     const int co0 = kg0 - tap * CO;
     const int kh = tap / KW;
     const int kw = tap - kh * KW;
-    // ---- stage Ad[m2][co within tap] with validity mask ----
-    {
-      bool valid = (m2 < M2);
-      int ho = 0, wo = 0;
+    bool valid = (m2 < M2);
+    int ho = 0, wo = 0;
+    if (valid) {
+      const int hh = hi - kh;
+      const int ww = wi - kw;
+      valid = hh >= 0 && ww >= 0 && (hh % STRIDE) == 0 && (ww % STRIDE) == 0;
       if (valid) {
-        const int hh = hi - kh;
-        const int ww = wi - kw;
-        valid = hh >= 0 && ww >= 0 && (hh % STRIDE) == 0 &&
-                (ww % STRIDE) == 0;
-        if (valid) {
-          ho = hh / STRIDE;
-          wo = ww / STRIDE;
-          valid = ho < HO && wo < WO;
-        }
-      }
-      if (valid) {
-        const bf16raw* src =
-            dy + ((long long)(n_idx * HO + ho) * WO + wo) * CO + co0 + a_k0;
-        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) =
-            *reinterpret_cast<const uint4*>(src);
-        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) =
-            *reinterpret_cast<const uint4*>(src + 8);
-      } else {
-        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) = uint4{0, 0, 0, 0};
-        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) = uint4{0, 0, 0, 0};
+        ho = hh / STRIDE;
+        wo = ww / STRIDE;
+        valid = ho < HO && wo < WO;
       }
     }
-    // ---- stage Bd[ci][co_local]: W[co0+col][tap*CI + ci] transposed ----
-    {
-      constexpr int CI_PER_T = (CI * BK) / 256;  // 4 (CI=32) or 8 (CI=64)
-      const int col = tid % BK;                  // co_local
-      const int ci0 = (tid / BK) * CI_PER_T;
-      const bf16raw* src = w + (long long)(co0 + col) * K + tap * CI + ci0;
-      for (int e = 0; e < CI_PER_T; ++e) {
-        Bd[ci0 + e][col] = src[e];
-      }
+    if (valid) {
+      const bf16raw* src =
+          dy + ((long long)(n_idx * HO + ho) * WO + wo) * CO + co0 + a_k0;
+      rda0 = *reinterpret_cast<const uint4*>(src);
+      rda1 = *reinterpret_cast<const uint4*>(src + 8);
+    } else {
+      rda0 = uint4{0, 0, 0, 0};
+      rda1 = uint4{0, 0, 0, 0};
     }
+    const bf16raw* src = w + (long long)(co0 + b_col) * K + tap * CI + b_ci0;
+#pragma unroll
+    for (int e = 0; e < CI_PER_T; ++e) rbw[e] = src[e];
+  };
+
+  auto store_chunk = [&]() {
+    *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) = rda0;
+    *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) = rda1;
+#pragma unroll
+    for (int e = 0; e < CI_PER_T; ++e) Bd[b_ci0 + e][b_col] = rbw[e];
+  };
+
+  load_chunk(0);
+  for (int kg0 = 0; kg0 < Kg; kg0 += BK) {
+    store_chunk();
     __syncthreads();
+    if (kg0 + BK < Kg) load_chunk(kg0 + BK);
 
     for (int mi = 0; mi < 2; ++mi) {
       const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
@@ -774,47 +812,57 @@ __device__ void conv_dgrad_s2_impl(const bf16raw* __restrict__ dy,
   const int hi = (rem / W2) * 2 + ph;
   const int wi = (rem - (rem / W2) * W2) * 2 + pw;
 
-  for (int it = 0; it < 4 * (CO / BK); ++it) {
+  uint4 rda0, rda1;
+  bf16raw rbw[8];
+  constexpr int CI_PER_T = (CI * BK) / 256;
+  const int b_col = tid % BK;
+  const int b_ci0 = (tid / BK) * CI_PER_T;
+  constexpr int NIT = 4 * (CO / BK);
+
+  auto load_chunk = [&](int it) {
     const int t = it / (CO / BK);           // live tap 0..3
     const int co0 = (it - t * (CO / BK)) * BK;
     const int kh = ph + 2 * (t >> 1);
     const int kw = pw + 2 * (t & 1);
-    {
-      bool valid = (mc < Mc);
-      int ho = 0, wo = 0;
+    bool valid = (mc < Mc);
+    int ho = 0, wo = 0;
+    if (valid) {
+      const int hh = hi - kh;
+      const int ww = wi - kw;
+      valid = hh >= 0 && ww >= 0;
       if (valid) {
-        const int hh = hi - kh;
-        const int ww = wi - kw;
-        valid = hh >= 0 && ww >= 0;
-        if (valid) {
-          ho = hh >> 1;
-          wo = ww >> 1;
-          valid = ho < HO && wo < WO;
-        }
-      }
-      if (valid) {
-        const bf16raw* src =
-            dy + ((long long)(n_idx * HO + ho) * WO + wo) * CO + co0 + a_k0;
-        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) =
-            *reinterpret_cast<const uint4*>(src);
-        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) =
-            *reinterpret_cast<const uint4*>(src + 8);
-      } else {
-        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) = uint4{0, 0, 0, 0};
-        *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) = uint4{0, 0, 0, 0};
+        ho = hh >> 1;
+        wo = ww >> 1;
+        valid = ho < HO && wo < WO;
       }
     }
-    {
-      constexpr int CI_PER_T = (CI * BK) / 256;
-      const int col = tid % BK;
-      const int ci0 = (tid / BK) * CI_PER_T;
-      const int tap = kh * KW + kw;
-      const bf16raw* src = w + (long long)(co0 + col) * K + tap * CI + ci0;
-      for (int e = 0; e < CI_PER_T; ++e) {
-        Bd[ci0 + e][col] = src[e];
-      }
+    if (valid) {
+      const bf16raw* src =
+          dy + ((long long)(n_idx * HO + ho) * WO + wo) * CO + co0 + a_k0;
+      rda0 = *reinterpret_cast<const uint4*>(src);
+      rda1 = *reinterpret_cast<const uint4*>(src + 8);
+    } else {
+      rda0 = uint4{0, 0, 0, 0};
+      rda1 = uint4{0, 0, 0, 0};
     }
+    const int tap = kh * KW + kw;
+    const bf16raw* src = w + (long long)(co0 + b_col) * K + tap * CI + b_ci0;
+#pragma unroll
+    for (int e = 0; e < CI_PER_T; ++e) rbw[e] = src[e];
+  };
+
+  auto store_chunk = [&]() {
+    *reinterpret_cast<uint4*>(&Ad[a_row][a_k0]) = rda0;
+    *reinterpret_cast<uint4*>(&Ad[a_row][a_k0 + 8]) = rda1;
+#pragma unroll
+    for (int e = 0; e < CI_PER_T; ++e) Bd[b_ci0 + e][b_col] = rbw[e];
+  };
+
+  load_chunk(0);
+  for (int it = 0; it < NIT; ++it) {
+    store_chunk();
     __syncthreads();
+    if (it + 1 < NIT) load_chunk(it + 1);
     for (int mi = 0; mi < 2; ++mi) {
       const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
           &Ad[wave * 32 + mi * 16 + (lane & 15)][(lane >> 4) * 8]);
