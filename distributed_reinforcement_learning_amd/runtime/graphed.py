@@ -67,6 +67,9 @@ class GraphedImpalaStep:
         self._consumed.record()
         self._uploaded = torch.cuda.Event()
         self._uploaded.record()
+        self._fwd_done = torch.cuda.Event()
+        self._copy_stream = torch.cuda.Stream()
+        self._primed = False
 
         # ---- warmup (eager, side stream), with state snapshot/restore -----
         opt = agent.optimizer
@@ -92,9 +95,17 @@ class GraphedImpalaStep:
             opt.flat_grads.zero_()
 
         # ---- capture ------------------------------------------------------
-        self.g_fwd_bwd = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_fwd_bwd):
-            self.losses = self._fwd_bwd()
+        # forward and backward are SEPARATE graphs so the next step's H2D
+        # upload (~8.4 MB pinned -> static inputs, ~110 us) can overlap the
+        # backward on a copy stream: inputs are only read by the forward.
+        self.g_fwd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_fwd):
+            self._total, self.losses = self._fwd()
+        self.g_bwd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_bwd, pool=self.g_fwd.pool()):
+            # retain_graph: the saved tensors live in the shared capture
+            # pool and are rewritten by every g_fwd replay
+            self._total.backward(retain_graph=True)
         # .grad now holds capture-pool tensors at replay-stable addresses
         opt.build_gather_table()
         from distributed_reinforcement_learning_amd.parallel.dist import (
@@ -102,7 +113,7 @@ class GraphedImpalaStep:
         )
         self._distributed = is_distributed()
         self.g_opt = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_opt):
+        with torch.cuda.graph(self.g_opt, pool=self.g_fwd.pool()):
             if not self._distributed:
                 # single-GPU: the gather rides inside the optimizer graph;
                 # distributed keeps it eager so the RCCL all-reduce sees
@@ -110,17 +121,23 @@ class GraphedImpalaStep:
                 opt.gather_grads()
             opt.step_tensor_lr(self.lr_buf)
 
-    def _fwd_bwd(self) -> Tuple[torch.Tensor, ...]:
+    def _fwd(self):
         agent = self.agent
-        if not agent.optimizer.scatter:
-            agent.optimizer.flat_grads.zero_()
         i = self.inputs
         s = agent.prepare_frames(i["state"])
         pi_loss, baseline_loss, entropy, total = agent.compute_losses(
             s, i["reward"], i["action"], i["done"], i["behavior_policy"],
             i["previous_action"], i["initial_h"], i["initial_c"])
+        return total, (pi_loss.detach(), baseline_loss.detach(),
+                       entropy.detach())
+
+    def _fwd_bwd(self) -> Tuple[torch.Tensor, ...]:
+        agent = self.agent
+        if not agent.optimizer.scatter:
+            agent.optimizer.flat_grads.zero_()
+        total, losses = self._fwd()
         total.backward()
-        return (pi_loss.detach(), baseline_loss.detach(), entropy.detach())
+        return losses
 
     # -- input staging -------------------------------------------------------
 
@@ -133,13 +150,22 @@ class GraphedImpalaStep:
 
     def upload_inputs(self, src: Optional[Dict[str, torch.Tensor]] = None
                       ) -> None:
-        """Async H2D pinned -> static graph inputs. Waits (usually no-op)
-        until the previous replay has consumed the static buffers."""
+        """Synchronous-path upload (priming / non-pipelined callers)."""
         self._consumed.synchronize()
         src = src or self.pinned
         for k, dst in self.inputs.items():
             dst.copy_(src[k], non_blocking=True)
         self._uploaded.record()
+        self._primed = True
+
+    def _upload_overlapped(self, src: Dict[str, torch.Tensor]) -> None:
+        """Queue the NEXT step's H2D on the copy stream, ordered after this
+        step's forward (the only consumer of the static inputs)."""
+        with torch.cuda.stream(self._copy_stream):
+            self._copy_stream.wait_event(self._fwd_done)
+            for k, dst in self.inputs.items():
+                dst.copy_(src[k], non_blocking=True)
+            self._uploaded.record(self._copy_stream)
 
     def wait_pinned_free(self) -> None:
         """Block until the last async H2D has finished reading the pinned
@@ -155,14 +181,31 @@ class GraphedImpalaStep:
         (numpy, staged through self.pinned), ``pinned_src`` (caller-owned
         pinned tensors), or self.pinned already filled. Never syncs; returns
         the loss TENSORS (device). Read them with last_losses() at logging
-        cadence."""
+        cadence.
+
+        Pipelined upload: the FIRST call uploads synchronously; later calls
+        replay on the inputs uploaded during the PREVIOUS step's backward
+        and queue this call's data on the copy stream (one-batch latency,
+        which an on-policy-ish learner does not notice). Callers that must
+        not pipeline can call upload_inputs() themselves each step."""
         agent = self.agent
         if batch is not None:
             self.stage_to_pinned(batch)
-        self.upload_inputs(pinned_src)
+        src = pinned_src if pinned_src is not None else self.pinned
+        main = torch.cuda.current_stream()
+        if not self._primed:
+            self.upload_inputs(src)
+            src = None
+        main.wait_event(self._uploaded)
         lr = agent.lr_at(agent.global_step)
         self.lr_buf.fill_(lr)
-        self.g_fwd_bwd.replay()
+        self.g_fwd.replay()
+        self._fwd_done.record(main)
+        if src is not None:
+            # overlap this call's H2D with the backward; the replay above
+            # consumed the PREVIOUS call's upload (one-batch pipeline)
+            self._upload_overlapped(src)
+        self.g_bwd.replay()
         if self._distributed:
             agent.optimizer.gather_grads()
             agent.reduce_gradients()
