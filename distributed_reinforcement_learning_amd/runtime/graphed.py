@@ -183,19 +183,25 @@ class GraphedImpalaStep:
         the loss TENSORS (device). Read them with last_losses() at logging
         cadence.
 
-        Pipelined upload: the FIRST call uploads synchronously; later calls
-        replay on the inputs uploaded during the PREVIOUS step's backward
-        and queue this call's data on the copy stream (one-batch latency,
-        which an on-policy-ish learner does not notice). Callers that must
-        not pipeline can call upload_inputs() themselves each step."""
+        Pipelined upload (pinned paths only): the FIRST call uploads
+        synchronously; later calls replay on the inputs uploaded during the
+        PREVIOUS step's backward and queue this call's data on the copy
+        stream — one-batch latency, which the asynchronous actor-learner
+        loop does not notice (losses/scalars lag one batch). The numpy
+        ``batch=`` path stays strictly synchronous."""
         agent = self.agent
-        if batch is not None:
-            self.stage_to_pinned(batch)
-        src = pinned_src if pinned_src is not None else self.pinned
         main = torch.cuda.current_stream()
-        if not self._primed:
-            self.upload_inputs(src)
+        if batch is not None:
+            # numpy path keeps same-batch semantics (no pipelining) — the
+            # eager-parity test and train()-style callers rely on it
+            self.stage_to_pinned(batch)
+            self.upload_inputs()
             src = None
+        else:
+            src = pinned_src if pinned_src is not None else self.pinned
+            if not self._primed:
+                self.upload_inputs(src)
+                src = None
         main.wait_event(self._uploaded)
         lr = agent.lr_at(agent.global_step)
         self.lr_buf.fill_(lr)
