@@ -432,47 +432,59 @@ extern "C" __global__ __launch_bounds__(256) void drla_heads_wgrad(
   f32x4 acc[4];
   for (int ni = 0; ni < 4; ++ni) acc[ni] = {0.f, 0.f, 0.f, 0.f};
 
-  for (int k0 = 0; k0 < N; k0 += 32) {
-    // stage: thread t owns global row k = k0 + t/8, columns (t%8)*8..+8
-    const int kk = tid >> 3;
-    const int c0 = (tid & 7) * 8;
-    const int k = k0 + kk;
-    // act is always [N,256] bf16, vec8
-    {
-      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (k < N) {
-        v = *reinterpret_cast<const bf16x8*>(
-            act + (long long)k * MH_HID + n0 + c0);
-      }
-      actT[c0 + 0][kk] = v[0]; actT[c0 + 1][kk] = v[1];
-      actT[c0 + 2][kk] = v[2]; actT[c0 + 3][kk] = v[3];
-      actT[c0 + 4][kk] = v[4]; actT[c0 + 5][kk] = v[5];
-      actT[c0 + 6][kk] = v[6]; actT[c0 + 7][kk] = v[7];
+  // software-pipelined K(=N)-loop: next chunk's globals prefetch into
+  // registers during this chunk's MFMA (same pattern as the conv wgrad)
+  const int st_kk = tid >> 3;
+  const int st_c0 = (tid & 7) * 8;
+  const bf16x8 z8 = {0, 0, 0, 0, 0, 0, 0, 0};
+  bf16x8 r_act = z8, r_dz = z8;
+
+  auto load_chunk = [&](int k0) {
+    const int k = k0 + st_kk;
+    r_act = z8;
+    if (k < N) {
+      r_act = *reinterpret_cast<const bf16x8*>(
+          act + (long long)k * MH_HID + n0 + st_c0);
     }
     if (job < 4) {           // dz [N,256] bf16, vec8
-      bf16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      r_dz = z8;
       if (k < N) {
-        v = *reinterpret_cast<const bf16x8*>(
-            dz + (long long)k * MH_HID + m0 + c0);
+        r_dz = *reinterpret_cast<const bf16x8*>(
+            dz + (long long)k * MH_HID + m0 + st_c0);
       }
-      dzT[c0 + 0][kk] = v[0]; dzT[c0 + 1][kk] = v[1];
-      dzT[c0 + 2][kk] = v[2]; dzT[c0 + 3][kk] = v[3];
-      dzT[c0 + 4][kk] = v[4]; dzT[c0 + 5][kk] = v[5];
-      dzT[c0 + 6][kk] = v[6]; dzT[c0 + 7][kk] = v[7];
     } else if (job == 4) {   // dlogits [N,A], scalar predicated
+#pragma unroll
       for (int e = 0; e < 8; ++e) {
-        const int m = c0 + e;
-        dzT[m][kk] = (k < N && m < A) ? dlogits[(long long)k * A + m]
-                                      : (bf16raw)0;
+        const int m = st_c0 + e;
+        r_dz[e] = (k < N && m < A)
+            ? (short)dlogits[(long long)k * A + m] : (short)0;
       }
     } else {                 // dvalue [N] f32, M = 1
+#pragma unroll
       for (int e = 0; e < 8; ++e) {
-        const int m = c0 + e;
-        dzT[m][kk] = (k < N && m == 0) ? drla_f32_to_bf16(dvalue[k])
-                                       : (bf16raw)0;
+        const int m = st_c0 + e;
+        r_dz[e] = (k < N && m == 0)
+            ? (short)drla_f32_to_bf16(dvalue[k]) : (short)0;
       }
     }
+  };
+
+  auto store_chunk = [&]() {
+    actT[st_c0 + 0][st_kk] = r_act[0]; actT[st_c0 + 1][st_kk] = r_act[1];
+    actT[st_c0 + 2][st_kk] = r_act[2]; actT[st_c0 + 3][st_kk] = r_act[3];
+    actT[st_c0 + 4][st_kk] = r_act[4]; actT[st_c0 + 5][st_kk] = r_act[5];
+    actT[st_c0 + 6][st_kk] = r_act[6]; actT[st_c0 + 7][st_kk] = r_act[7];
+    dzT[st_c0 + 0][st_kk] = r_dz[0]; dzT[st_c0 + 1][st_kk] = r_dz[1];
+    dzT[st_c0 + 2][st_kk] = r_dz[2]; dzT[st_c0 + 3][st_kk] = r_dz[3];
+    dzT[st_c0 + 4][st_kk] = r_dz[4]; dzT[st_c0 + 5][st_kk] = r_dz[5];
+    dzT[st_c0 + 6][st_kk] = r_dz[6]; dzT[st_c0 + 7][st_kk] = r_dz[7];
+  };
+
+  load_chunk(0);
+  for (int k0 = 0; k0 < N; k0 += 32) {
+    store_chunk();
     __syncthreads();
+    if (k0 + 32 < N) load_chunk(k0 + 32);
     // wave w: m rows [w*16, w*16+16); frags vec8 from LDS
     const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
         &dzT[wave * 16 + (lane & 15)][(lane >> 4) * 8]);
