@@ -34,3 +34,40 @@ def embed_lookup(table: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
     if table.is_cuda:
         return _EmbedLookup.apply(table, idx.long())
     return torch.nn.functional.embedding(idx.long(), table)
+
+
+class _FusedActionEmbed(torch.autograd.Function):
+    """Whole K2 block — one-hot -> 256 (+b, ReLU) -> 256 (+b, ReLU) — as
+    1 forward + 4 backward launches (pack, fused dgrad/masks/bias-sums,
+    dW2 GEMM, table scatter). Replaces ~12 torch launches (~70 us/step at
+    the flagship shape)."""
+
+    @staticmethod
+    def forward(ctx, idx, table, b1, w2, b2):
+        ext = _ops.require_ext()
+        out, a1 = ext.embed_mlp_fwd(idx, table, b1, w2, b2)
+        ctx.save_for_backward(idx, a1, out, w2)
+        ctx.A = table.shape[0]
+        ctx.set_materialize_grads(False)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        idx, a1, out, w2 = ctx.saved_tensors
+        ext = _ops.require_ext()
+        if dy is None:
+            return (None,) * 5
+        if not (dy.dim() == 2 and dy.stride(1) == 1):
+            dy = dy.contiguous()
+        dz2, da1, ws = ext.embed_mlp_bwd(dy.to(torch.bfloat16), out, a1, w2)
+        dw2 = dz2.t().mm(a1)
+        dtable = ext.embed_bwd(idx, da1, ctx.A, True)
+        dbs = ws.to(torch.bfloat16)
+        return None, dtable, dbs.narrow(0, 0, 256), dw2, \
+            dbs.narrow(0, 256, 256)
+
+
+def fused_action_embed(idx, table, b1, w2, b2):
+    """idx [N] long; table [A,256], b1 [256], w2 [256,256], b2 [256], all
+    bf16 CUDA -> [N,256] bf16 (post-ReLU)."""
+    return _FusedActionEmbed.apply(idx, table, b1, w2, b2)

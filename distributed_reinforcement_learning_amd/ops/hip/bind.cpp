@@ -137,6 +137,16 @@ extern "C" __global__ void drla_heads_wgrad(
     unsigned short*, unsigned short*, unsigned short*, const float*,
     unsigned short*, unsigned short*, unsigned short*, unsigned short*,
     unsigned short*, unsigned short*, int, int);
+extern "C" __global__ void drla_embed_mlp_fwd(
+    const long long*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, unsigned short*,
+    unsigned short*, int);
+extern "C" __global__ void drla_embed_w2t_pack(const unsigned short*,
+                                               unsigned short*);
+extern "C" __global__ void drla_embed_mlp_bwd(
+    const unsigned short*, long long, const unsigned short*,
+    const unsigned short*, const unsigned short*, unsigned short*,
+    unsigned short*, float*, int);
 extern "C" __global__ void drla_grad_gather(
     const unsigned long long*, const long long*, const long long*,
     unsigned short*, int, long long, float*, int);
@@ -736,6 +746,48 @@ std::vector<torch::Tensor> mlp_heads_bwd(
           db3v, ws};
 }
 
+std::tuple<torch::Tensor, torch::Tensor> embed_mlp_fwd(
+    torch::Tensor pa, torch::Tensor table, torch::Tensor b1,
+    torch::Tensor w2, torch::Tensor b2) {
+  for (auto* t : {&pa, &table, &b1, &w2, &b2})
+    check_gpu_contig(*t, "embed fwd input");
+  TORCH_CHECK(pa.scalar_type() == torch::kLong, "pa must be int64");
+  const int N = pa.numel();
+  auto bopt = table.options();
+  auto out = torch::empty({N, 256}, bopt);
+  auto a1 = torch::empty({N, 256}, bopt);
+  hipLaunchKernelGGL(drla_embed_mlp_fwd, dim3((N + 15) / 16), dim3(256), 0,
+                     cur_stream(),
+                     reinterpret_cast<const long long*>(pa.data_ptr<int64_t>()),
+                     u16p(table), u16p(b1), u16p(w2), u16p(b2), u16pm(out),
+                     u16pm(a1), N);
+  return {out, a1};
+}
+
+std::vector<torch::Tensor> embed_mlp_bwd(torch::Tensor dy, torch::Tensor out,
+                                         torch::Tensor a1,
+                                         torch::Tensor w2) {
+  for (auto* t : {&out, &a1, &w2}) check_gpu_contig(*t, "embed bwd input");
+  TORCH_CHECK(dy.is_cuda() && dy.dim() == 2 && dy.stride(1) == 1,
+              "dy must be row-contiguous");
+  const int N = out.size(0);
+  auto bopt = out.options();
+  // W2^T pack into a persistent buffer (overwritten fully each call)
+  static torch::Tensor w2t;
+  if (!w2t.defined()) w2t = torch::empty({65536}, bopt);
+  hipLaunchKernelGGL(drla_embed_w2t_pack, dim3(drla_grid(65536)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(), u16p(w2),
+                     u16pm(w2t));
+  auto dz2 = torch::empty({N, 256}, bopt);
+  auto da1 = torch::empty({N, 256}, bopt);
+  auto ws = torch::zeros({512}, bopt.dtype(torch::kFloat));
+  hipLaunchKernelGGL(drla_embed_mlp_bwd, dim3((N + 15) / 16), dim3(256), 0,
+                     cur_stream(), u16p(dy), (long long)dy.stride(0),
+                     u16p(out), u16p(a1), u16p(w2t), u16pm(dz2), u16pm(da1),
+                     ws.data_ptr<float>(), N);
+  return {dz2, da1, ws};
+}
+
 torch::Tensor mlp_heads_pack_wt(std::vector<torch::Tensor> weights,
                                 int64_t A) {
   TORCH_CHECK(weights.size() == 6);
@@ -1006,6 +1058,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused policy+value MLP heads forward (K4)");
   m.def("mlp_heads_bwd", &mlp_heads_bwd,
         "fused heads dgrad chain + ReLU masks + bias grads (K4 bwd)");
+  m.def("embed_mlp_fwd", &embed_mlp_fwd,
+        "fused action-embedding MLP forward (K2)");
+  m.def("embed_mlp_bwd", &embed_mlp_bwd,
+        "fused action-embedding MLP backward (K2 bwd)");
   m.def("mlp_heads_pack_wt", &mlp_heads_pack_wt,
         "one-kernel transposed-weight pack for the heads dgrad");
   m.def("mlp_heads_wgrad", &mlp_heads_wgrad,

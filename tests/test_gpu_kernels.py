@@ -303,3 +303,32 @@ def test_lstm_seq_kernel_matches_torch_loop():
         h, c = h * keep, c * keep
     assert torch.allclose(hb, h.float(), atol=3e-2)
     assert torch.allclose(cb, c.float(), atol=3e-2)
+
+
+def test_fused_action_embed_parity(ext):
+    from distributed_reinforcement_learning_amd.ops.embed_op import (
+        fused_action_embed,
+    )
+    torch.manual_seed(11)
+    N, A = 163, 18
+    idx = torch.randint(0, A, (N,), device="cuda")
+    mk = lambda *s: (torch.randn(*s, device="cuda") * 0.5).to(
+        torch.bfloat16).requires_grad_(True)
+    t1, b1, w2, b2 = mk(A, 256), mk(256), mk(256, 256), mk(256)
+    t2 = t1.detach().clone().requires_grad_(True)
+    b1r = b1.detach().clone().requires_grad_(True)
+    w2r = w2.detach().clone().requires_grad_(True)
+    b2r = b2.detach().clone().requires_grad_(True)
+
+    out_f = fused_action_embed(idx, t1, b1, w2, b2)
+    ref = torch.relu(torch.nn.functional.linear(
+        torch.relu(t2.index_select(0, idx) + b1r), w2r, b2r))
+    assert torch.allclose(out_f.float(), ref.float(), atol=0.05, rtol=0.05)
+
+    g = torch.randn(N, 256, device="cuda")
+    out_f.backward(g.to(torch.bfloat16))
+    ref.backward(g.to(torch.bfloat16))
+    for a, b, tol in [(t1.grad, t2.grad, 0.6), (b1.grad, b1r.grad, 0.5),
+                      (w2.grad, w2r.grad, 0.6), (b2.grad, b2r.grad, 0.5)]:
+        assert torch.allclose(a.float(), b.float(), atol=tol, rtol=0.05), \
+            (a.float() - b.float()).abs().max()
