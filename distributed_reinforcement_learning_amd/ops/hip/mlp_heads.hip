@@ -65,23 +65,41 @@ __device__ void mh_pass(const bf16raw (*act_in)[MH_LD],
 
   const bf16x8 zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
   if (live) {
-    for (int k0 = 0; k0 < kdim; k0 += 32) {
-      // a_frag per ni: M[out0 + ni*16 + (l&15)][k0 + (l>>4)*8 ..+8];
-      // rows beyond nout do not exist (lane-predicated load, guide:
-      // garbage rows can carry NaN bits and 0*NaN = NaN)
-      for (int ni = 0; ni < 4; ++ni) {
-        const int orow = out0 + ni * 16 + (lane & 15);
-        const bf16x8 a_frag =
-            (orow < nout)
-                ? *reinterpret_cast<const bf16x8*>(
-                      M + (long long)orow * kdim + k0 + (lane >> 4) * 8)
-                : zero8;
-        const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
-            &act_in[lane & 15][k0 + (lane >> 4) * 8]);
-        acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_frag, b_frag, acc[ni], 0, 0, 0);
-      }
+    // software-pipelined k-loop: the next chunk's global weight fragments
+    // load into registers while this chunk's MFMAs run (the serial
+    // load->mfma chain across ~8 chunks x 3 chained passes is what sets
+    // these kernels' time — 40-80 blocks can't hide it with occupancy)
+    bf16x8 a_cur0, a_cur1, a_cur2, a_cur3;
+    const int kf = (lane >> 4) * 8;
+#define DRLA_MH_LOAD(ni, dst, k0)                                       \
+    {                                                                   \
+      const int orow = out0 + (ni) * 16 + (lane & 15);                  \
+      dst = (orow < nout)                                               \
+          ? *reinterpret_cast<const bf16x8*>(                           \
+                M + (long long)orow * kdim + (k0) + kf)                 \
+          : zero8;                                                      \
     }
+    DRLA_MH_LOAD(0, a_cur0, 0); DRLA_MH_LOAD(1, a_cur1, 0);
+    DRLA_MH_LOAD(2, a_cur2, 0); DRLA_MH_LOAD(3, a_cur3, 0);
+    for (int k0 = 0; k0 < kdim; k0 += 32) {
+      bf16x8 a_nxt0, a_nxt1, a_nxt2, a_nxt3;
+      if (k0 + 32 < kdim) {
+        DRLA_MH_LOAD(0, a_nxt0, k0 + 32); DRLA_MH_LOAD(1, a_nxt1, k0 + 32);
+        DRLA_MH_LOAD(2, a_nxt2, k0 + 32); DRLA_MH_LOAD(3, a_nxt3, k0 + 32);
+      }
+      const bf16x8 b_frag =
+          *reinterpret_cast<const bf16x8*>(&act_in[lane & 15][k0 + kf]);
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur0, b_frag,
+                                                       acc[0], 0, 0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur1, b_frag,
+                                                       acc[1], 0, 0, 0);
+      acc[2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur2, b_frag,
+                                                       acc[2], 0, 0, 0);
+      acc[3] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_cur3, b_frag,
+                                                       acc[3], 0, 0, 0);
+      a_cur0 = a_nxt0; a_cur1 = a_nxt1; a_cur2 = a_nxt2; a_cur3 = a_nxt3;
+    }
+#undef DRLA_MH_LOAD
   }
   __syncthreads();
   // epilogue: D[out][row] un-transpose -> act_out[row][out] (+stash/dz/dh);
