@@ -59,12 +59,10 @@ class _FusedActionEmbed(torch.autograd.Function):
             return (None,) * 5
         if not (dy.dim() == 2 and dy.stride(1) == 1):
             dy = dy.contiguous()
-        dz2, da1, ws = ext.embed_mlp_bwd(dy.to(torch.bfloat16), out, a1, w2)
+        dz2, dtable, db1, db2 = ext.embed_mlp_bwd(
+            dy.to(torch.bfloat16), out, a1, w2, idx, ctx.A)
         dw2 = dz2.t().mm(a1)
-        dtable = ext.embed_bwd(idx, da1, ctx.A, True)
-        dbs = ws.to(torch.bfloat16)
-        return None, dtable, dbs.narrow(0, 0, 256), dw2, \
-            dbs.narrow(0, 256, 256)
+        return None, dtable, db1, dw2, db2
 
 
 def fused_action_embed(idx, table, b1, w2, b2):

@@ -147,6 +147,9 @@ extern "C" __global__ void drla_embed_mlp_bwd(
     const unsigned short*, long long, const unsigned short*,
     const unsigned short*, const unsigned short*, unsigned short*,
     unsigned short*, float*, int);
+extern "C" __global__ void drla_embed_finalize(float*, unsigned short*,
+                                               unsigned short*,
+                                               unsigned short*, long long);
 extern "C" __global__ void drla_grad_gather(
     const unsigned long long*, const long long*, const long long*,
     unsigned short*, int, long long, float*, int);
@@ -765,27 +768,46 @@ std::tuple<torch::Tensor, torch::Tensor> embed_mlp_fwd(
 }
 
 std::vector<torch::Tensor> embed_mlp_bwd(torch::Tensor dy, torch::Tensor out,
-                                         torch::Tensor a1,
-                                         torch::Tensor w2) {
+                                         torch::Tensor a1, torch::Tensor w2,
+                                         torch::Tensor idx, int64_t A) {
   for (auto* t : {&out, &a1, &w2}) check_gpu_contig(*t, "embed bwd input");
+  check_gpu_contig(idx, "idx");
   TORCH_CHECK(dy.is_cuda() && dy.dim() == 2 && dy.stride(1) == 1,
               "dy must be row-contiguous");
   const int N = out.size(0);
+  const long long n_table = (long long)A * 256;
   auto bopt = out.options();
-  // W2^T pack into a persistent buffer (overwritten fully each call)
-  static torch::Tensor w2t;
+  // persistent buffers: W2^T (fully overwritten) and the f32 partial
+  // scratch (zero-between-calls; drla_embed_finalize re-zeroes it)
+  static torch::Tensor w2t, scratch;
   if (!w2t.defined()) w2t = torch::empty({65536}, bopt);
+  if (!scratch.defined() || scratch.numel() != n_table + 512) {
+    scratch = torch::zeros({n_table + 512}, bopt.dtype(torch::kFloat));
+  }
   hipLaunchKernelGGL(drla_embed_w2t_pack, dim3(drla_grid(65536)),
                      dim3(DRLA_BLOCK), 0, cur_stream(), u16p(w2),
                      u16pm(w2t));
   auto dz2 = torch::empty({N, 256}, bopt);
   auto da1 = torch::empty({N, 256}, bopt);
-  auto ws = torch::zeros({512}, bopt.dtype(torch::kFloat));
+  float* bias_ws = scratch.data_ptr<float>() + n_table;
   hipLaunchKernelGGL(drla_embed_mlp_bwd, dim3((N + 15) / 16), dim3(256), 0,
                      cur_stream(), u16p(dy), (long long)dy.stride(0),
                      u16p(out), u16p(a1), u16p(w2t), u16pm(dz2), u16pm(da1),
-                     ws.data_ptr<float>(), N);
-  return {dz2, da1, ws};
+                     bias_ws, N);
+  // table scatter into the f32 scratch, then one 3-output finalize
+  hipLaunchKernelGGL(
+      drla_embed_bwd_scatter, dim3(drla_grid((long long)N * 256)),
+      dim3(DRLA_BLOCK), 0, cur_stream(),
+      reinterpret_cast<const long long*>(idx.data_ptr<int64_t>()),
+      u16p(da1), nullptr, scratch.data_ptr<float>(), N, 256);
+  auto dtable = torch::empty({A, 256}, bopt);
+  auto db1 = torch::empty({256}, bopt);
+  auto db2 = torch::empty({256}, bopt);
+  hipLaunchKernelGGL(drla_embed_finalize, dim3(drla_grid(n_table + 512)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     scratch.data_ptr<float>(), u16pm(dtable), u16pm(db1),
+                     u16pm(db2), n_table);
+  return {dz2, dtable, db1, db2};
 }
 
 torch::Tensor mlp_heads_pack_wt(std::vector<torch::Tensor> weights,

@@ -623,3 +623,21 @@ extern "C" __global__ __launch_bounds__(256) void drla_embed_mlp_bwd(
                         colsum, nullptr, nullptr, row0, N, MH_HID, MH_HID);
   if (tid < MH_HID) atomicAdd(&bias_ws[tid], colsum[tid]);
 }
+
+// finalize for the fused embed backward: cast the persistent f32 scratch
+// ([A*256] table-grad partials ++ [512] bias partials) to three contiguous
+// bf16 outputs, re-zeroing the scratch (zero-between-calls invariant).
+extern "C" __global__ void drla_embed_finalize(
+    float* __restrict__ src, bf16raw* __restrict__ dtable,
+    bf16raw* __restrict__ db1, bf16raw* __restrict__ db2,
+    long long n_table) {
+  const long long total = n_table + 512;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += gridDim.x * (long long)blockDim.x) {
+    const bf16raw v = drla_f32_to_bf16(src[i]);
+    src[i] = 0.0f;
+    if (i < n_table) dtable[i] = v;
+    else if (i < n_table + 256) db1[i - n_table] = v;
+    else db2[i - n_table - 256] = v;
+  }
+}
