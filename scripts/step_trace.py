@@ -52,6 +52,5 @@ with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
         one_step()
     torch.cuda.synchronize()
 
-print(prof.key_averages(group_by_input_shape=True).table(
-    sort_by="self_cuda_time_total", row_limit=60,
-    max_name_column_width=46))
+print(prof.key_averages().table(sort_by="self_cuda_time_total", row_limit=45,
+                                max_name_column_width=60))
