@@ -79,7 +79,7 @@ def bench_apex(steps=200, warmup=30):
         "algo": "apex", "ms_per_train_step": dt * 1000,
         "train_steps_per_s": 1 / dt,
         "transitions_per_s": cfg.batch_size / dt,
-        "batch": cfg.batch_size}))
+        "batch": cfg.batch_size}), flush=True)
 
 
 def bench_r2d2(steps=200, warmup=30):
@@ -104,8 +104,8 @@ def bench_r2d2(steps=200, warmup=30):
         "action": ((L,), torch.int32),
         "reward": ((L,), torch.float32),
         "done": ((L,), torch.bool),
-        "initial_h": ((cfg.lstm_size,), torch.float32),
-        "initial_c": ((cfg.lstm_size,), torch.float32)},
+        "initial_h": ((L, cfg.lstm_size), torch.float32),
+        "initial_c": ((L, cfg.lstm_size), torch.float32)},
         device=dev, seed=3)
     rng = np.random.default_rng(0)
     for _ in range(40):
@@ -120,12 +120,12 @@ def bench_r2d2(steps=200, warmup=30):
             "reward": torch.as_tensor(
                 rng.normal(size=(B, L)).astype(np.float32)).to(dev),
             "done": torch.as_tensor(rng.random((B, L)) < 0.02).to(dev),
-            "initial_h": torch.zeros(B, cfg.lstm_size, device=dev),
-            "initial_c": torch.zeros(B, cfg.lstm_size, device=dev)}
+            "initial_h": torch.zeros(B, L, cfg.lstm_size, device=dev),
+            "initial_c": torch.zeros(B, L, cfg.lstm_size, device=dev)}
         td = agent.get_td_error_batch(
             dev_u["state"], dev_u["previous_action"], dev_u["action"],
-            dev_u["initial_h"], dev_u["initial_c"], dev_u["reward"],
-            dev_u["done"], as_tensor=True)
+            dev_u["initial_h"][:, 0], dev_u["initial_c"][:, 0],
+            dev_u["reward"], dev_u["done"], as_tensor=True)
         mem.add_batch(td, dev_u)
 
     def one():
@@ -149,7 +149,8 @@ def bench_r2d2(steps=200, warmup=30):
         "algo": "r2d2", "ms_per_train_step": dt * 1000,
         "train_steps_per_s": 1 / dt,
         "sequences_per_s": cfg.batch_size / dt,
-        "batch": cfg.batch_size, "seq_len": L, "burn_in": cfg.burn_in}))
+        "batch": cfg.batch_size, "seq_len": L, "burn_in": cfg.burn_in}),
+        flush=True)
 
 
 if __name__ == "__main__":
