@@ -119,9 +119,15 @@ class Agent(AgentBase):
             )
             clipped_r = clip_rewards(r, self.reward_clipping)
             discounts = (~d).float() * self.discount_factor
-            main_q = self.model(s, pa)
+            # ONE main-net forward over [s ; ns] (the no-grad next-state
+            # eval rides in the same launches; at B=32 these kernels are
+            # latency-bound, so the extra backward rows are nearly free
+            # while a whole forward's worth of launches disappears)
+            B = s.shape[0]
+            qs = self.model(torch.cat([s, ns]), torch.cat([pa, a]))
+            main_q = qs[:B]
+            next_main_q = qs[B:].detach()
             with torch.no_grad():
-                next_main_q = self.model(ns, a)
                 next_target_q = self.target_model(ns, a)
             loss, td_signed = fused_dqn_loss(
                 main_q, next_main_q, next_target_q, a, clipped_r, discounts,
@@ -135,8 +141,10 @@ class Agent(AgentBase):
             self.num_env_frames += len(r)
             td_error = td_signed.detach().abs()
             if not as_tensor:
-                td_error = td_error.cpu().numpy()
-            return float(loss.detach()), td_error
+                return float(loss.detach()), td_error.cpu().numpy()
+            # as_tensor path: no host sync — callers float() the loss at
+            # logging cadence only
+            return loss.detach(), td_error
         target_value, sav = self._targets(s, ns, pa, a, r, d)
         td_sq = (target_value.detach() - sav) ** 2
         loss = (td_sq * w).mean()

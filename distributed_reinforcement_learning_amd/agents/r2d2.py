@@ -192,6 +192,8 @@ class Agent(AgentBase):
         td = (target_value - sav).mean(dim=1).abs().detach()
         if not as_tensor:
             td = td.cpu().numpy()
+        if as_tensor:
+            return loss.detach(), td
         return float(loss.detach()), td
 
     @torch.no_grad()
