@@ -75,10 +75,9 @@ class R2D2LstmQ(nn.Module):
             seq_prev_action.reshape(B * L),
         ).reshape(B, L, -1)
 
-        use_seq_kernel = (feat.is_cuda and not torch.is_grad_enabled()
-                          and self.lstm.weight.dtype == torch.bfloat16
-                          and 4 * self.lstm_size <= 1024)
-        if use_seq_kernel:
+        kern_ok = (feat.is_cuda and self.lstm.weight.dtype == torch.bfloat16
+                   and 4 * self.lstm_size <= 1024)
+        if kern_ok and not torch.is_grad_enabled():
             from distributed_reinforcement_learning_amd import ops as _o
             ext = _o.require_ext()
             F = feat.shape[-1]
@@ -89,6 +88,21 @@ class R2D2LstmQ(nn.Module):
                 w[F:].contiguous(), h0.float().contiguous(),
                 c0.float().contiguous(), seq_done.contiguous(),
                 self.lstm.forget_bias)
+            q = self._head(h_all.reshape(B * L, -1))
+            return q.reshape(B, L, -1)
+        if kern_ok:
+            # trained window: grad-carrying one-kernel recurrence
+            # (ops/lstm_op.lstm_seq_train); x-projection, dWh and bias
+            # grads are GEMM-shaped and ride on autograd/hipBLASLt
+            from distributed_reinforcement_learning_amd.ops.lstm_op import (
+                lstm_seq_train,
+            )
+            F = feat.shape[-1]
+            w = self.lstm.weight
+            xg = torch.addmm(self.lstm.bias, feat.reshape(B * L, F), w[:F])
+            h_all, _, _ = lstm_seq_train(
+                xg.reshape(B, L, -1), w[F:], h0.float(), c0.float(),
+                seq_done, self.lstm.forget_bias)
             q = self._head(h_all.reshape(B * L, -1))
             return q.reshape(B, L, -1)
 

@@ -153,6 +153,14 @@ extern "C" __global__ void drla_embed_finalize(float*, unsigned short*,
 extern "C" __global__ void drla_grad_gather(
     const unsigned long long*, const long long*, const long long*,
     unsigned short*, int, long long, float*, int);
+extern "C" __global__ void drla_lstm_seq_train_fwd(
+    const unsigned short*, const unsigned short*, const float*, const float*,
+    const unsigned char*, float*, float*, float*, float*, float*,
+    unsigned short*, float, int, int, int);
+extern "C" __global__ void drla_lstm_seq_train_bwd(
+    const float*, const float*, const float*, const float*, const float*,
+    const unsigned short*, const unsigned char*, unsigned short*, float*,
+    float*, int, int, int);
 extern "C" __global__ void drla_lstm_seq_fwd(
     const unsigned short*, const float*, const unsigned short*, const float*,
     const float*, const unsigned char*, float*, float*, float*, float, int,
@@ -914,6 +922,60 @@ void grad_gather(torch::Tensor srcs, torch::Tensor offs,
       u16pm(dst), (int)srcs.numel(), chunks, nw, nw_n);
 }
 
+std::vector<torch::Tensor> lstm_seq_train_fwd(
+    torch::Tensor xg, torch::Tensor Wh, torch::Tensor h0, torch::Tensor c0,
+    torch::Tensor done, double forget_bias) {
+  for (auto* t : {&xg, &Wh, &h0, &c0, &done})
+    check_gpu_contig(*t, "lstm_seq_train input");
+  TORCH_CHECK(xg.scalar_type() == torch::kBFloat16 &&
+              Wh.scalar_type() == torch::kBFloat16);
+  const int B = xg.size(0), L = xg.size(1), H = xg.size(2) / 4;
+  TORCH_CHECK(4 * H <= 1024 && (long long)H * 4 * H * 2 <= 120 * 1024,
+              "lstm seq-train caps: 4H<=1024 and Wh must fit LDS");
+  auto fopt = h0.options().dtype(torch::kFloat);
+  auto h_out = torch::empty({B, L, H}, fopt);
+  auto h_fin = torch::empty({B, H}, fopt);
+  auto c_fin = torch::empty({B, H}, fopt);
+  auto acts = torch::empty({B, L, 4 * H}, fopt);
+  auto c_prev = torch::empty({B, L, H}, fopt);
+  auto h_prev = torch::empty({B, L, H}, xg.options());
+  const int lds = H * 4 * H * 2 + 2 * H * 4 + 16;
+  hipLaunchKernelGGL(drla_lstm_seq_train_fwd, dim3(B), dim3(4 * H), lds,
+                     cur_stream(), u16p(xg), u16p(Wh),
+                     h0.data_ptr<float>(), c0.data_ptr<float>(),
+                     reinterpret_cast<const unsigned char*>(
+                         done.data_ptr<bool>()),
+                     h_out.data_ptr<float>(), h_fin.data_ptr<float>(),
+                     c_fin.data_ptr<float>(), acts.data_ptr<float>(),
+                     c_prev.data_ptr<float>(), u16pm(h_prev),
+                     static_cast<float>(forget_bias), B, L, H);
+  return {h_out, h_fin, c_fin, acts, c_prev, h_prev};
+}
+
+std::vector<torch::Tensor> lstm_seq_train_bwd(
+    torch::Tensor dh_out, c10::optional<torch::Tensor> dh_fin,
+    c10::optional<torch::Tensor> dc_fin, torch::Tensor acts,
+    torch::Tensor c_prev, torch::Tensor Wh, torch::Tensor done) {
+  for (auto* t : {&dh_out, &acts, &c_prev, &Wh, &done})
+    check_gpu_contig(*t, "lstm_seq_train bwd input");
+  const int B = acts.size(0), L = acts.size(1), H = acts.size(2) / 4;
+  auto fopt = dh_out.options().dtype(torch::kFloat);
+  auto dxg = torch::empty({B, L, 4 * H},
+                          dh_out.options().dtype(torch::kBFloat16));
+  auto dh0 = torch::empty({B, H}, fopt);
+  auto dc0 = torch::empty({B, H}, fopt);
+  const int lds = H * 4 * H * 2 + 2 * H * 4 + 16;
+  hipLaunchKernelGGL(
+      drla_lstm_seq_train_bwd, dim3(B), dim3(4 * H), lds, cur_stream(),
+      dh_out.data_ptr<float>(),
+      dh_fin.has_value() ? dh_fin->data_ptr<float>() : nullptr,
+      dc_fin.has_value() ? dc_fin->data_ptr<float>() : nullptr,
+      acts.data_ptr<float>(), c_prev.data_ptr<float>(), u16p(Wh),
+      reinterpret_cast<const unsigned char*>(done.data_ptr<bool>()),
+      u16pm(dxg), dh0.data_ptr<float>(), dc0.data_ptr<float>(), B, L, H);
+  return {dxg, dh0, dc0};
+}
+
 torch::Tensor sq_norm(torch::Tensor x) {
   check_gpu_contig(x, "x");
   TORCH_CHECK(x.scalar_type() == torch::kFloat, "sq_norm wants float32");
@@ -1088,6 +1150,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "one-kernel transposed-weight pack for the heads dgrad");
   m.def("mlp_heads_wgrad", &mlp_heads_wgrad,
         "all six head wgrads (dW = dz^T @ act) in one MFMA launch");
+  m.def("lstm_seq_train_fwd", &lstm_seq_train_fwd,
+        "grad-carrying whole-sequence LSTM forward (K3 seq, R2D2 train)");
+  m.def("lstm_seq_train_bwd", &lstm_seq_train_bwd,
+        "whole-sequence LSTM backward (one kernel over the recurrence)");
   m.def("lstm_seq_fwd", &lstm_seq_fwd,
         "whole no-grad LSTM unroll in one kernel (K3 seq / burn-in)");
   m.def("sq_norm", &sq_norm, "squared L2 norm of a flat tensor (K12)");

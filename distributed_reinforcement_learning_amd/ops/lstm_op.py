@@ -56,3 +56,43 @@ def lstm_fused_step(gates: torch.Tensor, c_prev: torch.Tensor,
         + torch.sigmoid(i) * torch.tanh(g)
     new_h = torch.sigmoid(o) * torch.tanh(new_c)
     return new_h, new_c
+
+
+class _LstmSeqTrain(torch.autograd.Function):
+    """Whole done-masked LSTM recurrence (R2D2 trained window) as ONE
+    kernel each way; dWh/dbias/x-projection grads are GEMM-shaped and stay
+    outside (ops/hip/lstm_gates.hip seq-train pair)."""
+
+    @staticmethod
+    def forward(ctx, xg, wh, h0, c0, done, forget_bias):
+        ext = _ops.require_ext()
+        h_out, h_fin, c_fin, acts, c_prev, h_prev = ext.lstm_seq_train_fwd(
+            xg.contiguous(), wh.contiguous(), h0.contiguous(),
+            c0.contiguous(), done.contiguous(), forget_bias)
+        ctx.save_for_backward(acts, c_prev, h_prev, wh, done)
+        ctx.set_materialize_grads(False)
+        return h_out, h_fin, c_fin
+
+    @staticmethod
+    def backward(ctx, dh_out, dh_fin, dc_fin):
+        acts, c_prev, h_prev, wh, done = ctx.saved_tensors
+        ext = _ops.require_ext()
+        B, L, H = c_prev.shape
+        if dh_out is None:
+            dh_out = torch.zeros(B, L, H, dtype=torch.float32,
+                                 device=c_prev.device)
+        dxg, dh0, dc0 = ext.lstm_seq_train_bwd(
+            dh_out.float().contiguous(),
+            None if dh_fin is None else dh_fin.float().contiguous(),
+            None if dc_fin is None else dc_fin.float().contiguous(),
+            acts, c_prev, wh, done)
+        # dWh = sum_t h_prev_t^T dgates_t — one GEMM over B*L rows
+        dwh = h_prev.reshape(B * L, H).t().mm(dxg.reshape(B * L, 4 * H))
+        return dxg, dwh, dh0, dc0, None, None
+
+
+def lstm_seq_train(xg, wh, h0, c0, done, forget_bias: float = 1.0):
+    """xg [B,L,4H] bf16 (x-projection + bias, grad ok), wh [H,4H] bf16
+    (grad ok), h0/c0 [B,H] f32, done [B,L] bool ->
+    (h_out [B,L,H] f32, h_fin, c_fin)."""
+    return _LstmSeqTrain.apply(xg, wh, h0, c0, done, float(forget_bias))

@@ -332,3 +332,51 @@ def test_fused_action_embed_parity(ext):
                       (w2.grad, w2r.grad, 0.6), (b2.grad, b2r.grad, 0.5)]:
         assert torch.allclose(a.float(), b.float(), atol=tol, rtol=0.05), \
             (a.float() - b.float()).abs().max()
+
+
+def test_lstm_seq_train_parity(ext):
+    from distributed_reinforcement_learning_amd.ops.lstm_op import (
+        lstm_seq_train,
+    )
+    torch.manual_seed(21)
+    B, L, H = 5, 9, 64
+    xg = (torch.randn(B, L, 4 * H, device="cuda") * 0.5).to(
+        torch.bfloat16).requires_grad_(True)
+    wh = (torch.randn(H, 4 * H, device="cuda") * 0.2).to(
+        torch.bfloat16).requires_grad_(True)
+    h0 = torch.randn(B, H, device="cuda").requires_grad_(True)
+    c0 = torch.randn(B, H, device="cuda").requires_grad_(True)
+    done = torch.rand(B, L, device="cuda") < 0.15
+
+    h_out, h_fin, c_fin = lstm_seq_train(xg, wh, h0, c0, done, 1.0)
+    g = torch.randn_like(h_out)
+    (h_out * g).sum().backward()
+
+    # python reference (fp32 math on the same bf16 inputs)
+    xg2 = xg.detach().clone().requires_grad_(True)
+    wh2 = wh.detach().clone().requires_grad_(True)
+    h02 = h0.detach().clone().requires_grad_(True)
+    c02 = c0.detach().clone().requires_grad_(True)
+    h, c = h02, c02
+    outs = []
+    for t in range(L):
+        gates = xg2[:, t].float() + h.float() @ wh2.float()
+        i, gg, f, o = gates.chunk(4, dim=1)
+        c_new = torch.sigmoid(f + 1.0) * c + torch.sigmoid(i) * torch.tanh(gg)
+        h_new = torch.sigmoid(o) * torch.tanh(c_new)
+        outs.append(h_new)
+        keep = (~done[:, t]).float().unsqueeze(1)
+        h, c = h_new * keep, c_new * keep
+    ref = torch.stack(outs, dim=1)
+    (ref * g).sum().backward()
+
+    assert torch.allclose(h_out, ref, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(h_fin, h, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(c_fin, c, atol=3e-2, rtol=3e-2)
+    assert torch.allclose(xg.grad.float(), xg2.grad.float(), atol=0.05,
+                          rtol=0.05)
+    assert torch.allclose(wh.grad.float(), wh2.grad.float(), atol=0.35,
+                          rtol=0.1), (wh.grad.float() -
+                                      wh2.grad.float()).abs().max()
+    assert torch.allclose(h0.grad, h02.grad, atol=0.05, rtol=0.05)
+    assert torch.allclose(c0.grad, c02.grad, atol=0.05, rtol=0.05)
