@@ -66,9 +66,11 @@ class _LstmSeqTrain(torch.autograd.Function):
     @staticmethod
     def forward(ctx, xg, wh, h0, c0, done, forget_bias):
         ext = _ops.require_ext()
+        wh = wh.contiguous()
+        done = done.contiguous()
         h_out, h_fin, c_fin, acts, c_prev, h_prev = ext.lstm_seq_train_fwd(
-            xg.contiguous(), wh.contiguous(), h0.contiguous(),
-            c0.contiguous(), done.contiguous(), forget_bias)
+            xg.contiguous(), wh, h0.contiguous(),
+            c0.contiguous(), done, forget_bias)
         ctx.save_for_backward(acts, c_prev, h_prev, wh, done)
         ctx.set_materialize_grads(False)
         return h_out, h_fin, c_fin
