@@ -57,12 +57,19 @@ class FlatAllReducer:
     def __init__(self, flat_grads: torch.Tensor):
         self.flat_grads = flat_grads
         self._inv_world = 1.0 / world_size()
+        # RCCL fuses the divide into the collective (ReduceOp.AVG);
+        # gloo (CPU tests) lacks AVG -> sum + scale fallback
+        self._use_avg = (is_distributed()
+                         and dist.get_backend() == "nccl")
 
     def all_reduce(self) -> None:
         if not is_distributed():
             return
-        dist.all_reduce(self.flat_grads, op=dist.ReduceOp.SUM)
-        self.flat_grads.mul_(self._inv_world)
+        if self._use_avg:
+            dist.all_reduce(self.flat_grads, op=dist.ReduceOp.AVG)
+        else:
+            dist.all_reduce(self.flat_grads, op=dist.ReduceOp.SUM)
+            self.flat_grads.mul_(self._inv_world)
 
 
 def broadcast_module(module: torch.nn.Module, src: int = 0) -> None:
