@@ -109,17 +109,35 @@ class GraphedImpalaStep:
         # .grad now holds capture-pool tensors at replay-stable addresses
         opt.build_gather_table()
         from distributed_reinforcement_learning_amd.parallel.dist import (
-            is_distributed,
+            is_distributed, world_size,
         )
-        self._distributed = is_distributed()
+        self._distributed = is_distributed() and world_size() > 1
+        self._eager_reduce = False
         self.g_opt = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_opt, pool=self.g_fwd.pool()):
-            if not self._distributed:
-                # single-GPU: the gather rides inside the optimizer graph;
-                # distributed keeps it eager so the RCCL all-reduce sees
-                # the packed bucket between the two replays
+        if self._distributed:
+            # capture gather + RCCL all-reduce + update as ONE graph
+            # (NCCL/RCCL collectives are capture-legal and every rank
+            # replays in lockstep); eager fallback if capture refuses —
+            # an eager all-reduce between replays costs ~0.4 ms/step of
+            # host latency (measured, gpurun_out/dist1.log)
+            try:
+                # prime the communicator outside capture
+                agent.reduce_gradients()
+                torch.cuda.synchronize()
+                with torch.cuda.graph(self.g_opt, pool=self.g_fwd.pool()):
+                    opt.gather_grads()
+                    agent.reduce_gradients()
+                    opt.step_tensor_lr(self.lr_buf)
+            except Exception:
+                self._eager_reduce = True
+                self.g_opt = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(self.g_opt, pool=self.g_fwd.pool()):
+                    opt.step_tensor_lr(self.lr_buf)
+        else:
+            with torch.cuda.graph(self.g_opt, pool=self.g_fwd.pool()):
+                # single-GPU: the gather rides inside the optimizer graph
                 opt.gather_grads()
-            opt.step_tensor_lr(self.lr_buf)
+                opt.step_tensor_lr(self.lr_buf)
 
     def _fwd(self):
         agent = self.agent
@@ -212,7 +230,7 @@ class GraphedImpalaStep:
             # consumed the PREVIOUS call's upload (one-batch pipeline)
             self._upload_overlapped(src)
         self.g_bwd.replay()
-        if self._distributed:
+        if self._distributed and self._eager_reduce:
             agent.optimizer.gather_grads()
             agent.reduce_gradients()
         self.g_opt.replay()
