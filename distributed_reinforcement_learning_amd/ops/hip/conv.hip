@@ -339,33 +339,6 @@ extern "C" __global__ void drla_relu_mask_bwd(
   }
 }
 
-// dbias[co] = sum_m dY[m][co]  (f32 out, zeroed by caller)
-// grid: (ceil(CO/64), M_SPLIT) — blockIdx.y slices M so hundreds of blocks
-// stream dY; partials atomicAdd per block.
-extern "C" __global__ void drla_bias_grad(
-    const bf16raw* __restrict__ dy, float* __restrict__ dbias, long long M,
-    int CO) {
-  const int co = blockIdx.x * 64 + (threadIdx.x % 64);
-  const bool live = co < CO;
-  const int mslice = threadIdx.x / 64;     // 4 slices within the block
-  const long long per = (M + gridDim.y - 1) / gridDim.y;
-  const long long m_begin = blockIdx.y * per;
-  const long long m_end = min(M, m_begin + per);
-  float acc = 0.0f;
-  if (live) {
-    for (long long m = m_begin + mslice; m < m_end; m += 4) {
-      acc += cv_bf2f(dy[m * CO + co]);
-    }
-  }
-  __shared__ float red[4][64];
-  red[mslice][threadIdx.x % 64] = acc;
-  __syncthreads();
-  if (mslice == 0 && live) {
-    atomicAdd(&dbias[co], red[0][threadIdx.x] + red[1][threadIdx.x] +
-                              red[2][threadIdx.x] + red[3][threadIdx.x]);
-  }
-}
-
 // cast + transpose the f32 [K][CO] wgrad scratch into bf16 [CO][K]
 // (the channels_last weight-grad layout), re-zeroing the persistent
 // scratch as it reads (zero-between-calls invariant, no fill kernel).
