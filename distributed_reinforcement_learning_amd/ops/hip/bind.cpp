@@ -83,9 +83,6 @@ extern "C" __global__ void drla_per_sample(const float*, const float*,
 extern "C" __global__ void drla_embed_bwd_scatter(
     const long long*, const unsigned short*, const float*, float*,
     long long, int);
-extern "C" __global__ void drla_f32_to_bf16_kernel(const float*,
-                                                   unsigned short*,
-                                                   long long);
 extern "C" __global__ void drla_f32_to_bf16_zero_kernel(float*,
                                                         unsigned short*,
                                                         long long);

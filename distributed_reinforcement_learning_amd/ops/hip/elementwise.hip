@@ -79,15 +79,6 @@ extern "C" __global__ void drla_embed_bwd_scatter(
   }
 }
 
-extern "C" __global__ void drla_f32_to_bf16_kernel(
-    const float* __restrict__ in, unsigned short* __restrict__ out,
-    long long n) {
-  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-  const long long stride = gridDim.x * (long long)blockDim.x;
-  for (; i < n; i += stride) out[i] = drla_f32_to_bf16(in[i]);
-}
-
-
 // variant that zeroes the f32 source as it reads (persistent-scratch
 // consumers: embed_bwd keeps one zero-between-calls scratch)
 extern "C" __global__ void drla_f32_to_bf16_zero_kernel(
