@@ -103,6 +103,29 @@ class Agent(AgentBase):
         td = (target_value - sav).abs()
         return td if as_tensor else td.cpu().numpy()
 
+    def compute_distributed_loss(self, s, ns, pa, a, r, d, w):
+        """Pure GPU loss body (device tensors in, no optimizer/python
+        state) — shared by the eager distributed_train and the hipGraph-
+        captured replay step (runtime/replay_graphed.py).
+
+        Fused K8 path: double-DQN target + IS-weighted TD loss and its
+        closed-form backward in one kernel each (ops/hip/dqn_loss.hip);
+        ONE main-net forward over [s ; ns] (the no-grad next-state eval
+        rides in the same launches)."""
+        from distributed_reinforcement_learning_amd.ops.dqn_op import (
+            fused_dqn_loss,
+        )
+        clipped_r = clip_rewards(r, self.reward_clipping)
+        discounts = (~d).float() * self.discount_factor
+        B = s.shape[0]
+        qs = self.model(torch.cat([s, ns]), torch.cat([pa, a]))
+        main_q = qs[:B]
+        next_main_q = qs[B:].detach()
+        with torch.no_grad():
+            next_target_q = self.target_model(ns, a)
+        return fused_dqn_loss(main_q, next_main_q, next_target_q, a,
+                              clipped_r, discounts, w)
+
     def distributed_train(self, state, next_state, previous_action, action,
                           reward, done, is_weight, as_tensor: bool = False
                           ) -> Tuple[float, np.ndarray]:
@@ -112,26 +135,8 @@ class Agent(AgentBase):
                                         action, reward, done)
         w = self.to_device(is_weight, torch.float32)
         if self.device.type == "cuda":
-            # fused K8 path: double-DQN target + IS-weighted TD loss and its
-            # closed-form backward in one kernel each (ops/hip/dqn_loss.hip)
-            from distributed_reinforcement_learning_amd.ops.dqn_op import (
-                fused_dqn_loss,
-            )
-            clipped_r = clip_rewards(r, self.reward_clipping)
-            discounts = (~d).float() * self.discount_factor
-            # ONE main-net forward over [s ; ns] (the no-grad next-state
-            # eval rides in the same launches; at B=32 these kernels are
-            # latency-bound, so the extra backward rows are nearly free
-            # while a whole forward's worth of launches disappears)
-            B = s.shape[0]
-            qs = self.model(torch.cat([s, ns]), torch.cat([pa, a]))
-            main_q = qs[:B]
-            next_main_q = qs[B:].detach()
-            with torch.no_grad():
-                next_target_q = self.target_model(ns, a)
-            loss, td_signed = fused_dqn_loss(
-                main_q, next_main_q, next_target_q, a, clipped_r, discounts,
-                w)
+            loss, td_signed = self.compute_distributed_loss(s, ns, pa, a, r,
+                                                            d, w)
             self.optimizer.zero_grad()
             loss.backward()
             self.reduce_gradients()

@@ -169,6 +169,17 @@ class Agent(AgentBase):
         td = (target_value - sav).mean(dim=1).abs()
         return td if as_tensor else td.cpu().numpy()
 
+    def compute_sequence_loss(self, state, previous_action, action, h0,
+                              c0, reward, done, w):
+        """Pure loss body (shared by train and the graphed replay step):
+        returns (weighted scalar loss, per-sequence |mean TD| tensor)."""
+        unweighted, target_value, sav = self._sequence_losses(
+            state, previous_action, action, h0, c0, reward, done,
+            with_grad=True)
+        loss = (unweighted * w).mean()
+        td = (target_value - sav).mean(dim=1).abs().detach()
+        return loss, td
+
     def train(self, state, previous_action, action, h, c, reward, done,
               weight, as_tensor: bool = False) -> Tuple[float, np.ndarray]:
         """IS-weighted batch update; h/c [B, L, H] stored per step — the
@@ -176,11 +187,9 @@ class Agent(AgentBase):
         Returns (loss, per-sequence |mean TD|)."""
         h0 = h[:, 0] if isinstance(h, torch.Tensor) else np.asarray(h)[:, 0]
         c0 = c[:, 0] if isinstance(c, torch.Tensor) else np.asarray(c)[:, 0]
-        unweighted, target_value, sav = self._sequence_losses(
-            state, previous_action, action, h0, c0, reward, done,
-            with_grad=True)
         w = self.to_device(weight, torch.float32)
-        loss = (unweighted * w).mean()
+        loss, td_t = self.compute_sequence_loss(
+            state, previous_action, action, h0, c0, reward, done, w)
 
         self.optimizer.zero_grad()
         loss.backward()
@@ -189,12 +198,9 @@ class Agent(AgentBase):
         self.optimizer.step(lr=lr)
         self.global_step += 1
         self.num_env_frames += int(np.prod(np.shape(reward)))
-        td = (target_value - sav).mean(dim=1).abs().detach()
-        if not as_tensor:
-            td = td.cpu().numpy()
         if as_tensor:
-            return loss.detach(), td
-        return float(loss.detach()), td
+            return loss.detach(), td_t
+        return float(loss.detach()), td_t.cpu().numpy()
 
     @torch.no_grad()
     def main_q_value_test(self, state, h, c, done, previous_action):

@@ -44,6 +44,10 @@ class GpuMemory:
         self.write = 0
         self.n_entries = 0
         self.beta = self.beta_start
+        # β and n_entries live in device buffers so the sample math is
+        # hipGraph-capturable (the host advances them; replays read them)
+        self.beta_buf = torch.tensor([self.beta], device=device)
+        self.n_entries_buf = torch.zeros(1, device=device)
         self.gen = torch.Generator(device=self.device)
         if seed is not None:
             self.gen.manual_seed(seed)
@@ -66,11 +70,26 @@ class GpuMemory:
                        self._prio(errors).contiguous(), self.capacity)
         self.write = int((self.write + n) % self.capacity)
         self.n_entries = min(self.n_entries + n, self.capacity)
+        self.n_entries_buf.fill_(float(self.n_entries))
 
     @torch.no_grad()
     def sample(self, n: int):
         """Returns (rows [n] i64, tree idxs [n] i64, is_weight [n] f32),
         all on-device; no host sync."""
+        ext = _ops.require_ext()
+        self.advance_beta()
+        return self.sample_static(n)
+
+    def advance_beta(self) -> None:
+        """Host-side β annealing (reference buffer_queue.py:398): call once
+        per sample — the graphed learner calls it before each replay."""
+        self.beta = min(1.0, self.beta + self.beta_increment_per_sampling)
+        self.beta_buf.fill_(self.beta)
+
+    @torch.no_grad()
+    def sample_static(self, n: int):
+        """Capture-safe sample body: every input that changes over training
+        (β, n_entries, priorities) is read from device memory."""
         ext = _ops.require_ext()
         total = self.tree[0]
         u = torch.rand(n, device=self.device, generator=self.gen)
@@ -78,9 +97,8 @@ class GpuMemory:
             * (total / n)
         idxs, prios = ext.per_sample(self.tree, s.contiguous(),
                                      self.capacity)
-        self.beta = min(1.0, self.beta + self.beta_increment_per_sampling)
         probs = prios / total
-        w = (self.n_entries * probs).pow(-self.beta)
+        w = (self.n_entries_buf * probs).pow(-self.beta_buf)
         w = w / w.max()
         rows = idxs - (self.capacity - 1)
         return rows, idxs, w
@@ -113,3 +131,5 @@ class GpuMemory:
         self.write = sd["write"]
         self.n_entries = sd["n_entries"]
         self.beta = sd["beta"]
+        self.beta_buf.fill_(self.beta)
+        self.n_entries_buf.fill_(float(self.n_entries))

@@ -1,0 +1,117 @@
+"""hipGraph-captured PER learner step (Ape-X / R2D2).
+
+One graph replays sample -> gather -> loss forward+backward -> priority
+update; a second replays grad-gather (+ captured RCCL all-reduce in DP) +
+the fused Adam update. Everything that changes across steps is read from
+device memory at replay time: the segment tree and payloads (mutated by the
+eager ingest between steps), beta / n_entries (device buffers the host
+advances), the RNG state (philox offsets are graph-managed), and lr_t (a
+1-element buffer).
+
+Same warmup snapshot/restore discipline as runtime/graphed.py: weights,
+optimizer state, the priority tree and beta are restored after the eager
+warmup iterations so graph construction never perturbs training."""
+
+from __future__ import annotations
+
+import os
+from typing import Callable, Dict, Tuple
+
+import torch
+
+
+class GraphedReplayStep:
+    def __init__(self, agent, memory, batch_size: int,
+                 loss_fn: Callable[[Dict[str, torch.Tensor], torch.Tensor],
+                                   Tuple[torch.Tensor, torch.Tensor]],
+                 warmup_iters: int = 3):
+        assert agent.device.type == "cuda"
+        self.agent = agent
+        self.memory = memory
+        self.batch_size = batch_size
+        self._loss_fn = loss_fn
+        dev = agent.device
+        self.lr_buf = torch.zeros(1, dtype=torch.float32, device=dev)
+
+        opt = agent.optimizer
+        opt.enable_scatter_grads()
+        snap_params = opt.flat_params.detach().clone()
+        snap_state = {k: v.detach().clone()
+                      for k, v in opt._state_tensors().items()}
+        snap_tree = memory.tree.detach().clone()
+        snap_beta = memory.beta
+
+        def _iter():
+            rows, idxs, w = memory.sample_static(batch_size)
+            b = memory.gather(rows)
+            loss, td = loss_fn(b, w)
+            loss.backward()
+            memory.update_batch(idxs, td)
+            return loss.detach(), td
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(warmup_iters):
+                _iter()
+                opt.gather_grads_eager()
+                opt.step_tensor_lr(self.lr_buf)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        with torch.no_grad():
+            opt.flat_params.copy_(snap_params)
+            for k, v in opt._state_tensors().items():
+                v.copy_(snap_state[k])
+            opt.flat_grads.zero_()
+            memory.tree.copy_(snap_tree)
+        memory.beta = snap_beta
+        memory.beta_buf.fill_(snap_beta)
+
+        self.g_main = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_main):
+            self.loss, self.td = _iter()
+        opt.build_gather_table()
+
+        from distributed_reinforcement_learning_amd.parallel.dist import (
+            is_distributed, world_size,
+        )
+        self._distributed = is_distributed() and (
+            world_size() > 1
+            or bool(os.environ.get("DRLA_FORCE_DIST_GRAPH")))
+        self._eager_reduce = False
+        self.g_opt = torch.cuda.CUDAGraph()
+        if self._distributed:
+            try:
+                agent.reduce_gradients()
+                torch.cuda.synchronize()
+                with torch.cuda.graph(self.g_opt, pool=self.g_main.pool()):
+                    opt.gather_grads()
+                    agent.reduce_gradients()
+                    opt.step_tensor_lr(self.lr_buf)
+            except Exception:
+                self._eager_reduce = True
+                self.g_opt = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(self.g_opt, pool=self.g_main.pool()):
+                    opt.step_tensor_lr(self.lr_buf)
+        else:
+            with torch.cuda.graph(self.g_opt, pool=self.g_main.pool()):
+                opt.gather_grads()
+                opt.step_tensor_lr(self.lr_buf)
+
+    def step(self) -> torch.Tensor:
+        """One sample+train+update step; never syncs. Returns the loss
+        TENSOR (float() it only at logging cadence)."""
+        agent = self.agent
+        opt = agent.optimizer
+        lr = agent.lr_at(agent.global_step)
+        opt.step_count += 1
+        self.lr_buf.fill_(opt.lr_t_for(lr, opt.step_count)
+                          if hasattr(opt, "lr_t_for") else lr)
+        self.memory.advance_beta()
+        self.g_main.replay()
+        if self._distributed and self._eager_reduce:
+            opt.gather_grads()
+            agent.reduce_gradients()
+        self.g_opt.replay()
+        agent.global_step += 1
+        return self.loss

@@ -59,13 +59,21 @@ def bench_apex(steps=200, warmup=30):
             dev_u["action"], dev_u["reward"], dev_u["done"], as_tensor=True)
         mem.add_batch(td, dev_u)
 
+    from distributed_reinforcement_learning_amd.runtime import (
+        GraphedReplayStep,
+    )
+
+    def loss_fn(b, w):
+        return agent.compute_distributed_loss(
+            agent.frames_to_device(b["state"]),
+            agent.frames_to_device(b["next_state"]),
+            b["previous_action"].long(), b["action"].long(),
+            b["reward"], b["done"], w)
+
+    graphed = GraphedReplayStep(agent, mem, cfg.batch_size, loss_fn)
+
     def one():
-        rows, idxs, w = mem.sample(cfg.batch_size)
-        b = mem.gather(rows)
-        loss, td = agent.distributed_train(
-            b["state"], b["next_state"], b["previous_action"], b["action"],
-            b["reward"], b["done"], w, as_tensor=True)
-        mem.update_batch(idxs, td)
+        graphed.step()
 
     for _ in range(warmup):
         one()
@@ -128,14 +136,21 @@ def bench_r2d2(steps=200, warmup=30):
             dev_u["reward"], dev_u["done"], as_tensor=True)
         mem.add_batch(td, dev_u)
 
+    from distributed_reinforcement_learning_amd.runtime import (
+        GraphedReplayStep,
+    )
+
+    def loss_fn(b, w):
+        # compute_sequence_loss preps device tensors internally
+        return agent.compute_sequence_loss(
+            b["state"], b["previous_action"], b["action"],
+            b["initial_h"][:, 0], b["initial_c"][:, 0], b["reward"],
+            b["done"], w)
+
+    graphed = GraphedReplayStep(agent, mem, cfg.batch_size, loss_fn)
+
     def one():
-        rows, idxs, w = mem.sample(cfg.batch_size)
-        b = mem.gather(rows)
-        loss, td = agent.train(
-            state=b["state"], previous_action=b["previous_action"],
-            action=b["action"], h=b["initial_h"], c=b["initial_c"],
-            reward=b["reward"], done=b["done"], weight=w, as_tensor=True)
-        mem.update_batch(idxs, td)
+        graphed.step()
 
     for _ in range(warmup):
         one()
