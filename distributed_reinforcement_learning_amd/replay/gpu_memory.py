@@ -48,9 +48,13 @@ class GpuMemory:
         # hipGraph-capturable (the host advances them; replays read them)
         self.beta_buf = torch.tensor([self.beta], device=device)
         self.n_entries_buf = torch.zeros(1, device=device)
-        self.gen = torch.Generator(device=self.device)
+        # sampling uses the DEFAULT cuda generator: custom Generator
+        # objects are not registered with hipGraph capture ("Attempt to
+        # increase offset for a CUDA generator not in capture mode"); the
+        # default generator's philox offsets are graph-managed
+        self.gen = None
         if seed is not None:
-            self.gen.manual_seed(seed)
+            torch.cuda.manual_seed(seed)
 
     def _prio(self, errors: torch.Tensor) -> torch.Tensor:
         return (errors.abs().float() + self.e) ** self.a
@@ -92,7 +96,7 @@ class GpuMemory:
         (β, n_entries, priorities) is read from device memory."""
         ext = _ops.require_ext()
         total = self.tree[0]
-        u = torch.rand(n, device=self.device, generator=self.gen)
+        u = torch.rand(n, device=self.device)
         s = (torch.arange(n, device=self.device, dtype=torch.float32) + u) \
             * (total / n)
         idxs, prios = ext.per_sample(self.tree, s.contiguous(),
