@@ -89,6 +89,7 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
     else:
         memory = Memory(MEMORY_CAPACITY, seed=args.seed)
     train_step, buffer_steps = 0, 0
+    graphed = None
     min_warm = 2 * cfg.batch_size  # reference :122
     try:
         while args.max_steps <= 0 or train_step < args.max_steps:
@@ -116,18 +117,35 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
                 continue
             t0 = time.time()
             if use_gpu_replay:
-                with timer.track("sample"):
-                    rows, idxs, is_weight = memory.sample(cfg.batch_size)
-                    b = memory.gather(rows)
-                with timer.track("train"):
-                    loss, td_error = agent.train(
-                        state=b["state"],
-                        previous_action=b["previous_action"],
-                        action=b["action"], h=b["initial_h"],
-                        c=b["initial_c"], reward=b["reward"],
-                        done=b["done"], weight=is_weight, as_tensor=True)
-                with timer.track("per_update"):
-                    memory.update_batch(idxs, td_error)
+                if graphed is None and not getattr(args, "no_graph", False):
+                    from distributed_reinforcement_learning_amd.runtime \
+                        import GraphedReplayStep
+
+                    def _loss_fn(b, w):
+                        return agent.compute_sequence_loss(
+                            b["state"], b["previous_action"], b["action"],
+                            b["initial_h"][:, 0], b["initial_c"][:, 0],
+                            b["reward"], b["done"], w)
+                    graphed = GraphedReplayStep(agent, memory,
+                                                cfg.batch_size, _loss_fn)
+                if graphed is not None:
+                    with timer.track("train"):
+                        loss = graphed.step()
+                else:
+                    with timer.track("sample"):
+                        rows, idxs, is_weight = memory.sample(
+                            cfg.batch_size)
+                        b = memory.gather(rows)
+                    with timer.track("train"):
+                        loss, td_error = agent.train(
+                            state=b["state"],
+                            previous_action=b["previous_action"],
+                            action=b["action"], h=b["initial_h"],
+                            c=b["initial_c"], reward=b["reward"],
+                            done=b["done"], weight=is_weight,
+                            as_tensor=True)
+                    with timer.track("per_update"):
+                        memory.update_batch(idxs, td_error)
             else:
                 with timer.track("sample"):
                     batch, idxs, is_weight = memory.sample(cfg.batch_size)
