@@ -118,3 +118,65 @@ def test_fused_dqn_loss_bf16_path():
     loss.backward()
     assert mq.grad.dtype == torch.bfloat16
     assert torch.isfinite(loss)
+
+
+def test_graphed_replay_step_trains():
+    """GraphedReplayStep (Ape-X flavor): replays must train (weights move),
+    refresh priorities (tree root changes) and produce finite losses."""
+    import numpy as np
+    from distributed_reinforcement_learning_amd.agents import apex
+    from distributed_reinforcement_learning_amd.replay.gpu_memory import (
+        GpuMemory,
+    )
+    from distributed_reinforcement_learning_amd.runtime import (
+        GraphedReplayStep,
+    )
+    agent = apex.Agent(
+        input_shape=[84, 84, 4], num_action=6, discount_factor=0.99,
+        gradient_clip_norm=40.0, reward_clipping="abs_one",
+        start_learning_rate=1e-4, end_learning_rate=0.0,
+        learning_frame=10 ** 9, device="cuda:0", build_optimizer=True,
+        seed=5)
+    mem = GpuMemory(4096, fields={
+        "state": ((84, 84, 4), torch.uint8),
+        "next_state": ((84, 84, 4), torch.uint8),
+        "previous_action": ((), torch.int32),
+        "action": ((), torch.int32),
+        "reward": ((), torch.float32),
+        "done": ((), torch.bool)}, device="cuda:0", seed=5)
+    rng = np.random.default_rng(2)
+    T = 64
+    u = {
+        "state": torch.as_tensor(rng.integers(0, 255, (T, 84, 84, 4),
+                                              dtype=np.uint8)).cuda(),
+        "next_state": torch.as_tensor(
+            rng.integers(0, 255, (T, 84, 84, 4), dtype=np.uint8)).cuda(),
+        "previous_action": torch.as_tensor(
+            rng.integers(0, 6, T).astype(np.int32)).cuda(),
+        "action": torch.as_tensor(
+            rng.integers(0, 6, T).astype(np.int32)).cuda(),
+        "reward": torch.as_tensor(
+            rng.normal(size=T).astype(np.float32)).cuda(),
+        "done": torch.as_tensor(rng.random(T) < 0.05).cuda()}
+    td = agent.get_td_error(u["state"], u["next_state"],
+                            u["previous_action"], u["action"], u["reward"],
+                            u["done"], as_tensor=True)
+    mem.add_batch(td, u)
+
+    def loss_fn(b, w):
+        return agent.compute_distributed_loss(
+            agent.frames_to_device(b["state"]),
+            agent.frames_to_device(b["next_state"]),
+            b["previous_action"].long(), b["action"].long(),
+            b["reward"], b["done"], w)
+
+    p_before = agent.optimizer.flat_params.detach().clone()
+    tree_before = float(mem.tree[0])
+    g = GraphedReplayStep(agent, mem, 16, loss_fn)
+    # construction must not perturb weights or priorities (snapshot/restore)
+    assert torch.equal(agent.optimizer.flat_params, p_before)
+    assert float(mem.tree[0]) == pytest.approx(tree_before, rel=1e-4)
+    losses = [float(g.step()) for _ in range(5)]
+    assert all(np.isfinite(losses))
+    assert not torch.equal(agent.optimizer.flat_params, p_before)
+    assert float(mem.tree[0]) != pytest.approx(tree_before, rel=1e-6)
