@@ -115,3 +115,37 @@ class GraphedReplayStep:
         self.g_opt.replay()
         agent.global_step += 1
         return self.loss
+
+
+class GraphedTdScore:
+    """hipGraph-captured ingest-side TD scoring (Ape-X get_td_error / R2D2
+    get_td_error_batch): arriving unrolls have a fixed shape, and the
+    no-grad scoring forward is ~half the learner loop's eager launches.
+    ``score_fn(inputs) -> td tensor`` must be pure no-grad compute over the
+    static input buffers."""
+
+    def __init__(self, agent, example: Dict[str, torch.Tensor],
+                 score_fn: Callable[[Dict[str, torch.Tensor]],
+                                    torch.Tensor]):
+        assert agent.device.type == "cuda"
+        self.inputs = {k: torch.zeros_like(v) for k, v in example.items()}
+        self._score_fn = score_fn
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):
+                score_fn(self.inputs)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.td = score_fn(self.inputs)
+
+    def score(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
+        """Copies ``batch`` into the static buffers, replays, and returns
+        the TD tensor (plus self.inputs holds the staged batch — feed BOTH
+        straight into GpuMemory.add_batch)."""
+        for k, dst in self.inputs.items():
+            dst.copy_(batch[k], non_blocking=True)
+        self.graph.replay()
+        return self.td

@@ -93,6 +93,7 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
         memory = Memory(MEMORY_CAPACITY, seed=args.seed)
     train_step, buffer_steps = 0, 0
     graphed = None
+    td_scorer = None
     try:
         while args.max_steps <= 0 or train_step < args.max_steps:
             # Phase A: ingest one arrived unroll per iteration (blocking
@@ -106,11 +107,25 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
                     T = u["state"].shape[1]
                     if use_gpu_replay:
                         dev = {k: agent.to_device(v[0]) for k, v in u.items()}
-                        td = agent.get_td_error(
-                            dev["state"], dev["next_state"],
-                            dev["previous_action"], dev["action"],
-                            dev["reward"], dev["done"], as_tensor=True)
-                        memory.add_batch(td, dev)
+                        if td_scorer is None and not getattr(
+                                args, "no_graph", False):
+                            from distributed_reinforcement_learning_amd \
+                                .runtime.replay_graphed import GraphedTdScore
+                            td_scorer = GraphedTdScore(
+                                agent, dev,
+                                lambda i: agent.get_td_error(
+                                    i["state"], i["next_state"],
+                                    i["previous_action"], i["action"],
+                                    i["reward"], i["done"], as_tensor=True))
+                        if td_scorer is not None:
+                            td = td_scorer.score(dev)
+                            memory.add_batch(td, td_scorer.inputs)
+                        else:
+                            td = agent.get_td_error(
+                                dev["state"], dev["next_state"],
+                                dev["previous_action"], dev["action"],
+                                dev["reward"], dev["done"], as_tensor=True)
+                            memory.add_batch(td, dev)
                     else:
                         td = agent.get_td_error(
                             u["state"][0], u["next_state"][0],

@@ -90,6 +90,7 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
         memory = Memory(MEMORY_CAPACITY, seed=args.seed)
     train_step, buffer_steps = 0, 0
     graphed = None
+    td_scorer = None
     min_warm = 2 * cfg.batch_size  # reference :122
     try:
         while args.max_steps <= 0 or train_step < args.max_steps:
@@ -100,12 +101,27 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
                     u = queue.sample_batch(1)
                     if use_gpu_replay:
                         dev = {k: agent.to_device(v) for k, v in u.items()}
-                        td = agent.get_td_error_batch(
-                            dev["state"], dev["previous_action"],
-                            dev["action"], dev["initial_h"][:, 0],
-                            dev["initial_c"][:, 0], dev["reward"],
-                            dev["done"], as_tensor=True)
-                        memory.add_batch(td, dev)
+                        if td_scorer is None and not getattr(
+                                args, "no_graph", False):
+                            from distributed_reinforcement_learning_amd \
+                                .runtime.replay_graphed import GraphedTdScore
+                            td_scorer = GraphedTdScore(
+                                agent, dev,
+                                lambda i: agent.get_td_error_batch(
+                                    i["state"], i["previous_action"],
+                                    i["action"], i["initial_h"][:, 0],
+                                    i["initial_c"][:, 0], i["reward"],
+                                    i["done"], as_tensor=True))
+                        if td_scorer is not None:
+                            td = td_scorer.score(dev)
+                            memory.add_batch(td, td_scorer.inputs)
+                        else:
+                            td = agent.get_td_error_batch(
+                                dev["state"], dev["previous_action"],
+                                dev["action"], dev["initial_h"][:, 0],
+                                dev["initial_c"][:, 0], dev["reward"],
+                                dev["done"], as_tensor=True)
+                            memory.add_batch(td, dev)
                     else:
                         td = agent.get_td_error(
                             u["state"][0], u["previous_action"][0],
