@@ -191,3 +191,21 @@ def test_r2d2_act():
     a, q, h, c = agent.get_action(s, np.zeros(8, np.float32),
                                   np.zeros(8, np.float32), 0, epsilon=0.0)
     assert 0 <= a < 4 and h.shape == (8,)
+
+
+def test_r2d2_baseline_scale_config():
+    """BASELINE's R2D2 target config (burn-in 40, seq 80) must run
+    end-to-end on CPU: train step, TD scoring, burn-in recompute."""
+    agent = r2d2_agent.Agent(
+        seq_len=80, burn_in=40, input_shape=[84, 84, 1], num_action=4,
+        lstm_size=8, discount_factor=0.997, start_learning_rate=1e-4,
+        end_learning_rate=0.0, learning_frame=10 ** 9,
+        gradient_clip_norm=40.0, seed=0)
+    batch = _r2d2_batch(B=2, L=80)
+    loss, td = agent.train(**batch, weight=np.ones(2, np.float32))
+    assert np.isfinite(loss) and td.shape == (2,)
+    td2 = agent.get_td_error(
+        batch["state"][0], batch["previous_action"][0],
+        batch["action"][0], batch["h"][0], batch["c"][0],
+        batch["reward"][0], batch["done"][0])
+    assert np.isfinite(td2)
