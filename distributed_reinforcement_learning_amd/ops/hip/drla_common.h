@@ -25,9 +25,18 @@ __device__ __forceinline__ float drla_sigmoid(float x) {
   return 1.0f / (1.0f + __expf(-x));
 }
 
-// round-to-nearest-even f32 -> bf16 raw bits (inputs here are finite)
+// round-to-nearest-even f32 -> bf16 raw bits. gfx950 has a hardware
+// packed convert (v_cvt_pk_bf16_f32, 1 VALU op); the bit-math expansion
+// costs ~6 ops per value and showed up in the staging-heavy kernels
+// (PMC p31: conv_fwd_l1 at 3.7e8 VALU insts).
 __device__ __forceinline__ unsigned short drla_f32_to_bf16(float f) {
-  unsigned int u = __float_as_uint(f);
-  unsigned int rounding = 0x7FFFu + ((u >> 16) & 1u);
-  return static_cast<unsigned short>((u + rounding) >> 16);
+  __hip_bfloat162 v = __float22bfloat162_rn(float2{f, f});
+  return *reinterpret_cast<unsigned short*>(&v);
+}
+
+// packed pair -> one u32 (low = a, high = b); single v_cvt_pk_bf16_f32
+__device__ __forceinline__ unsigned int drla_f32x2_to_bf16x2(float a,
+                                                             float b) {
+  __hip_bfloat162 v = __float22bfloat162_rn(float2{a, b});
+  return *reinterpret_cast<unsigned int*>(&v);
 }
