@@ -26,7 +26,7 @@ for layer, (ishape, oshape) in shapes.items():
     e.record(); torch.cuda.synchronize()
     print(f"L{layer}: {s.elapsed_time(e)/100*1000:.1f} us")
 '''
-for split in ["64", "128", "256", "512"]:
+for split in ["256", "512", "768", "1024", "1536"]:
     env = dict(os.environ, DRLA_WGRAD_SPLIT=split)
     r = subprocess.run(["python", "-c", CODE], env=env, capture_output=True,
                        text=True, timeout=300)
