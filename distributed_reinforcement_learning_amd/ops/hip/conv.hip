@@ -396,8 +396,12 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
   // into registers while chunk m's MFMA runs — the per-chunk load-latency
   // chain (the measured bottleneck: time scaled with M/split, not with
   // atomics or LDS ops) overlaps compute.
-  __shared__ bf16raw Am[BKK][BKM + PAD];  // A image: [k][m]
-  __shared__ bf16raw BmT[CO][BKM + PAD];  // dY image: [co][m]
+  // DOUBLE-BUFFERED images: chunk t+1's LDS stores overlap chunk t's
+  // MFMAs (different buffer), leaving ONE barrier per chunk instead of
+  // two — with the register prefetch this removes the last serial
+  // stage of the per-chunk latency chain.
+  __shared__ bf16raw Am[2][BKK][BKM + PAD];   // A image: [k][m]
+  __shared__ bf16raw BmT[2][CO][BKM + PAD];   // dY image: [co][m]
 
   const int tid = threadIdx.x;
   const int wave = tid / 64;
@@ -513,18 +517,20 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
     }
   };
 
-  auto store_chunk = [&]() {
+  auto store_chunk = [&](int buf) {
+    bf16raw (*AmB)[BKM + PAD] = Am[buf];
+    bf16raw (*BmTB)[BKM + PAD] = BmT[buf];
     const int lm = a_m;
     if (a_live) {
       if constexpr (CI == 4) {
         const float sc = 1.0f / 255.0f;
-#define DRLA_WG_PUT(st, v)                                   \
-        {                                                    \
-          const int o = a_k + ((st) >> 1) * 8 + ((st) & 1) * 4; \
-          Am[o + 0][lm] = drla_f32_to_bf16(v.x * sc);        \
-          Am[o + 1][lm] = drla_f32_to_bf16(v.y * sc);        \
-          Am[o + 2][lm] = drla_f32_to_bf16(v.z * sc);        \
-          Am[o + 3][lm] = drla_f32_to_bf16(v.w * sc);        \
+#define DRLA_WG_PUT(st, v)                                          \
+        {                                                           \
+          const int o = a_k + ((st) >> 1) * 8 + ((st) & 1) * 4;     \
+          AmB[o + 0][lm] = drla_f32_to_bf16(v.x * sc);              \
+          AmB[o + 1][lm] = drla_f32_to_bf16(v.y * sc);              \
+          AmB[o + 2][lm] = drla_f32_to_bf16(v.z * sc);              \
+          AmB[o + 3][lm] = drla_f32_to_bf16(v.w * sc);              \
         }
         DRLA_WG_PUT(0, ra4_0); DRLA_WG_PUT(1, ra4_1);
         DRLA_WG_PUT(2, ra4_2); DRLA_WG_PUT(3, ra4_3);
@@ -532,63 +538,66 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
       } else if constexpr (CI == 1) {
 #pragma unroll
         for (int t = 0; t < 16; ++t) {
-          Am[a_k + t][lm] = drla_f32_to_bf16(ra1[t] * (1.0f / 255.0f));
+          AmB[a_k + t][lm] = drla_f32_to_bf16(ra1[t] * (1.0f / 255.0f));
         }
       } else {
-        Am[a_k + 0][lm] = rab0[0]; Am[a_k + 1][lm] = rab0[1];
-        Am[a_k + 2][lm] = rab0[2]; Am[a_k + 3][lm] = rab0[3];
-        Am[a_k + 4][lm] = rab0[4]; Am[a_k + 5][lm] = rab0[5];
-        Am[a_k + 6][lm] = rab0[6]; Am[a_k + 7][lm] = rab0[7];
-        Am[a_k + 8][lm] = rab1[0]; Am[a_k + 9][lm] = rab1[1];
-        Am[a_k + 10][lm] = rab1[2]; Am[a_k + 11][lm] = rab1[3];
-        Am[a_k + 12][lm] = rab1[4]; Am[a_k + 13][lm] = rab1[5];
-        Am[a_k + 14][lm] = rab1[6]; Am[a_k + 15][lm] = rab1[7];
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          AmB[a_k + e][lm] = rab0[e];
+          AmB[a_k + 8 + e][lm] = rab1[e];
+        }
       }
     } else {
 #pragma unroll
-      for (int t = 0; t < 16; ++t) Am[a_k + t][lm] = 0;
+      for (int t = 0; t < 16; ++t) AmB[a_k + t][lm] = 0;
     }
     if constexpr (CO_PER_T == 8) {
       const bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
       const bf16x8 v0 = b_live0 ? rb8_0 : z;
-      BmT[b_co0 + 0][b_lm0] = v0[0]; BmT[b_co0 + 1][b_lm0] = v0[1];
-      BmT[b_co0 + 2][b_lm0] = v0[2]; BmT[b_co0 + 3][b_lm0] = v0[3];
-      BmT[b_co0 + 4][b_lm0] = v0[4]; BmT[b_co0 + 5][b_lm0] = v0[5];
-      BmT[b_co0 + 6][b_lm0] = v0[6]; BmT[b_co0 + 7][b_lm0] = v0[7];
       const bf16x8 v1 = b_live1 ? rb8_1 : z;
       const int lm1 = b_lm0 + 32;
-      BmT[b_co0 + 0][lm1] = v1[0]; BmT[b_co0 + 1][lm1] = v1[1];
-      BmT[b_co0 + 2][lm1] = v1[2]; BmT[b_co0 + 3][lm1] = v1[3];
-      BmT[b_co0 + 4][lm1] = v1[4]; BmT[b_co0 + 5][lm1] = v1[5];
-      BmT[b_co0 + 6][lm1] = v1[6]; BmT[b_co0 + 7][lm1] = v1[7];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        BmTB[b_co0 + e][b_lm0] = v0[e];
+        BmTB[b_co0 + e][lm1] = v1[e];
+      }
     } else {
       const uint2 z = {0, 0};
       const uint2 v0 = b_live0 ? rb4_0 : z;
-      BmT[b_co0 + 0][b_lm0] = (bf16raw)(v0.x & 0xFFFF);
-      BmT[b_co0 + 1][b_lm0] = (bf16raw)(v0.x >> 16);
-      BmT[b_co0 + 2][b_lm0] = (bf16raw)(v0.y & 0xFFFF);
-      BmT[b_co0 + 3][b_lm0] = (bf16raw)(v0.y >> 16);
       const uint2 v1 = b_live1 ? rb4_1 : z;
       const int lm1 = b_lm0 + 32;
-      BmT[b_co0 + 0][lm1] = (bf16raw)(v1.x & 0xFFFF);
-      BmT[b_co0 + 1][lm1] = (bf16raw)(v1.x >> 16);
-      BmT[b_co0 + 2][lm1] = (bf16raw)(v1.y & 0xFFFF);
-      BmT[b_co0 + 3][lm1] = (bf16raw)(v1.y >> 16);
+      BmTB[b_co0 + 0][b_lm0] = (bf16raw)(v0.x & 0xFFFF);
+      BmTB[b_co0 + 1][b_lm0] = (bf16raw)(v0.x >> 16);
+      BmTB[b_co0 + 2][b_lm0] = (bf16raw)(v0.y & 0xFFFF);
+      BmTB[b_co0 + 3][b_lm0] = (bf16raw)(v0.y >> 16);
+      BmTB[b_co0 + 0][lm1] = (bf16raw)(v1.x & 0xFFFF);
+      BmTB[b_co0 + 1][lm1] = (bf16raw)(v1.x >> 16);
+      BmTB[b_co0 + 2][lm1] = (bf16raw)(v1.y & 0xFFFF);
+      BmTB[b_co0 + 3][lm1] = (bf16raw)(v1.y >> 16);
     }
   };
 
+  // ping-pong: store chunk t+1 into the other buffer while chunk t's
+  // MFMAs read this one; ONE barrier per chunk
   load_chunk(m_begin);
-  for (int m0 = m_begin; m0 < m_end; m0 += BKM) {
-    store_chunk();
-    __syncthreads();
-    if (m0 + BKM < m_end) load_chunk(m0 + BKM);  // prefetch next chunk
+  store_chunk(0);
+  load_chunk(m_begin + BKM);
+  __syncthreads();
+  int nchunks = 0;
+  for (int m0 = m_begin; m0 < m_end; m0 += BKM) ++nchunks;
+  for (int t = 0; t < nchunks; ++t) {
+    const int cur = t & 1;
+    if (t + 1 < nchunks) {
+      store_chunk(1 - cur);
+      load_chunk(m_begin + (long long)(t + 2) * BKM);
+    }
     for (int kk = 0; kk < BKM; kk += 32) {
       const bf16x8 a_frag = *reinterpret_cast<const bf16x8*>(
-          &Am[wave * 16 + (lane & 15)][kk + (lane >> 4) * 8]);
+          &Am[cur][wave * 16 + (lane & 15)][kk + (lane >> 4) * 8]);
       for (int ni = 0; ni < NFRAG; ++ni) {
         const int co = ni * 16 + (lane & 15);
         const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
-            &BmT[co][kk + (lane >> 4) * 8]);
+            &BmT[cur][co][kk + (lane >> 4) * 8]);
         acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag,
                                                           acc[ni], 0, 0, 0);
       }
