@@ -51,6 +51,27 @@ class Agent(AgentBase):
                                        lr=start_learning_rate,
                                        clip_norm=gradient_clip_norm)
 
+    def compute_a2c_losses(self, s, ns, pa, a, r, d):
+        """Pure loss body over device tensors (shared by train and the
+        graphed learner step): (pi, baseline, entropy, total)."""
+        clipped_r = clip_rewards(r, self.reward_clipping)
+        discounts = (~d).float() * self.discount_factor
+        with self.autocast():
+            policy, value = self.model(s, pa)
+            # next-state eval: prev_action for s' is the current action
+            # (reference train_a3c.py feed: npa_ph = action)
+            _, next_value = self.model(ns, a)
+        policy, value = policy.float(), value.float()
+        next_value = next_value.float().detach()
+        pi_loss = a2c.compute_policy_loss(policy, a, value, next_value,
+                                          discounts, clipped_r)
+        baseline_loss = a2c.compute_baseline_loss(value, next_value,
+                                                  discounts, clipped_r)
+        entropy = a2c.compute_entropy_loss(policy)
+        total = (pi_loss + baseline_loss * self.baseline_loss_coef
+                 + entropy * self.entropy_coef)
+        return pi_loss, baseline_loss, entropy, total
+
     def train(self, state, next_state, previous_action, action, reward,
               done) -> Tuple[float, float, float, float]:
         s = self.frames_to_device(state)
@@ -60,24 +81,8 @@ class Agent(AgentBase):
         r = self.to_device(reward, torch.float32)
         d = self.to_device(done, torch.bool)
 
-        clipped_r = clip_rewards(r, self.reward_clipping)
-        discounts = (~d).float() * self.discount_factor
-
-        with self.autocast():
-            policy, value = self.model(s, pa)
-            # next-state eval: prev_action for s' is the current action
-            # (reference train_a3c.py feed: npa_ph = action)
-            _, next_value = self.model(ns, a)
-        policy, value = policy.float(), value.float()
-        next_value = next_value.float().detach()
-
-        pi_loss = a2c.compute_policy_loss(policy, a, value, next_value,
-                                          discounts, clipped_r)
-        baseline_loss = a2c.compute_baseline_loss(value, next_value,
-                                                  discounts, clipped_r)
-        entropy = a2c.compute_entropy_loss(policy)
-        total = (pi_loss + baseline_loss * self.baseline_loss_coef
-                 + entropy * self.entropy_coef)
+        pi_loss, baseline_loss, entropy, total = self.compute_a2c_losses(
+            s, ns, pa, a, r, d)
 
         self.optimizer.zero_grad()
         total.backward()

@@ -149,3 +149,97 @@ class GraphedTdScore:
             dst.copy_(batch[k], non_blocking=True)
         self.graph.replay()
         return self.td
+
+
+class GraphedTrainStep:
+    """hipGraph-captured plain learner step (A3C flavor): static input
+    buffers -> loss forward+backward as one graph, grad-gather (+ captured
+    RCCL all-reduce in DP) + fused optimizer as a second. Same scatter-grad
+    and snapshot/restore discipline as GraphedReplayStep."""
+
+    def __init__(self, agent, example: Dict[str, torch.Tensor],
+                 loss_fn: Callable[[Dict[str, torch.Tensor]],
+                                   Tuple[torch.Tensor, ...]],
+                 warmup_iters: int = 3):
+        assert agent.device.type == "cuda"
+        self.agent = agent
+        self._loss_fn = loss_fn
+        dev = agent.device
+        self.inputs = {k: v.detach().clone() for k, v in example.items()}
+        self.lr_buf = torch.zeros(1, dtype=torch.float32, device=dev)
+
+        opt = agent.optimizer
+        opt.enable_scatter_grads()
+        snap_params = opt.flat_params.detach().clone()
+        snap_state = {k: v.detach().clone()
+                      for k, v in opt._state_tensors().items()}
+
+        def _iter():
+            losses = loss_fn(self.inputs)
+            losses[-1].backward()
+            return tuple(x.detach() for x in losses[:-1])
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(warmup_iters):
+                _iter()
+                opt.gather_grads_eager()
+                opt.step_tensor_lr(self.lr_buf)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        with torch.no_grad():
+            opt.flat_params.copy_(snap_params)
+            for k, v in opt._state_tensors().items():
+                v.copy_(snap_state[k])
+            opt.flat_grads.zero_()
+
+        self.g_main = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_main):
+            self.losses = _iter()
+        opt.build_gather_table()
+
+        from distributed_reinforcement_learning_amd.parallel.dist import (
+            is_distributed, world_size,
+        )
+        self._distributed = is_distributed() and (
+            world_size() > 1
+            or bool(os.environ.get("DRLA_FORCE_DIST_GRAPH")))
+        self._eager_reduce = False
+        self.g_opt = torch.cuda.CUDAGraph()
+        if self._distributed:
+            try:
+                agent.reduce_gradients()
+                torch.cuda.synchronize()
+                with torch.cuda.graph(self.g_opt, pool=self.g_main.pool()):
+                    opt.gather_grads()
+                    agent.reduce_gradients()
+                    opt.step_tensor_lr(self.lr_buf)
+            except Exception:
+                self._eager_reduce = True
+                self.g_opt = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(self.g_opt, pool=self.g_main.pool()):
+                    opt.step_tensor_lr(self.lr_buf)
+        else:
+            with torch.cuda.graph(self.g_opt, pool=self.g_main.pool()):
+                opt.gather_grads()
+                opt.step_tensor_lr(self.lr_buf)
+
+    def step(self, batch: Dict[str, torch.Tensor]):
+        """Copy batch into the static inputs and replay; returns the loss
+        TENSORS plus this step's lr (floats only at logging cadence)."""
+        agent = self.agent
+        opt = agent.optimizer
+        for k, dst in self.inputs.items():
+            dst.copy_(batch[k], non_blocking=True)
+        lr = agent.lr_at(agent.global_step)
+        opt.step_count += 1
+        self.lr_buf.fill_(opt.lr_t_for(lr, opt.step_count)
+                          if hasattr(opt, "lr_t_for") else lr)
+        self.g_main.replay()
+        if self._distributed and self._eager_reduce:
+            opt.gather_grads()
+            agent.reduce_gradients()
+        self.g_opt.replay()
+        agent.global_step += 1
+        return self.losses + (lr,)
