@@ -14,6 +14,10 @@ timeout 900 python bench.py --steps 300 --warmup 50 > gpurun_out/bench_graph.log
 echo "bench rc=$?"; grep -o '"value": [0-9.]*' gpurun_out/bench_graph.log; grep -o '"ms_per_step": [0-9.]*' gpurun_out/bench_graph.log
 timeout 900 python bench.py --steps 60 --warmup 15 --model resnet > gpurun_out/bench_resnet.log 2>&1
 echo "resnet rc=$?"; grep -o '"value": [0-9.]*' gpurun_out/bench_resnet.log
+timeout 600 python -u scripts/bench_algos.py > gpurun_out/bench_algos.log 2>&1
+echo "algos rc=$?"; cat gpurun_out/bench_algos.log | grep algo
+timeout 600 python -m torch.distributed.run --nnodes=1 --nproc-per-node 1 --master-addr 127.0.0.1 --master-port 29571 bench.py --gpus 1 --steps 60 --warmup 15 > gpurun_out/bench_dist1.log 2>&1
+echo "dist canary rc=$?"; grep -o '"ms_per_step": [0-9.]*' gpurun_out/bench_dist1.log
 timeout 600 python -c "import __graft_entry__ as g; g.build(); g.smoke(); print('graft clean')" > gpurun_out/smoke.log 2>&1
 echo "smoke rc=$?"; tail -2 gpurun_out/smoke.log
-bash scripts/prof_bench.sh r32
+bash scripts/prof_bench.sh r34
