@@ -14,15 +14,19 @@ Two measured host-side traps shape this class (gpu_triage.py r03/r04):
     can produce data straight into ``self.pinned`` (the trajectory queue, the
     bench pool) skip it entirely.
 
-Step structure:
-  host:    wait on the consume-event (no-op once the pipeline is primed),
-           async H2D pinned -> static inputs, write decayed LR into a device
-           buffer
-  graph 1: zero flat grads, normalize frames, batched unroll (bf16),
-           V-trace, losses, backward (into the flat grad bucket)
-  eager:   single fused RCCL all-reduce of the bucket (world > 1)
-  graph 2: global-norm clip + RMSProp update reading the LR buffer
-  host:    record consume-event
+Step structure (three graphs + a copy stream):
+  host:     write decayed LR into a device buffer (never blocks)
+  graph 1:  normalize frames, batched unroll (bf16), fused V-trace losses
+            (forward only)
+  copy str: the NEXT batch's pinned -> static-input H2D, ordered after
+            graph 1 via an event (inputs are only read by the forward), so
+            the ~8.4 MB upload hides behind the backward
+  graph 2:  backward (autograd ASSIGNS scatter-mode grads — no
+            AccumulateGrad adds)
+  graph 3:  one-kernel grad gather + (world > 1: CAPTURED RCCL all-reduce)
+            + fused global-norm clip + RMSProp update reading the LR buffer
+  host:     record consume-event; losses stay on-device until
+            last_losses()
 
 Weights and optimizer state are snapshotted before the warmup iterations and
 restored before capture, so graphing never perturbs training state.
