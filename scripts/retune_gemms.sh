@@ -13,4 +13,6 @@ timeout 900 python bench.py --steps 60 --warmup 40 > gpurun_out/retune_bench.log
 echo "tune rc=$?"
 timeout 300 python bench.py --model resnet --steps 20 --warmup 10 >> gpurun_out/retune_bench.log 2>&1
 echo "resnet tune rc=$?"
+timeout 600 python -u scripts/bench_algos.py >> gpurun_out/retune_bench.log 2>&1
+echo "algos tune rc=$?"
 wc -l gpurun_out/tunableop_gfx950_0.csv
