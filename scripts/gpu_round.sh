@@ -20,4 +20,4 @@ timeout 600 python -m torch.distributed.run --nnodes=1 --nproc-per-node 1 --mast
 echo "dist canary rc=$?"; grep -o '"ms_per_step": [0-9.]*' gpurun_out/bench_dist1.log
 timeout 600 python -c "import __graft_entry__ as g; g.build(); g.smoke(); print('graft clean')" > gpurun_out/smoke.log 2>&1
 echo "smoke rc=$?"; tail -2 gpurun_out/smoke.log
-bash scripts/prof_bench.sh r34
+bash scripts/prof_bench.sh r35
