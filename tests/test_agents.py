@@ -209,3 +209,48 @@ def test_r2d2_baseline_scale_config():
         batch["action"][0], batch["h"][0], batch["c"][0],
         batch["reward"][0], batch["done"][0])
     assert np.isfinite(td2)
+
+
+def _apex():
+    return apex_agent.Agent(
+        input_shape=[84, 84, 4], num_action=4, discount_factor=0.99,
+        gradient_clip_norm=40.0, reward_clipping="abs_one",
+        start_learning_rate=1e-4, end_learning_rate=0.0,
+        learning_frame=10 ** 9, seed=0)
+
+
+def test_apex_checkpoint_roundtrip(tmp_path):
+    agent = _apex()
+    rng = np.random.default_rng(0)
+    N = 8
+    batch = (
+        rng.integers(0, 255, (N, 84, 84, 4), dtype=np.uint8),
+        rng.integers(0, 255, (N, 84, 84, 4), dtype=np.uint8),
+        rng.integers(0, 4, N), rng.integers(0, 4, N),
+        rng.normal(size=N).astype(np.float32), np.zeros(N, bool))
+    agent.distributed_train(*batch, np.ones(N, np.float32))
+    path = str(tmp_path / "ck.pt")
+    agent.save_weights(path)
+    agent2 = _apex()
+    agent2.load_weights(path)
+    assert agent2.global_step == 1
+    for a, b in zip(agent.model.parameters(), agent2.model.parameters()):
+        assert torch.equal(a, b)
+    # the target net travels too (reference keeps main+target)
+    for a, b in zip(agent.target_model.parameters(),
+                    agent2.target_model.parameters()):
+        assert torch.equal(a, b)
+
+
+def test_r2d2_checkpoint_roundtrip(tmp_path):
+    agent = _r2d2()
+    batch = _r2d2_batch()
+    agent.train(**batch, weight=np.ones(2, np.float32))
+    path = str(tmp_path / "ck.pt")
+    agent.save_weights(path)
+    agent2 = _r2d2()
+    agent2.load_weights(path)
+    assert agent2.global_step == 1
+    for a, b in zip(agent.model.parameters(), agent2.model.parameters()):
+        assert torch.equal(a, b)
+    assert torch.equal(agent.optimizer.m, agent2.optimizer.m)
