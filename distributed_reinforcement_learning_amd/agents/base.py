@@ -14,7 +14,6 @@ from typing import Dict, Optional
 import numpy as np
 import torch
 
-from distributed_reinforcement_learning_amd.parallel import dist as _dist_mod
 from distributed_reinforcement_learning_amd.parallel.dist import (
     FlatAllReducer, is_distributed,
 )

@@ -14,7 +14,7 @@ from __future__ import annotations
 import json
 import os
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 _VALID_REWARD_CLIPPING = ("abs_one", "soft_asymmetric", "none")
 

@@ -13,7 +13,6 @@ from __future__ import annotations
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from distributed_reinforcement_learning_amd.models.blocks import (
     ActionEmbedding, AtariConvStack, MLPHead,

@@ -16,7 +16,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from distributed_reinforcement_learning_amd.models.blocks import (
-    ActionEmbedding, AtariConvStack, LSTMCellTF, MLPHead,
+    ActionEmbedding, AtariConvStack, LSTMCellTF,
 )
 
 

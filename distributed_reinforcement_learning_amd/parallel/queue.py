@@ -23,7 +23,7 @@ from __future__ import annotations
 import struct
 import time
 from multiprocessing import shared_memory
-from typing import Dict, Iterable, List, Optional, Sequence, Tuple
+from typing import Dict, Optional, Tuple
 
 import numpy as np
 

@@ -15,7 +15,7 @@ from __future__ import annotations
 import struct
 import time
 from multiprocessing import shared_memory
-from typing import Dict, Iterable, List, Optional, Tuple
+from typing import Dict, Optional, Tuple
 
 import numpy as np
 import torch

@@ -15,8 +15,6 @@ priorities never round-trip to the host.
 
 from __future__ import annotations
 
-from typing import Tuple
-
 import numpy as np
 
 
