@@ -103,7 +103,7 @@ __device__ void conv_fwd_impl(const IN_T* __restrict__ in,
   // software-pipelined K-loop: chunk k0+BK's global loads issue into
   // registers while chunk k0's MFMA runs (same structure as the wgrad —
   // the per-chunk load-latency chain is what sets kernel time here)
-  uchar4 ra4_0, ra4_1, ra4_2, ra4_3;
+  unsigned int ra4u_0, ra4u_1, ra4u_2, ra4u_3;
   unsigned char ra1[16];
   bf16x8 rab0 = {0, 0, 0, 0, 0, 0, 0, 0}, rab1 = rab0;
   uint4 rbw4;
@@ -116,17 +116,16 @@ __device__ void conv_fwd_impl(const IN_T* __restrict__ in,
   auto load_chunk = [&](int k0) {
     if (gm < M) {
       if constexpr (CI == 4) {
-#define DRLA_CF_TAP(t, dst)                                       \
-        {                                                         \
-          const int kk = k0 + a_k0 + (t) * 4;                     \
-          const int kh = kk / (KW * CI);                          \
-          const int kw = (kk - kh * KW * CI) / CI;                \
-          dst = *reinterpret_cast<const uchar4*>(                 \
-              in + (in_base + (long long)kh * WI + kw) * CI);     \
-        }
-        DRLA_CF_TAP(0, ra4_0); DRLA_CF_TAP(1, ra4_1);
-        DRLA_CF_TAP(2, ra4_2); DRLA_CF_TAP(3, ra4_3);
-#undef DRLA_CF_TAP
+        // the thread's 4 taps share kh and have CONSECUTIVE kw (16
+        // k-values = 4 taps x 4 ci; tap0 % 8 is always 0 or 4), so the
+        // four uchar4 gathers are 16 contiguous 16 B-aligned bytes
+        const int kk = k0 + a_k0;
+        const int kh = kk / (KW * CI);
+        const int kw = (kk - kh * KW * CI) / CI;
+        const uint4 packed = *reinterpret_cast<const uint4*>(
+            in + (in_base + (long long)kh * WI + kw) * CI);
+        ra4u_0 = packed.x; ra4u_1 = packed.y;
+        ra4u_2 = packed.z; ra4u_3 = packed.w;
       } else if constexpr (CI == 1) {
 #pragma unroll
         for (int t = 0; t < 16; ++t) {
@@ -158,16 +157,18 @@ __device__ void conv_fwd_impl(const IN_T* __restrict__ in,
     if (gm < M) {
       if constexpr (CI == 4) {
         const float sc = 1.0f / 255.0f;
-#define DRLA_CF_PUT(t, v)                                        \
-        {                                                        \
-          const int o = a_k0 + (t) * 4;                          \
-          Abuf[a_row][o + 0] = drla_f32_to_bf16(v.x * sc);       \
-          Abuf[a_row][o + 1] = drla_f32_to_bf16(v.y * sc);       \
-          Abuf[a_row][o + 2] = drla_f32_to_bf16(v.z * sc);       \
-          Abuf[a_row][o + 3] = drla_f32_to_bf16(v.w * sc);       \
+#define DRLA_CF_PUT(t, u)                                               \
+        {                                                               \
+          const int o = a_k0 + (t) * 4;                                 \
+          Abuf[a_row][o + 0] = drla_f32_to_bf16((float)((u) & 0xFF) * sc); \
+          Abuf[a_row][o + 1] =                                          \
+              drla_f32_to_bf16((float)(((u) >> 8) & 0xFF) * sc);        \
+          Abuf[a_row][o + 2] =                                          \
+              drla_f32_to_bf16((float)(((u) >> 16) & 0xFF) * sc);       \
+          Abuf[a_row][o + 3] = drla_f32_to_bf16((float)((u) >> 24) * sc); \
         }
-        DRLA_CF_PUT(0, ra4_0); DRLA_CF_PUT(1, ra4_1);
-        DRLA_CF_PUT(2, ra4_2); DRLA_CF_PUT(3, ra4_3);
+        DRLA_CF_PUT(0, ra4u_0); DRLA_CF_PUT(1, ra4u_1);
+        DRLA_CF_PUT(2, ra4u_2); DRLA_CF_PUT(3, ra4u_3);
 #undef DRLA_CF_PUT
       } else if constexpr (CI == 1) {
 #pragma unroll
@@ -456,7 +457,7 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
   const int b_co0 = (tid % (CO / CO_PER_T)) * CO_PER_T;
 
   // register prefetch state
-  uchar4 ra4_0, ra4_1, ra4_2, ra4_3;
+  unsigned int ra4u_0, ra4u_1, ra4u_2, ra4u_3;
   unsigned char ra1[16];
   bf16x8 rab0 = {0, 0, 0, 0, 0, 0, 0, 0}, rab1 = rab0;
   bf16x8 rb8_0 = rab0, rb8_1 = rab0;
@@ -474,10 +475,13 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
       const long long base =
           ((long long)n_idx * HI + ho * STRIDE) * WI + wo * STRIDE;
       if constexpr (CI == 4) {
-        ra4_0 = *reinterpret_cast<const uchar4*>(in + (base + a_off0) * CI);
-        ra4_1 = *reinterpret_cast<const uchar4*>(in + (base + a_off1) * CI);
-        ra4_2 = *reinterpret_cast<const uchar4*>(in + (base + a_off2) * CI);
-        ra4_3 = *reinterpret_cast<const uchar4*>(in + (base + a_off3) * CI);
+        // taps 0..3 of this thread share kh with consecutive kw (see the
+        // fwd note): one contiguous 16 B-aligned uint4 replaces the four
+        // uchar4 gathers
+        const uint4 packed = *reinterpret_cast<const uint4*>(
+            in + (base + a_off0) * CI);
+        ra4u_0 = packed.x; ra4u_1 = packed.y;
+        ra4u_2 = packed.z; ra4u_3 = packed.w;
       } else if constexpr (CI == 1) {
 #pragma unroll
         for (int t = 0; t < 16; ++t) {
@@ -524,16 +528,18 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
     if (a_live) {
       if constexpr (CI == 4) {
         const float sc = 1.0f / 255.0f;
-#define DRLA_WG_PUT(st, v)                                          \
-        {                                                           \
-          const int o = a_k + ((st) >> 1) * 8 + ((st) & 1) * 4;     \
-          AmB[o + 0][lm] = drla_f32_to_bf16(v.x * sc);              \
-          AmB[o + 1][lm] = drla_f32_to_bf16(v.y * sc);              \
-          AmB[o + 2][lm] = drla_f32_to_bf16(v.z * sc);              \
-          AmB[o + 3][lm] = drla_f32_to_bf16(v.w * sc);              \
+#define DRLA_WG_PUT(st, u)                                              \
+        {                                                               \
+          const int o = a_k + ((st) >> 1) * 8 + ((st) & 1) * 4;         \
+          AmB[o + 0][lm] = drla_f32_to_bf16((float)((u) & 0xFF) * sc);  \
+          AmB[o + 1][lm] =                                              \
+              drla_f32_to_bf16((float)(((u) >> 8) & 0xFF) * sc);        \
+          AmB[o + 2][lm] =                                              \
+              drla_f32_to_bf16((float)(((u) >> 16) & 0xFF) * sc);       \
+          AmB[o + 3][lm] = drla_f32_to_bf16((float)((u) >> 24) * sc);   \
         }
-        DRLA_WG_PUT(0, ra4_0); DRLA_WG_PUT(1, ra4_1);
-        DRLA_WG_PUT(2, ra4_2); DRLA_WG_PUT(3, ra4_3);
+        DRLA_WG_PUT(0, ra4u_0); DRLA_WG_PUT(1, ra4u_1);
+        DRLA_WG_PUT(2, ra4u_2); DRLA_WG_PUT(3, ra4u_3);
 #undef DRLA_WG_PUT
       } else if constexpr (CI == 1) {
 #pragma unroll
