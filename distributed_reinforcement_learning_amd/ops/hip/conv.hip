@@ -434,20 +434,14 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
   // k-position decode is m-invariant, so tap offsets precompute once
   const int a_m = tid >> 2;
   const int a_k = (tid & 3) * 16;
-  long long a_off0 = 0, a_off1 = 0, a_off2 = 0, a_off3 = 0;
+  long long a_off0 = 0, a_off1 = 0;
   long long a_offc[16];
   if constexpr (CI == 4) {
+    // the thread's 4 taps are pixel-consecutive within one kh row
+    // (tap0 % 8 is 0 or 4), so one base offset covers the uint4 load
     const int kb = k_row0 + a_k;
-#define DRLA_WG_TAP(st, dst)                                   \
-    {                                                          \
-      const int k2 = kb + ((st) >> 1) * 8 + ((st) & 1) * 4;    \
-      const int kh = k2 / (KW * CI);                           \
-      const int kw = (k2 - kh * KW * CI) / CI;                 \
-      dst = (long long)kh * WI + kw;                           \
-    }
-    DRLA_WG_TAP(0, a_off0); DRLA_WG_TAP(1, a_off1);
-    DRLA_WG_TAP(2, a_off2); DRLA_WG_TAP(3, a_off3);
-#undef DRLA_WG_TAP
+    const int kh = kb / (KW * CI);
+    a_off0 = (long long)kh * WI + (kb - kh * KW * CI) / CI;
   } else if constexpr (CI == 1) {
 #pragma unroll
     for (int t = 0; t < 16; ++t) {
