@@ -86,8 +86,8 @@ class Agent(AgentBase):
         trained window unrolls with gradient."""
         if with_grad and not self.burn_in_gradient and self.burn_in > 0:
             b = self.burn_in
-            h, c = self.model.burn_in_states(s[:, :b], pa[:, :b], h0, c0,
-                                             d[:, :b])
+            h, c = model.burn_in_states(s[:, :b], pa[:, :b], h0, c0,
+                                        d[:, :b])
             q_rest = model.unroll_sequence(s[:, b:], pa[:, b:], h.detach(),
                                            c.detach(), d[:, b:])
             # burn-in Q values never enter the loss (it slices [:, b:]);

@@ -80,6 +80,8 @@ extern "C" __global__ void drla_per_update(float*, const long long*,
 extern "C" __global__ void drla_per_sample(const float*, const float*,
                                            long long*, float*, int,
                                            long long);
+extern "C" __global__ void drla_per_rebuild_level(float*, long long,
+                                                  long long);
 extern "C" __global__ void drla_embed_bwd_scatter(
     const long long*, const unsigned short*, const float*, float*,
     long long, int);
@@ -498,6 +500,25 @@ void per_update(torch::Tensor tree, torch::Tensor idxs,
                      reinterpret_cast<const long long*>(
                          idxs.data_ptr<int64_t>()),
                      prios.data_ptr<float>(), n, cap);
+}
+
+void per_rebuild(torch::Tensor tree, int64_t cap) {
+  // bottom-up level-by-level recompute of the interior sums from the
+  // leaves (drift repair for the float32 atomicAdd delta chains); the
+  // level loop lives here so Python pays one call
+  check_gpu_contig(tree, "tree");
+  // interior nodes are [0, cap-2]; process descending ranges [hi>>1, hi):
+  // any i >= hi>>1 has children 2i+1 >= hi, already final from the prior
+  // launch — correct for any capacity, ~log2(cap) launches
+  long long hi = cap - 1;
+  while (hi > 0) {
+    const long long lo = hi >> 1;
+    const long long count = hi - lo;
+    hipLaunchKernelGGL(drla_per_rebuild_level,
+                       dim3((count + 255) / 256), dim3(256), 0,
+                       cur_stream(), tree.data_ptr<float>(), lo, count);
+    hi = lo;
+  }
 }
 
 std::tuple<torch::Tensor, torch::Tensor> per_sample(torch::Tensor tree,
@@ -1131,6 +1152,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dqn_loss_bwd", &dqn_loss_bwd, "closed-form K8 backward");
   m.def("per_update", &per_update, "GPU PER segment-tree batched update");
   m.def("per_sample", &per_sample, "GPU PER stratified sample descent");
+  m.def("per_rebuild", &per_rebuild,
+        "GPU PER interior-sum rebuild (float32 drift repair)");
   m.def("embed_bwd", &embed_bwd,
         "action-embedding table gradient (K2 backward)");
   m.def("lstm_tail_fwd", &lstm_tail_fwd, "fused LSTM gate tail fwd (K3)");

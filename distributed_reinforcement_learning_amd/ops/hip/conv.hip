@@ -443,6 +443,11 @@ __device__ void conv_wgrad_impl(const IN_T* __restrict__ in,
     const int kh = kb / (KW * CI);
     a_off0 = (long long)kh * WI + (kb - kh * KW * CI) / CI;
   } else if constexpr (CI == 1) {
+    // the contiguous uint row loads below read all 16 taps without a
+    // per-tap range check — only sound when every k-block is fully
+    // in-range (the 8x8x1 instantiation: K=64 fills one BKK block)
+    static_assert(K % BKK == 0,
+                  "CI==1 contiguous row staging requires K % BKK == 0");
 #pragma unroll
     for (int t = 0; t < 16; ++t) {
       const int k2 = k_row0 + a_k + t;

@@ -28,6 +28,20 @@ extern "C" __global__ void drla_per_update(
   }
 }
 
+// Periodic interior rebuild: millions of float32 atomicAdd delta
+// propagations drift the interior sums away from the true leaf sums
+// (the CPU twin replay/sum_tree.py uses float64), slowly biasing the
+// stratified sampling. One launch per tree level, host-driven
+// (replay/gpu_memory.py rebuild()): tree[i] = tree[2i+1] + tree[2i+2]
+// for i in [first, first+count).
+extern "C" __global__ void drla_per_rebuild_level(
+    float* __restrict__ tree, long long first, long long count) {
+  long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+  if (i >= count) return;
+  const long long idx = first + i;
+  tree[idx] = tree[2 * idx + 1] + tree[2 * idx + 2];
+}
+
 extern "C" __global__ void drla_per_sample(
     const float* __restrict__ tree, const float* __restrict__ s,
     long long* __restrict__ out_idx, float* __restrict__ out_prio, int n,

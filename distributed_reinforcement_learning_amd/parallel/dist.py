@@ -72,6 +72,30 @@ class FlatAllReducer:
             self.flat_grads.mul_(self._inv_world)
 
 
+def handle_capture_failure(exc: BaseException) -> None:
+    """Policy for a failed hipGraph capture of the distributed optimizer
+    step (gather + RCCL all-reduce + update). A rank that silently falls
+    back to an eager all-reduce while its peers replay a captured one
+    deadlocks the lockstep replay, so the default is
+    FAIL FAST with the traceback. Set ``DRLA_ALLOW_EAGER_REDUCE=1`` to
+    accept the eager path (~0.4 ms/step of host latency, measured) — only
+    sound when EVERY rank takes it, e.g. a single-process debug run."""
+    import sys
+    import traceback
+    traceback.print_exc()
+    if os.environ.get("DRLA_ALLOW_EAGER_REDUCE") == "1":
+        print("[drla] WARNING: distributed optimizer-graph capture failed "
+              "(traceback above); DRLA_ALLOW_EAGER_REDUCE=1 -> continuing "
+              "with the eager all-reduce. All ranks must take this path.",
+              file=sys.stderr, flush=True)
+        return
+    print("[drla] FATAL: hipGraph capture of the distributed optimizer "
+          "step failed (traceback above). Refusing the silent eager "
+          "fallback; set DRLA_ALLOW_EAGER_REDUCE=1 to accept it on every "
+          "rank.", file=sys.stderr, flush=True)
+    raise exc
+
+
 def broadcast_module(module: torch.nn.Module, src: int = 0) -> None:
     """Rank-0 init + broadcast (fixes the reference's re-init race, C5 in
     SURVEY.md §2.4)."""

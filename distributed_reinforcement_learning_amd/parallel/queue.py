@@ -16,6 +16,15 @@ reference queue API: ``append_to_queue(task, **fields)`` /
 
 Each ring header also carries the producer's heartbeat timestamp
 (failure detection — SURVEY §5.3).
+
+Memory-ordering contract (x86-64 ONLY): the producer publishes a slot with a
+plain u64 ``tail`` store AFTER the payload stores, and the consumer reads
+``tail`` before the payload — correct under x86 TSO (stores retire in program
+order, loads are not reordered with older loads), which is the only host ISA
+this framework targets (MI355X nodes are EPYC hosts). On a weakly-ordered
+host (aarch64) the payload/tail ordering would need explicit release/acquire
+fences; none are emitted. tests/tsan/ring_tsan.cc carries a C++ re-statement
+of this protocol checked under ThreadSanitizer.
 """
 
 from __future__ import annotations

@@ -118,6 +118,16 @@ class GpuMemory:
         ext.per_update(self.tree, idxs.contiguous(),
                        self._prio(errors).contiguous(), self.capacity)
 
+    @torch.no_grad()
+    def rebuild(self) -> None:
+        """Recompute every interior sum from the leaves (~log2(cap)
+        launches). The float32 atomicAdd delta chains in drla_per_update
+        drift the interior sums over millions of updates (the CPU twin
+        replay/sum_tree.py carries float64); the graphed learner calls
+        this every few thousand steps, eagerly between graph replays."""
+        ext = _ops.require_ext()
+        ext.per_rebuild(self.tree, self.capacity)
+
     def total(self) -> float:
         return float(self.tree[0])
 

@@ -123,9 +123,9 @@ class GraphedImpalaStep:
         if self._distributed:
             # capture gather + RCCL all-reduce + update as ONE graph
             # (NCCL/RCCL collectives are capture-legal and every rank
-            # replays in lockstep); eager fallback if capture refuses —
-            # an eager all-reduce between replays costs ~0.4 ms/step of
-            # host latency (measured, gpurun_out/dist1.log)
+            # replays in lockstep). Capture failure is FATAL by default —
+            # a silent per-rank eager fallback deadlocks the lockstep
+            # replay (parallel/dist.py handle_capture_failure).
             try:
                 # prime the communicator outside capture
                 agent.reduce_gradients()
@@ -134,7 +134,11 @@ class GraphedImpalaStep:
                     opt.gather_grads()
                     agent.reduce_gradients()
                     opt.step_tensor_lr(self.lr_buf)
-            except Exception:
+            except Exception as exc:
+                from distributed_reinforcement_learning_amd.parallel.dist import (
+                    handle_capture_failure,
+                )
+                handle_capture_failure(exc)
                 self._eager_reduce = True
                 self.g_opt = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(self.g_opt, pool=self.g_fwd.pool()):

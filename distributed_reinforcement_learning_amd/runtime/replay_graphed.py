@@ -88,7 +88,11 @@ class GraphedReplayStep:
                     opt.gather_grads()
                     agent.reduce_gradients()
                     opt.step_tensor_lr(self.lr_buf)
-            except Exception:
+            except Exception as exc:
+                from distributed_reinforcement_learning_amd.parallel.dist import (
+                    handle_capture_failure,
+                )
+                handle_capture_failure(exc)
                 self._eager_reduce = True
                 self.g_opt = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(self.g_opt, pool=self.g_main.pool()):
@@ -98,11 +102,17 @@ class GraphedReplayStep:
                 opt.gather_grads()
                 opt.step_tensor_lr(self.lr_buf)
 
+    REBUILD_EVERY = 2048  # steps between interior-sum drift repairs
+
     def step(self) -> torch.Tensor:
         """One sample+train+update step; never syncs. Returns the loss
         TENSOR (float() it only at logging cadence)."""
         agent = self.agent
         opt = agent.optimizer
+        self._steps = getattr(self, "_steps", 0) + 1
+        if self._steps % self.REBUILD_EVERY == 0:
+            # eager, between replays: repair float32 tree drift (ADVICE r1)
+            self.memory.rebuild()
         lr = agent.lr_at(agent.global_step)
         opt.step_count += 1
         self.lr_buf.fill_(opt.lr_t_for(lr, opt.step_count)
@@ -215,7 +225,11 @@ class GraphedTrainStep:
                     opt.gather_grads()
                     agent.reduce_gradients()
                     opt.step_tensor_lr(self.lr_buf)
-            except Exception:
+            except Exception as exc:
+                from distributed_reinforcement_learning_amd.parallel.dist import (
+                    handle_capture_failure,
+                )
+                handle_capture_failure(exc)
                 self._eager_reduce = True
                 self.g_opt = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(self.g_opt, pool=self.g_main.pool()):

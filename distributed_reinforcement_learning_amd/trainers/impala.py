@@ -81,6 +81,20 @@ def learner(ctx: common.TrainerContext, supervisor=None) -> None:
         pinned_views = {k: v.numpy() for k, v in graphed.pinned.items()}
     train_step = 0
     log_every = max(1, args.publish_every) if graphed is None else 25
+    # warm-up depth gate (reference train_impala.py:94-95: train only once
+    # size > 3*batch_size). sample_batch blocks, so in steady state the
+    # behavior is identical — the gate only delays the FIRST updates until
+    # enough off-policy data has accumulated, matching the reference's
+    # warm-up off-policy lag. GPU path only (the tiny CPU e2e runs would
+    # wait out the deadline for depth their 2 actors never build), with a
+    # deadline so a starved run degrades to blocking instead of hanging.
+    if torch.cuda.is_available():
+        gate_deadline = time.time() + 60.0
+        while (queue.get_size() <= 3 * cfg.batch_size
+               and time.time() < gate_deadline):
+            if monitor is not None:
+                supervisor.check()
+            time.sleep(0.05)
     try:
         while args.max_steps <= 0 or train_step < args.max_steps:
             with timer.track("ingest"):
