@@ -1,5 +1,6 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: IMPALA deep-conv learner, env-frames/sec (whole job).
+"""Driver benchmark entry. Default (the driver contract): IMPALA deep-conv
+learner, env-frames/sec (whole job).
 
 BASELINE.json metric: "learner env-frames/sec (whole node), IMPALA deep-conv
 84x84x4 at 1/2/4/8 MI355X". One train step consumes B*T = 32*20 = 640 env
@@ -9,10 +10,18 @@ batch = 32*N.
 
 Each timed step is the FULL learner update on synthetic uint8 frames with
 random-init weights: H2D upload of the uint8 batch, on-device /255 normalize,
-batched conv+LSTM+heads forward (bf16 autocast), V-trace, backward, flat
-all-reduce (N>1, RCCL), fused global-norm-clip + RMSProp update. Nothing is
-cached or skipped; data is synthetic because this image has no network
-(BASELINE.json mandates synthetic frames).
+batched conv+LSTM+heads forward (bf16), V-trace, backward, flat all-reduce
+(N>1, RCCL), fused global-norm-clip + RMSProp update. Nothing is cached or
+skipped; data is synthetic because this image has no network (BASELINE.json
+mandates synthetic frames).
+
+--algo {impala,apex,r2d2} additionally covers BASELINE configs #3/#4
+(VERDICT r1 item 3): the Ape-X and R2D2 runs time the full
+sample -> gather -> loss fwd+bwd -> priority-update -> optimizer replay
+iteration against a device-resident PER shard (reference loops
+train_apex.py:124-155 / train_r2d2.py:122-162). R2D2 defaults to the
+BASELINE shape seq_len=80 / burn_in=40 (config #4); --seq-len 15
+--burn-in 7 reproduces the reference config.json:88-101 shape.
 
 Launch (driver contract):
     python bench.py --gpus N --steps K --warmup W
@@ -37,16 +46,27 @@ import numpy as np
 import torch
 
 
-def main(argv=None) -> None:
+def parse_args(argv):
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch", type=int, default=32,
-                   help="per-rank batch (reference config: 32)")
+    p.add_argument("--algo", default="impala",
+                   choices=["impala", "apex", "r2d2"],
+                   help="impala = the driver's headline metric; apex/r2d2 "
+                        "cover BASELINE configs #3/#4 (PER replay step)")
+    p.add_argument("--batch", type=int, default=None,
+                   help="per-rank batch (reference: impala/apex 32, "
+                        "r2d2 16)")
     p.add_argument("--trajectory", type=int, default=20)
     p.add_argument("--num-action", type=int, default=18)
     p.add_argument("--lstm", type=int, default=256)
+    p.add_argument("--seq-len", type=int, default=80,
+                   help="r2d2 sequence length (BASELINE config #4: 80; "
+                        "reference config.json: 15)")
+    p.add_argument("--burn-in", type=int, default=40,
+                   help="r2d2 burn-in (BASELINE config #4: 40; "
+                        "reference: 7)")
     p.add_argument("--pool", type=int, default=8,
                    help="distinct synthetic batches cycled through")
     p.add_argument("--no-graph", action="store_true",
@@ -55,18 +75,100 @@ def main(argv=None) -> None:
                    choices=["deep_conv", "resnet"],
                    help="flagship (deep_conv) or IMPALA ResNet-large "
                         "(BASELINE config #5, stresses the all-reduce)")
-    args = p.parse_args(argv)
+    p.add_argument("--min-warm-s", type=float, default=1.25,
+                   help="minimum UNTIMED busy time before the timed region: "
+                        "keeps stepping past --warmup until the GPU has run "
+                        "this long, so short-run contracts (K=20, W=5) "
+                        "measure boost-clock steady state, not DVFS ramp")
+    p.add_argument("--step-times", action="store_true",
+                   help="print per-step GPU ms to stderr (diagnostics)")
+    return p.parse_args(argv)
 
+
+def run_timed(step, args, world: int, have_gpu: bool, device: str) -> float:
+    """Warmup (+ untimed clock priming), then time EXACTLY args.steps steps
+    bracketed by barrier+synchronize; returns MAX-over-ranks elapsed
+    seconds. ``step(i)`` must run one full train step."""
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if have_gpu:
+            torch.cuda.synchronize()
+
+    t_warm0 = time.perf_counter()
+    i = 0
+    for _ in range(args.warmup):
+        step(i)
+        i += 1
+    if have_gpu and args.min_warm_s > 0:
+        # clock priming, still untimed: a fresh MI355X ramps its boost
+        # clocks over O(100 ms) of sustained load; W=5 warmup steps
+        # (~4 ms of GPU work) leave the first timed steps measuring the
+        # DVFS ramp (BENCH_r01: 1.07 ms/step at K=20 vs 0.71 steady)
+        torch.cuda.synchronize()
+        while time.perf_counter() - t_warm0 < args.min_warm_s:
+            step(i)
+            i += 1
+    barrier_sync()
+
+    events = None
+    if args.step_times and have_gpu:
+        events = [torch.cuda.Event(enable_timing=True)
+                  for _ in range(args.steps + 1)]
+    t0 = time.perf_counter()
+    if events:
+        events[0].record()
+    for k in range(args.steps):
+        step(i)
+        i += 1
+        if events:
+            events[k + 1].record()
+    barrier_sync()
+    dt = time.perf_counter() - t0
+    if events:
+        times = [events[k].elapsed_time(events[k + 1])
+                 for k in range(args.steps)]
+        print("step_ms=" + json.dumps([round(x, 3) for x in times]),
+              file=sys.stderr, flush=True)
+
+    if world > 1:
+        dt_t = torch.tensor([dt], dtype=torch.float64,
+                            device=device if have_gpu else "cpu")
+        torch.distributed.all_reduce(dt_t,
+                                     op=torch.distributed.ReduceOp.MAX)
+        dt = float(dt_t.item())
+    return dt
+
+
+def emit(rank: int, metric: str, value: float, unit: str, world: int,
+         args, dt: float, config: dict, dtype: str) -> None:
+    if rank != 0:
+        return
+    print(json.dumps({
+        "metric": metric,
+        "value": value,
+        "unit": unit,
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": dt / args.steps * 1e3,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": dtype,
+        "data": "synthetic",
+        "config": config,
+    }), flush=True)
+
+
+# --------------------------------------------------------------- impala --
+
+def bench_impala(args, rank, world, local_rank, have_gpu, device):
     from distributed_reinforcement_learning_amd.agents import impala
-    from distributed_reinforcement_learning_amd.parallel import dist as pdist
 
-    local_rank = pdist.init_distributed()
-    world = pdist.world_size()
-    rank = pdist.rank()
-    have_gpu = torch.cuda.is_available()
-    device = f"cuda:{local_rank}" if have_gpu else "cpu"
-
-    B, T, A, H = args.batch, args.trajectory, args.num_action, args.lstm
+    B = args.batch or 32
+    T, A, H = args.trajectory, args.num_action, args.lstm
     agent = impala.Agent(
         trajectory=T, input_shape=[84, 84, 4], num_action=A,
         lstm_hidden_size=H, discount_factor=0.99, start_learning_rate=6e-4,
@@ -123,59 +225,266 @@ def main(argv=None) -> None:
             previous_action=b["previous_action"], initial_h=b["initial_h"],
             initial_c=b["initial_c"])
 
-    def barrier_sync():
-        if world > 1:
-            torch.distributed.barrier()
-        if have_gpu:
-            torch.cuda.synchronize()
-
-    for i in range(args.warmup):
-        step(i)
-    barrier_sync()
-    t0 = time.perf_counter()
-    for i in range(args.steps):
-        step(args.warmup + i)
-    barrier_sync()
-    dt = time.perf_counter() - t0
-
-    # MAX over ranks (slowest rank defines job time)
-    if world > 1:
-        dt_t = torch.tensor([dt], dtype=torch.float64,
-                            device=device if have_gpu else "cpu")
-        torch.distributed.all_reduce(dt_t,
-                                     op=torch.distributed.ReduceOp.MAX)
-        dt = float(dt_t.item())
-
+    dt = run_timed(step, args, world, have_gpu, device)
     frames = args.steps * B * T * world
-    value = frames / dt
-    ms_per_step = dt / args.steps * 1e3
-    if rank == 0:
-        print(json.dumps({
-            "metric": "learner env-frames/sec (whole node), IMPALA "
-                      "deep-conv 84x84x4",
-            "value": value,
-            "unit": "frames/s",
-            "n_gpus": world,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": ms_per_step,
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": None,
-            "dtype": "bf16" if have_gpu else "fp32",
-            "data": "synthetic",
-            "config": {
-                "model": ("impala deep-conv 84x84x4 + LSTM256 V-trace"
-                          if args.model == "deep_conv"
-                          else "impala resnet-large 84x84x4 + LSTM256 "
-                               "V-trace"),
-                "global_batch": B * world,
-                "seq_len": T,
-                "parallelism": f"dp{world}",
-                "num_action": A,
-                "frames_per_step_per_rank": B * T,
-            },
-        }), flush=True)
+    emit(rank,
+         "learner env-frames/sec (whole node), IMPALA deep-conv 84x84x4",
+         frames / dt, "frames/s", world, args, dt,
+         {
+             "model": ("impala deep-conv 84x84x4 + LSTM256 V-trace"
+                       if args.model == "deep_conv"
+                       else "impala resnet-large 84x84x4 + LSTM256 V-trace"),
+             "global_batch": B * world,
+             "seq_len": T,
+             "parallelism": f"dp{world}",
+             "num_action": A,
+             "frames_per_step_per_rank": B * T,
+         },
+         "bf16" if have_gpu else "fp32")
+
+
+# ----------------------------------------------------------- apex/r2d2 --
+
+def _dist_setup(agent, world):
+    if world > 1:
+        from distributed_reinforcement_learning_amd.parallel.dist import (
+            broadcast_module,
+        )
+        broadcast_module(agent.model)
+        agent.setup_all_reduce()
+
+
+def bench_apex(args, rank, world, local_rank, have_gpu, device):
+    """BASELINE config #3: Ape-X DQN, PER shard on-GPU, DP learners.
+    Timed region = the reference's PER train iteration
+    (train_apex.py:124-155): stratified sample -> payload gather -> fused
+    double-DQN loss fwd+bwd -> priority update -> all-reduce (N>1) ->
+    fused Adam — one hipGraph replay pair per step."""
+    from distributed_reinforcement_learning_amd.agents import apex as apex_agent
+
+    B = args.batch or 32
+    H, W, C = 84, 84, 4
+    A = 4  # reference config.json apex model_output (Breakout)
+    T = 32
+    agent = apex_agent.Agent(
+        input_shape=[H, W, C], num_action=A, discount_factor=0.99,
+        gradient_clip_norm=40.0, reward_clipping="abs_one",
+        start_learning_rate=1e-4, end_learning_rate=0.0,
+        learning_frame=10 ** 14, device=device, build_optimizer=True,
+        seed=3 + rank)
+    _dist_setup(agent, world)
+    rng = np.random.default_rng(7 + rank)
+
+    def synth_unroll():
+        return {
+            "state": rng.integers(0, 255, (T, H, W, C), dtype=np.uint8),
+            "next_state": rng.integers(0, 255, (T, H, W, C),
+                                       dtype=np.uint8),
+            "previous_action": rng.integers(0, A, T).astype(np.int32),
+            "action": rng.integers(0, A, T).astype(np.int32),
+            "reward": rng.normal(size=T).astype(np.float32),
+            "done": (rng.random(T) < 0.02),
+        }
+
+    if have_gpu:
+        from distributed_reinforcement_learning_amd.replay.gpu_memory import (
+            GpuMemory,
+        )
+        from distributed_reinforcement_learning_amd.runtime import (
+            GraphedReplayStep,
+        )
+        mem = GpuMemory(100_000, fields={
+            "state": ((H, W, C), torch.uint8),
+            "next_state": ((H, W, C), torch.uint8),
+            "previous_action": ((), torch.int32),
+            "action": ((), torch.int32),
+            "reward": ((), torch.float32),
+            "done": ((), torch.bool)}, device=device, seed=3 + rank)
+        for _ in range(40):
+            u = {k: torch.as_tensor(v).to(device)
+                 for k, v in synth_unroll().items()}
+            td = agent.get_td_error(
+                u["state"], u["next_state"], u["previous_action"],
+                u["action"], u["reward"], u["done"], as_tensor=True)
+            mem.add_batch(td, u)
+
+        def loss_fn(b, w):
+            return agent.compute_distributed_loss(
+                agent.frames_to_device(b["state"]),
+                agent.frames_to_device(b["next_state"]),
+                b["previous_action"].long(), b["action"].long(),
+                b["reward"], b["done"], w)
+
+        graphed = GraphedReplayStep(agent, mem, B, loss_fn)
+
+        def step(i: int):
+            graphed.step()
+        dtype = "bf16"
+    else:
+        # CPU fallback (contract tests only): same sample -> train ->
+        # priority-update loop against the float64 CPU Memory
+        from distributed_reinforcement_learning_amd.replay.memory import (
+            Memory,
+        )
+        mem = Memory(10_000, seed=3 + rank)
+        for _ in range(4):
+            u = synth_unroll()
+            for t in range(T):
+                mem.add(float(rng.random()),
+                        {k: v[t] for k, v in u.items()})
+
+        def step(i: int):
+            samples, idxs, w = mem.sample(B)
+            batch = {k: np.stack([s[k] for s in samples])
+                     for k in samples[0]}
+            _, td = agent.distributed_train(
+                batch["state"], batch["next_state"],
+                batch["previous_action"], batch["action"],
+                batch["reward"], batch["done"], w)
+            mem.update_batch(idxs, np.abs(td))
+        dtype = "fp32"
+
+    dt = run_timed(step, args, world, have_gpu, device)
+    emit(rank,
+         "Ape-X learner transitions/sec (whole node), PER replay step "
+         "84x84x4 dueling double-DQN",
+         args.steps * B * world / dt, "transitions/s", world, args, dt,
+         {
+             "model": "apex dueling double-DQN 84x84x4",
+             "global_batch": B * world,
+             "parallelism": f"dp{world}",
+             "num_action": A,
+             "replay": "GPU PER shard 1e5" if have_gpu else "CPU PER",
+             "frames_per_step_per_rank": B,
+         },
+         dtype)
+
+
+def bench_r2d2(args, rank, world, local_rank, have_gpu, device):
+    """BASELINE config #4: R2D2 LSTM POMDP, burn-in=40 seq=80 (default;
+    --seq-len 15 --burn-in 7 reproduces reference config.json:88-101).
+    Timed region = the reference's recurrent PER train iteration
+    (train_r2d2.py:122-162): sample -> gather -> burn-in recompute +
+    trained-window unroll fwd+bwd -> per-sequence priority update ->
+    all-reduce (N>1) -> fused Adam."""
+    from distributed_reinforcement_learning_amd.agents import r2d2 as r2d2_agent
+
+    B = args.batch or 16
+    L, BI = args.seq_len, args.burn_in
+    H, W, C = 84, 84, 1
+    A = 4
+    LSTM = 64
+    agent = r2d2_agent.Agent(
+        seq_len=L, burn_in=BI, input_shape=[H, W, C], num_action=A,
+        lstm_size=LSTM, discount_factor=0.997, start_learning_rate=1e-4,
+        end_learning_rate=0.0, learning_frame=10 ** 9,
+        gradient_clip_norm=40.0, reward_clipping="abs_one", device=device,
+        build_optimizer=True, seed=3 + rank)
+    _dist_setup(agent, world)
+    rng = np.random.default_rng(11 + rank)
+
+    def synth_seqs(n):
+        return {
+            "state": rng.integers(0, 255, (n, L, H, W, C), dtype=np.uint8),
+            "previous_action": rng.integers(0, A, (n, L)).astype(np.int32),
+            "action": rng.integers(0, A, (n, L)).astype(np.int32),
+            "reward": rng.normal(size=(n, L)).astype(np.float32),
+            "done": (rng.random((n, L)) < 0.02),
+            "initial_h": (rng.normal(size=(n, L, LSTM)) * 0.1
+                          ).astype(np.float32),
+            "initial_c": (rng.normal(size=(n, L, LSTM)) * 0.1
+                          ).astype(np.float32),
+        }
+
+    if have_gpu:
+        from distributed_reinforcement_learning_amd.replay.gpu_memory import (
+            GpuMemory,
+        )
+        from distributed_reinforcement_learning_amd.runtime import (
+            GraphedReplayStep,
+        )
+        mem = GpuMemory(100_000, fields={
+            "state": ((L, H, W, C), torch.uint8),
+            "previous_action": ((L,), torch.int32),
+            "action": ((L,), torch.int32),
+            "reward": ((L,), torch.float32),
+            "done": ((L,), torch.bool),
+            "initial_h": ((L, LSTM), torch.float32),
+            "initial_c": ((L, LSTM), torch.float32)},
+            device=device, seed=3 + rank)
+        for _ in range(12):
+            u = {k: torch.as_tensor(v).to(device)
+                 for k, v in synth_seqs(4).items()}
+            td = agent.get_td_error_batch(
+                u["state"], u["previous_action"], u["action"],
+                u["initial_h"][:, 0], u["initial_c"][:, 0], u["reward"],
+                u["done"], as_tensor=True)
+            mem.add_batch(td, u)
+
+        def loss_fn(b, w):
+            return agent.compute_sequence_loss(
+                b["state"], b["previous_action"], b["action"],
+                b["initial_h"][:, 0], b["initial_c"][:, 0], b["reward"],
+                b["done"], w)
+
+        graphed = GraphedReplayStep(agent, mem, B, loss_fn)
+
+        def step(i: int):
+            graphed.step()
+        dtype = "bf16"
+    else:
+        from distributed_reinforcement_learning_amd.replay.memory import (
+            Memory,
+        )
+        mem = Memory(1_000, seed=3 + rank)
+        seqs = synth_seqs(8)
+        for s in range(8):
+            mem.add(float(rng.random()),
+                    {k: v[s] for k, v in seqs.items()})
+
+        def step(i: int):
+            samples, idxs, w = mem.sample(B)
+            batch = {k: np.stack([s[k] for s in samples])
+                     for k in samples[0]}
+            _, td = agent.train(
+                batch["state"], batch["previous_action"], batch["action"],
+                batch["initial_h"], batch["initial_c"], batch["reward"],
+                batch["done"], w)
+            mem.update_batch(idxs, np.abs(td))
+        dtype = "fp32"
+
+    dt = run_timed(step, args, world, have_gpu, device)
+    emit(rank,
+         "R2D2 learner sequences/sec (whole node), recurrent PER "
+         f"seq={L} burn-in={BI} POMDP 84x84x1",
+         args.steps * B * world / dt, "sequences/s", world, args, dt,
+         {
+             "model": f"r2d2 conv+LSTM{LSTM} dueling 84x84x1",
+             "global_batch": B * world,
+             "seq_len": L,
+             "burn_in": BI,
+             "parallelism": f"dp{world}",
+             "num_action": A,
+             "replay": ("GPU recurrent PER shard 1e5" if have_gpu
+                        else "CPU PER"),
+             "frames_per_step_per_rank": B * L,
+         },
+         dtype)
+
+
+def main(argv=None) -> None:
+    args = parse_args(argv)
+    from distributed_reinforcement_learning_amd.parallel import dist as pdist
+
+    local_rank = pdist.init_distributed()
+    world = pdist.world_size()
+    rank = pdist.rank()
+    have_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+
+    bench = {"impala": bench_impala, "apex": bench_apex,
+             "r2d2": bench_r2d2}[args.algo]
+    bench(args, rank, world, local_rank, have_gpu, device)
 
 
 if __name__ == "__main__":

@@ -334,7 +334,7 @@ def test_fused_action_embed_parity(ext):
             (a.float() - b.float()).abs().max()
 
 
-@pytest.mark.parametrize("L", [9, 40])  # 40 = BASELINE burn-in scale
+@pytest.mark.parametrize("L", [9, 40, 80])  # 80 = BASELINE seq_len (config #4)
 def test_lstm_seq_train_parity(ext, L):
     from distributed_reinforcement_learning_amd.ops.lstm_op import (
         lstm_seq_train,
