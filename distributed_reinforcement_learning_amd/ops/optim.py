@@ -31,12 +31,30 @@ import torch
 from distributed_reinforcement_learning_amd import ops as _ops
 
 
+def _is_cl4(p: torch.Tensor) -> bool:
+    """4D channels_last param (conv weights): flat-buffer slots store its
+    PHYSICAL [CO,KH,KW,CI] order so the custom conv path's
+    permute(0,2,3,1).reshape view is copy-free both ways (the strided
+    permute copies were ~6 kernels = ~30 us/step, trace r02)."""
+    return p.dim() == 4 and p.is_contiguous(
+        memory_format=torch.channels_last) and not p.is_contiguous()
+
+
+def _phys_flat(p: torch.Tensor) -> torch.Tensor:
+    """Flat view of p in its physical memory order."""
+    if _is_cl4(p):
+        return p.permute(0, 2, 3, 1).reshape(-1)
+    return p.reshape(-1)
+
+
 def flatten_dense_params(params: List[torch.Tensor]):
     """Re-home ``params`` as views into one new contiguous flat buffer.
 
     Slots are 8-element aligned (small zero holes between params) so the
     scatter-mode gather kernel can use vec8 loads/stores; returns
-    (flat, slots) with slots = [(offset, numel), ...]."""
+    (flat, slots) with slots = [(offset, numel), ...]. Channels_last 4D
+    params keep their PHYSICAL layout in the slot (p stays a
+    channels_last strided view into the flat buffer)."""
     slots = []
     offset = 0
     for p in params:
@@ -45,8 +63,13 @@ def flatten_dense_params(params: List[torch.Tensor]):
     flat = torch.zeros(offset, dtype=params[0].dtype,
                        device=params[0].device)
     for p, (off, n) in zip(params, slots):
-        flat[off:off + n].copy_(p.detach().reshape(-1))
-        p.data = flat[off:off + n].view_as(p)
+        flat[off:off + n].copy_(_phys_flat(p.detach()))
+        if _is_cl4(p):
+            co, ci, kh, kw = p.shape
+            p.data = flat[off:off + n].view(co, kh, kw, ci) \
+                .permute(0, 3, 1, 2)
+        else:
+            p.data = flat[off:off + n].view_as(p)
     return flat, slots
 
 
@@ -68,12 +91,22 @@ class _FlatOptimizerBase:
                        if self.mixed else self.flat_params)
         self.scatter = False
         self._gather_table = None
-        # route autograd into the flat grad buffer
+        # route autograd into the flat grad buffer (channels_last params
+        # get a matching strided view so accumulation lands in physical
+        # slot order)
         for p, (off, n) in zip(self.params, self.slots):
-            p.grad = self.flat_grads[off:off + n].view_as(p)
+            p.grad = self._slot_view(p, self.flat_grads, off, n)
         self.lr = lr
         self.clip_norm = clip_norm
         self.step_count = 0
+
+    @staticmethod
+    def _slot_view(p: torch.Tensor, flat: torch.Tensor, off: int,
+                   n: int) -> torch.Tensor:
+        if _is_cl4(p):
+            co, ci, kh, kw = p.shape
+            return flat[off:off + n].view(co, kh, kw, ci).permute(0, 3, 1, 2)
+        return flat[off:off + n].view_as(p)
 
     def zero_grad(self) -> None:
         self.flat_grads.zero_()
@@ -96,7 +129,7 @@ class _FlatOptimizerBase:
         """Eager/warmup-path gather: per-param copies, then detach again."""
         with torch.no_grad():
             for p, (off, n) in zip(self.params, self.slots):
-                self.flat_grads[off:off + n].copy_(p.grad.reshape(-1))
+                self.flat_grads[off:off + n].copy_(_phys_flat(p.grad))
                 p.grad = None
 
     def build_gather_table(self) -> None:
@@ -106,7 +139,10 @@ class _FlatOptimizerBase:
         ptrs, offs, sizes = [], [], []
         for p, (off, n) in zip(self.params, self.slots):
             g = p.grad
-            if g is None or not g.is_contiguous() or g.numel() != n \
+            # channels_last grads are one contiguous memory block in the
+            # slot's (physical) order — the linear gather copy is correct
+            dense = g is not None and (g.is_contiguous() or _is_cl4(g))
+            if not dense or g.numel() != n \
                     or g.dtype != self.flat_grads.dtype:
                 raise RuntimeError(
                     f"scatter-grad table: bad grad for slot {off} "

@@ -98,8 +98,16 @@ def handle_capture_failure(exc: BaseException) -> None:
 
 def broadcast_module(module: torch.nn.Module, src: int = 0) -> None:
     """Rank-0 init + broadcast (fixes the reference's re-init race, C5 in
-    SURVEY.md §2.4)."""
+    SURVEY.md §2.4). Strided params (channels_last conv weights re-homed
+    into the flat optimizer buffer) bounce through a contiguous staging
+    tensor — collectives require dense tensors."""
     if not is_distributed():
         return
-    for t in module.state_dict().values():
-        dist.broadcast(t, src=src)
+    with torch.no_grad():
+        for t in module.state_dict().values():
+            if t.is_contiguous():
+                dist.broadcast(t, src=src)
+            else:
+                tmp = t.contiguous()
+                dist.broadcast(tmp, src=src)
+                t.copy_(tmp)

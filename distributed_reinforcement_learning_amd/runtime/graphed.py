@@ -55,7 +55,10 @@ class GraphedImpalaStep:
         self.inputs: Dict[str, torch.Tensor] = {
             "state": z((B, T, HH, WW, C), torch.uint8),
             "reward": z((B, T), torch.float32),
-            "action": z((B, T), torch.int64),
+            # int32: the fused V-trace kernel consumes i32 and the ring
+            # stores i32 — an i64 static input cost one [B,T] cast per
+            # replay (actions feed nothing else in the captured step)
+            "action": z((B, T), torch.int32),
             "done": z((B, T), torch.bool),
             "behavior_policy": z((B, T, A), torch.float32),
             "previous_action": z((B, T), torch.int64),
