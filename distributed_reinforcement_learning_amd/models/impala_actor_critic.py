@@ -60,11 +60,11 @@ class ImpalaActorCritic(nn.Module):
         flat_c = traj_c.reshape(B * T, -1)
         conv_out = self.conv(flat_state)
         emb = self.action_emb(flat_pa)
-        # ONE cat builds the gate-GEMM input directly (conv || emb || h);
-        # backward slices it in place — the strided-view support in the
-        # conv/LSTM backward kernels means no .contiguous() copies
-        xh = torch.cat([conv_out, emb, flat_h.to(conv_out.dtype)], dim=1)
-        new_h, _ = self.lstm.forward_xh(xh, flat_c)
+        # ONE cat builds the gate-GEMM input directly (conv || emb || h
+        # [|| ones — bias augmentation, blocks._AugGateWeight]); backward
+        # slices it in place — the strided-view support in the conv/LSTM
+        # backward kernels means no .contiguous() copies
+        new_h, _ = self.lstm.forward_cat([conv_out, emb, flat_h], flat_c)
         return new_h, B, T
 
     def unroll(self, traj_state: torch.Tensor, traj_prev_action: torch.Tensor,

@@ -672,10 +672,17 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> lstm_tail_fwd(
 }
 
 std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
-    torch::Tensor grad_h, torch::Tensor grad_c, torch::Tensor stash,
-    torch::Tensor c_prev, torch::Tensor new_c, bool bf16_gates) {
-  for (auto* t : {&grad_c, &stash, &c_prev, &new_c})
+    torch::Tensor grad_h, c10::optional<torch::Tensor> grad_c_opt,
+    torch::Tensor stash, torch::Tensor c_prev, torch::Tensor new_c,
+    bool bf16_gates) {
+  for (auto* t : {&stash, &c_prev, &new_c})
     check_gpu_contig(*t, "lstm bwd input");
+  // grad_c is null when new_c is unused downstream (saves the zero-fill)
+  const float* grad_c_ptr = nullptr;
+  if (grad_c_opt.has_value()) {
+    check_gpu_contig(*grad_c_opt, "grad_c");
+    grad_c_ptr = grad_c_opt->data_ptr<float>();
+  }
   // grad_h may be an N-strided [N,H] view (slice of the xh gradient)
   TORCH_CHECK(grad_h.is_cuda() && grad_h.dim() == 2 &&
               grad_h.stride(1) == 1, "grad_h must be row-contiguous");
@@ -689,7 +696,7 @@ std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
         stash.options().dtype(torch::kBFloat16));
     hipLaunchKernelGGL(drla_lstm_tail_bwd_bf16, dim3(drla_grid(N * H)),
                        dim3(DRLA_BLOCK), 0, cur_stream(),
-                       grad_h.data_ptr<float>(), grad_c.data_ptr<float>(),
+                       grad_h.data_ptr<float>(), grad_c_ptr,
                        stash.data_ptr<float>(), c_prev.data_ptr<float>(),
                        new_c.data_ptr<float>(), u16pm(grad_gates),
                        grad_c_prev.data_ptr<float>(), N, H, gh_stride);
@@ -698,7 +705,7 @@ std::tuple<torch::Tensor, torch::Tensor> lstm_tail_bwd(
   auto grad_gates = torch::empty_like(stash);
   hipLaunchKernelGGL(drla_lstm_tail_bwd, dim3(drla_grid(N * H)),
                      dim3(DRLA_BLOCK), 0, cur_stream(),
-                     grad_h.data_ptr<float>(), grad_c.data_ptr<float>(),
+                     grad_h.data_ptr<float>(), grad_c_ptr,
                      stash.data_ptr<float>(), c_prev.data_ptr<float>(),
                      new_c.data_ptr<float>(), grad_gates.data_ptr<float>(),
                      grad_c_prev.data_ptr<float>(), N, H, gh_stride);

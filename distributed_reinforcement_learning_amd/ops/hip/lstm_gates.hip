@@ -45,6 +45,9 @@ extern "C" __global__ void drla_lstm_tail_fwd(
   }
 }
 
+// grad_c may be null (the cell's new_c is unused downstream — IMPALA's
+// batched unroll consumes only new_h — so autograd has no c-grad; the
+// null saves a [N,H] zero-fill per step)
 extern "C" __global__ void drla_lstm_tail_bwd(
     const float* __restrict__ grad_h, const float* __restrict__ grad_c,
     const float* __restrict__ stash, const float* __restrict__ c_prev,
@@ -64,7 +67,8 @@ extern "C" __global__ void drla_lstm_tail_bwd(
     const float o_s = stash[g0 + 3 * H];
     const float tc = tanhf(new_c[idx]);
     const float dh = grad_h[n * gh_stride + h];
-    const float d_tc = dh * o_s * (1.0f - tc * tc) + grad_c[idx];
+    const float gc = grad_c ? grad_c[idx] : 0.0f;
+    const float d_tc = dh * o_s * (1.0f - tc * tc) + gc;
     grad_c_prev[idx] = d_tc * f_s;
     const float di = d_tc * g_t;
     const float dg = d_tc * i_s;
@@ -127,7 +131,8 @@ extern "C" __global__ void drla_lstm_tail_bwd_bf16(
     const float o_s = stash[g0 + 3 * H];
     const float tc = tanhf(new_c[idx]);
     const float dh = grad_h[n * gh_stride + h];
-    const float d_tc = dh * o_s * (1.0f - tc * tc) + grad_c[idx];
+    const float gc = grad_c ? grad_c[idx] : 0.0f;
+    const float d_tc = dh * o_s * (1.0f - tc * tc) + gc;
     grad_c_prev[idx] = d_tc * f_s;
     const float di = d_tc * g_t;
     const float dg = d_tc * i_s;

@@ -29,17 +29,23 @@ class _FusedLstmTail(torch.autograd.Function):
                                                 forget_bias)
         ctx.save_for_backward(stash, c_prev, new_c)
         ctx.bf16_gates = gates.dtype == torch.bfloat16
+        # no materialized grad_c: the batched unroll consumes only new_h,
+        # so autograd would otherwise zero-fill a [N,H] every step
+        ctx.set_materialize_grads(False)
         return new_h, new_c
 
     @staticmethod
     def backward(ctx, grad_h: torch.Tensor, grad_c: torch.Tensor):
         stash, c_prev, new_c = ctx.saved_tensors
         ext = _ops.require_ext()
+        if grad_h is None:
+            grad_h = torch.zeros(c_prev.shape, dtype=torch.float32,
+                                 device=c_prev.device)
         if not (grad_h.dim() == 2 and grad_h.stride(1) == 1):
             grad_h = grad_h.contiguous()
         grad_gates, grad_c_prev = ext.lstm_tail_bwd(
-            grad_h, grad_c.contiguous(), stash,
-            c_prev.contiguous(), new_c, ctx.bf16_gates)
+            grad_h, None if grad_c is None else grad_c.contiguous(),
+            stash, c_prev.contiguous(), new_c, ctx.bf16_gates)
         return grad_gates, grad_c_prev, None
 
 

@@ -108,11 +108,16 @@ class GraphedImpalaStep:
         self.g_fwd = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_fwd):
             self._total, self.losses = self._fwd()
+        # preallocated backward seed: backward() without an explicit
+        # gradient fills a scalar ones(()) INSIDE the graph (one fill
+        # kernel per replay)
+        self._bwd_seed = torch.ones((), device=dev)
         self.g_bwd = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_bwd, pool=self.g_fwd.pool()):
             # retain_graph: the saved tensors live in the shared capture
             # pool and are rewritten by every g_fwd replay
-            self._total.backward(retain_graph=True)
+            self._total.backward(gradient=self._bwd_seed,
+                                 retain_graph=True)
         # .grad now holds capture-pool tensors at replay-stable addresses
         opt.build_gather_table()
         from distributed_reinforcement_learning_amd.parallel.dist import (
