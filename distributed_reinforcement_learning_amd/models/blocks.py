@@ -142,18 +142,27 @@ class _AugGateWeight(torch.autograd.Function):
     ADJACENTLY in its flat parameter buffer, so the augmented matrix is an
     as_strided view over the same storage; pairing it with a ones-column on
     the xh input folds the bias add into the GEMM and — the real win — the
-    bias GRADIENT into the existing dW GEMM (dW_aug's last row), deleting
+    bias GRADIENT into the existing dW GEMM (dW_aug's bias row), deleting
     the separate [N,4H] column-sum reduce from the captured step (~13 us,
-    trace r02)."""
+    trace r02).
+
+    The augmentation is PAD=8 rows, not 1, so the GEMM's K dimension stays
+    8-aligned (odd K forces hipBLASLt off its fast MFMA tilings — measured
+    +7 us on the forward). Row K is the bias; rows K+1..K+7 alias whatever
+    follows in the flat buffer and meet only the ZERO columns of the xh
+    padding, so they contribute nothing and their grad rows are dropped."""
+
+    PAD = 8
 
     @staticmethod
     def forward(ctx, weight: torch.Tensor, bias: torch.Tensor):
         K, G = weight.shape
-        return weight.as_strided((K + 1, G), (G, 1))
+        ctx.K = K
+        return weight.as_strided((K + _AugGateWeight.PAD, G), (G, 1))
 
     @staticmethod
     def backward(ctx, d_aug: torch.Tensor):
-        return d_aug[:-1], d_aug[-1]
+        return d_aug[:ctx.K], d_aug[ctx.K]
 
 
 class LSTMCellTF(nn.Module):
@@ -194,20 +203,25 @@ class LSTMCellTF(nn.Module):
                 and w.data_ptr() + w.numel() * w.element_size()
                 == b.data_ptr()):
             return False
-        # the augmented view must stay inside w's storage (true for the
-        # flat optimizer buffer; guards against accidental adjacency of
-        # separate allocations)
+        # the augmented view (PAD=8 rows past the weight) must stay inside
+        # w's storage (true for the flat optimizer buffer; guards against
+        # accidental adjacency of separate allocations)
         need = (w.storage_offset() + w.numel()
-                + b.numel()) * w.element_size()
+                + _AugGateWeight.PAD * w.shape[1]) * w.element_size()
         return w.untyped_storage().nbytes() >= need
 
     def _ones_col(self, n: int, xh_dtype, dev) -> torch.Tensor:
+        """[n, PAD] padding block: column 0 is ones (meets the bias row),
+        columns 1..PAD-1 are zeros (annihilate the alias rows)."""
         cache = getattr(self, "_ones_cache", None)
         if cache is None:
             cache = self._ones_cache = {}
         key = (n, xh_dtype, dev)
         if key not in cache:
-            cache[key] = torch.ones(n, 1, dtype=xh_dtype, device=dev)
+            pad = torch.zeros(n, _AugGateWeight.PAD, dtype=xh_dtype,
+                              device=dev)
+            pad[:, 0] = 1
+            cache[key] = pad
         return cache[key]
 
     def forward_cat(self, parts, c: torch.Tensor):

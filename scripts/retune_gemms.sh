@@ -13,6 +13,10 @@ timeout 900 python bench.py --steps 60 --warmup 40 > gpurun_out/retune_bench.log
 echo "tune rc=$?"
 timeout 300 python bench.py --model resnet --steps 20 --warmup 10 >> gpurun_out/retune_bench.log 2>&1
 echo "resnet tune rc=$?"
-timeout 600 python -u scripts/bench_algos.py >> gpurun_out/retune_bench.log 2>&1
-echo "algos tune rc=$?"
+timeout 300 python bench.py --algo apex --steps 40 --warmup 10 >> gpurun_out/retune_bench.log 2>&1
+echo "apex tune rc=$?"
+timeout 400 python bench.py --algo r2d2 --steps 20 --warmup 5 >> gpurun_out/retune_bench.log 2>&1
+echo "r2d2-80 tune rc=$?"
+timeout 300 python bench.py --algo r2d2 --seq-len 15 --burn-in 7 --steps 40 --warmup 10 >> gpurun_out/retune_bench.log 2>&1
+echo "r2d2-15 tune rc=$?"
 wc -l gpurun_out/tunableop_gfx950_0.csv
