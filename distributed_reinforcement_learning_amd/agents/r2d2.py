@@ -100,6 +100,61 @@ class Agent(AgentBase):
         with ctx:
             return model.unroll_sequence(s, pa, h0, c0, d)
 
+    def _prep_seq(self, state, previous_action, action, h0, c0, reward,
+                  done):
+        s = self.frames_to_device(state)
+        pa = self.to_device(previous_action, torch.int64)
+        a = self.to_device(action, torch.int64)
+        r = self.to_device(reward, torch.float32)
+        d = self.to_device(done, torch.bool)
+        h = self.to_device(h0, torch.float32)
+        c = self.to_device(c0, torch.float32)
+        return s, pa, a, r, d, h, c
+
+    def _use_fused_tail(self) -> bool:
+        from distributed_reinforcement_learning_amd import ops as _ops
+        return (self.device.type == "cuda" and not self.burn_in_gradient
+                and self.seq_len - self.burn_in >= 2 and _ops.available())
+
+    def _window_qs(self, s, pa, d, h, c, with_grad: bool):
+        """Post-burn-in Q windows [B, L-burn_in, A]: main (grad per
+        ``with_grad``), target (never). Both nets burn in via the fused
+        no-grad state recompute and unroll only the trained window — no
+        burn-in head evaluations, no zero-pad/cat (the eager _unroll_q
+        padded to keep the [B,L] interface)."""
+        b = self.burn_in
+        with torch.no_grad():
+            if b > 0:
+                hm, cm = self.model.burn_in_states(s[:, :b], pa[:, :b],
+                                                   h, c, d[:, :b])
+                ht, ct = self.target_model.burn_in_states(
+                    s[:, :b], pa[:, :b], h, c, d[:, :b])
+            else:
+                hm = ht = h
+                cm = ct = c
+            tgt_w = self.target_model.unroll_sequence(
+                s[:, b:], pa[:, b:], ht, ct, d[:, b:])
+        ctxm = torch.enable_grad() if with_grad else torch.no_grad()
+        with ctxm:
+            main_w = self.model.unroll_sequence(
+                s[:, b:], pa[:, b:], hm.detach(), cm.detach(), d[:, b:])
+        return main_w, tgt_w
+
+    def _fused_seq_loss(self, state, previous_action, action, h0, c0,
+                        reward, done, w, with_grad: bool):
+        """GPU fast path: one-kernel TD tail (K9, ops/hip/r2d2_loss.hip)
+        over the trained window. Returns (weighted loss, |mean td| [B])."""
+        from distributed_reinforcement_learning_amd.ops.r2d2_op import (
+            fused_r2d2_loss,
+        )
+        s, pa, a, r, d, h, c = self._prep_seq(
+            state, previous_action, action, h0, c0, reward, done)
+        main_w, tgt_w = self._window_qs(s, pa, d, h, c, with_grad)
+        b = self.burn_in
+        return fused_r2d2_loss(main_w, tgt_w, a[:, b:], r[:, b:],
+                               d[:, b:], w, self.discount_factor,
+                               self.reward_clipping)
+
     def _sequence_losses(self, state, previous_action, action, h0, c0,
                          reward, done, with_grad: bool):
         """Returns (per-sequence unweighted loss [B], target_value,
@@ -163,6 +218,13 @@ class Agent(AgentBase):
                            reward, done, as_tensor: bool = False):
         """Per-sequence |mean TD| for a whole [B, L] batch in one forward
         (device tensors ok; feeds the GPU replay shard directly)."""
+        if self._use_fused_tail():
+            B = state.shape[0]
+            w1 = torch.ones(B, dtype=torch.float32, device=self.device)
+            _, td = self._fused_seq_loss(state, previous_action, action,
+                                         h0, c0, reward, done, w1,
+                                         with_grad=False)
+            return td if as_tensor else td.cpu().numpy()
         _, target_value, sav = self._sequence_losses(
             state, previous_action, action, h0, c0, reward, done,
             with_grad=False)
@@ -173,6 +235,11 @@ class Agent(AgentBase):
                               c0, reward, done, w):
         """Pure loss body (shared by train and the graphed replay step):
         returns (weighted scalar loss, per-sequence |mean TD| tensor)."""
+        if self._use_fused_tail():
+            w_t = self.to_device(w, torch.float32)
+            return self._fused_seq_loss(state, previous_action, action,
+                                        h0, c0, reward, done, w_t,
+                                        with_grad=True)
         unweighted, target_value, sav = self._sequence_losses(
             state, previous_action, action, h0, c0, reward, done,
             with_grad=True)

@@ -75,6 +75,13 @@ extern "C" __global__ void drla_dqn_loss_fwd(
 extern "C" __global__ void drla_dqn_loss_bwd(
     const float*, const int*, const float*, const float*, unsigned short*,
     float*, int, int);
+extern "C" __global__ void drla_r2d2_loss_fwd(
+    const unsigned short*, const float*, const unsigned short*,
+    const float*, const int*, const float*, const unsigned char*,
+    const float*, float, int, float*, float*, float*, int, int, int);
+extern "C" __global__ void drla_r2d2_loss_bwd(
+    const float*, const int*, const float*, const float*, unsigned short*,
+    float*, int, int, int);
 extern "C" __global__ void drla_per_update(float*, const long long*,
                                            const float*, int, long long);
 extern "C" __global__ void drla_per_sample(const float*, const float*,
@@ -487,6 +494,62 @@ torch::Tensor dqn_loss_bwd(torch::Tensor td, torch::Tensor actions,
       weights.data_ptr<float>(), gloss.data_ptr<float>(),
       want_bf16 ? u16pm(dmq) : nullptr,
       want_bf16 ? nullptr : dmq.data_ptr<float>(), B, (int)A);
+  return dmq;
+}
+
+static int next_pow2(int x) {
+  int p = 1;
+  while (p < x) p <<= 1;
+  return p;
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> r2d2_loss_fwd(
+    torch::Tensor mq, torch::Tensor tq, torch::Tensor actions,
+    torch::Tensor rewards, torch::Tensor done, torch::Tensor weights,
+    double gamma, int64_t clip_mode) {
+  for (auto* t : {&mq, &tq, &actions, &rewards, &done, &weights})
+    check_gpu_contig(*t, "r2d2 loss input");
+  TORCH_CHECK(actions.scalar_type() == torch::kInt, "actions must be i32");
+  TORCH_CHECK(done.scalar_type() == torch::kBool, "done must be bool");
+  const int B = mq.size(0), W = mq.size(1), A = mq.size(2);
+  TORCH_CHECK(W >= 2 && W <= 1024, "window out of range");
+  const bool bf16 = mq.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(tq.scalar_type() == mq.scalar_type(), "mq/tq dtype mismatch");
+  auto fopt = rewards.options().dtype(torch::kFloat);
+  auto loss = torch::zeros({1}, fopt);
+  auto td_st = torch::empty({B, W - 1}, fopt);
+  auto td_out = torch::empty({B}, fopt);
+  const int block = next_pow2(W - 1);
+  hipLaunchKernelGGL(
+      drla_r2d2_loss_fwd, dim3(B), dim3(block), 0, cur_stream(),
+      bf16 ? u16p(mq) : nullptr, bf16 ? nullptr : mq.data_ptr<float>(),
+      bf16 ? u16p(tq) : nullptr, bf16 ? nullptr : tq.data_ptr<float>(),
+      actions.data_ptr<int>(), rewards.data_ptr<float>(),
+      reinterpret_cast<const unsigned char*>(done.data_ptr<bool>()),
+      weights.data_ptr<float>(),
+      static_cast<float>(gamma), static_cast<int>(clip_mode),
+      loss.data_ptr<float>(), td_st.data_ptr<float>(),
+      td_out.data_ptr<float>(), B, W, A);
+  return {loss, td_st, td_out};
+}
+
+torch::Tensor r2d2_loss_bwd(torch::Tensor td_st, torch::Tensor actions,
+                            torch::Tensor weights, torch::Tensor gloss,
+                            int64_t W, int64_t A, bool want_bf16) {
+  for (auto* t : {&td_st, &actions, &weights, &gloss})
+    check_gpu_contig(*t, "r2d2 bwd input");
+  const int B = td_st.size(0);
+  auto dmq = torch::empty(
+      {B, W, A}, td_st.options().dtype(want_bf16 ? torch::kBFloat16
+                                                 : torch::kFloat));
+  const long long total = (long long)B * W * A;
+  hipLaunchKernelGGL(drla_r2d2_loss_bwd, dim3(drla_grid(total)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     td_st.data_ptr<float>(), actions.data_ptr<int>(),
+                     weights.data_ptr<float>(), gloss.data_ptr<float>(),
+                     want_bf16 ? u16pm(dmq) : nullptr,
+                     want_bf16 ? nullptr : dmq.data_ptr<float>(), B,
+                     (int)W, (int)A);
   return dmq;
 }
 
@@ -1157,6 +1220,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dqn_loss_fwd", &dqn_loss_fwd,
         "fused double-DQN target + IS-weighted TD loss (K8)");
   m.def("dqn_loss_bwd", &dqn_loss_bwd, "closed-form K8 backward");
+  m.def("r2d2_loss_fwd", &r2d2_loss_fwd,
+        "fused R2D2 sequence-TD tail fwd (K9)");
+  m.def("r2d2_loss_bwd", &r2d2_loss_bwd,
+        "fused R2D2 sequence-TD tail bwd (K9)");
   m.def("per_update", &per_update, "GPU PER segment-tree batched update");
   m.def("per_sample", &per_sample, "GPU PER stratified sample descent");
   m.def("per_rebuild", &per_rebuild,
