@@ -34,6 +34,7 @@ class _FusedR2d2Loss(torch.autograd.Function):
         ctx.shape = (main_q.shape[1], main_q.shape[2])
         ctx.want_bf16 = main_q.dtype == torch.bfloat16
         ctx.set_materialize_grads(False)
+        ctx.mark_non_differentiable(td_out)
         return loss[0], td_out
 
     @staticmethod
