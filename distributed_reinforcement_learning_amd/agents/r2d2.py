@@ -124,20 +124,21 @@ class Agent(AgentBase):
         padded to keep the [B,L] interface)."""
         b = self.burn_in
         with torch.no_grad():
+            # target net: ONE full-sequence pass (conv/embed/Wx once, one
+            # recurrence kernel, head on the window only)
+            tgt_w = self.target_model.q_window(s, pa, h, c, d, b)
+            if not with_grad:
+                main_w = self.model.q_window(s, pa, h, c, d, b)
+                return main_w, tgt_w
             if b > 0:
                 hm, cm = self.model.burn_in_states(s[:, :b], pa[:, :b],
                                                    h, c, d[:, :b])
-                ht, ct = self.target_model.burn_in_states(
-                    s[:, :b], pa[:, :b], h, c, d[:, :b])
             else:
-                hm = ht = h
-                cm = ct = c
-            tgt_w = self.target_model.unroll_sequence(
-                s[:, b:], pa[:, b:], ht, ct, d[:, b:])
-        ctxm = torch.enable_grad() if with_grad else torch.no_grad()
-        with ctxm:
-            main_w = self.model.unroll_sequence(
-                s[:, b:], pa[:, b:], hm.detach(), cm.detach(), d[:, b:])
+                hm, cm = h, c
+        # main net: separate no-grad burn-in keeps the trained window's
+        # backward from sweeping the burn-in frames
+        main_w = self.model.unroll_sequence(
+            s[:, b:], pa[:, b:], hm.detach(), cm.detach(), d[:, b:])
         return main_w, tgt_w
 
     def _fused_seq_loss(self, state, previous_action, action, h0, c0,

@@ -117,6 +117,43 @@ class R2D2LstmQ(nn.Module):
         return torch.stack(qs, dim=1)
 
     @torch.no_grad()
+    def q_window(self, seq_state: torch.Tensor,
+                 seq_prev_action: torch.Tensor, h0: torch.Tensor,
+                 c0: torch.Tensor, seq_done: torch.Tensor, burn: int):
+        """No-grad post-burn-in Q window in ONE front-end pass: conv/
+        embed/x-projection over the FULL [B,L] sequence, one seq-recurrence
+        kernel, head only on the trained window. Replaces the
+        burn_in_states + unroll_sequence pair for the target net and the
+        scoring path (same math, half the launches)."""
+        B, L = seq_state.shape[:2]
+        feat = self.features(
+            seq_state.reshape(B * L, *seq_state.shape[2:]),
+            seq_prev_action.reshape(B * L)).reshape(B, L, -1)
+        if (feat.is_cuda and self.lstm.weight.dtype == torch.bfloat16
+                and 4 * self.lstm_size <= 1024):
+            from distributed_reinforcement_learning_amd import ops as _o
+            ext = _o.require_ext()
+            F_ = feat.shape[-1]
+            w = self.lstm.weight
+            xg = feat.reshape(B * L, F_) @ w[:F_] + self.lstm.bias
+            h_all, _, _ = ext.lstm_seq_fwd(
+                xg.reshape(B, L, -1).contiguous(), w[F_:].contiguous(),
+                h0.float().contiguous(), c0.float().contiguous(),
+                seq_done.contiguous(), self.lstm.forget_bias)
+            # the head's dtype cast absorbs the window slice's strides
+            return self._head(h_all[:, burn:])
+        h, c = h0, c0
+        qs = []
+        for i in range(L):
+            h, c = self.lstm(feat[:, i], h, c)
+            if i >= burn:
+                qs.append(self._head(h))
+            keep = (~seq_done[:, i]).to(h.dtype).unsqueeze(1)
+            h = h * keep
+            c = c * keep
+        return torch.stack(qs, dim=1)
+
+    @torch.no_grad()
     def burn_in_states(self, seq_state: torch.Tensor,
                        seq_prev_action: torch.Tensor, h0: torch.Tensor,
                        c0: torch.Tensor, seq_done: torch.Tensor):

@@ -89,6 +89,9 @@ extern "C" __global__ void drla_per_sample(const float*, const float*,
                                            long long);
 extern "C" __global__ void drla_per_rebuild_level(float*, long long,
                                                   long long);
+extern "C" __global__ void drla_multi_gather(
+    const long long*, const unsigned long long*, const unsigned long long*,
+    const long long*, int);
 extern "C" __global__ void drla_embed_bwd_scatter(
     const long long*, const unsigned short*, const float*, float*,
     long long, int);
@@ -582,6 +585,28 @@ void per_rebuild(torch::Tensor tree, int64_t cap) {
                        cur_stream(), tree.data_ptr<float>(), lo, count);
     hi = lo;
   }
+}
+
+void multi_gather(torch::Tensor rows, torch::Tensor srcs,
+                  torch::Tensor dsts, torch::Tensor fbytes,
+                  int64_t max_chunks) {
+  for (auto* t : {&rows, &srcs, &dsts, &fbytes})
+    check_gpu_contig(*t, "multi_gather table");
+  TORCH_CHECK(rows.scalar_type() == torch::kLong, "rows must be i64");
+  const int B = rows.numel();
+  const int F = srcs.numel();
+  const long long total = (long long)B * max_chunks;
+  const int gx = (int)std::min<long long>((total + 255) / 256, 4096);
+  hipLaunchKernelGGL(drla_multi_gather, dim3(gx, F), dim3(256), 0,
+                     cur_stream(),
+                     reinterpret_cast<const long long*>(
+                         rows.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(
+                         srcs.data_ptr<int64_t>()),
+                     reinterpret_cast<const unsigned long long*>(
+                         dsts.data_ptr<int64_t>()),
+                     reinterpret_cast<const long long*>(
+                         fbytes.data_ptr<int64_t>()), B);
 }
 
 std::tuple<torch::Tensor, torch::Tensor> per_sample(torch::Tensor tree,
@@ -1226,6 +1251,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused R2D2 sequence-TD tail bwd (K9)");
   m.def("per_update", &per_update, "GPU PER segment-tree batched update");
   m.def("per_sample", &per_sample, "GPU PER stratified sample descent");
+  m.def("multi_gather", &multi_gather,
+        "one-kernel replay batch gather (all fields)");
   m.def("per_rebuild", &per_rebuild,
         "GPU PER interior-sum rebuild (float32 drift repair)");
   m.def("embed_bwd", &embed_bwd,

@@ -42,6 +42,39 @@ extern "C" __global__ void drla_per_rebuild_level(
   tree[idx] = tree[2 * idx + 1] + tree[2 * idx + 2];
 }
 
+// One-kernel replay-batch gather: replaces the per-field index_select
+// chain (7 launches for an R2D2 sample at ~5-9 us each) with a single
+// launch copying every field of every sampled row. blockIdx.y = field;
+// fields whose row stride is 16-byte-divisible copy as uint4 (the frame /
+// hidden-state payloads), the tiny remainder fields copy bytewise.
+extern "C" __global__ void drla_multi_gather(
+    const long long* __restrict__ rows,            // [B] sampled rows
+    const unsigned long long* __restrict__ srcs,   // [F] payload ptrs
+    const unsigned long long* __restrict__ dsts,   // [F] output ptrs
+    const long long* __restrict__ fbytes,          // [F] bytes per row
+    int B) {
+  const int f = blockIdx.y;
+  const long long fb = fbytes[f];
+  const char* __restrict__ src = reinterpret_cast<const char*>(srcs[f]);
+  char* __restrict__ dst = reinterpret_cast<char*>(dsts[f]);
+  const long long cpr = (fb + 15) >> 4;  // 16B chunks per row
+  const long long total = (long long)B * cpr;
+  const bool vec = (fb & 15) == 0;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += gridDim.x * (long long)blockDim.x) {
+    const long long r = i / cpr;
+    const long long off = (i - r * cpr) << 4;
+    const char* s = src + rows[r] * fb + off;
+    char* d = dst + r * fb + off;
+    if (vec) {
+      *reinterpret_cast<uint4*>(d) = *reinterpret_cast<const uint4*>(s);
+    } else {
+      const int n = (int)(fb - off < 16 ? fb - off : 16);
+      for (int k = 0; k < n; ++k) d[k] = s[k];
+    }
+  }
+}
+
 extern "C" __global__ void drla_per_sample(
     const float* __restrict__ tree, const float* __restrict__ s,
     long long* __restrict__ out_idx, float* __restrict__ out_prio, int n,

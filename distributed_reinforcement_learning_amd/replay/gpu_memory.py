@@ -107,9 +107,41 @@ class GpuMemory:
         rows = idxs - (self.capacity - 1)
         return rows, idxs, w
 
+    def _ensure_gather(self, n: int) -> None:
+        """Persistent gather outputs + device descriptor tables for
+        drla_multi_gather (one kernel replaces the 7-launch index_select
+        chain). Built lazily per batch size, BEFORE any hipGraph capture
+        (the graphed learner's eager warmup triggers it)."""
+        if getattr(self, "_g_n", None) == n:
+            return
+        out = {k: torch.empty((n, *buf.shape[1:]), dtype=buf.dtype,
+                              device=self.device)
+               for k, buf in self.data.items()}
+        fbytes = [buf.element_size() * int(buf[0].numel()) if buf.dim() > 1
+                  else buf.element_size()
+                  for buf in self.data.values()]
+        i64 = dict(dtype=torch.int64, device=self.device)
+        self._g_out = out
+        self._g_srcs = torch.tensor(
+            [buf.data_ptr() for buf in self.data.values()], **i64)
+        self._g_dsts = torch.tensor(
+            [t.data_ptr() for t in out.values()], **i64)
+        self._g_fbytes = torch.tensor(fbytes, **i64)
+        self._g_maxchunks = max((fb + 15) // 16 for fb in fbytes)
+        self._g_n = n
+
     @torch.no_grad()
     def gather(self, rows: torch.Tensor) -> Dict[str, torch.Tensor]:
-        return {k: buf.index_select(0, rows) for k, buf in self.data.items()}
+        """All fields of the sampled rows in ONE kernel; returns views of
+        persistent per-batch-size output buffers (callers consume them
+        within the step — the graphed learner overwrites them per replay,
+        exactly like its other static buffers)."""
+        ext = _ops.require_ext()
+        n = rows.numel()
+        self._ensure_gather(n)
+        ext.multi_gather(rows.contiguous(), self._g_srcs, self._g_dsts,
+                         self._g_fbytes, self._g_maxchunks)
+        return dict(self._g_out)
 
     @torch.no_grad()
     def update_batch(self, idxs: torch.Tensor,
