@@ -53,7 +53,26 @@ class Agent(AgentBase):
 
     def compute_a2c_losses(self, s, ns, pa, a, r, d):
         """Pure loss body over device tensors (shared by train and the
-        graphed learner step): (pi, baseline, entropy, total)."""
+        graphed learner step): (pi, baseline, entropy, total).
+
+        On GPU the whole post-network pipeline (softmax, 1-step-TD
+        advantage, all three losses + backward) is the fused K6 kernel
+        pair (ops/hip/a2c_loss.hip); the CPU composition below is the
+        golden the GPU parity test compares against."""
+        from distributed_reinforcement_learning_amd import ops as _ops
+        if s.is_cuda and _ops.available():
+            from distributed_reinforcement_learning_amd.ops.a2c_op import (
+                fused_a2c_loss,
+            )
+            logits, value = self.model.logits_value(s, pa)
+            with torch.no_grad():
+                # next-state eval: prev_action for s' is the current
+                # action (reference train_a3c.py feed: npa_ph = action)
+                _, next_value = self.model.logits_value(ns, a)
+            return fused_a2c_loss(
+                logits, value.float(), next_value.float(), a, r, d,
+                self.discount_factor, self.reward_clipping,
+                self.baseline_loss_coef, self.entropy_coef)
         clipped_r = clip_rewards(r, self.reward_clipping)
         discounts = (~d).float() * self.discount_factor
         with self.autocast():

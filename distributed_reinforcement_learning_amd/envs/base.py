@@ -64,23 +64,40 @@ class ObservationWrapper(Wrapper):
 
 
 class GymAdapter(Env):
-    """Wrap a real gym env (old 4-tuple or new 5-tuple API) if gym exists."""
+    """Wrap a real gym env (old 4-tuple or new 5-tuple API) if gym exists.
+
+    Translates ALE's life counter into the ``life_lost`` flag the trainer
+    loops read (trainers/*.py ``info.get("life_lost")``): the reference
+    actors track ``info['ale.lives']`` across steps and shape
+    reward=-1/done=True on a decrease (/root/reference/train_impala.py:
+    147-154); this adapter performs the tracking so every env — real ALE
+    or synthetic — exposes the same flag."""
 
     def __init__(self, gym_env):
         self._env = gym_env
         self.action_space_n = gym_env.action_space.n
         self.observation_shape = tuple(gym_env.observation_space.shape)
+        self._lives = None
 
     def reset(self):
         out = self._env.reset()
+        self._lives = None
         return out[0] if isinstance(out, tuple) else out
 
     def step(self, action):
         out = self._env.step(action)
         if len(out) == 5:  # gymnasium API
             obs, r, term, trunc, info = out
-            return obs, r, term or trunc, info
-        return out
+            done = term or trunc
+        else:
+            obs, r, done, info = out
+        lives = info.get("ale.lives", info.get("lives"))
+        if lives is not None:
+            if self._lives is not None and lives < self._lives:
+                info = dict(info)
+                info["life_lost"] = True
+            self._lives = lives
+        return obs, r, done, info
 
     def get_action_meanings(self):
         try:

@@ -39,6 +39,14 @@ class ActorCritic(nn.Module):
         value = self.value_head(feat).squeeze(-1)
         return policy, value
 
+    def logits_value(self, state: torch.Tensor, prev_action: torch.Tensor):
+        """Pre-softmax logits + value — the fused A2C loss kernel's input
+        (ops/a2c_op.py)."""
+        feat = torch.cat(
+            [self.conv(state), self.action_emb(prev_action)], dim=1)
+        return self.policy_head.logits(feat), \
+            self.value_head(feat).squeeze(-1)
+
 
 class VectorActorCritic(nn.Module):
     """MLP actor-critic over 1-D observations (CartPole plumbing config)."""
@@ -62,3 +70,10 @@ class VectorActorCritic(nn.Module):
         policy = self.policy_head(feat)
         value = self.value_head(feat).squeeze(-1)
         return policy, value
+
+    def logits_value(self, state: torch.Tensor, prev_action: torch.Tensor):
+        state = state.to(self.obs_mlp[0].weight.dtype)
+        feat = torch.cat(
+            [self.obs_mlp(state), self.action_emb(prev_action)], dim=1)
+        return self.policy_head.logits(feat), \
+            self.value_head(feat).squeeze(-1)

@@ -33,6 +33,7 @@ SOURCES = [
     "conv.hip",
     "per_tree.hip",
     "dqn_loss.hip",
+    "a2c_loss.hip",
     "r2d2_loss.hip",
     "mlp_heads.hip",
 ]

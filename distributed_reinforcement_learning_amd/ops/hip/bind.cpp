@@ -75,6 +75,13 @@ extern "C" __global__ void drla_dqn_loss_fwd(
 extern "C" __global__ void drla_dqn_loss_bwd(
     const float*, const int*, const float*, const float*, unsigned short*,
     float*, int, int);
+extern "C" __global__ void drla_a2c_loss_fwd(
+    const unsigned short*, const float*, const float*, const float*,
+    const int*, const float*, const unsigned char*, float, int, float,
+    float, float*, float*, float*, int, int);
+extern "C" __global__ void drla_a2c_loss_bwd(
+    const float*, const float*, const int*, const float*, int, float,
+    float, unsigned short*, float*, float*, int, int);
 extern "C" __global__ void drla_r2d2_loss_fwd(
     const unsigned short*, const float*, const unsigned short*,
     const float*, const int*, const float*, const unsigned char*,
@@ -498,6 +505,56 @@ torch::Tensor dqn_loss_bwd(torch::Tensor td, torch::Tensor actions,
       want_bf16 ? u16pm(dmq) : nullptr,
       want_bf16 ? nullptr : dmq.data_ptr<float>(), B, (int)A);
   return dmq;
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> a2c_loss_fwd(
+    torch::Tensor logits, torch::Tensor value, torch::Tensor next_value,
+    torch::Tensor actions, torch::Tensor rewards, torch::Tensor done,
+    double gamma, int64_t clip_mode, double c_bl, double c_ent) {
+  for (auto* t : {&logits, &value, &next_value, &actions, &rewards, &done})
+    check_gpu_contig(*t, "a2c loss input");
+  TORCH_CHECK(actions.scalar_type() == torch::kInt, "actions must be i32");
+  TORCH_CHECK(done.scalar_type() == torch::kBool, "done must be bool");
+  const int N = logits.size(0), A = logits.size(1);
+  const bool bf16 = logits.scalar_type() == torch::kBFloat16;
+  auto fopt = value.options().dtype(torch::kFloat);
+  auto losses = torch::zeros({4}, fopt);
+  auto p_stash = torch::empty({N, A}, fopt);
+  auto adv_st = torch::empty({N}, fopt);
+  hipLaunchKernelGGL(
+      drla_a2c_loss_fwd, dim3((N + 255) / 256), dim3(256), 0, cur_stream(),
+      bf16 ? u16p(logits) : nullptr,
+      bf16 ? nullptr : logits.data_ptr<float>(), value.data_ptr<float>(),
+      next_value.data_ptr<float>(), actions.data_ptr<int>(),
+      rewards.data_ptr<float>(),
+      reinterpret_cast<const unsigned char*>(done.data_ptr<bool>()),
+      static_cast<float>(gamma), static_cast<int>(clip_mode),
+      static_cast<float>(c_bl), static_cast<float>(c_ent),
+      losses.data_ptr<float>(), p_stash.data_ptr<float>(),
+      adv_st.data_ptr<float>(), N, A);
+  return {losses, p_stash, adv_st};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> a2c_loss_bwd(
+    torch::Tensor p_stash, torch::Tensor adv_st, torch::Tensor actions,
+    torch::Tensor grad3, bool from_total, double c_bl, double c_ent,
+    bool want_bf16) {
+  for (auto* t : {&p_stash, &adv_st, &actions, &grad3})
+    check_gpu_contig(*t, "a2c bwd input");
+  const int N = p_stash.size(0), A = p_stash.size(1);
+  auto dlg = torch::empty(
+      {N, A}, p_stash.options().dtype(want_bf16 ? torch::kBFloat16
+                                                : torch::kFloat));
+  auto dvalue = torch::empty({N}, p_stash.options());
+  hipLaunchKernelGGL(
+      drla_a2c_loss_bwd, dim3((N + 255) / 256), dim3(256), 0, cur_stream(),
+      p_stash.data_ptr<float>(), adv_st.data_ptr<float>(),
+      actions.data_ptr<int>(), grad3.data_ptr<float>(),
+      from_total ? 1 : 0, static_cast<float>(c_bl),
+      static_cast<float>(c_ent), want_bf16 ? u16pm(dlg) : nullptr,
+      want_bf16 ? nullptr : dlg.data_ptr<float>(),
+      dvalue.data_ptr<float>(), N, A);
+  return {dlg, dvalue};
 }
 
 static int next_pow2(int x) {
@@ -1245,6 +1302,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dqn_loss_fwd", &dqn_loss_fwd,
         "fused double-DQN target + IS-weighted TD loss (K8)");
   m.def("dqn_loss_bwd", &dqn_loss_bwd, "closed-form K8 backward");
+  m.def("a2c_loss_fwd", &a2c_loss_fwd, "fused A2C loss fwd (K6)");
+  m.def("a2c_loss_bwd", &a2c_loss_bwd, "fused A2C loss bwd (K6)");
   m.def("r2d2_loss_fwd", &r2d2_loss_fwd,
         "fused R2D2 sequence-TD tail fwd (K9)");
   m.def("r2d2_loss_bwd", &r2d2_loss_bwd,
