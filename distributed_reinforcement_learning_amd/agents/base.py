@@ -140,11 +140,10 @@ class AgentBase:
         None."""
         if self.weight_subscriber is None:
             return None
+        # state_dict() returns references to the live tensors; pull copies
+        # the snapshot into them in place — no load_state_dict round trip
         sd = self.model.state_dict()
-        step = self.weight_subscriber.pull(sd)
-        if step is not None:
-            self.model.load_state_dict(sd)
-        return step
+        return self.weight_subscriber.pull(sd)
 
     def publish_weights(self) -> None:
         if self.weight_publisher is not None:

@@ -154,6 +154,8 @@ def actor(ctx: common.TrainerContext, task: int) -> None:
     torch.set_num_threads(1)  # batch-1 CPU inference; also avoids
     # the forked-child OpenMP deadlock (see trainers/common.py)
     cfg, args = ctx.cfg, ctx.args
+    if int(cfg.get("envs_per_actor", 1)) > 1:
+        return vector_actor(ctx, task)
     env_name = cfg.env[task]
     available_action = cfg.available_action[task]
     env = make_uint8_env(env_name, num_actions=cfg.model_output,
@@ -211,6 +213,89 @@ def actor(ctx: common.TrainerContext, task: int) -> None:
                     c = np.zeros(cfg.lstm_size, dtype=np.float32)
             queue.append_to_queue(task, **traj.stacked())
             unrolls += 1
+    finally:
+        writer.close()
+        queue.close()
+
+
+def vector_actor(ctx: common.TrainerContext, task: int) -> None:
+    """E envs per actor process, ONE batched inference per tick (VERDICT
+    r1 item 8: the batch-1 python loop feeds ~0.1% of learner capacity).
+    Same behavior per env as the scalar loop — unrolls of length T with
+    stored per-step LSTM state, life-loss shaping, weight pull once per
+    unroll period — but the model forward amortizes over E envs
+    (config key ``envs_per_actor``)."""
+    import torch
+    torch.set_num_threads(1)
+    cfg, args = ctx.cfg, ctx.args
+    E = int(cfg.get("envs_per_actor", 1))
+    env_name = cfg.env[task]
+    available_action = cfg.available_action[task]
+    envs = [make_uint8_env(env_name, num_actions=cfg.model_output,
+                           seed=(args.seed or 0) + task * 1000 + e)
+            for e in range(E)]
+    queue = TrajectoryQueue(
+        queue_schema_for("impala", cfg), cfg.num_actors, cfg.queue_size,
+        role="actor", namespace=ctx.namespace, actor_task=task,
+        world_size=ctx.world_size)
+    agent = build_agent(ctx, "cpu", build_optimizer=False,
+                        seed=(args.seed or 0) + 1000 + task)
+    agent.weight_subscriber = WeightSubscriber(ctx.weights_name,
+                                               agent.model.state_dict())
+    agent.weight_subscriber.wait_for_first()
+    writer = SummaryWriter(ctx.actor_logdir(task))
+
+    H = cfg.lstm_size
+    states = np.stack([env.reset() for env in envs])
+    prev_action = np.zeros(E, dtype=np.int64)
+    h = np.zeros((E, H), dtype=np.float32)
+    c = np.zeros((E, H), dtype=np.float32)
+    trajs = [UnrolledTrajectory() for _ in range(E)]
+    for t in trajs:
+        t.initialize()
+    episode = 0
+    score = np.zeros(E)
+    episode_step = np.zeros(E, dtype=np.int64)
+    unrolls = 0
+    steps_in_unroll = 0
+    try:
+        while args.max_unrolls <= 0 or unrolls < args.max_unrolls * E:
+            if steps_in_unroll == 0:
+                agent.parameter_sync()
+            actions, policies, max_probs, nh, nc = \
+                agent.get_policy_and_action_batch(states, prev_action, h, c)
+            for e, env in enumerate(envs):
+                env_action = int(actions[e]) % available_action
+                next_state, reward, done, info = env.step(env_action)
+                if info.get("life_lost"):
+                    reward, done = -1.0, True
+                score[e] += reward
+                episode_step[e] += 1
+                trajs[e].append(
+                    state=states[e], next_state=next_state,
+                    previous_action=int(prev_action[e]),
+                    action=int(actions[e]), reward=reward, done=done,
+                    behavior_policy=policies[e], initial_h=h[e],
+                    initial_c=c[e])
+                states[e], prev_action[e] = next_state, actions[e]
+                h[e], c[e] = nh[e], nc[e]
+                if done:
+                    writer.add_scalar("data/score", score[e], episode)
+                    writer.add_scalar("data/episode_step",
+                                      episode_step[e], episode)
+                    episode += 1
+                    score[e], episode_step[e] = 0.0, 0
+                    states[e] = env.reset()
+                    prev_action[e] = 0
+                    h[e] = 0.0
+                    c[e] = 0.0
+            steps_in_unroll += 1
+            if steps_in_unroll == cfg.trajectory:
+                for e in range(E):
+                    queue.append_to_queue(task, **trajs[e].stacked())
+                    trajs[e].initialize()
+                unrolls += E
+                steps_in_unroll = 0
     finally:
         writer.close()
         queue.close()

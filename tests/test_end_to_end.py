@@ -29,6 +29,11 @@ def _tiny_config(tmp_path):
         "impala": block("SyntheticAtari", 2, 4, model_input=[84, 84, 4],
                         model_output=4, queue_size=16, batch_size=2,
                         trajectory=6, lstm_size=8),
+        # vectorized actors: 1 process x 4 envs, batched inference
+        "impala_vec": block("SyntheticAtari", 1, 4,
+                            model_input=[84, 84, 4], model_output=4,
+                            queue_size=16, batch_size=2, trajectory=6,
+                            lstm_size=8, envs_per_actor=4),
         "apex": block("SyntheticAtari", 1, 4, model_input=[84, 84, 4],
                       model_output=4, queue_size=8, batch_size=4,
                       trajectory=4),
@@ -61,6 +66,7 @@ def _run_main(algo, block, cfg_path, max_steps, q):
 @pytest.mark.parametrize("algo,block,max_steps,timeout", [
     ("a3c", "a3c_cartpole", 4, 120),
     ("impala", "impala", 3, 240),
+    ("impala", "impala_vec", 3, 240),
     ("apex", "apex", 3, 300),
     ("r2d2", "r2d2", 3, 300),
 ])
