@@ -224,9 +224,11 @@ def vector_actor(ctx: common.TrainerContext, task: int) -> None:
     Same behavior per env as the scalar loop — unrolls of length T with
     stored per-step LSTM state, life-loss shaping, weight pull once per
     unroll period — but the model forward amortizes over E envs
-    (config key ``envs_per_actor``)."""
+    (config key ``envs_per_actor``; ``actor_threads`` sizes the torch
+    intra-op pool for the batched forward — a 256-core EPYC host has
+    ~12 cores per actor process to spare at the reference's 20 actors)."""
     import torch
-    torch.set_num_threads(1)
+    torch.set_num_threads(int(ctx.cfg.get("actor_threads", 1)))
     cfg, args = ctx.cfg, ctx.args
     E = int(cfg.get("envs_per_actor", 1))
     env_name = cfg.env[task]

@@ -15,7 +15,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
 import torch
 
-torch.set_num_threads(1)
+torch.set_num_threads(int(os.environ.get("ACTOR_THREADS", "1")))
 
 from distributed_reinforcement_learning_amd.agents import impala
 from distributed_reinforcement_learning_amd.envs import make_uint8_env
@@ -68,10 +68,9 @@ def bench_vector(E, ticks=120):
 
 
 if __name__ == "__main__":
-    scalar = bench_scalar()
-    out = {"scalar_frames_per_s": scalar}
-    for E in (8, 16, 32):
-        out[f"vector{E}_frames_per_s"] = bench_vector(E)
-    out["speedup_16"] = out["vector16_frames_per_s"] / scalar
-    print(json.dumps({k: round(v, 1) for k, v in out.items()}),
-          flush=True)
+    out = {"threads": torch.get_num_threads()}
+    if os.environ.get("SKIP_SCALAR") != "1":
+        out["scalar_frames_per_s"] = round(bench_scalar(), 1)
+    for E in (16, 32, 64):
+        out[f"vector{E}_frames_per_s"] = round(bench_vector(E), 1)
+    print(json.dumps(out), flush=True)
