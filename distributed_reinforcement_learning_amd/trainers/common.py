@@ -113,6 +113,16 @@ def _actor_process_entry(algorithm: str, actor_fn: Callable, args,
     for var in ("WORLD_SIZE", "RANK", "LOCAL_RANK"):
         os.environ.pop(var, None)
     torch.set_num_threads(1)
+    # numpy's BLAS (the env pipeline's resize matmuls) defaults its pool
+    # to nproc (256 on the EPYC hosts) PER PROCESS; with 20 actors inside
+    # a 16-core cgroup quota that oversubscription collapses throughput
+    # ~8x (measured r2). Clamp every pool to the actor's thread budget.
+    try:
+        import threadpoolctl
+        global _TP_LIMITS  # keep alive: limits restore when GC'd
+        _TP_LIMITS = threadpoolctl.threadpool_limits(1)
+    except Exception:
+        pass
     actor_args = argparse.Namespace(**vars(args))
     actor_args.spawn = False
     actor_args.job_name = "actor"

@@ -228,7 +228,13 @@ def vector_actor(ctx: common.TrainerContext, task: int) -> None:
     intra-op pool for the batched forward — a 256-core EPYC host has
     ~12 cores per actor process to spare at the reference's 20 actors)."""
     import torch
-    torch.set_num_threads(int(ctx.cfg.get("actor_threads", 1)))
+    threads = int(ctx.cfg.get("actor_threads", 1))
+    torch.set_num_threads(threads)
+    try:
+        import threadpoolctl
+        ctx._tp_limits = threadpoolctl.threadpool_limits(threads)
+    except Exception:
+        pass
     cfg, args = ctx.cfg, ctx.args
     E = int(cfg.get("envs_per_actor", 1))
     env_name = cfg.env[task]

@@ -15,7 +15,13 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
 import torch
 
-torch.set_num_threads(int(os.environ.get("ACTOR_THREADS", "1")))
+_THREADS = int(os.environ.get("ACTOR_THREADS", "1"))
+torch.set_num_threads(_THREADS)
+try:
+    import threadpoolctl
+    _TP = threadpoolctl.threadpool_limits(_THREADS)
+except Exception:
+    pass
 
 from distributed_reinforcement_learning_amd.agents import impala
 from distributed_reinforcement_learning_amd.envs import make_uint8_env
