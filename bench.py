@@ -251,6 +251,8 @@ def _dist_setup(agent, world):
             broadcast_module,
         )
         broadcast_module(agent.model)
+        if getattr(agent, "target_model", None) is not None:
+            broadcast_module(agent.target_model)
         agent.setup_all_reduce()
 
 

@@ -36,6 +36,28 @@ def test_bench_single_process():
     _check_line(r.stdout.strip().splitlines()[-1], 1)
 
 
+@pytest.mark.parametrize("algo,extra", [
+    ("apex", ["--batch", "4"]),
+    ("r2d2", ["--batch", "2", "--seq-len", "6", "--burn-in", "2"]),
+])
+def test_bench_algo_flags(algo, extra):
+    """--algo apex/r2d2 (BASELINE configs #3/#4) emit well-formed contract
+    lines; CPU fallback exercises the same sample->train->priority-update
+    loop against the float64 Memory."""
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--algo", algo, "--steps", "2",
+         "--warmup", "1"] + extra,
+        capture_output=True, text=True, cwd=REPO, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    out = json.loads(r.stdout.strip().splitlines()[-1])
+    assert algo in out["metric"].lower().replace("-", "")
+    assert out["value"] > 0 and out["ms_per_step"] > 0
+    assert out["data"] == "synthetic"
+    if algo == "r2d2":
+        assert out["config"]["seq_len"] == 6
+        assert out["config"]["burn_in"] == 2
+
+
 def test_bench_torchrun_two_ranks():
     """Exactly the driver's multi-GPU launch shape, on CPU/gloo."""
     env = dict(os.environ)
