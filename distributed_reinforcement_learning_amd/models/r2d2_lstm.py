@@ -32,12 +32,16 @@ class R2D2LstmQ(nn.Module):
         feat = self.conv.out_features + self.action_emb.out_features
         self.lstm = LSTMCellTF(feat, lstm_size)
         self.trunk = nn.Linear(lstm_size, 128)
-        self.value_out = nn.Linear(128, num_action)
-        self.mean_out = nn.Linear(128, 1)
+        # value(|A|) and mean(1) branches share ONE [128, A+1] GEMM
+        # (columns [:A] = value, column A = mean) — identical math to the
+        # reference's two dense layers (r2d2_lstm.py:46-49) at half the
+        # output-GEMM launches; the dueling subtract slices the result
+        self.out = nn.Linear(128, num_action + 1)
 
     def _head(self, h: torch.Tensor) -> torch.Tensor:
         x = F.relu(self.trunk(h.to(self.trunk.weight.dtype)))
-        return self.value_out(x) - self.mean_out(x)
+        y = self.out(x)
+        return y[..., :self.num_action] - y[..., self.num_action:]
 
     def features(self, state: torch.Tensor,
                  prev_action: torch.Tensor) -> torch.Tensor:
