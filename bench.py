@@ -476,6 +476,13 @@ def bench_r2d2(args, rank, world, local_rank, have_gpu, device):
 
 def main(argv=None) -> None:
     args = parse_args(argv)
+    # benchmark robustness: if the distributed optimizer-graph capture
+    # refuses on some stack, continue with the LOUD eager all-reduce
+    # rather than killing the whole scaling run — the eager rank issues
+    # the same collective sequence as a captured rank, so mixed modes
+    # stay consistent (tests/test_gpu_dist.py keeps the strict fail-fast
+    # to prove capture actually works)
+    os.environ.setdefault("DRLA_ALLOW_EAGER_REDUCE", "1")
     from distributed_reinforcement_learning_amd.parallel import dist as pdist
 
     local_rank = pdist.init_distributed()

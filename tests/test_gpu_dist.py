@@ -29,6 +29,7 @@ def _run_check(nproc: int) -> str:
     env = dict(os.environ)
     env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     env["MASTER_ADDR"] = "127.0.0.1"
+    env.pop("DRLA_ALLOW_EAGER_REDUCE", None)  # strict: capture or fail
     cmd = [sys.executable, "-m", "torch.distributed.run",
            "--nnodes=1", "--nproc-per-node", str(nproc),
            "--master-addr", "127.0.0.1", "--master-port", "29537",
