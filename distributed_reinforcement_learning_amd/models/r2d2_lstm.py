@@ -144,6 +144,16 @@ class R2D2LstmQ(nn.Module):
                 xg.reshape(B, L, -1).contiguous(), w[F_:].contiguous(),
                 h0.float().contiguous(), c0.float().contiguous(),
                 seq_done.contiguous(), self.lstm.forget_bias)
+            if (self.out.weight.dtype == torch.bfloat16
+                    and self.trunk.weight.dtype == torch.bfloat16):
+                # whole dueling head in one launch (window slice resolved
+                # by index math in-kernel; replaces cast/addmm/relu/addmm/
+                # slice-sub)
+                return ext.dueling_head_fwd(
+                    h_all, self.trunk.weight.contiguous(),
+                    self.trunk.bias.contiguous(),
+                    self.out.weight.contiguous(),
+                    self.out.bias.contiguous(), burn)
             # the head's dtype cast absorbs the window slice's strides
             return self._head(h_all[:, burn:])
         h, c = h0, c0

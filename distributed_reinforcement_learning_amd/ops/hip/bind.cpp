@@ -86,6 +86,10 @@ extern "C" __global__ void drla_r2d2_loss_fwd(
     const unsigned short*, const float*, const unsigned short*,
     const float*, const int*, const float*, const unsigned char*,
     const float*, float, int, float*, float*, float*, int, int, int);
+extern "C" __global__ void drla_dueling_head_fwd(
+    const float*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, unsigned short*, int,
+    int, int, int, int, int);
 extern "C" __global__ void drla_r2d2_loss_bwd(
     const float*, const int*, const float*, const float*, unsigned short*,
     float*, int, int, int);
@@ -591,6 +595,32 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> r2d2_loss_fwd(
       loss.data_ptr<float>(), td_st.data_ptr<float>(),
       td_out.data_ptr<float>(), B, W, A);
   return {loss, td_st, td_out};
+}
+
+torch::Tensor dueling_head_fwd(torch::Tensor h, torch::Tensor Wt,
+                               torch::Tensor bt, torch::Tensor Wo,
+                               torch::Tensor bo, int64_t burn) {
+  for (auto* t : {&h, &Wt, &bt, &Wo, &bo})
+    check_gpu_contig(*t, "dueling head input");
+  TORCH_CHECK(h.dim() == 3 && h.scalar_type() == torch::kFloat,
+              "h must be [B,L,IN] f32");
+  const int B = h.size(0), L = h.size(1), IN = h.size(2);
+  const int MID = Wt.size(0), AO = Wo.size(0);
+  TORCH_CHECK(Wt.size(1) == IN && Wo.size(1) == MID, "weight shape");
+  TORCH_CHECK(IN <= 256 && MID <= 512 && AO <= 64 && AO >= 2,
+              "dueling head dims out of range");
+  const int W = L - (int)burn;
+  auto q = torch::empty({B, W, AO - 1},
+                        h.options().dtype(torch::kBFloat16));
+  const size_t lds = (size_t)(MID * IN + AO * MID) * 2 +
+                     (size_t)(2 * MID + 2 * IN + 2 * AO) * 4;
+  const long long npairs = ((long long)B * W + 1) / 2;
+  const int gx = (int)std::min<long long>(npairs, 1024);
+  hipLaunchKernelGGL(drla_dueling_head_fwd, dim3(gx), dim3(256), lds,
+                     cur_stream(), h.data_ptr<float>(), u16p(Wt),
+                     u16p(bt), u16p(Wo), u16p(bo), u16pm(q), B, L,
+                     (int)burn, IN, MID, AO);
+  return q;
 }
 
 torch::Tensor r2d2_loss_bwd(torch::Tensor td_st, torch::Tensor actions,
@@ -1306,6 +1336,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("a2c_loss_bwd", &a2c_loss_bwd, "fused A2C loss bwd (K6)");
   m.def("r2d2_loss_fwd", &r2d2_loss_fwd,
         "fused R2D2 sequence-TD tail fwd (K9)");
+  m.def("dueling_head_fwd", &dueling_head_fwd,
+        "no-grad dueling head over the post-burn-in window");
   m.def("r2d2_loss_bwd", &r2d2_loss_bwd,
         "fused R2D2 sequence-TD tail bwd (K9)");
   m.def("per_update", &per_update, "GPU PER segment-tree batched update");

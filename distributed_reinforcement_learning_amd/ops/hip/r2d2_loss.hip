@@ -109,6 +109,75 @@ extern "C" __global__ void drla_r2d2_loss_fwd(
   }
 }
 
+// No-grad dueling head over the post-burn-in window, ONE launch:
+//   x = relu(h @ Wt^T + bt);  y = x @ Wo^T + bo;  q = y[:A] - y[A]
+// replacing the 5-kernel torch chain (cast, trunk addmm, relu, out addmm,
+// slice-sub) on the R2D2 target-net / TD-scoring paths. Weights stream
+// through LDS once per block; h rows come straight from the seq-recurrence
+// kernel's [B,L,H] output with the window slice resolved by index math
+// (srow = b*L + burn + t — no strided-copy kernel).
+// blockDim 256 = 2 rows in flight x 128 lanes; nn.Linear weight layout
+// [out][in] row-major.
+extern "C" __global__ void drla_dueling_head_fwd(
+    const float* __restrict__ h,       // [B, L, IN]
+    const r2_bf16* __restrict__ Wt,    // [MID][IN]
+    const r2_bf16* __restrict__ bt,    // [MID]
+    const r2_bf16* __restrict__ Wo,    // [AO][MID]
+    const r2_bf16* __restrict__ bo,    // [AO]
+    r2_bf16* __restrict__ q,           // [B, W, AO-1]
+    int B, int L, int burn, int IN, int MID, int AO) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  r2_bf16* wt = reinterpret_cast<r2_bf16*>(smem);          // [MID*IN]
+  r2_bf16* wo = wt + MID * IN;                             // [AO*MID]
+  float* xb = reinterpret_cast<float*>(wo + AO * MID);     // [2][MID]
+  float* hb = xb + 2 * MID;                                // [2][IN]
+  float* yb = hb + 2 * IN;                                 // [2][AO]
+
+  const int tid = threadIdx.x;
+  for (int i = tid; i < MID * IN; i += blockDim.x) wt[i] = Wt[i];
+  for (int i = tid; i < AO * MID; i += blockDim.x) wo[i] = Wo[i];
+  __syncthreads();
+
+  const int W = L - burn;
+  const long long npairs = ((long long)B * W + 1) / 2;
+  const int half = tid / 128;   // which of the 2 in-flight rows
+  const int lane = tid % 128;
+  const int A = AO - 1;
+  for (long long p = blockIdx.x; p < npairs; p += gridDim.x) {
+    const long long n = 2 * p + half;
+    const bool live = n < (long long)B * W;
+    if (live) {
+      const long long b = n / W;
+      const long long srow = (b * L + burn + (n - b * W)) * IN;
+      for (int i = lane; i < IN; i += 128) hb[half * IN + i] = h[srow + i];
+    }
+    __syncthreads();
+    if (live) {
+      for (int o = lane; o < MID; o += 128) {
+        float acc = r2_ld(bt, nullptr, o);
+        const r2_bf16* wrow = wt + o * IN;
+        for (int i = 0; i < IN; ++i)
+          acc = fmaf(hb[half * IN + i], r2_ld(wrow, nullptr, i), acc);
+        xb[half * MID + o] = fmaxf(acc, 0.0f);
+      }
+    }
+    __syncthreads();
+    if (live && lane < AO) {
+      float acc = r2_ld(bo, nullptr, lane);
+      const r2_bf16* wrow = wo + lane * MID;
+      for (int i = 0; i < MID; ++i)
+        acc = fmaf(xb[half * MID + i], r2_ld(wrow, nullptr, i), acc);
+      yb[half * AO + lane] = acc;
+    }
+    __syncthreads();
+    if (live && lane < A) {
+      q[n * A + lane] =
+          drla_f32_to_bf16(yb[half * AO + lane] - yb[half * AO + A]);
+    }
+    __syncthreads();
+  }
+}
+
 extern "C" __global__ void drla_r2d2_loss_bwd(
     const float* __restrict__ td_st,   // [B,W-1]
     const int* __restrict__ actions,   // [B,W]

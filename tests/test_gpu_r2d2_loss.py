@@ -118,3 +118,23 @@ def test_agent_fused_path_matches_eager_composition():
 
     assert float(loss_f) == pytest.approx(float(loss_e), rel=0.05)
     assert torch.allclose(td_f, td_e, atol=0.05, rtol=0.05)
+
+
+def test_dueling_head_kernel_matches_torch():
+    """drla_dueling_head_fwd (one-launch no-grad head over the window)
+    vs the torch composition on the same bf16 weights."""
+    from distributed_reinforcement_learning_amd.models import R2D2LstmQ
+    torch.manual_seed(13)
+    m = R2D2LstmQ([84, 84, 1], 4, 64).cuda().bfloat16()
+    from distributed_reinforcement_learning_amd import ops as _o
+    ext = _o.require_ext()
+    B, L, burn = 3, 12, 5
+    h_all = torch.randn(B, L, 64, device="cuda")
+    q_k = ext.dueling_head_fwd(
+        h_all, m.trunk.weight.contiguous(), m.trunk.bias.contiguous(),
+        m.out.weight.contiguous(), m.out.bias.contiguous(), burn)
+    with torch.no_grad():
+        q_t = m._head(h_all[:, burn:])
+    assert q_k.shape == (B, L - burn, 4)
+    assert torch.allclose(q_k.float(), q_t.float(), atol=3e-2, rtol=3e-2), \
+        (q_k.float() - q_t.float()).abs().max()
