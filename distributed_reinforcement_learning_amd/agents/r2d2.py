@@ -123,12 +123,32 @@ class Agent(AgentBase):
         burn-in head evaluations, no zero-pad/cat (the eager _unroll_q
         padded to keep the [B,L] interface)."""
         b = self.burn_in
+        # the target-net evaluation is independent of every main-net pass
+        # and the seq-recurrence kernels run one block per sequence (B=16
+        # workgroups on a 256-CU chip) — overlap it on a side stream.
+        # Stream forks are capture-legal: hipGraph capture spans streams
+        # and records the fork/join edges.
+        side = getattr(self, "_tgt_stream", None)
+        if side is None and self.device.type == "cuda":
+            side = self._tgt_stream = torch.cuda.Stream()
+        main_stream = torch.cuda.current_stream() if side else None
         with torch.no_grad():
-            # target net: ONE full-sequence pass (conv/embed/Wx once, one
-            # recurrence kernel, head on the window only)
-            tgt_w = self.target_model.q_window(s, pa, h, c, d, b)
+            if side is not None:
+                side.wait_stream(main_stream)
+                with torch.cuda.stream(side):
+                    # ONE full-sequence pass (conv/embed/Wx once, one
+                    # recurrence kernel, fused head on the window only)
+                    tgt_w = self.target_model.q_window(s, pa, h, c, d, b)
+            else:
+                tgt_w = self.target_model.q_window(s, pa, h, c, d, b)
             if not with_grad:
                 main_w = self.model.q_window(s, pa, h, c, d, b)
+                if side is not None:
+                    main_stream.wait_stream(side)
+                    # allocated on side, consumed on main (eager only:
+                    # inside capture the graph pool owns lifetimes)
+                    if not torch.cuda.is_current_stream_capturing():
+                        tgt_w.record_stream(main_stream)
                 return main_w, tgt_w
             if b > 0:
                 hm, cm = self.model.burn_in_states(s[:, :b], pa[:, :b],
@@ -139,6 +159,10 @@ class Agent(AgentBase):
         # backward from sweeping the burn-in frames
         main_w = self.model.unroll_sequence(
             s[:, b:], pa[:, b:], hm.detach(), cm.detach(), d[:, b:])
+        if side is not None:
+            main_stream.wait_stream(side)
+            if not torch.cuda.is_current_stream_capturing():
+                tgt_w.record_stream(main_stream)
         return main_w, tgt_w
 
     def _fused_seq_loss(self, state, previous_action, action, h0, c0,
