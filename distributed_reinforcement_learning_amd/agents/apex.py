@@ -118,11 +118,27 @@ class Agent(AgentBase):
         clipped_r = clip_rewards(r, self.reward_clipping)
         discounts = (~d).float() * self.discount_factor
         B = s.shape[0]
+        # target-net eval is independent of the main forward — overlap it
+        # on a side stream (capture-legal fork/join, same pattern as the
+        # R2D2 agent's _window_qs)
+        side = getattr(self, "_tgt_stream", None)
+        if side is None and self.device.type == "cuda":
+            side = self._tgt_stream = torch.cuda.Stream()
+        if side is not None:
+            main_stream = torch.cuda.current_stream()
+            side.wait_stream(main_stream)
+            with torch.cuda.stream(side), torch.no_grad():
+                next_target_q = self.target_model(ns, a)
+        else:
+            with torch.no_grad():
+                next_target_q = self.target_model(ns, a)
         qs = self.model(torch.cat([s, ns]), torch.cat([pa, a]))
         main_q = qs[:B]
         next_main_q = qs[B:].detach()
-        with torch.no_grad():
-            next_target_q = self.target_model(ns, a)
+        if side is not None:
+            main_stream.wait_stream(side)
+            if not torch.cuda.is_current_stream_capturing():
+                next_target_q.record_stream(main_stream)
         return fused_dqn_loss(main_q, next_main_q, next_target_q, a,
                               clipped_r, discounts, w)
 
