@@ -17,6 +17,29 @@ from distributed_reinforcement_learning_amd import ops as _ops
 
 _LAYER_CO = {0: 32, 1: 32, 2: 64, 3: 64}
 
+# The weight gradients feed ONLY the optimizer (after the whole backward),
+# while dx continues the critical dgrad chain — so the wgrad+finalize pair
+# runs on a dedicated side stream and overlaps the rest of the backward
+# (~95 us of MFMA wgrad work hidden behind dgrad/heads/embed). Under
+# hipGraph capture the fork becomes graph edges and the graph owner joins
+# once at capture end (join_wgrad_stream); eagerly each backward joins
+# immediately (correctness over overlap — warmup only).
+_WGRAD_STREAM = None
+
+
+def wgrad_stream() -> torch.cuda.Stream:
+    global _WGRAD_STREAM
+    if _WGRAD_STREAM is None:
+        _WGRAD_STREAM = torch.cuda.Stream()
+    return _WGRAD_STREAM
+
+
+def join_wgrad_stream() -> None:
+    """Order the current stream after all pending side-stream wgrads —
+    call at the END of a captured backward region."""
+    if _WGRAD_STREAM is not None:
+        torch.cuda.current_stream().wait_stream(_WGRAD_STREAM)
+
 
 class _ConvLayer(torch.autograd.Function):
     @staticmethod
@@ -40,8 +63,19 @@ class _ConvLayer(torch.autograd.Function):
         if not (dy.is_contiguous() or (dy.stride(-1) == 1
                                         and dy.dim() >= 2)):
             dy = dy.contiguous()
-        dy_m = ext.relu_mask_bwd(dy, y, co)
-        dw, dbias = ext.conv_wgrad(layer, x, dy_m)
+        dy_m = ext.relu_mask_bwd(dy, y, co, layer)
+        cur = torch.cuda.current_stream()
+        side = wgrad_stream()
+        side.wait_stream(cur)
+        with torch.cuda.stream(side):
+            dw, dbias = ext.conv_wgrad(layer, x, dy_m)
+        if not torch.cuda.is_current_stream_capturing():
+            # eager: join now and pin allocator lifetimes across streams
+            cur.wait_stream(side)
+            for t in (dw, dbias):
+                t.record_stream(cur)
+            for t in (x, dy_m):
+                t.record_stream(side)
         dx = None
         if layer >= 2 and ctx.needs_input_grad[0]:
             dx = ext.conv_dgrad(layer, dy_m, w_flat)

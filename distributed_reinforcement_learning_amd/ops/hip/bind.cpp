@@ -341,17 +341,23 @@ torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
 }
 
 // persistent [16][64] bias-grad partial slots shared by relu_mask_bwd
-// (atomic producers) and wgrad_finalize (sum + re-zero consumer); the conv
-// backward always runs them as a mask -> wgrad pair on one stream.
-static torch::Tensor& dbias_slot_buf(const torch::Tensor& like) {
-  static torch::Tensor buf;
-  if (!buf.defined()) {
-    buf = torch::zeros({16 * 64}, like.options().dtype(torch::kFloat));
+// (atomic producers, compute stream) and wgrad_finalize (sum + re-zero
+// consumer — which may run on the SIDE wgrad stream, see ops/conv_op.py),
+// so each conv layer owns its own slot set: a later layer's mask must not
+// touch slots a pending finalize still reads.
+static torch::Tensor& dbias_slot_buf(const torch::Tensor& like,
+                                     int64_t slot_id) {
+  static torch::Tensor bufs[4];
+  TORCH_CHECK(slot_id >= 0 && slot_id < 4, "bad slot id");
+  if (!bufs[slot_id].defined()) {
+    bufs[slot_id] =
+        torch::zeros({16 * 64}, like.options().dtype(torch::kFloat));
   }
-  return buf;
+  return bufs[slot_id];
 }
 
-torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y, int64_t CO) {
+torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y, int64_t CO,
+                            int64_t slot_id) {
   check_gpu_contig(y, "y");
   TORCH_CHECK(dy.is_cuda(), "dy must be on GPU");
   // dy may be an N-strided view (slice of the fused xh grad): inner dims
@@ -375,7 +381,7 @@ torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y, int64_t CO) {
     n_stride = dy.stride(0);
   }
   auto out = torch::empty(dy.sizes(), dy.options());
-  auto& slots = dbias_slot_buf(dy);
+  auto& slots = dbias_slot_buf(dy, slot_id);
   const long long n = dy.numel();
   TORCH_CHECK(n % 8 == 0, "relu_mask_bwd wants numel % 8 == 0");
   int grid = drla_grid(n / 8);
@@ -437,7 +443,7 @@ std::tuple<torch::Tensor, torch::Tensor> conv_wgrad(int layer,
   }
   auto dw = torch::empty({cfg.co, K}, dy.options().dtype(torch::kBFloat16));
   auto dbias = torch::empty({cfg.co}, dy.options().dtype(torch::kBFloat16));
-  auto& slots = dbias_slot_buf(dy);
+  auto& slots = dbias_slot_buf(dy, layer);
   hipLaunchKernelGGL(drla_wgrad_finalize,
                      dim3(drla_grid((long long)K * cfg.co)),
                      dim3(DRLA_BLOCK), 0, cur_stream(),

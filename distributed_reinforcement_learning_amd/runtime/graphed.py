@@ -118,6 +118,12 @@ class GraphedImpalaStep:
             # pool and are rewritten by every g_fwd replay
             self._total.backward(gradient=self._bwd_seed,
                                  retain_graph=True)
+            # join the side-stream conv wgrads (ops/conv_op.py) so the
+            # capture region ends with a single ordered stream
+            from distributed_reinforcement_learning_amd.ops.conv_op import (
+                join_wgrad_stream,
+            )
+            join_wgrad_stream()
         # .grad now holds capture-pool tensors at replay-stable addresses
         opt.build_gather_table()
         from distributed_reinforcement_learning_amd.parallel.dist import (
@@ -159,7 +165,15 @@ class GraphedImpalaStep:
 
     def _fwd(self):
         agent = self.agent
-        i = self.inputs
+        i = dict(self.inputs)
+        # The overlapped H2D upload (copy stream) is ordered only after
+        # the FORWARD, but four inputs are re-read by backward kernels
+        # (conv-l1 wgrad: state; embed scatter: previous_action; V-trace
+        # bwd: action; LSTM tail bwd: initial_c) — clone those INSIDE the
+        # forward graph so every read of a static buffer happens before
+        # the upload barrier. ~19 MB through the capture pool, ~6 us.
+        for k in ("state", "previous_action", "action", "initial_c"):
+            i[k] = i[k].clone()
         s = agent.prepare_frames(i["state"])
         pi_loss, baseline_loss, entropy, total = agent.compute_losses(
             s, i["reward"], i["action"], i["done"], i["behavior_policy"],

@@ -46,6 +46,10 @@ class GraphedReplayStep:
             b = memory.gather(rows)
             loss, td = loss_fn(b, w)
             loss.backward()
+            from distributed_reinforcement_learning_amd.ops.conv_op import (
+                join_wgrad_stream,
+            )
+            join_wgrad_stream()  # side-stream conv wgrads (capture-safe)
             memory.update_batch(idxs, td)
             return loss.detach(), td
 
@@ -187,6 +191,10 @@ class GraphedTrainStep:
         def _iter():
             losses = loss_fn(self.inputs)
             losses[-1].backward()
+            from distributed_reinforcement_learning_amd.ops.conv_op import (
+                join_wgrad_stream,
+            )
+            join_wgrad_stream()  # side-stream conv wgrads (capture-safe)
             return tuple(x.detach() for x in losses[:-1])
 
         side = torch.cuda.Stream()
