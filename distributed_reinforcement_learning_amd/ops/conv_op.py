@@ -27,6 +27,14 @@ _LAYER_CO = {0: 32, 1: 32, 2: 64, 3: 64}
 _WGRAD_STREAM = None
 
 
+def _side_enabled() -> bool:
+    # measured SLOWER inside the captured step on ROCm 7.2 (cross-stream
+    # graph edges cost more than the overlap wins — 0.59 -> 1.01 ms/step,
+    # r2); opt-in for experiments
+    import os
+    return os.environ.get("DRLA_WGRAD_SIDE") == "1"
+
+
 def wgrad_stream() -> torch.cuda.Stream:
     global _WGRAD_STREAM
     if _WGRAD_STREAM is None:
@@ -64,18 +72,21 @@ class _ConvLayer(torch.autograd.Function):
                                         and dy.dim() >= 2)):
             dy = dy.contiguous()
         dy_m = ext.relu_mask_bwd(dy, y, co, layer)
-        cur = torch.cuda.current_stream()
-        side = wgrad_stream()
-        side.wait_stream(cur)
-        with torch.cuda.stream(side):
+        if _side_enabled():
+            cur = torch.cuda.current_stream()
+            side = wgrad_stream()
+            side.wait_stream(cur)
+            with torch.cuda.stream(side):
+                dw, dbias = ext.conv_wgrad(layer, x, dy_m)
+            if not torch.cuda.is_current_stream_capturing():
+                # eager: join now, pin allocator lifetimes across streams
+                cur.wait_stream(side)
+                for t in (dw, dbias):
+                    t.record_stream(cur)
+                for t in (x, dy_m):
+                    t.record_stream(side)
+        else:
             dw, dbias = ext.conv_wgrad(layer, x, dy_m)
-        if not torch.cuda.is_current_stream_capturing():
-            # eager: join now and pin allocator lifetimes across streams
-            cur.wait_stream(side)
-            for t in (dw, dbias):
-                t.record_stream(cur)
-            for t in (x, dy_m):
-                t.record_stream(side)
         dx = None
         if layer >= 2 and ctx.needs_input_grad[0]:
             dx = ext.conv_dgrad(layer, dy_m, w_flat)
