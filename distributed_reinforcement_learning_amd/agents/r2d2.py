@@ -128,8 +128,10 @@ class Agent(AgentBase):
         # workgroups on a 256-CU chip) — overlap it on a side stream.
         # Stream forks are capture-legal: hipGraph capture spans streams
         # and records the fork/join edges.
+        import os as _os
         side = getattr(self, "_tgt_stream", None)
-        if side is None and self.device.type == "cuda":
+        if (side is None and self.device.type == "cuda"
+                and _os.environ.get("DRLA_NO_TGT_STREAM") != "1"):
             side = self._tgt_stream = torch.cuda.Stream()
         main_stream = torch.cuda.current_stream() if side else None
         with torch.no_grad():

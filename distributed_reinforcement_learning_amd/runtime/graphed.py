@@ -172,8 +172,10 @@ class GraphedImpalaStep:
         # bwd: action; LSTM tail bwd: initial_c) — clone those INSIDE the
         # forward graph so every read of a static buffer happens before
         # the upload barrier. ~19 MB through the capture pool, ~6 us.
-        for k in ("state", "previous_action", "action", "initial_c"):
-            i[k] = i[k].clone()
+        import os as _os
+        if _os.environ.get("DRLA_NO_INPUT_CLONE") != "1":  # measurement-only escape
+            for k in ("state", "previous_action", "action", "initial_c"):
+                i[k] = i[k].clone()
         s = agent.prepare_frames(i["state"])
         pi_loss, baseline_loss, entropy, total = agent.compute_losses(
             s, i["reward"], i["action"], i["done"], i["behavior_policy"],
