@@ -49,14 +49,22 @@ def join_wgrad_stream() -> None:
         torch.cuda.current_stream().wait_stream(_WGRAD_STREAM)
 
 
+# when True (set by GraphedImpalaStep around its warmup+captures), the u8
+# layer-1 forward kernel bundles a pass-through copy of its input and the
+# backward consumes THAT — the graphed step's overlapped H2D rewrites the
+# static input during the backward otherwise (torn l1 wgrad reads)
+STASH_INPUTS = False
+
+
 class _ConvLayer(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, w_flat: torch.Tensor,
                 bias_f32: torch.Tensor, layer: int):
         ext = _ops.require_ext()
-        y = ext.conv_fwd(layer, x.contiguous(), w_flat.contiguous(),
-                         bias_f32.contiguous())
-        ctx.save_for_backward(x, w_flat, y)
+        stash = STASH_INPUTS and layer <= 1
+        y, x_st = ext.conv_fwd(layer, x.contiguous(), w_flat.contiguous(),
+                               bias_f32.contiguous(), stash)
+        ctx.save_for_backward(x_st if stash else x, w_flat, y)
         ctx.layer = layer
         return y
 

@@ -29,11 +29,13 @@ extern "C" __global__ void drla_mfma_probe(const unsigned short*,
 extern "C" __global__ void drla_conv_fwd_l1(const unsigned char*,
                                             const unsigned short*,
                                             const unsigned short*,
-                                            unsigned short*, int);
+                                            unsigned short*, int,
+                                            unsigned char*);
 extern "C" __global__ void drla_conv_fwd_l1_c1(const unsigned char*,
                                                const unsigned short*,
                                                const unsigned short*,
-                                               unsigned short*, int);
+                                               unsigned short*, int,
+                                               unsigned char*);
 extern "C" __global__ void drla_conv_fwd_l2(const unsigned short*,
                                             const unsigned short*,
                                             const unsigned short*,
@@ -299,8 +301,9 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
   return D;
 }
 
-torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
-                       torch::Tensor bias) {
+std::tuple<torch::Tensor, torch::Tensor> conv_fwd(
+    int layer, torch::Tensor in, torch::Tensor w, torch::Tensor bias,
+    bool stash_input) {
   check_gpu_contig(in, "in");
   check_gpu_contig(w, "w");
   check_gpu_contig(bias, "bias");
@@ -313,16 +316,27 @@ torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
                           in.options().dtype(torch::kBFloat16));
   const int grid = (M + 127) / 128;
   auto* outp = u16pm(out);
+  // stashed-input mode (u8 layers only): the kernel bundles a
+  // pass-through copy of the input so the backward never re-reads the
+  // (possibly upload-overlapped) source buffer
+  torch::Tensor x_stash;
+  unsigned char* stashp = nullptr;
+  if (stash_input) {
+    TORCH_CHECK(layer <= 1, "input stash is a u8-layer feature");
+    TORCH_CHECK((in.numel() % 16) == 0, "stash wants 16-divisible bytes");
+    x_stash = torch::empty_like(in);
+    stashp = x_stash.data_ptr<uint8_t>();
+  }
   switch (layer) {
     case 0:
       hipLaunchKernelGGL(drla_conv_fwd_l1, dim3(grid), dim3(256), 0,
                          cur_stream(), in.data_ptr<uint8_t>(), u16p(w),
-                         u16p(bias), outp, batch);
+                         u16p(bias), outp, batch, stashp);
       break;
     case 1:
       hipLaunchKernelGGL(drla_conv_fwd_l1_c1, dim3(grid), dim3(256), 0,
                          cur_stream(), in.data_ptr<uint8_t>(), u16p(w),
-                         u16p(bias), outp, batch);
+                         u16p(bias), outp, batch, stashp);
       break;
     case 2:
       hipLaunchKernelGGL(drla_conv_fwd_l2, dim3(grid), dim3(256), 0,
@@ -337,7 +351,7 @@ torch::Tensor conv_fwd(int layer, torch::Tensor in, torch::Tensor w,
     default:
       TORCH_CHECK(false, "bad layer");
   }
-  return out;
+  return {out, x_stash};
 }
 
 // persistent [16][64] bias-grad partial slots shared by relu_mask_bwd

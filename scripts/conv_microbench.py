@@ -60,7 +60,7 @@ def main():
         bias = conv_bf.bias.contiguous()
         xc = x.contiguous()
 
-        t_custom = t_ms(lambda: ext.conv_fwd(layer, xc, w_flat, bias))
+        t_custom = t_ms(lambda: ext.conv_fwd(layer, xc, w_flat, bias, False))
         t_torch = t_ms(lambda: F.relu(conv_ref(x_ref)))
         flops = 2.0 * N * ho * ho * co * k * k * ci
         results.append(
@@ -71,7 +71,7 @@ def main():
 
         # backward A/B: wgrad+dgrad via our kernels vs autograd
         dy = torch.randn(N, ho, ho, co, device="cuda").bfloat16().contiguous()
-        y = ext.conv_fwd(layer, xc, w_flat, bias)
+        y, _ = ext.conv_fwd(layer, xc, w_flat, bias, False)
         def custom_bwd():
             dy_m = ext.relu_mask_bwd(dy, y, co)
             ext.conv_wgrad(layer, xc, dy_m)

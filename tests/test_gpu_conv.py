@@ -62,8 +62,8 @@ def test_conv_fwd_layer_parity(ext, layer, ci, co, k, s, hi, ho, u8):
         x_in = torch.randn(N, hi, hi, ci, device="cuda").bfloat16() * 0.5
         x_ref = x_in.float()
     w_flat = conv_bf.weight.permute(0, 2, 3, 1).reshape(co, -1).contiguous()
-    y = ext.conv_fwd(layer, x_in.contiguous(), w_flat,
-                     conv_bf.bias.contiguous())
+    y, _ = ext.conv_fwd(layer, x_in.contiguous(), w_flat,
+                        conv_bf.bias.contiguous(), False)
     ref = _torch_conv_nhwc(x_ref, conv)
     err = (y.float() - ref).abs()
     scale = ref.abs().mean().clamp(min=1e-3)
