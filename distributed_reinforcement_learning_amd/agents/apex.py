@@ -201,6 +201,20 @@ class Agent(AgentBase):
         return next_main, target
 
     @torch.no_grad()
+    def get_actions_batch(self, states, previous_actions, epsilons):
+        """Vectorized ε-greedy over E envs in ONE forward (vector actors,
+        trainers/apex.py). Returns (actions [E], q [E,A], q_a [E])."""
+        s = self.frames_to_device(np.asarray(states))
+        pa = self.to_device(np.asarray(previous_actions), torch.int64)
+        q = self.model(s, pa).float().cpu().numpy()
+        E = len(q)
+        greedy = q.argmax(axis=1)
+        rand = self.rng.integers(self.num_action, size=E)
+        explore = self.rng.random(E) <= np.asarray(epsilons)
+        actions = np.where(explore, rand, greedy).astype(np.int64)
+        return actions, q, q[np.arange(E), actions].astype(np.float32)
+
+    @torch.no_grad()
     def get_policy_and_action(self, state, previous_action, epsilon: float
                               ) -> Tuple[int, np.ndarray, float]:
         s = self.frames_to_device(np.asarray(state)[None])

@@ -321,6 +321,25 @@ class Agent(AgentBase):
         return q[0].float().cpu().numpy()
 
     @torch.no_grad()
+    def get_actions_batch(self, states, h, c, previous_actions, epsilons):
+        """Vectorized ε-greedy single step over E envs in ONE forward
+        (vector actors, trainers/r2d2.py). Returns (actions [E], q_a [E],
+        h' [E,H], c' [E,H])."""
+        s = self.frames_to_device(np.asarray(states))
+        pa = self.to_device(np.asarray(previous_actions), torch.int64)
+        ht = self.to_device(np.asarray(h, dtype=np.float32), torch.float32)
+        ct = self.to_device(np.asarray(c, dtype=np.float32), torch.float32)
+        q, nh, nc = self.model.single_step(s, pa, ht, ct)
+        qv = q.float().cpu().numpy()
+        E = len(qv)
+        greedy = qv.argmax(axis=1)
+        rand = self.rng.integers(self.num_action, size=E)
+        explore = self.rng.random(E) <= np.asarray(epsilons)
+        actions = np.where(explore, rand, greedy).astype(np.int64)
+        return (actions, qv[np.arange(E), actions].astype(np.float32),
+                nh.float().cpu().numpy(), nc.float().cpu().numpy())
+
+    @torch.no_grad()
     def get_action(self, state, h, c, previous_action, epsilon: float):
         """Epsilon-greedy single-step acting (reference agent/r2d2.py:166-186).
         Returns (action, q[action], h', c')."""

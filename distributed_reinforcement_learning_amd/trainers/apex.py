@@ -214,6 +214,8 @@ def actor(ctx: common.TrainerContext, task: int) -> None:
     torch.set_num_threads(1)  # batch-1 CPU inference; also avoids
     # the forked-child OpenMP deadlock (see trainers/common.py)
     cfg, args = ctx.cfg, ctx.args
+    if int(cfg.get("envs_per_actor", 1)) > 1:
+        return vector_actor(ctx, task)
     env = make_uint8_env(cfg.env[task], num_actions=cfg.model_output,
                          seed=(args.seed or 0) + task)
     available_action = cfg.available_action[task]
@@ -273,6 +275,99 @@ def actor(ctx: common.TrainerContext, task: int) -> None:
                 state = env.reset()
                 previous_action = 0
                 agent.parameter_sync()  # reference pulls per episode (:177)
+    finally:
+        writer.close()
+        queue.close()
+
+
+def vector_actor(ctx: common.TrainerContext, task: int) -> None:
+    """E envs per actor process with ONE batched ε-greedy forward per tick
+    (``envs_per_actor`` — same redesign as trainers/impala.vector_actor);
+    per-env LocalBuffer/episode state, same enqueue-a-random-sample
+    behavior as the scalar loop (reference train_apex.py:157-231)."""
+    import torch
+    threads = int(ctx.cfg.get("actor_threads", 1))
+    torch.set_num_threads(threads)
+    try:
+        import threadpoolctl
+        ctx._tp_limits = threadpoolctl.threadpool_limits(threads)
+    except Exception:
+        pass
+    cfg, args = ctx.cfg, ctx.args
+    E = int(cfg.get("envs_per_actor", 1))
+    available_action = cfg.available_action[task]
+    envs = [make_uint8_env(cfg.env[task], num_actions=cfg.model_output,
+                           seed=(args.seed or 0) + task * 1000 + e)
+            for e in range(E)]
+    queue = TrajectoryQueue(
+        queue_schema_for("apex", cfg), cfg.num_actors, cfg.queue_size,
+        role="actor", namespace=ctx.namespace, actor_task=task,
+        world_size=ctx.world_size)
+    agent = build_agent(ctx, "cpu", False, (args.seed or 0) + 1000 + task)
+    agent.weight_subscriber = WeightSubscriber(ctx.weights_name,
+                                               agent.model.state_dict())
+    agent.weight_subscriber.wait_for_first()
+    writer = SummaryWriter(ctx.actor_logdir(task))
+    locals_ = [LocalBuffer(LOCAL_BUFFER_CAPACITY,
+                           seed=(args.seed or 0) + task * 1000 + e)
+               for e in range(E)]
+
+    states = np.stack([env.reset() for env in envs])
+    prev_action = np.zeros(E, dtype=np.int64)
+    episode_n = np.zeros(E, dtype=np.int64)
+    episode = 0
+    score = np.zeros(E)
+    episode_step = np.zeros(E, dtype=np.int64)
+    q_sum = np.zeros(E)
+    enqueued = 0
+    T = cfg.trajectory
+    sync_every = 0
+    try:
+        while args.max_unrolls <= 0 or enqueued < args.max_unrolls * E:
+            eps = 1.0 / (0.05 * episode_n + 1)  # reference :229, per env
+            actions, _, q_a = agent.get_actions_batch(states, prev_action,
+                                                      eps)
+            for e, env in enumerate(envs):
+                next_state, reward, done, info = env.step(
+                    int(actions[e]) % available_action)
+                if info.get("life_lost"):
+                    reward, done = -1.0, True
+                score[e] += reward
+                episode_step[e] += 1
+                q_sum[e] += q_a[e]
+                locals_[e].append(states[e], next_state,
+                                  int(prev_action[e]), int(actions[e]),
+                                  reward, done)
+                states[e], prev_action[e] = next_state, actions[e]
+                if len(locals_[e]) > 3 * T:
+                    s = locals_[e].sample(T)
+                    queue.append_to_queue(
+                        task,
+                        state=np.stack(s["state"]),
+                        next_state=np.stack(s["next_state"]),
+                        previous_action=np.asarray(s["previous_action"],
+                                                   np.int32),
+                        action=np.asarray(s["action"], np.int32),
+                        reward=np.asarray(s["reward"], np.float32),
+                        done=np.asarray(s["done"], np.bool_))
+                    enqueued += 1
+                if done:
+                    writer.add_scalar("data/score", score[e], episode)
+                    writer.add_scalar("data/episode_step",
+                                      episode_step[e], episode)
+                    writer.add_scalar("data/epsilon", eps[e], episode)
+                    writer.add_scalar("data/avg_q",
+                                      q_sum[e] / max(episode_step[e], 1),
+                                      episode)
+                    episode += 1
+                    episode_n[e] += 1
+                    score[e], episode_step[e], q_sum[e] = 0.0, 0, 0.0
+                    states[e] = env.reset()
+                    prev_action[e] = 0
+            sync_every += 1
+            if sync_every >= T:  # amortized weight pull (per-episode in
+                sync_every = 0   # the scalar loop)
+                agent.parameter_sync()
     finally:
         writer.close()
         queue.close()

@@ -206,6 +206,8 @@ def actor(ctx: common.TrainerContext, task: int) -> None:
     torch.set_num_threads(1)  # batch-1 CPU inference; also avoids
     # the forked-child OpenMP deadlock (see trainers/common.py)
     cfg, args = ctx.cfg, ctx.args
+    if int(cfg.get("envs_per_actor", 1)) > 1:
+        return vector_actor(ctx, task)
     env = pomdp_uint8_env(cfg.env[task], num_actions=cfg.model_output,
                           seed=(args.seed or 0) + task)
     queue = TrajectoryQueue(
@@ -259,6 +261,92 @@ def actor(ctx: common.TrainerContext, task: int) -> None:
                 previous_action = 0
                 h = np.zeros(cfg.lstm_size, dtype=np.float32)
                 c = np.zeros(cfg.lstm_size, dtype=np.float32)
+    finally:
+        writer.close()
+        queue.close()
+
+
+def vector_actor(ctx: common.TrainerContext, task: int) -> None:
+    """E POMDP envs per actor process, ONE batched ε-greedy LSTM step per
+    tick (``envs_per_actor``); per-env recurrent state carry, per-env
+    sequence accumulators, same per-sequence enqueue + weight pull as the
+    scalar loop (reference train_r2d2.py:167-251)."""
+    import torch
+    threads = int(ctx.cfg.get("actor_threads", 1))
+    torch.set_num_threads(threads)
+    try:
+        import threadpoolctl
+        ctx._tp_limits = threadpoolctl.threadpool_limits(threads)
+    except Exception:
+        pass
+    cfg, args = ctx.cfg, ctx.args
+    E = int(cfg.get("envs_per_actor", 1))
+    envs = [pomdp_uint8_env(cfg.env[task], num_actions=cfg.model_output,
+                            seed=(args.seed or 0) + task * 1000 + e)
+            for e in range(E)]
+    queue = TrajectoryQueue(
+        queue_schema_for("r2d2", cfg), cfg.num_actors, cfg.queue_size,
+        role="actor", namespace=ctx.namespace, actor_task=task,
+        world_size=ctx.world_size)
+    agent = build_agent(ctx, "cpu", False, (args.seed or 0) + 1000 + task)
+    agent.weight_subscriber = WeightSubscriber(ctx.weights_name,
+                                               agent.model.state_dict())
+    agent.weight_subscriber.wait_for_first()
+    writer = SummaryWriter(ctx.actor_logdir(task))
+    H = cfg.lstm_size
+    seqs = [FieldTrajectory(["state", "previous_action", "action",
+                             "reward", "done", "initial_h", "initial_c"])
+            for _ in range(E)]
+
+    states = np.stack([env.reset() for env in envs])
+    prev_action = np.zeros(E, dtype=np.int64)
+    h = np.zeros((E, H), dtype=np.float32)
+    c = np.zeros((E, H), dtype=np.float32)
+    episode_n = np.zeros(E, dtype=np.int64)
+    episode = 0
+    score = np.zeros(E)
+    episode_step = np.zeros(E, dtype=np.int64)
+    q_sum = np.zeros(E)
+    enqueued = 0
+    try:
+        while args.max_unrolls <= 0 or enqueued < args.max_unrolls * E:
+            eps = 1.0 / (0.01 * episode_n + 1)  # reference :233, per env
+            actions, q_a, nh, nc = agent.get_actions_batch(
+                states, h, c, prev_action, eps)
+            for e, env in enumerate(envs):
+                next_state, reward, done, info = env.step(int(actions[e]))
+                if info.get("life_lost"):
+                    reward, done = -1.0, True
+                score[e] += reward
+                episode_step[e] += 1
+                q_sum[e] += q_a[e]
+                seqs[e].append(state=states[e],
+                               previous_action=int(prev_action[e]),
+                               action=int(actions[e]), reward=reward,
+                               done=done, initial_h=h[e], initial_c=c[e])
+                states[e], prev_action[e] = next_state, actions[e]
+                h[e], c[e] = nh[e], nc[e]
+                if len(seqs[e]) == cfg.seq_len:
+                    queue.append_to_queue(task, **seqs[e].stacked())
+                    seqs[e].initialize()
+                    enqueued += 1
+                    if enqueued % E == 0:
+                        agent.parameter_sync()
+                if done:
+                    writer.add_scalar("data/score", score[e], episode)
+                    writer.add_scalar("data/episode_step",
+                                      episode_step[e], episode)
+                    writer.add_scalar("data/epsilon", eps[e], episode)
+                    writer.add_scalar("data/avg_q",
+                                      q_sum[e] / max(episode_step[e], 1),
+                                      episode)
+                    episode += 1
+                    episode_n[e] += 1
+                    score[e], episode_step[e], q_sum[e] = 0.0, 0, 0.0
+                    states[e] = env.reset()
+                    prev_action[e] = 0
+                    h[e] = 0.0
+                    c[e] = 0.0
     finally:
         writer.close()
         queue.close()

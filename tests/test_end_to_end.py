@@ -37,9 +37,16 @@ def _tiny_config(tmp_path):
         "apex": block("SyntheticAtari", 1, 4, model_input=[84, 84, 4],
                       model_output=4, queue_size=8, batch_size=4,
                       trajectory=4),
+        "apex_vec": block("SyntheticAtari", 1, 4, model_input=[84, 84, 4],
+                          model_output=4, queue_size=8, batch_size=4,
+                          trajectory=4, envs_per_actor=3),
         "r2d2": block("SyntheticAtari", 1, 4, model_input=[84, 84, 1],
                       model_output=4, queue_size=8, batch_size=2,
                       seq_len=5, burn_in=2, lstm_size=8),
+        "r2d2_vec": block("SyntheticAtari", 1, 4, model_input=[84, 84, 1],
+                          model_output=4, queue_size=8, batch_size=2,
+                          seq_len=5, burn_in=2, lstm_size=8,
+                          envs_per_actor=3),
     }
     p = tmp_path / "config.json"
     p.write_text(json.dumps(cfg))
@@ -67,6 +74,8 @@ def _run_main(algo, block, cfg_path, max_steps, q):
     ("a3c", "a3c_cartpole", 4, 120),
     ("impala", "impala", 3, 240),
     ("impala", "impala_vec", 3, 240),
+    ("apex", "apex_vec", 3, 300),
+    ("r2d2", "r2d2_vec", 3, 300),
     ("apex", "apex", 3, 300),
     ("r2d2", "r2d2", 3, 300),
 ])
