@@ -634,6 +634,8 @@ torch::Tensor dueling_head_fwd(torch::Tensor h, torch::Tensor Wt,
                         h.options().dtype(torch::kBFloat16));
   const size_t lds = (size_t)(MID * IN + AO * MID) * 2 +
                      (size_t)(2 * MID + 2 * IN + 2 * AO) * 4;
+  TORCH_CHECK(lds <= 160 * 1024,
+              "dueling head weights exceed the 160 KB LDS budget");
   const long long npairs = ((long long)B * W + 1) / 2;
   const int gx = (int)std::min<long long>(npairs, 1024);
   hipLaunchKernelGGL(drla_dueling_head_fwd, dim3(gx), dim3(256), lds,

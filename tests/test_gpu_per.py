@@ -180,3 +180,54 @@ def test_graphed_replay_step_trains():
     assert all(np.isfinite(losses))
     assert not torch.equal(agent.optimizer.flat_params, p_before)
     assert float(mem.tree[0]) != pytest.approx(tree_before, rel=1e-6)
+
+
+def test_interior_rebuild_repairs_drift():
+    """drla_per_rebuild_level: corrupt the interior sums, rebuild, and the
+    tree must again be consistent bottom-up (the float32 atomicAdd delta
+    chains drift over millions of updates — ADVICE r1)."""
+    import torch
+    m = _gpu_memory(cap=64, seed=1)
+    rng = np.random.default_rng(2)
+    m.add_batch(torch.as_tensor(rng.random(50).astype(np.float32) * 3,
+                                device="cuda"),
+                {"x": torch.zeros(50, 3, device="cuda")})
+    leaves = m.tree[m.capacity - 1:].clone()
+    # corrupt every interior node
+    m.tree[:m.capacity - 1] += torch.rand(m.capacity - 1, device="cuda")
+    m.rebuild()
+    assert torch.equal(m.tree[m.capacity - 1:], leaves)
+    # every interior node equals the sum of its children
+    tree = m.tree.cpu()
+    for i in range(m.capacity - 1):
+        assert float(tree[i]) == pytest.approx(
+            float(tree[2 * i + 1] + tree[2 * i + 2]), rel=1e-6, abs=1e-6)
+
+
+def test_multi_gather_matches_index_select():
+    """drla_multi_gather (one-kernel all-field batch gather) vs per-field
+    index_select, incl. non-16-divisible row strides."""
+    import torch
+    from distributed_reinforcement_learning_amd.replay.gpu_memory import (
+        GpuMemory,
+    )
+    rng = np.random.default_rng(3)
+    m = GpuMemory(32, fields={
+        "frames": ((5, 16), torch.uint8),     # 80 B rows (16-divisible)
+        "vec": ((15,), torch.float32),        # 60 B rows (byte path)
+        "flag": ((7,), torch.bool),           # 7 B rows (byte path)
+        "idx": ((), torch.int32),             # 4 B rows
+    }, device="cuda:0", seed=4)
+    n = 20
+    m.add_batch(
+        torch.as_tensor(rng.random(n).astype(np.float32), device="cuda"),
+        {"frames": torch.randint(0, 255, (n, 5, 16), dtype=torch.uint8,
+                                 device="cuda"),
+         "vec": torch.randn(n, 15, device="cuda"),
+         "flag": torch.rand(n, 7, device="cuda") > 0.5,
+         "idx": torch.arange(n, dtype=torch.int32, device="cuda")})
+    rows = torch.as_tensor(rng.integers(0, n, 12), device="cuda")
+    out = m.gather(rows)
+    for k, buf in m.data.items():
+        ref = buf.index_select(0, rows)
+        assert torch.equal(out[k], ref), k
