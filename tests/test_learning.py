@@ -52,3 +52,56 @@ def test_a2c_learns_cartpole():
     # clear improvement (the run reaches ~50-90 by 300 updates)
     assert last > max(40.0, 1.5 * first), \
         f"no learning: first10={first:.1f} last10={last:.1f}"
+
+
+def test_dqn_learns_cartpole():
+    """Value-based path end-to-end: ε-greedy double-DQN + prioritized
+    replay (CPU Memory) on CartPole — episode length must improve
+    substantially (the reference's simple_network sanity config,
+    apex_value.py:67-100)."""
+    from distributed_reinforcement_learning_amd.agents import apex as apex_agent
+    from distributed_reinforcement_learning_amd.replay.memory import Memory
+
+    agent = apex_agent.Agent(
+        input_shape=[4], num_action=2, discount_factor=0.99,
+        gradient_clip_norm=40.0, reward_clipping="none",
+        start_learning_rate=1e-3, end_learning_rate=1e-3,
+        learning_frame=10 ** 9, seed=0)
+    env = CartPoleEnv(seed=0)
+    mem = Memory(20_000, seed=0)
+    rng = np.random.default_rng(0)
+
+    episode_lengths = []
+    state = env.reset()
+    prev_action, ep_len, episode = 0, 0, 0
+    B = 64
+    for step in range(6000):
+        eps = max(0.05, 1.0 - episode * 0.02)
+        action, q, q_a = agent.get_policy_and_action(state, prev_action,
+                                                     eps)
+        nstate, r, done, _ = env.step(action)
+        ep_len += 1
+        mem.add(1.0, {"s": state, "ns": nstate, "pa": prev_action,
+                      "a": action, "r": r, "d": done})
+        state, prev_action = nstate, action
+        if done:
+            episode_lengths.append(ep_len)
+            ep_len, episode = 0, episode + 1
+            state = env.reset()
+            prev_action = 0
+            agent.parameter_sync()
+        if len(mem) > 4 * B and step % 4 == 0:
+            samples, idxs, w = mem.sample(B)
+            batch = {k: np.stack([s[k] for s in samples])
+                     for k in samples[0]}
+            _, td = agent.distributed_train(
+                batch["s"], batch["ns"], batch["pa"], batch["a"],
+                batch["r"].astype(np.float32), batch["d"], w)
+            mem.update_batch(idxs, np.abs(td))
+        if step % 200 == 0:
+            agent.target_to_main()
+
+    early = np.mean(episode_lengths[:10])
+    late = np.mean(episode_lengths[-10:])
+    assert late > 2.5 * early, \
+        f"no learning: early {early:.1f} -> late {late:.1f}"
