@@ -126,6 +126,11 @@ class AgentBase:
 
     def setup_all_reduce(self) -> None:
         if is_distributed():
+            # the rank-0 broadcast that precedes this rewrote the bf16
+            # model copy out-of-band; the fp32 master must follow or the
+            # first update reverts the broadcast and diverges the ranks
+            if self.optimizer is not None:
+                self.optimizer.refresh_master()
             self._all_reducer = FlatAllReducer(self.optimizer.flat_grads)
 
     def reduce_gradients(self) -> None:
@@ -172,6 +177,10 @@ class AgentBase:
         self.model.load_state_dict(blob["model"])
         if self.optimizer is not None and blob.get("optimizer") is not None:
             self.optimizer.load_state_dict(blob["optimizer"])
+        elif self.optimizer is not None:
+            # checkpoint without optimizer state: re-derive the fp32
+            # master from the restored model copy
+            self.optimizer.refresh_master()
         self.global_step = blob.get("global_step", 0)
         self.num_env_frames = blob.get("num_env_frames", 0)
         self._load_checkpoint_extra(blob)

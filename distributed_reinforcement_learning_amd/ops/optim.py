@@ -185,6 +185,16 @@ class _FlatOptimizerBase:
             with torch.no_grad():
                 self.flat_params.copy_(self.master)
 
+    def refresh_master(self) -> None:
+        """Re-derive the fp32 master from the (bf16) model copy — REQUIRED
+        after any out-of-band rewrite of the parameters (rank-0 broadcast,
+        checkpoint restore without optimizer state): the update math runs
+        on the master, so a stale master would silently revert the rewrite
+        on the next step and diverge DP ranks."""
+        if self.mixed:
+            with torch.no_grad():
+                self.master.copy_(self.flat_params.float())
+
     def state_dict(self) -> dict:
         sd = {"step_count": self.step_count,
               "state": {k: v for k, v in self._state_tensors().items()}}

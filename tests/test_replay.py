@@ -122,3 +122,26 @@ def test_memory_state_dict_roundtrip():
     m2.load_state_dict(sd)
     assert m2.tree.total() == pytest.approx(m.tree.total())
     assert m2.tree.n_entries == m.tree.n_entries
+
+
+def test_refresh_master_after_out_of_band_param_rewrite():
+    """bf16 model copy + fp32 master: an out-of-band parameter rewrite
+    (rank-0 broadcast, restore) must be followed by refresh_master(), or
+    the next update runs on the stale master and silently reverts the
+    rewrite (the world>1 divergence bug fixed in r2 — setup_all_reduce
+    and load_weights now call it)."""
+    import torch
+    from distributed_reinforcement_learning_amd.ops.optim import FusedRMSProp
+
+    p = torch.nn.Parameter(torch.randn(64).bfloat16())
+    opt = FusedRMSProp([p], lr=0.0, clip_norm=None)  # lr=0: pure sync test
+    assert opt.mixed
+    # simulate a broadcast: overwrite the bf16 copy out-of-band
+    with torch.no_grad():
+        new = torch.randn(64).bfloat16()
+        opt.flat_params[:64].copy_(new)
+    opt.refresh_master()
+    opt.flat_grads.zero_()
+    opt.step(lr=0.0)  # lr 0: step must be a no-op on the params
+    assert torch.equal(p.detach(), new), \
+        "stale master reverted the out-of-band rewrite"
