@@ -98,8 +98,8 @@ extern "C" __global__ void drla_r2d2_loss_bwd(
 extern "C" __global__ void drla_per_update(float*, const long long*,
                                            const float*, int, long long);
 extern "C" __global__ void drla_per_sample(const float*, const float*,
-                                           long long*, float*, int,
-                                           long long);
+                                           const float*, long long*,
+                                           float*, int, long long);
 extern "C" __global__ void drla_per_rebuild_level(float*, long long,
                                                   long long);
 extern "C" __global__ void drla_multi_gather(
@@ -720,15 +720,17 @@ void multi_gather(torch::Tensor rows, torch::Tensor srcs,
 
 std::tuple<torch::Tensor, torch::Tensor> per_sample(torch::Tensor tree,
                                                     torch::Tensor s,
+                                                    torch::Tensor n_entries,
                                                     int64_t cap) {
   check_gpu_contig(tree, "tree");
   check_gpu_contig(s, "s");
+  check_gpu_contig(n_entries, "n_entries");
   const int n = s.numel();
   auto idx = torch::empty({n}, s.options().dtype(torch::kLong));
   auto prio = torch::empty({n}, s.options().dtype(torch::kFloat));
   hipLaunchKernelGGL(drla_per_sample, dim3((n + 255) / 256), dim3(256), 0,
                      cur_stream(), tree.data_ptr<float>(),
-                     s.data_ptr<float>(),
+                     s.data_ptr<float>(), n_entries.data_ptr<float>(),
                      reinterpret_cast<long long*>(idx.data_ptr<int64_t>()),
                      prio.data_ptr<float>(), n, cap);
   return {idx, prio};

@@ -77,6 +77,8 @@ extern "C" __global__ void drla_multi_gather(
 
 extern "C" __global__ void drla_per_sample(
     const float* __restrict__ tree, const float* __restrict__ s,
+    const float* __restrict__ n_entries,  // [1] device buffer (the host
+                                          // advances it; capture-safe)
     long long* __restrict__ out_idx, float* __restrict__ out_prio, int n,
     long long cap) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -95,6 +97,11 @@ extern "C" __global__ void drla_per_sample(
       idx = left + 1;
     }
   }
+  // float32 rounding at a segment boundary can walk past the last
+  // WRITTEN leaf onto a zero-priority slot -> stale payload and an
+  // infinite IS weight ((n*0)^-beta); clamp into the populated range
+  const long long last = cap - 2 + (long long)n_entries[0];
+  if (idx > last) idx = last;
   out_idx[i] = idx;
   out_prio[i] = tree[idx];
 }

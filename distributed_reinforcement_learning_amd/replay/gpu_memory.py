@@ -100,7 +100,7 @@ class GpuMemory:
         s = (torch.arange(n, device=self.device, dtype=torch.float32) + u) \
             * (total / n)
         idxs, prios = ext.per_sample(self.tree, s.contiguous(),
-                                     self.capacity)
+                                     self.n_entries_buf, self.capacity)
         probs = prios / total
         w = (self.n_entries_buf * probs).pow(-self.beta_buf)
         w = w / w.max()

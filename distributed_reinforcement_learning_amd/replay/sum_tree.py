@@ -97,7 +97,11 @@ class SumTree:
             s[interior] = np.where(go_right, s[interior] - left_sums,
                                    s[interior])
             idx[interior] = np.where(go_right, li + 1, li)
-        return idx
+        # rounding at a segment boundary can walk onto a zero-priority
+        # never-written leaf (stale payload, infinite IS weight); clamp
+        # into the populated range (ring writes fill the low slots first)
+        last = self.capacity - 2 + max(self.n_entries, 1)
+        return np.minimum(idx, last)
 
     def leaf_priorities(self, idxs: np.ndarray) -> np.ndarray:
         return self.tree[np.asarray(idxs, dtype=np.int64)]
