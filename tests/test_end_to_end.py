@@ -117,3 +117,35 @@ def test_trainer_end_to_end(tmp_path, algo, block, max_steps, timeout):
         assert blob["global_step"] == max_steps
     finally:
         os.chdir(cwd)
+
+
+def test_evaluate_script_on_trained_checkpoint(tmp_path):
+    """scripts/evaluate.py loads a checkpoint written by a spawn run and
+    reports episode scores (standalone evaluation entrypoint)."""
+    import subprocess
+    import sys
+    cfg_path = _tiny_config(tmp_path)
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        p = ctx.Process(target=_run_main,
+                        args=("a3c", "a3c_cartpole", cfg_path, 3, q))
+        p.start()
+        result = q.get(timeout=120)
+        p.join(timeout=30)
+        assert result == "ok", result
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        run = subprocess.run(
+            [sys.executable, os.path.join(repo, "scripts", "evaluate.py"),
+             "--algorithm", "a3c", "--algorithm_block", "a3c_cartpole",
+             "--checkpoint", str(tmp_path / "checkpoints"
+                                 / "a3c_cartpole.pt"),
+             "--config", cfg_path, "--episodes", "2"],
+            capture_output=True, text=True, timeout=120)
+        assert run.returncode == 0, run.stderr[-2000:]
+        out = json.loads(run.stdout.strip().splitlines()[-1])
+        assert out["episodes"] == 2 and out["mean_episode_steps"] > 0
+    finally:
+        os.chdir(cwd)
