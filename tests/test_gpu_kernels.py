@@ -395,7 +395,12 @@ def test_aug_gate_gemm_matches_addmm(ext):
     )
     torch.manual_seed(3)
     cell = LSTMCellTF(24, 16).cuda().bfloat16()
-    opt = FusedRMSProp(cell.parameters(), lr=1e-3)  # re-homes into flat
+    # heads follow the cell in the real models — the augmented view reads
+    # PAD-1 alias rows past the bias, so the flat buffer must extend
+    # (the storage guard rejects a cell-only buffer, by design)
+    tail = torch.nn.Linear(64, 64).cuda().bfloat16()
+    opt = FusedRMSProp(list(cell.parameters()) + list(tail.parameters()),
+                       lr=1e-3)  # re-homes into one flat buffer
     assert cell._aug_weight_ok()
     N = 33
     xh = (torch.randn(N, 40, device="cuda") * 0.5).to(torch.bfloat16)
