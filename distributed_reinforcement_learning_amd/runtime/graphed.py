@@ -19,8 +19,11 @@ Step structure (three graphs + a copy stream):
   graph 1:  normalize frames, batched unroll (bf16), fused V-trace losses
             (forward only)
   copy str: the NEXT batch's pinned -> static-input H2D, ordered after
-            graph 1 via an event (inputs are only read by the forward), so
-            the ~8.4 MB upload hides behind the backward
+            graph 1 via an event, so the ~18 MB upload hides behind the
+            backward. Every read of a static input happens INSIDE graph 1:
+            the conv-l1 kernel stashes its u8 input in passing and the
+            other backward-read inputs (previous_action / action /
+            initial_c) are cloned at the top of the forward
   graph 2:  backward (autograd ASSIGNS scatter-mode grads — no
             AccumulateGrad adds)
   graph 3:  one-kernel grad gather + (world > 1: CAPTURED RCCL all-reduce)
@@ -108,8 +111,10 @@ class GraphedImpalaStep:
 
         # ---- capture ------------------------------------------------------
         # forward and backward are SEPARATE graphs so the next step's H2D
-        # upload (~8.4 MB pinned -> static inputs, ~110 us) can overlap the
-        # backward on a copy stream: inputs are only read by the forward.
+        # upload (pinned -> static inputs, ~110 us) can overlap the
+        # backward on a copy stream; _fwd() confines every input read to
+        # graph 1 (conv-l1 input stash + clones of the backward-read
+        # fields), so the overlap cannot tear the backward's reads.
         self.g_fwd = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_fwd):
             self._total, self.losses = self._fwd()
