@@ -70,7 +70,9 @@ extern "C" __global__ void drla_a2c_loss_fwd(
   const float adv = a2c_clip(rewards[n], clip_mode)
                     + disc * next_value[n] - value[n];
   adv_st[n] = adv;
-  const float pi = -adv * __logf(p_stash[base + actions[n]] + 1e-8f) / N;
+  const float pi =
+      -adv * __logf(p_stash[base + drla_clamp_idx(actions[n], A)]
+                    + 1e-8f) / N;
   const float bl = adv * adv / N;
   const float en = ent / N;
   atomicAdd(&losses[0], pi);
@@ -91,7 +93,7 @@ extern "C" __global__ void drla_a2c_loss_bwd(
   const int n = blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= N) return;
   const long long base = (long long)n * A;
-  const int a = actions[n];
+  const int a = drla_clamp_idx(actions[n], A);
   const float adv = adv_st[n];
   const float sa = p_stash[base + a];
   const float w = sa / (sa + 1e-8f);

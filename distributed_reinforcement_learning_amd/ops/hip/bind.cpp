@@ -107,7 +107,7 @@ extern "C" __global__ void drla_multi_gather(
     const long long*, int);
 extern "C" __global__ void drla_embed_bwd_scatter(
     const long long*, const unsigned short*, const float*, float*,
-    long long, int);
+    long long, int, long long);
 extern "C" __global__ void drla_f32_to_bf16_zero_kernel(float*,
                                                         unsigned short*,
                                                         long long);
@@ -162,7 +162,7 @@ extern "C" __global__ void drla_heads_wgrad(
 extern "C" __global__ void drla_embed_mlp_fwd(
     const long long*, const unsigned short*, const unsigned short*,
     const unsigned short*, const unsigned short*, unsigned short*,
-    unsigned short*, int);
+    unsigned short*, int, int);
 extern "C" __global__ void drla_embed_w2t_pack(const unsigned short*,
                                                unsigned short*);
 extern "C" __global__ void drla_embed_mlp_bwd(
@@ -767,7 +767,7 @@ torch::Tensor embed_bwd(torch::Tensor indices, torch::Tensor grad_out,
       in16 ? reinterpret_cast<const unsigned short*>(grad_out.data_ptr())
            : nullptr,
       in16 ? nullptr : grad_out.data_ptr<float>(), scratch.data_ptr<float>(),
-      N, H);
+      N, H, (long long)num_rows);
   if (!want_bf16) return scratch;
   auto out = torch::empty({num_rows, H},
                           grad_out.options().dtype(torch::kBFloat16));
@@ -995,7 +995,7 @@ std::tuple<torch::Tensor, torch::Tensor> embed_mlp_fwd(
                      cur_stream(),
                      reinterpret_cast<const long long*>(pa.data_ptr<int64_t>()),
                      u16p(table), u16p(b1), u16p(w2), u16p(b2), u16pm(out),
-                     u16pm(a1), N);
+                     u16pm(a1), N, (int)table.size(0));
   return {out, a1};
 }
 
@@ -1031,7 +1031,8 @@ std::vector<torch::Tensor> embed_mlp_bwd(torch::Tensor dy, torch::Tensor out,
       drla_embed_bwd_scatter, dim3(drla_grid((long long)N * 256)),
       dim3(DRLA_BLOCK), 0, cur_stream(),
       reinterpret_cast<const long long*>(idx.data_ptr<int64_t>()),
-      u16p(da1), nullptr, scratch.data_ptr<float>(), N, 256);
+      u16p(da1), nullptr, scratch.data_ptr<float>(), N, 256,
+      (long long)A);
   auto dtable = torch::empty({A, 256}, bopt);
   auto db1 = torch::empty({256}, bopt);
   auto db2 = torch::empty({256}, bopt);

@@ -39,7 +39,8 @@ extern "C" __global__ void drla_dqn_loss_fwd(
     if (v > best) { best = v; astar = k; }
   }
   const float y = rewards[b] + discounts[b] * next_tgt[base + astar];
-  const float sav = dq_ld(mq16, mq32, base + actions[b]);
+  const float sav = dq_ld(mq16, mq32,
+                          base + drla_clamp_idx(actions[b], A));
   const float td = y - sav;
   td_out[b] = td;
   atomicAdd(loss, weights[b] * td * td / B);
@@ -53,7 +54,7 @@ extern "C" __global__ void drla_dqn_loss_bwd(
   if (i >= (long long)B * A) return;
   const int b = i / A;
   const int k = i - (long long)b * A;
-  const float g = (k == actions[b])
+  const float g = (k == drla_clamp_idx(actions[b], A))
                       ? gloss[0] * weights[b] * (-2.0f * td[b]) / B
                       : 0.0f;
   if (dmq16) dmq16[i] = drla_f32_to_bf16(g);

@@ -21,6 +21,14 @@ static inline int drla_grid(long long work_items, int block = DRLA_BLOCK) {
   return static_cast<int>(blocks);
 }
 
+// Defensive clamp for indices that arrive from EXTERNAL data (actor
+// rings, replay payloads): an out-of-range action must not become a GPU
+// aperture fault that takes the whole learner down — corrupt input
+// yields a wrong-but-bounded read instead.
+__device__ __forceinline__ int drla_clamp_idx(int a, int n) {
+  return a < 0 ? 0 : (a >= n ? n - 1 : a);
+}
+
 __device__ __forceinline__ float drla_sigmoid(float x) {
   return 1.0f / (1.0f + __expf(-x));
 }

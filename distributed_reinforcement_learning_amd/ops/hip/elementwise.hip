@@ -61,7 +61,7 @@ extern "C" __global__ void drla_embed_bwd_scatter(
     const unsigned short* __restrict__ gout16,  // [N,H] bf16 (nullable)
     const float* __restrict__ gout32,           // [N,H] f32 (nullable)
     float* __restrict__ scratch,                // [A,H] f32, zeroed
-    long long N, int H) {
+    long long N, int H, long long nrows) {
   long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
   const long long total = N * H;
   const long long stride = gridDim.x * (long long)blockDim.x;
@@ -75,7 +75,9 @@ extern "C" __global__ void drla_embed_bwd_scatter(
     } else {
       g = gout32[i];
     }
-    atomicAdd(&scratch[indices[n] * H + h], g);
+    // clamp external indices (see drla_clamp_idx rationale)
+    const long long row = drla_clamp_idx((int)indices[n], (int)nrows);
+    atomicAdd(&scratch[row * H + h], g);
   }
 }
 

@@ -540,7 +540,7 @@ extern "C" __global__ __launch_bounds__(256) void drla_embed_mlp_fwd(
     const bf16raw* __restrict__ b2,       // [256]
     bf16raw* __restrict__ out,            // [N,256] (post-ReLU2)
     bf16raw* __restrict__ a1stash,        // [N,256] (post-ReLU1)
-    int N) {
+    int N, int A) {
   __shared__ bf16raw a1img[MH_BM][MH_LD];
   __shared__ bf16raw outimg[MH_BM][MH_LD];
   const int tid = threadIdx.x;
@@ -549,7 +549,10 @@ extern "C" __global__ __launch_bounds__(256) void drla_embed_mlp_fwd(
     const int r = tid >> 4;
     const int c0 = (tid & 15) * 16;
     const int grow = row0 + r;
-    const long long idx = (grow < N) ? pa[grow] : 0;
+    // clamp: pa arrives from external actor data — an out-of-range
+    // index must not become an aperture fault (drla_common.h)
+    const long long idx =
+        (grow < N) ? drla_clamp_idx((int)pa[grow], A) : 0;
     const bf16raw* trow = table + idx * MH_HID + c0;
     for (int c = 0; c < 16; ++c) {
       const float v =

@@ -92,7 +92,8 @@ extern "C" __global__ void drla_r2d2_loss_fwd(
     const float y = r2_h(r2_hinv(nsav) * disc
                          + r2_clip(rewards[bt], clip_mode));
     const float sav = r2_ld(mq16, mq32,
-                            ((long long)b * W + t) * A + actions[bt]);
+                            ((long long)b * W + t) * A
+                                + drla_clamp_idx(actions[bt], A));
     td = y - sav;
     td_st[(long long)b * T1 + t] = td;
   }
@@ -194,7 +195,7 @@ extern "C" __global__ void drla_r2d2_loss_bwd(
   const int b = bt / W;
   const int T1 = W - 1;
   float g = 0.0f;
-  if (t < T1 && a == actions[bt]) {
+  if (t < T1 && a == drla_clamp_idx(actions[bt], A)) {
     g = gloss[0] * weights[b]
         * (-2.0f * td_st[(long long)b * T1 + t]) / T1 / B;
   }

@@ -97,7 +97,7 @@ void drla_vtrace_loss_fwd(
                                   : logits_f32[base + k];
       p_stash[base + k] = __expf(x - mx) * inv;
     }
-    const int a = actions[row + t];
+    const int a = drla_clamp_idx(actions[row + t], A);
     rho[t] = p_stash[base + a] / mu[base + a];
   }
   __syncthreads();
@@ -139,7 +139,8 @@ void drla_vtrace_loss_fwd(
     ent_l += p * __logf(p);  // negative entropy (reference vtrace.py:120)
     if (k == 0) {
       const float adv = adv_stash[b * Tp + t];
-      pi_l -= __logf(p_stash[pbase + actions[row + t]] + 1e-8f) * adv;
+      pi_l -= __logf(p_stash[pbase + drla_clamp_idx(
+                  actions[row + t], A)] + 1e-8f) * adv;
       const float diff = vs_stash[b * Tp + t] - value[row + t];
       base_l += 0.5f * diff * diff;
     }
@@ -204,7 +205,7 @@ extern "C" __global__ void drla_vtrace_loss_bwd(
       continue;
     }
     const int wi = b * Tp + t;
-    const int a = actions[i];
+    const int a = drla_clamp_idx(actions[i], A);
     const float adv = adv_stash[wi];
     const float sa = p_stash[pbase + a];
     const float w = sa / (sa + 1e-8f);

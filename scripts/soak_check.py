@@ -35,8 +35,17 @@ def soak_impala(steps=20000):
         device="cuda:0", seed=1)
     g = GraphedImpalaStep(agent, 32)
     rng = np.random.default_rng(0)
-    g.pinned["state"].copy_(torch.as_tensor(
-        rng.integers(0, 255, g.pinned["state"].shape, dtype=np.uint8)))
+    B, T, A, H = 32, 20, 18, 256
+    batch = dict(
+        state=rng.integers(0, 255, (B, T, 84, 84, 4), dtype=np.uint8),
+        reward=rng.normal(size=(B, T)).astype(np.float32),
+        action=rng.integers(0, A, (B, T)).astype(np.int32),
+        done=(rng.random((B, T)) < 0.02),
+        behavior_policy=np.full((B, T, A), 1.0 / A, dtype=np.float32),
+        previous_action=rng.integers(0, A, (B, T)).astype(np.int32),
+        initial_h=(rng.normal(size=(B, T, H)) * 0.1).astype(np.float32),
+        initial_c=(rng.normal(size=(B, T, H)) * 0.1).astype(np.float32))
+    g.stage_to_pinned(batch)
     for _ in range(50):
         g.step()
     torch.cuda.synchronize()
