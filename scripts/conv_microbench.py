@@ -73,7 +73,7 @@ def main():
         dy = torch.randn(N, ho, ho, co, device="cuda").bfloat16().contiguous()
         y, _ = ext.conv_fwd(layer, xc, w_flat, bias, False)
         def custom_bwd():
-            dy_m = ext.relu_mask_bwd(dy, y, co)
+            dy_m = ext.relu_mask_bwd(dy, y, co, layer)
             ext.conv_wgrad(layer, xc, dy_m)
             if layer >= 2:
                 ext.conv_dgrad(layer, dy_m, w_flat)
