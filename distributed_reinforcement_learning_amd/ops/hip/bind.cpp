@@ -92,6 +92,17 @@ extern "C" __global__ void drla_dueling_head_fwd(
     const float*, const unsigned short*, const unsigned short*,
     const unsigned short*, const unsigned short*, unsigned short*, int,
     int, int, int, int, int);
+extern "C" __global__ void drla_dhead_train_fwd(
+    const float*, const unsigned short*, const unsigned short*,
+    const unsigned short*, const unsigned short*, unsigned short*,
+    unsigned short*, int, int, int, int);
+extern "C" __global__ void drla_dhead_train_bwd(
+    const unsigned short*, const unsigned short*, const float*,
+    const unsigned short*, const unsigned short*, float*, float*, int,
+    int, int, int);
+extern "C" __global__ void drla_dhead_train_fin(
+    const float*, unsigned short*, unsigned short*, unsigned short*,
+    unsigned short*, int, int, int);
 extern "C" __global__ void drla_r2d2_loss_bwd(
     const float*, const int*, const float*, const float*, unsigned short*,
     float*, int, int, int);
@@ -643,6 +654,68 @@ torch::Tensor dueling_head_fwd(torch::Tensor h, torch::Tensor Wt,
                      u16p(bt), u16p(Wo), u16p(bo), u16pm(q), B, L,
                      (int)burn, IN, MID, AO);
   return q;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> dhead_train_fwd(
+    torch::Tensor h, torch::Tensor Wt, torch::Tensor bt, torch::Tensor Wo,
+    torch::Tensor bo) {
+  for (auto* t : {&h, &Wt, &bt, &Wo, &bo})
+    check_gpu_contig(*t, "dhead fwd input");
+  TORCH_CHECK(h.dim() == 2 && h.scalar_type() == torch::kFloat,
+              "h must be [N,IN] f32");
+  const int N = h.size(0), IN = h.size(1);
+  const int MID = Wt.size(0), AO = Wo.size(0);
+  TORCH_CHECK(Wt.size(1) == IN && Wo.size(1) == MID && AO >= 2,
+              "dhead weight shape");
+  const size_t lds = (size_t)(MID * IN + AO * MID) * 2 +
+                     (size_t)(2 * MID + 2 * IN + 2 * AO) * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "dhead fwd LDS budget");
+  auto q = torch::empty({N, AO - 1},
+                        h.options().dtype(torch::kBFloat16));
+  auto x_st = torch::empty({N, MID},
+                           h.options().dtype(torch::kBFloat16));
+  const long long npairs = ((long long)N + 1) / 2;
+  const int gx = (int)std::min<long long>(npairs, 1024);
+  hipLaunchKernelGGL(drla_dhead_train_fwd, dim3(gx), dim3(256), lds,
+                     cur_stream(), h.data_ptr<float>(), u16p(Wt), u16p(bt),
+                     u16p(Wo), u16p(bo), u16pm(q), u16pm(x_st), N, IN, MID,
+                     AO);
+  return {q, x_st};
+}
+
+std::vector<torch::Tensor> dhead_train_bwd(
+    torch::Tensor dq, torch::Tensor x_st, torch::Tensor h,
+    torch::Tensor Wt, torch::Tensor Wo) {
+  for (auto* t : {&dq, &x_st, &h, &Wt, &Wo})
+    check_gpu_contig(*t, "dhead bwd input");
+  const int N = h.size(0), IN = h.size(1);
+  const int MID = Wt.size(0), AO = Wo.size(0);
+  TORCH_CHECK(dq.scalar_type() == torch::kBFloat16, "dq must be bf16");
+  const size_t lds = (size_t)(MID * IN + AO * MID) * 2 +
+                     (size_t)(2 * (MID * IN + AO * MID)
+                              + 2 * MID + 2 * AO + MID + IN) * 4;
+  TORCH_CHECK(lds <= 160 * 1024, "dhead bwd LDS budget");
+  auto fopt = h.options();
+  auto bopt = h.options().dtype(torch::kBFloat16);
+  auto dh = torch::empty({N, IN}, fopt);
+  const int ws_n = MID * IN + MID + AO * MID + AO;
+  auto ws = torch::zeros({ws_n}, fopt);
+  const int gx =
+      (int)std::min<long long>(((long long)N + 7) / 8, 64);
+  hipLaunchKernelGGL(drla_dhead_train_bwd, dim3(gx), dim3(256), lds,
+                     cur_stream(), u16p(dq), u16p(x_st),
+                     h.data_ptr<float>(), u16p(Wt), u16p(Wo),
+                     dh.data_ptr<float>(), ws.data_ptr<float>(), N, IN,
+                     MID, AO);
+  auto dWt = torch::empty({MID, IN}, bopt);
+  auto dbt = torch::empty({MID}, bopt);
+  auto dWo = torch::empty({AO, MID}, bopt);
+  auto dbo = torch::empty({AO}, bopt);
+  hipLaunchKernelGGL(drla_dhead_train_fin, dim3(drla_grid(ws_n)),
+                     dim3(DRLA_BLOCK), 0, cur_stream(),
+                     ws.data_ptr<float>(), u16pm(dWt), u16pm(dbt),
+                     u16pm(dWo), u16pm(dbo), IN, MID, AO);
+  return {dh, dWt, dbt, dWo, dbo};
 }
 
 torch::Tensor r2d2_loss_bwd(torch::Tensor td_st, torch::Tensor actions,
@@ -1361,6 +1434,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("a2c_loss_bwd", &a2c_loss_bwd, "fused A2C loss bwd (K6)");
   m.def("r2d2_loss_fwd", &r2d2_loss_fwd,
         "fused R2D2 sequence-TD tail fwd (K9)");
+  m.def("dhead_train_fwd", &dhead_train_fwd,
+        "grad-carrying dueling head fwd (R2D2 trained window)");
+  m.def("dhead_train_bwd", &dhead_train_bwd,
+        "grad-carrying dueling head bwd chain + weight grads");
   m.def("dueling_head_fwd", &dueling_head_fwd,
         "no-grad dueling head over the post-burn-in window");
   m.def("r2d2_loss_bwd", &r2d2_loss_bwd,

@@ -202,3 +202,194 @@ extern "C" __global__ void drla_r2d2_loss_bwd(
   if (dmq16) dmq16[i] = drla_f32_to_bf16(g);
   else dmq32[i] = g;
 }
+
+// ---------------------------------------------------------------------------
+// GRAD-CARRYING dueling head (the R2D2 trained window): q = y[:A] - y[A],
+// y = relu(h @ Wt^T + bt) @ Wo^T + bo. The torch chain costs ~80 us/step
+// in ~13 launches at [B*W=128..640, 64]; this trio does fwd in 1 launch
+// and bwd in 2 (chain + finalize).
+//
+// bwd math (dq [N,A] from the K9 kernel):
+//   dz[:A] = dq,  dz[A] = -sum_a dq[a]          (the dueling subtract)
+//   dx     = (dz @ Wo) * [x > 0]
+//   dh     = dx @ Wt
+//   dWo    = dz^T @ x,  dbo = colsum dz
+//   dWt    = dx^T @ h,  dbt = colsum dx
+// Weight-grad partials accumulate in per-block LDS tiles (the whole
+// dWt/dWo fit: [MID][IN]+[AO][MID] f32 <= ~50 KB at the R2D2 shape) and
+// merge with ONE atomicAdd per element per block into an f32 workspace;
+// the finalize kernel emits the four bf16 grads.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void drla_dhead_train_fwd(
+    const float* __restrict__ h,       // [N, IN]
+    const r2_bf16* __restrict__ Wt,    // [MID][IN]
+    const r2_bf16* __restrict__ bt,    // [MID]
+    const r2_bf16* __restrict__ Wo,    // [AO][MID]
+    const r2_bf16* __restrict__ bo,    // [AO]
+    r2_bf16* __restrict__ q,           // [N, AO-1]
+    r2_bf16* __restrict__ x_st,        // [N, MID] post-ReLU stash
+    int N, int IN, int MID, int AO) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  r2_bf16* wt = reinterpret_cast<r2_bf16*>(smem);
+  r2_bf16* wo = wt + MID * IN;
+  float* xb = reinterpret_cast<float*>(wo + AO * MID);     // [2][MID]
+  float* hb = xb + 2 * MID;                                // [2][IN]
+  float* yb = hb + 2 * IN;                                 // [2][AO]
+
+  const int tid = threadIdx.x;
+  for (int i = tid; i < MID * IN; i += blockDim.x) wt[i] = Wt[i];
+  for (int i = tid; i < AO * MID; i += blockDim.x) wo[i] = Wo[i];
+  __syncthreads();
+
+  const long long npairs = ((long long)N + 1) / 2;
+  const int half = tid / 128;
+  const int lane = tid % 128;
+  const int A = AO - 1;
+  for (long long p = blockIdx.x; p < npairs; p += gridDim.x) {
+    const long long n = 2 * p + half;
+    const bool live = n < (long long)N;
+    if (live) {
+      for (int i = lane; i < IN; i += 128)
+        hb[half * IN + i] = h[n * IN + i];
+    }
+    __syncthreads();
+    if (live) {
+      for (int o = lane; o < MID; o += 128) {
+        float acc = r2_ld(bt, nullptr, o);
+        const r2_bf16* wrow = wt + o * IN;
+        for (int i = 0; i < IN; ++i)
+          acc = fmaf(hb[half * IN + i], r2_ld(wrow, nullptr, i), acc);
+        const float x = fmaxf(acc, 0.0f);
+        xb[half * MID + o] = x;
+        x_st[n * MID + o] = drla_f32_to_bf16(x);
+      }
+    }
+    __syncthreads();
+    if (live && lane < AO) {
+      float acc = r2_ld(bo, nullptr, lane);
+      const r2_bf16* wrow = wo + lane * MID;
+      for (int i = 0; i < MID; ++i)
+        acc = fmaf(xb[half * MID + i], r2_ld(wrow, nullptr, i), acc);
+      yb[half * AO + lane] = acc;
+    }
+    __syncthreads();
+    if (live && lane < A) {
+      q[n * A + lane] =
+          drla_f32_to_bf16(yb[half * AO + lane] - yb[half * AO + A]);
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" __global__ void drla_dhead_train_bwd(
+    const r2_bf16* __restrict__ dq,    // [N, AO-1]
+    const r2_bf16* __restrict__ x_st,  // [N, MID]
+    const float* __restrict__ h,       // [N, IN]
+    const r2_bf16* __restrict__ Wt,    // [MID][IN]
+    const r2_bf16* __restrict__ Wo,    // [AO][MID]
+    float* __restrict__ dh,            // [N, IN] f32
+    float* __restrict__ ws,            // [MID*IN + MID + AO*MID + AO],
+                                       // zeroed: dWt | dbt | dWo | dbo
+    int N, int IN, int MID, int AO) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  r2_bf16* wt = reinterpret_cast<r2_bf16*>(smem);          // [MID*IN]
+  r2_bf16* wo = wt + MID * IN;                             // [AO*MID]
+  float* lWt = reinterpret_cast<float*>(wo + AO * MID);    // [MID*IN]
+  float* lWo = lWt + MID * IN;                             // [AO*MID]
+  float* lbt = lWo + AO * MID;                             // [MID]
+  float* lbo = lbt + MID;                                  // [AO]
+  float* dzb = lbo + AO;                                   // [AO]
+  float* dxb = dzb + AO;                                   // [MID]
+  float* xb = dxb + MID;                                   // [MID]
+  float* hbuf = xb + MID;                                  // [IN]
+
+  const int tid = threadIdx.x;
+  for (int i = tid; i < MID * IN; i += blockDim.x) {
+    wt[i] = Wt[i];
+    lWt[i] = 0.0f;
+  }
+  for (int i = tid; i < AO * MID; i += blockDim.x) {
+    wo[i] = Wo[i];
+    lWo[i] = 0.0f;
+  }
+  for (int i = tid; i < MID; i += blockDim.x) lbt[i] = 0.0f;
+  for (int i = tid; i < AO; i += blockDim.x) lbo[i] = 0.0f;
+  __syncthreads();
+
+  const int A = AO - 1;
+  const int rows_per_block = (N + gridDim.x - 1) / gridDim.x;
+  const long long n0 = (long long)blockIdx.x * rows_per_block;
+  const long long n1 = min((long long)N, n0 + rows_per_block);
+  for (long long n = n0; n < n1; ++n) {
+    // dz (+ dbo) and stage x/h
+    if (tid < AO) {
+      float v;
+      if (tid < A) {
+        v = r2_ld(dq, nullptr, n * A + tid);
+      } else {
+        float s = 0.0f;
+        for (int a = 0; a < A; ++a)
+          s += r2_ld(dq, nullptr, n * A + a);
+        v = -s;
+      }
+      dzb[tid] = v;
+      lbo[tid] += v;
+    }
+    for (int o = tid; o < MID; o += blockDim.x)
+      xb[o] = r2_ld(x_st, nullptr, n * MID + o);
+    for (int i = tid; i < IN; i += blockDim.x) hbuf[i] = h[n * IN + i];
+    __syncthreads();
+    // dx (+ dbt, dWo)
+    for (int o = tid; o < MID; o += blockDim.x) {
+      float acc = 0.0f;
+      for (int a = 0; a < AO; ++a) {
+        const float dz = dzb[a];
+        acc = fmaf(dz, r2_ld(wo, nullptr, a * MID + o), acc);
+        lWo[a * MID + o] = fmaf(dz, xb[o], lWo[a * MID + o]);
+      }
+      const float dx = (xb[o] > 0.0f) ? acc : 0.0f;
+      dxb[o] = dx;
+      lbt[o] += dx;
+    }
+    __syncthreads();
+    // dh and the dWt rank-1 update
+    for (int i = tid; i < IN; i += blockDim.x) {
+      float acc = 0.0f;
+      for (int o = 0; o < MID; ++o)
+        acc = fmaf(dxb[o], r2_ld(wt, nullptr, o * IN + i), acc);
+      dh[n * IN + i] = acc;
+    }
+    for (int oi = tid; oi < MID * IN; oi += blockDim.x) {
+      const int o = oi / IN;
+      const int i = oi - o * IN;
+      lWt[oi] = fmaf(dxb[o], hbuf[i], lWt[oi]);
+    }
+    __syncthreads();
+  }
+  // merge block-local partials: one atomicAdd per element per block
+  for (int i = tid; i < MID * IN; i += blockDim.x)
+    atomicAdd(&ws[i], lWt[i]);
+  for (int i = tid; i < MID; i += blockDim.x)
+    atomicAdd(&ws[MID * IN + i], lbt[i]);
+  for (int i = tid; i < AO * MID; i += blockDim.x)
+    atomicAdd(&ws[MID * IN + MID + i], lWo[i]);
+  for (int i = tid; i < AO; i += blockDim.x)
+    atomicAdd(&ws[MID * IN + MID + AO * MID + i], lbo[i]);
+}
+
+extern "C" __global__ void drla_dhead_train_fin(
+    const float* __restrict__ ws, r2_bf16* __restrict__ dWt,
+    r2_bf16* __restrict__ dbt, r2_bf16* __restrict__ dWo,
+    r2_bf16* __restrict__ dbo, int IN, int MID, int AO) {
+  const int total = MID * IN + MID + AO * MID + AO;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    const r2_bf16 v = drla_f32_to_bf16(ws[i]);
+    if (i < MID * IN) dWt[i] = v;
+    else if (i < MID * IN + MID) dbt[i - MID * IN] = v;
+    else if (i < MID * IN + MID + AO * MID)
+      dWo[i - MID * IN - MID] = v;
+    else dbo[i - MID * IN - MID - AO * MID] = v;
+  }
+}

@@ -138,3 +138,44 @@ def test_dueling_head_kernel_matches_torch():
     assert q_k.shape == (B, L - burn, 4)
     assert torch.allclose(q_k.float(), q_t.float(), atol=3e-2, rtol=3e-2), \
         (q_k.float() - q_t.float()).abs().max()
+
+
+def test_dueling_head_train_matches_torch():
+    """Grad-carrying fused dueling head (drla_dhead_train_*) vs the torch
+    composition: forward values and ALL five grads (dh, dWt, dbt, dWo,
+    dbo)."""
+    from distributed_reinforcement_learning_amd.ops.r2d2_op import (
+        fused_dueling_head_train,
+    )
+    torch.manual_seed(17)
+    N, IN, MID, A = 123, 64, 128, 4
+    trunk = torch.nn.Linear(IN, MID).cuda().bfloat16()
+    out = torch.nn.Linear(MID, A + 1).cuda().bfloat16()
+    h = (torch.randn(N, IN, device="cuda")).requires_grad_(True)
+
+    q = fused_dueling_head_train(h, trunk, out)
+    g = torch.randn(N, A, device="cuda").to(torch.bfloat16)
+    q.backward(g)
+
+    h2 = h.detach().clone().requires_grad_(True)
+    trunk2 = torch.nn.Linear(IN, MID).cuda().bfloat16()
+    out2 = torch.nn.Linear(MID, A + 1).cuda().bfloat16()
+    trunk2.load_state_dict(trunk.state_dict())
+    out2.load_state_dict(out.state_dict())
+    x = torch.relu(trunk2(h2.to(torch.bfloat16)))
+    y = out2(x)
+    q2 = y[:, :A] - y[:, A:]
+    q2.backward(g)
+
+    assert torch.allclose(q.float(), q2.float(), atol=3e-2, rtol=3e-2), \
+        (q.float() - q2.float()).abs().max()
+    assert torch.allclose(h.grad, h2.grad, atol=3e-2, rtol=3e-2), \
+        (h.grad - h2.grad).abs().max()
+    pairs = [(trunk.weight.grad, trunk2.weight.grad, 0.3),
+             (trunk.bias.grad, trunk2.bias.grad, 0.2),
+             (out.weight.grad, out2.weight.grad, 0.3),
+             (out.bias.grad, out2.bias.grad, 0.2)]
+    for a, b, tol in pairs:
+        assert a is not None and b is not None
+        assert torch.allclose(a.float(), b.float(), atol=tol, rtol=0.05), \
+            (a.float() - b.float()).abs().max()
