@@ -75,3 +75,22 @@ def test_bench_torchrun_two_ranks():
     # whole-job aggregate: 2 ranks x 2 batch x 4 steps... value is frames/s;
     # just confirm the frames accounting doubled via global_batch above
     assert out["config"]["frames_per_step_per_rank"] == 8
+
+
+def test_bench_torchrun_four_ranks():
+    """The driver's N=4 launch shape (gloo on CPU): rank logic, pool
+    seeding, MAX-over-ranks timing and the single rank-0 JSON line must
+    hold beyond 2 ranks."""
+    env = dict(os.environ)
+    env.pop("WORLD_SIZE", None)
+    env.pop("RANK", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29513", "bench.py"] + BENCH_ARGS,
+        capture_output=True, text=True, cwd=REPO, timeout=600, env=env)
+    assert r.returncode == 0, r.stderr[-3000:]
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1
+    out = _check_line(json_lines[-1], 4)
+    assert out["config"]["global_batch"] == 8
