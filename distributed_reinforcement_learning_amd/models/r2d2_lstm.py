@@ -107,8 +107,10 @@ class R2D2LstmQ(nn.Module):
             h_all, _, _ = lstm_seq_train(
                 xg.reshape(B, L, -1), w[F:], h0.float(), c0.float(),
                 seq_done, self.lstm.forget_bias)
+            import os as _os
             if (self.out.weight.dtype == torch.bfloat16
-                    and self.trunk.weight.dtype == torch.bfloat16):
+                    and self.trunk.weight.dtype == torch.bfloat16
+                    and _os.environ.get("DRLA_NO_FUSED_DHEAD") != "1"):
                 from distributed_reinforcement_learning_amd.ops import (
                     r2d2_op,
                 )

@@ -648,7 +648,9 @@ torch::Tensor dueling_head_fwd(torch::Tensor h, torch::Tensor Wt,
   TORCH_CHECK(lds <= 160 * 1024,
               "dueling head weights exceed the 160 KB LDS budget");
   const long long npairs = ((long long)B * W + 1) / 2;
-  const int gx = (int)std::min<long long>(npairs, 1024);
+  // few blocks, many row-pairs each: every block stages the full weight
+  // set in LDS, so the grid must amortize that (~24 KB/block) over rows
+  const int gx = (int)std::min<long long>((npairs + 7) / 8, 64);
   hipLaunchKernelGGL(drla_dueling_head_fwd, dim3(gx), dim3(256), lds,
                      cur_stream(), h.data_ptr<float>(), u16p(Wt),
                      u16p(bt), u16p(Wo), u16p(bo), u16pm(q), B, L,
@@ -675,7 +677,7 @@ std::tuple<torch::Tensor, torch::Tensor> dhead_train_fwd(
   auto x_st = torch::empty({N, MID},
                            h.options().dtype(torch::kBFloat16));
   const long long npairs = ((long long)N + 1) / 2;
-  const int gx = (int)std::min<long long>(npairs, 1024);
+  const int gx = (int)std::min<long long>((npairs + 7) / 8, 64);
   hipLaunchKernelGGL(drla_dhead_train_fwd, dim3(gx), dim3(256), lds,
                      cur_stream(), h.data_ptr<float>(), u16p(Wt), u16p(bt),
                      u16p(Wo), u16p(bo), u16pm(q), u16pm(x_st), N, IN, MID,
