@@ -108,9 +108,13 @@ class R2D2LstmQ(nn.Module):
                 xg.reshape(B, L, -1), w[F:], h0.float(), c0.float(),
                 seq_done, self.lstm.forget_bias)
             import os as _os
+            # fused train-head measured NET SLOWER than the torch chain
+            # (A/B same box: 22.2k -> 20.0k seq/s at 15/7 — the fused
+            # backward's serial per-row phases cost more than the ~13
+            # launches it saves); opt-in for round-3 tuning
             if (self.out.weight.dtype == torch.bfloat16
                     and self.trunk.weight.dtype == torch.bfloat16
-                    and _os.environ.get("DRLA_NO_FUSED_DHEAD") != "1"):
+                    and _os.environ.get("DRLA_FUSED_DHEAD") == "1"):
                 from distributed_reinforcement_learning_amd.ops import (
                     r2d2_op,
                 )
