@@ -74,3 +74,51 @@ def test_graphed_step_updates_weights_every_replay():
     assert not torch.equal(w0, w1)
     assert not torch.equal(w1, w2)
     assert agent.global_step == 2
+
+
+def test_graphed_a3c_train_step():
+    """GraphedTrainStep (the A3C GPU learner's captured loss+optimizer
+    step, runtime/replay_graphed.py) over the fused K6 loss: capture must
+    not perturb weights, replays must train, losses stay finite."""
+    from distributed_reinforcement_learning_amd.agents import a3c
+    from distributed_reinforcement_learning_amd.runtime.replay_graphed import (
+        GraphedTrainStep,
+    )
+    rng = np.random.default_rng(4)
+    N, A = 8, 4
+    agent = a3c.Agent(
+        input_shape=[84, 84, 4], num_action=A, discount_factor=0.997,
+        baseline_loss_coef=1.0, entropy_coef=0.05,
+        start_learning_rate=1e-3, end_learning_rate=0.0,
+        learning_frame=10 ** 9, gradient_clip_norm=40.0,
+        reward_clipping="abs_one", device="cuda:0", seed=2)
+
+    def mk():
+        return {
+            "state": torch.as_tensor(rng.integers(
+                0, 255, (N, 84, 84, 4), dtype=np.uint8)).cuda(),
+            "next_state": torch.as_tensor(rng.integers(
+                0, 255, (N, 84, 84, 4), dtype=np.uint8)).cuda(),
+            "pa": torch.as_tensor(rng.integers(0, A, N)).cuda(),
+            "a": torch.as_tensor(rng.integers(0, A, N)).cuda(),
+            "r": torch.as_tensor(
+                rng.normal(size=N).astype(np.float32)).cuda(),
+            "d": torch.as_tensor(rng.random(N) < 0.1).cuda(),
+        }
+
+    def loss_fn(i):
+        pi, bl, ent, total = agent.compute_a2c_losses(
+            agent.frames_to_device(i["state"]),
+            agent.frames_to_device(i["next_state"]),
+            i["pa"].long(), i["a"].long(), i["r"], i["d"])
+        return pi, bl, ent, total
+
+    p_before = agent.optimizer.flat_params.detach().clone()
+    g = GraphedTrainStep(agent, mk(), loss_fn)
+    assert torch.equal(agent.optimizer.flat_params, p_before)
+    for _ in range(4):
+        out = g.step(mk())
+        assert len(out) == 4
+    torch.cuda.synchronize()
+    assert all(np.isfinite(float(x)) for x in out[:3])
+    assert not torch.equal(agent.optimizer.flat_params, p_before)
