@@ -1,4 +1,8 @@
-"""Per-algorithm GPU learner microbenches (Ape-X and R2D2): synthetic
+"""DEPRECATED in favor of ``python bench.py --algo {apex,r2d2}`` (the
+driver-contract harness with warmup/clock-priming and the standard JSON
+line); kept for r1 comparability.
+
+Per-algorithm GPU learner microbenches (Ape-X and R2D2): synthetic
 replay contents, device-resident PER shard, reference configs. Reports
 train-steps/s and transitions/s per algorithm (one JSON line each).
 
