@@ -52,3 +52,24 @@ def test_impala_two_rank_learner_spawn(tmp_path):
     assert (tmp_path / "runs" / "learner" / "scalars.jsonl").exists()
     assert (tmp_path / "runs" / "learner_rank1").exists()
     assert (tmp_path / "checkpoints" / "impala.pt").exists()
+
+
+def test_capture_failure_policy(monkeypatch):
+    """handle_capture_failure: FAIL-FAST by default (re-raises), continue
+    with a loud warning only under DRLA_ALLOW_EAGER_REDUCE=1."""
+    import pytest
+    from distributed_reinforcement_learning_amd.parallel.dist import (
+        handle_capture_failure,
+    )
+    err = RuntimeError("capture refused")
+    monkeypatch.delenv("DRLA_ALLOW_EAGER_REDUCE", raising=False)
+    try:
+        raise err
+    except RuntimeError:
+        with pytest.raises(RuntimeError):
+            handle_capture_failure(err)
+    monkeypatch.setenv("DRLA_ALLOW_EAGER_REDUCE", "1")
+    try:
+        raise err
+    except RuntimeError:
+        handle_capture_failure(err)  # must not raise

@@ -77,6 +77,25 @@ def test_adapter_translates_ale_lives(fake_gym):
     assert not info.get("life_lost")
 
 
+def test_adapter_accepts_gymnasium_lives_key(fake_gym):
+    """Newer ALE builds expose the counter as info['lives'] (no 'ale.'
+    prefix) — the adapter must track either spelling."""
+    import gym
+    from distributed_reinforcement_learning_amd.envs.base import GymAdapter
+    base = gym.make("x")
+    orig_step = base.step
+
+    def step(a):
+        obs, r, done, info = orig_step(a)
+        return obs, r, done, {"lives": info["ale.lives"]}
+
+    base.step = step
+    env = GymAdapter(base)
+    env.reset()
+    flags = [bool(env.step(0)[3].get("life_lost")) for _ in range(8)]
+    assert flags.count(True) == 1 and flags[4]
+
+
 def test_full_pipeline_over_fake_gym(fake_gym):
     """make_uint8_env resolves to the GymAdapter (not the synthetic
     fallback) and the preprocessing pipeline produces the reference's
