@@ -17,7 +17,9 @@ import torch
 
 def compute_entropy_loss(policy: torch.Tensor) -> torch.Tensor:
     """-mean_t entropy_t over the batch ([N,A] softmax)."""
-    entropy = (-policy * torch.log(policy)).sum(dim=1)
+    # clamp inside the log: softmax underflow gives policy == 0 exactly
+    # and 0 * log(0) is NaN; the correct limit of p*log p is 0
+    entropy = (-policy * torch.log(policy.clamp_min(1e-30))).sum(dim=1)
     return -entropy.mean()
 
 

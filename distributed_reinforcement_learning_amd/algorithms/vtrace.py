@@ -117,5 +117,7 @@ def compute_baseline_loss(vs: torch.Tensor, value: torch.Tensor) -> torch.Tensor
 def compute_entropy_loss(softmax: torch.Tensor) -> torch.Tensor:
     """-sum_t entropy_t (reference vtrace.py:120-126: the *negative* entropy,
     added to the total loss with entropy_coef)."""
-    entropy_per_step = (-softmax * torch.log(softmax)).sum(dim=-1)
+    # clamp inside the log: 0 * log(0) would be NaN (underflowed softmax)
+    entropy_per_step = (
+        -softmax * torch.log(softmax.clamp_min(1e-30))).sum(dim=-1)
     return -entropy_per_step.sum()

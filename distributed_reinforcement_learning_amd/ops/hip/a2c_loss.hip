@@ -64,7 +64,9 @@ extern "C" __global__ void drla_a2c_loss_fwd(
   for (int k = 0; k < A; ++k) {
     const float p = __expf(a2c_ld(lg16, lg32, base + k) - mx) * inv_z;
     p_stash[base + k] = p;
-    ent += p * __logf(p);
+    // clamp inside the log: p == 0 after softmax underflow would make
+    // 0 * logf(0) NaN; the correct limit is 0
+    ent += p * __logf(fmaxf(p, 1e-30f));
   }
   const float disc = done[n] ? 0.0f : gamma;
   const float adv = a2c_clip(rewards[n], clip_mode)
@@ -100,13 +102,13 @@ extern "C" __global__ void drla_a2c_loss_bwd(
   float E = 0.0f;
   for (int k = 0; k < A; ++k) {
     const float p = p_stash[base + k];
-    E += p * __logf(p);
+    E += p * __logf(fmaxf(p, 1e-30f));  // 0*log(0) -> 0, not NaN
   }
   for (int k = 0; k < A; ++k) {
     const float p = p_stash[base + k];
     const float onehot = (k == a) ? 1.0f : 0.0f;
     const float d = (gpi * (-adv * w * (onehot - p))
-                     + ge * p * (__logf(p) - E)) / N;
+                     + ge * p * (__logf(fmaxf(p, 1e-30f)) - E)) / N;
     if (dlg16) dlg16[base + k] = drla_f32_to_bf16(d);
     else dlg32[base + k] = d;
   }

@@ -136,7 +136,10 @@ void drla_vtrace_loss_fwd(
     const int k = i - t * A;
     const long long pbase = (row + t) * (long long)A;
     const float p = p_stash[pbase + k];
-    ent_l += p * __logf(p);  // negative entropy (reference vtrace.py:120)
+    // clamp inside the log: softmax underflow gives p == 0 exactly and
+    // 0 * logf(0) is NaN — the correct limit of p*log p is 0
+    ent_l += p * __logf(fmaxf(p, 1e-30f));  // negative entropy
+                                            // (reference vtrace.py:120)
     if (k == 0) {
       const float adv = adv_stash[b * Tp + t];
       pi_l -= __logf(p_stash[pbase + drla_clamp_idx(
@@ -213,13 +216,13 @@ extern "C" __global__ void drla_vtrace_loss_bwd(
     float E = 0.0f;
     for (int k = 0; k < A; ++k) {
       const float s = p_stash[pbase + k];
-      E += s * __logf(s);
+      E += s * __logf(fmaxf(s, 1e-30f));  // 0*log(0) -> 0, not NaN
     }
     for (int k = 0; k < A; ++k) {
       const float s = p_stash[pbase + k];
       const float onehot = (k == a) ? 1.0f : 0.0f;
       const float d_pi = -adv * w * (onehot - s);
-      const float d_ent = s * (__logf(s) - E);
+      const float d_ent = s * (__logf(fmaxf(s, 1e-30f)) - E);
       const float d = gpi * d_pi + ge * d_ent;
       if (dlogits_bf16) dlogits_bf16[pbase + k] = drla_f32_to_bf16(d);
       else dlogits_f32[pbase + k] = d;
